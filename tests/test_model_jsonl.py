@@ -1,0 +1,135 @@
+"""JSONL schema + Go-compatible encoding tests (reference model/data.go)."""
+import datetime as dt
+import json
+
+from crawler_amd.models import (
+    ChannelData,
+    Comment,
+    EngagementData,
+    PerformanceScores,
+    Post,
+    format_go_time,
+    go_json_escape,
+)
+
+UTC = dt.timezone.utc
+
+
+def test_escape_basic():
+    assert go_json_escape('he said "hi"') == 'he said \\"hi\\"'
+    assert go_json_escape("a\\b") == "a\\\\b"
+    assert go_json_escape("l1\nl2\tend\r") == "l1\\nl2\\tend\\r"
+
+
+def test_escape_html_unsafe_matches_go():
+    # Go encoding/json escapes <, >, & as < etc. by default
+    assert go_json_escape("<b>&") == "\\u003cb\\u003e\\u0026"
+
+
+def test_escape_control_chars():
+    assert go_json_escape("\x00\x01\x1f") == "\\u0000\\u0001\\u001f"
+
+
+def test_escape_line_separators():
+    assert go_json_escape("a b ") == "a\\u2028b\\u2029"
+
+
+def test_escape_unicode_passthrough():
+    # Non-ASCII is NOT escaped (Go writes UTF-8 bytes through)
+    assert go_json_escape("привет мир 🚀") == "привет мир 🚀"
+
+
+def test_go_time_format():
+    t = dt.datetime(2024, 1, 2, 3, 4, 5, tzinfo=UTC)
+    assert format_go_time(t) == "2024-01-02T03:04:05Z"
+
+
+def test_go_time_zero_value():
+    assert format_go_time(None) == "0001-01-01T00:00:00Z"
+
+
+def test_go_time_fractional_trimmed():
+    t = dt.datetime(2024, 1, 2, 3, 4, 5, 120000, tzinfo=UTC)
+    assert format_go_time(t) == "2024-01-02T03:04:05.12Z"
+
+
+def test_go_time_offset():
+    t = dt.datetime(2024, 1, 2, 3, 4, 5, tzinfo=dt.timezone(dt.timedelta(hours=2)))
+    assert format_go_time(t) == "2024-01-02T03:04:05+02:00"
+
+
+def test_default_post_is_valid_json_with_exact_field_order():
+    p = Post()
+    s = p.to_json()
+    obj = json.loads(s)
+    keys = list(obj.keys())
+    # First and last few fields in Go struct declaration order
+    assert keys[:6] == [
+        "post_link", "channel_id", "post_uid", "url", "published_at",
+        "created_at",
+    ]
+    assert keys[-6:] == [
+        "media_url", "comments", "reactions", "outlinks", "capture_time",
+        "handle",
+    ]
+    assert len(keys) == 65  # reference model/data.go Post json tag count
+
+
+def test_nil_slices_marshal_null_empty_marshal_brackets():
+    p = Post()
+    obj = json.loads(p.to_json())
+    assert obj["list_ids"] is None         # nil slice -> null
+    assert obj["outlinks"] is None
+    p2 = Post(outlinks=[], comments=[], reactions={})
+    obj2 = json.loads(p2.to_json())
+    assert obj2["outlinks"] == []
+    assert obj2["comments"] == []
+    assert obj2["reactions"] == {}
+
+
+def test_reactions_sorted_like_go_maps():
+    p = Post(reactions={"👍": 5, "🔥": 2, "a": 1})
+    s = p.to_json()
+    # Go sorts map keys lexicographically (by UTF-8 bytes)
+    ordered = [k for k in json.loads(s)["reactions"]]
+    assert ordered == sorted(["👍", "🔥", "a"])
+
+
+def test_nested_channel_data_shape():
+    cd = ChannelData(
+        channel_id="123", channel_name="Test",
+        channel_engagement_data=EngagementData(follower_count=10, post_count=3),
+        channel_url="https://t.me/c/test",
+    )
+    obj = json.loads(cd.to_json())
+    assert obj["channel_engagement_data"]["follower_count"] == 10
+    assert obj["published_at"] == "0001-01-01T00:00:00Z"
+
+
+def test_performance_scores_null_pattern():
+    ps = PerformanceScores()
+    assert ps.to_json() == '{"likes":null,"shares":null,"comments":null,"views":0}'
+
+
+def test_comment_encoding():
+    c = Comment(text='say "x"', reactions={"👍": 1}, view_count=2, handle="u")
+    obj = json.loads(c.to_json())
+    assert obj["text"] == 'say "x"'
+    assert obj["reactions"] == {"👍": 1}
+
+
+def test_jsonl_line_roundtrip():
+    p = Post(
+        post_link="https://t.me/chan/5",
+        channel_id="42",
+        published_at=dt.datetime(2024, 5, 1, tzinfo=UTC),
+        description="hello <world> & 'stuff'\nnewline",
+        outlinks=["abcde", "fghij"],
+        post_type=["messageText"],
+    )
+    line = p.to_jsonl()
+    assert line.endswith("\n")
+    obj = json.loads(line)
+    assert obj["description"] == "hello <world> & 'stuff'\nnewline"
+    assert "\\u003cworld\\u003e" in line  # raw bytes show Go-style escapes
+    assert obj["outlinks"] == ["abcde", "fghij"]
