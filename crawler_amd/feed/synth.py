@@ -1,0 +1,553 @@
+"""Synthetic Telegram feed engine — the TDLib stand-in.
+
+Generates deterministic, TDLib-shaped message batches directly in the packed
+SoA format (ops/batch.py) that the HIP parse/encode kernels consume. The
+design goal is vectorized generation (numpy, no per-message Python) so a
+10M-post corpus materializes in seconds at bench setup.
+
+Mechanics:
+- channel usernames are FIXED WIDTH: "c" + 10 digits (11 bytes), so link /
+  mention slots inside text templates sit at constant byte+UTF-16 offsets;
+- each message picks one of ~16 byte templates (weighted); a template
+  carries: pool bytes (text + optional url tail for text_url entities),
+  entity rows at constant offsets, digit slots to patch with the target
+  channel number, its content type, and aux strings;
+- all scalar fields derive from splitmix64 hashes of (seed, msg index), so
+  any message is reproducible in isolation (GPU and CPU agree);
+- templates include non-ASCII (Cyrillic, emoji incl. surrogate pairs),
+  JSON-escaping hazards (quotes, backslash, <, >, &, newline, U+2028) and
+  reserved t.me paths, to exercise the UTF-16 entity math and the Go-style
+  escaper end to end.
+
+FLOOD_WAIT / TDLib-400 injection and cache-vs-server latency classes are
+implemented at the client facade level (feed/client.py), not in the batch
+data itself (reference: telegramhelper/rate_limiter.go, crawl/runner.go:55-104).
+"""
+from __future__ import annotations
+
+import dataclasses
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..ops import batch as B
+
+USERNAME_WIDTH = 11  # "c" + 10 digits
+HANDLE_WIDTH = 9     # "u" + 8 digits
+
+
+def _utf16_len(s: str) -> int:
+    return sum(2 if ord(c) >= 0x10000 else 1 for c in s)
+
+
+@dataclasses.dataclass
+class Template:
+    name: str
+    content_type: str
+    text: str                      # with one "{U}" per link slot (11-wide when filled)
+    entity_specs: List[Tuple[str, int]] = dataclasses.field(default_factory=list)
+    # entity_specs: (etype, slot_index) — offsets computed after slot fill
+    url_slots: int = 0             # text_url URL-tail slots ("https://t.me/" + user)
+    aux: str = ""                  # emoji / poll question / prize / doc name
+    has_thumb: bool = False
+    has_video: bool = False
+    weight: float = 1.0
+
+    # computed by _compile:
+    pool: bytes = b""
+    text_len: int = 0
+    slot_positions: List[int] = dataclasses.field(default_factory=list)
+    entities: List[Tuple[int, int, int, int, int]] = dataclasses.field(
+        default_factory=list
+    )
+
+
+# Placeholder username used during compilation; every real username has the
+# same width so offsets survive substitution.
+_PH = "c" + "0" * 10
+
+
+def _compile(t: Template) -> Template:
+    """Fill slots with the placeholder, compute byte/UTF-16 entity offsets."""
+    parts = t.text.split("{U}")
+    n_slots = len(parts) - 1
+    text = _PH.join(parts)
+    tb = text.encode("utf-8")
+
+    # byte position of each slot
+    slot_pos: List[int] = []
+    bpos = 0
+    for k, p in enumerate(parts[:-1]):
+        bpos += len(p.encode("utf-8"))
+        slot_pos.append(bpos + 1)  # +1 skips the 'c' prefix; digits only
+        bpos += len(_PH.encode("utf-8"))
+
+    # utf16 position of each slot start (including the char before, e.g. '@')
+    # entity offsets are computed from entity_specs: each spec names a slot
+    # and a kind; kinds determine how the entity covers the slot.
+    def u16_at(byte_off: int) -> int:
+        return _utf16_len(tb[:byte_off].decode("utf-8"))
+
+    entities = []
+    url_tail = b""
+    url_off_base = len(tb)
+    url_k = 0
+    for (etype, slot) in t.entity_specs:
+        # text_url URLs live in the tail, not at a text slot
+        spos = (slot_pos[slot] - 1) if etype != "text_url" else -1
+        if etype == "mention":
+            # template text must have '@' right before the slot
+            assert tb[spos - 1:spos] == b"@"
+            off16 = u16_at(spos - 1)
+            entities.append((B.ENTITY_TYPE_IDX["mention"], off16,
+                             1 + USERNAME_WIDTH, 0, 0))
+        elif etype == "url":
+            # entity covers "https://t.me/<user>" present in the text
+            prefix = b"https://t.me/"
+            assert tb[spos - len(prefix):spos] == prefix
+            off16 = u16_at(spos - len(prefix))
+            entities.append((B.ENTITY_TYPE_IDX["url"], off16,
+                             len(prefix) + USERNAME_WIDTH, 0, 0))
+        elif etype == "text_url":
+            # URL lives in the tail, not the text; entity covers the anchor
+            # text between «» markers placed in the template around a word.
+            anchor = "click here"
+            a = text.index(anchor)
+            off16 = _utf16_len(text[:a])
+            url = "https://t.me/" + _PH
+            uoff = url_off_base + len(url_tail) + len(b"https://t.me/")
+            slot_pos.append(uoff + 1)  # url slot also patched with digits
+            entities.append((B.ENTITY_TYPE_IDX["text_url"], off16,
+                             _utf16_len(anchor),
+                             url_off_base + len(url_tail), len(url)))
+            url_tail += url.encode("utf-8")
+            url_k += 1
+        else:
+            raise ValueError(etype)
+
+    out = dataclasses.replace(t)
+    out.pool = tb + url_tail
+    out.text_len = len(tb)
+    out.slot_positions = slot_pos
+    out.entities = entities
+    return out
+
+
+def default_templates() -> List[Template]:
+    """The standard synthetic corpus template set."""
+    ts = [
+        Template(
+            name="plain_short", content_type="messageText", weight=3.0,
+            text="Сегодня отличные новости по рынку. Подробности позже.",
+        ),
+        Template(
+            name="plain_long", content_type="messageText", weight=2.0,
+            text=(
+                "Аналитика за неделю: рост продолжился, несмотря на "
+                "волатильность. Ключевые уровни удержались, объём вырос на "
+                "12%. Detailed breakdown & charts attached — see the thread "
+                "below for \"context\" and <notes>.\nStay tuned! 🚀🚀"
+            ),
+        ),
+        Template(
+            name="one_link", content_type="messageText", weight=2.5,
+            text="Подписывайтесь на наш резервный канал: https://t.me/{U} 🔥",
+        ),
+        Template(
+            name="bare_link", content_type="messageText", weight=1.5,
+            text="репост от t.me/{U} — читайте первыми",
+        ),
+        Template(
+            name="mention", content_type="messageText", weight=2.0,
+            text="спасибо @{U} за наводку 🙏 детали в закрепе",
+            entity_specs=[("mention", 0)],
+        ),
+        Template(
+            name="mention_emoji_prefix", content_type="messageText", weight=1.0,
+            text="🤯🤯 срочно! @{U} опубликовал данные",
+            entity_specs=[("mention", 0)],
+        ),
+        Template(
+            name="url_entity", content_type="messageText", weight=1.0,
+            text="источник: https://t.me/{U} (проверено)",
+            entity_specs=[("url", 0)],
+        ),
+        Template(
+            name="text_url", content_type="messageText", weight=1.0,
+            text="Новый разбор — click here чтобы открыть",
+            entity_specs=[("text_url", 0)],
+        ),
+        Template(
+            name="two_links", content_type="messageText", weight=1.0,
+            text="зеркала: t.me/{U} и https://t.me/{U}, сохраняйте",
+        ),
+        Template(
+            name="dup_link", content_type="messageText", weight=0.5,
+            text="канал https://t.me/{U} — да, именно https://t.me/{U}",
+        ),
+        Template(
+            name="reserved_path", content_type="messageText", weight=0.5,
+            text="вход: https://t.me/joinchat/AbCdEf123 или t.me/share/url",
+        ),
+        Template(
+            name="escapes", content_type="messageText", weight=0.5,
+            text="a<b & b>c \\ \"quoted\"\nline2\ttab after-ls",
+        ),
+        Template(
+            name="photo_caption", content_type="messagePhoto", weight=1.5,
+            text="фото дня 📸 подпись с ссылкой t.me/{U}",
+            has_thumb=True,
+        ),
+        Template(
+            name="video_caption", content_type="messageVideo", weight=1.0,
+            text="видео: главное за день",
+            has_thumb=True,
+        ),
+        Template(
+            name="document", content_type="messageDocument", weight=0.5,
+            text="отчёт во вложении", aux="report_2024_final.pdf",
+            has_thumb=True, has_video=True,
+        ),
+        Template(
+            name="sticker", content_type="messageSticker", weight=0.5,
+            text="", has_thumb=True,
+        ),
+        Template(
+            name="animated_emoji", content_type="messageAnimatedEmoji",
+            weight=0.3, text="", aux="🎉",
+        ),
+        Template(
+            name="poll", content_type="messagePoll", weight=0.3,
+            text="", aux="Как вам обновление?",
+        ),
+        Template(
+            name="giveaway", content_type="messageGiveaway", weight=0.2,
+            text="", aux="premium",
+        ),
+    ]
+    return [_compile(t) for t in ts]
+
+
+COMMENT_TEXTS = [
+    "согласен полностью", "интересно 🤔", "first!", "спасибо за инфу",
+    "это уже было вчера", "не согласен, но ок", "топ контент 🔥🔥",
+    "а пруфы будут?",
+]
+
+
+def _splitmix64(x: np.ndarray) -> np.ndarray:
+    """Vectorized splitmix64 over uint64 arrays."""
+    z = (x + np.uint64(0x9E3779B97F4A7C15)).astype(np.uint64)
+    z = (z ^ (z >> np.uint64(30))) * np.uint64(0xBF58476D1CE4E5B9)
+    z = (z ^ (z >> np.uint64(27))) * np.uint64(0x94D049BB133111EB)
+    return z ^ (z >> np.uint64(31))
+
+
+@dataclasses.dataclass
+class FeedConfig:
+    seed: int = 1234
+    universe: int = 1_000_000       # channel-id namespace for outlink targets
+    posts_per_channel: int = 10_000
+    base_date: int = 1_700_000_000  # unix seconds of post 0
+    date_step: int = 60             # seconds between posts in a channel
+    comment_rate: float = 0.02      # fraction of posts with comments
+    max_comments_per_post: int = 3
+
+
+class SyntheticFeed:
+    """Deterministic channel/history generator in packed-batch form."""
+
+    def __init__(self, cfg: Optional[FeedConfig] = None):
+        self.cfg = cfg or FeedConfig()
+        self.templates = default_templates()
+        w = np.array([t.weight for t in self.templates])
+        self._cdf = np.cumsum(w) / w.sum()
+        # Precompute template arrays
+        self._pool_bytes = [np.frombuffer(t.pool, dtype=np.uint8)
+                            for t in self.templates]
+        self._pool_len = np.array([len(t.pool) for t in self.templates])
+        self._text_len = np.array([t.text_len for t in self.templates])
+        self._ctype = np.array(
+            [B.CONTENT_TYPE_IDX[t.content_type] for t in self.templates]
+        )
+        self._flags = np.array(
+            [(B.FLAG_HAS_THUMB if t.has_thumb else 0)
+             | (B.FLAG_HAS_VIDEO if t.has_video else 0)
+             for t in self.templates]
+        )
+        self._aux = [t.aux.encode("utf-8") for t in self.templates]
+        self._ents = [np.array(t.entities, dtype=np.int32).reshape(-1, 5)
+                      for t in self.templates]
+        self._n_slots = np.array([len(t.slot_positions)
+                                  for t in self.templates])
+        self._comment_bytes = [c.encode("utf-8") for c in COMMENT_TEXTS]
+
+    # ---- channel metadata ----
+
+    @staticmethod
+    def username_of(cid: int) -> str:
+        return "c%010d" % cid
+
+    def chat_id_of(self, cid: int) -> int:
+        return -1001000000000 - cid
+
+    def channel_rows(self, cids: np.ndarray,
+                     posts_per_channel: Optional[int] = None) -> List[B.ChannelRow]:
+        h = _splitmix64(np.uint64(self.cfg.seed) ^ (cids.astype(np.uint64)
+                                                    * np.uint64(0x51ED)))
+        rows = []
+        for k, cid in enumerate(cids):
+            rows.append(B.ChannelRow(
+                chat_id=self.chat_id_of(int(cid)),
+                username=self.username_of(int(cid)),
+                title="Synthetic Channel %d" % int(cid),
+                member_count=int(h[k] % np.uint64(500_000)) + 100,
+                post_count=posts_per_channel or self.cfg.posts_per_channel,
+                total_views=int(h[k] % np.uint64(10_000_000)),
+            ))
+        return rows
+
+    # ---- batched history generation (the hot input path) ----
+
+    def build_batch(
+        self, channel_ids: np.ndarray, posts_per_channel: Optional[int] = None
+    ) -> B.MessageBatch:
+        """Generate all messages for `channel_ids` as one packed batch.
+
+        Layout: messages grouped by channel, newest-last; channel_idx points
+        into the batch channel table (ordered as channel_ids).
+        """
+        cfg = self.cfg
+        P = posts_per_channel or cfg.posts_per_channel
+        K = len(channel_ids)
+        N = K * P
+        seed = np.uint64(cfg.seed)
+
+        cid = np.repeat(channel_ids.astype(np.int64), P)
+        pidx = np.tile(np.arange(P, dtype=np.int64), K)
+        gidx = (cid.astype(np.uint64) * np.uint64(1_000_003)
+                + pidx.astype(np.uint64))
+        h0 = _splitmix64(seed ^ gidx)
+        h1 = _splitmix64(h0)
+        h2 = _splitmix64(h1)
+        h3 = _splitmix64(h2)
+
+        # template choice
+        u = (h0 >> np.uint64(11)).astype(np.float64) / float(1 << 53)
+        tidx = np.searchsorted(self._cdf, u, side="right").astype(np.int32)
+        tidx = np.minimum(tidx, len(self.templates) - 1)
+
+        # fixed-size meta
+        meta = {f: np.zeros(N, dtype=np.int32) for f in B._META_FIELDS_I32}
+        meta["content_type"] = self._ctype[tidx].astype(np.int32)
+        meta["flags"] = self._flags[tidx].astype(np.int32)
+        meta["views"] = (h1 % np.uint64(100_000)).astype(np.int32)
+        meta["forwards"] = ((h1 >> np.uint64(17)) % np.uint64(1000)).astype(np.int32)
+        meta["media_album_id"] = np.where(
+            (h2 % np.uint64(16)) == 0, (h2 % np.uint64(1 << 31)).astype(np.int64), 0
+        ).astype(np.int32)
+        meta["channel_idx"] = np.repeat(
+            np.arange(K, dtype=np.int32), P
+        )
+        meta["date"] = (
+            cfg.base_date + pidx * cfg.date_step + (h2 % np.uint64(30)).astype(np.int64)
+        ).astype(np.int32)
+
+        msg_id = ((pidx + 1) << 20).astype(np.int64)
+        chat_id = -1001000000000 - cid
+
+        # ---- pool assembly ----
+        # per-message pool block = template pool + aux + handle(9B)
+        aux_len = np.array([len(a) for a in self._aux])[tidx]
+        block_len = self._pool_len[tidx] + aux_len + HANDLE_WIDTH
+        block_off = np.zeros(N + 1, dtype=np.int64)
+        np.cumsum(block_len, out=block_off[1:])
+        total = int(block_off[-1])
+        pool = np.zeros(total, dtype=np.uint8)
+
+        if total >= (1 << 31):
+            raise ValueError("pool exceeds int32 offsets; shrink the batch")
+        text_off = block_off[:-1].copy()
+        meta["text_len"] = self._text_len[tidx].astype(np.int32)
+        meta["aux_off"] = (block_off[:-1] + self._pool_len[tidx]).astype(np.int32)
+        meta["aux_len"] = aux_len.astype(np.int32)
+        meta["poster_off"] = (block_off[:-1] + self._pool_len[tidx] + aux_len).astype(np.int32)
+        meta["poster_len"] = np.full(N, HANDLE_WIDTH, dtype=np.int32)
+
+        # write template+aux bytes per template id (vectorized per template)
+        for ti, tmpl in enumerate(self.templates):
+            rows = np.nonzero(tidx == ti)[0]
+            if len(rows) == 0:
+                continue
+            tb = self._pool_bytes[ti]
+            if len(tb):
+                idx = block_off[rows][:, None] + np.arange(len(tb))[None, :]
+                pool[idx] = tb[None, :]
+            ab = np.frombuffer(self._aux[ti], dtype=np.uint8)
+            if len(ab):
+                idx = (block_off[rows] + len(tb))[:, None] + np.arange(len(ab))[None, :]
+                pool[idx] = ab[None, :]
+            # handle: "u" + 8 digits of (h3 % 1e8)
+            hoff = block_off[rows] + len(tb) + len(ab)
+            pool[hoff] = ord("u")
+            hv = (h3[rows] % np.uint64(100_000_000)).astype(np.int64)
+            for d in range(8):
+                digit = (hv // (10 ** (7 - d))) % 10
+                pool[hoff + 1 + d] = (48 + digit).astype(np.uint8)
+            # patch link slots with target digits
+            for s, spos in enumerate(tmpl.slot_positions):
+                tgt = (_splitmix64(h0[rows] + np.uint64(7919 * (s + 1)))
+                       % np.uint64(self.cfg.universe)).astype(np.int64)
+                for d in range(10):
+                    digit = (tgt // (10 ** (9 - d))) % 10
+                    pool[block_off[rows] + spos + d] = (48 + digit).astype(np.uint8)
+
+        # ---- entities ----
+        ent_cnt = np.array([len(e) for e in self._ents])[tidx]
+        ent_off = np.zeros(N + 1, dtype=np.int64)
+        np.cumsum(ent_cnt, out=ent_off[1:])
+        E = int(ent_off[-1])
+        entities = np.zeros((E, 5), dtype=np.int32)
+        for ti, tmpl in enumerate(self.templates):
+            te = self._ents[ti]
+            if len(te) == 0:
+                continue
+            rows = np.nonzero(tidx == ti)[0]
+            if len(rows) == 0:
+                continue
+            dst = (ent_off[rows][:, None] + np.arange(len(te))[None, :]).ravel()
+            tiled = np.tile(te, (len(rows), 1))
+            # url offsets are relative to the message block start
+            tiled_abs = tiled.copy()
+            has_url = tiled[:, 4] > 0
+            tiled_abs[:, 3] = tiled[:, 3] + np.where(
+                has_url, np.repeat(block_off[rows], len(te)), 0
+            ).astype(np.int64)
+            entities[dst] = tiled_abs
+        meta["ent_off"] = ent_off[:-1].astype(np.int32)
+        meta["ent_cnt"] = ent_cnt.astype(np.int32)
+
+        # ---- reactions ----
+        nem = len(B.EMOJI_TABLE)
+        rmask = (h2 & h2 >> np.uint64(13) & np.uint64((1 << nem) - 1)).astype(
+            np.int64
+        )
+        bits = ((rmask[:, None] >> np.arange(nem)[None, :]) & 1).astype(bool)
+        react_cnt = bits.sum(axis=1).astype(np.int32)
+        react_off = np.zeros(N + 1, dtype=np.int64)
+        np.cumsum(react_cnt, out=react_off[1:])
+        rmsg, remoji = np.nonzero(bits)  # ordered by (msg, emoji idx asc)
+        react_emoji = remoji.astype(np.int32)
+        rh = _splitmix64(h2[rmsg] + remoji.astype(np.uint64))
+        react_count = ((rh % np.uint64(500)) + np.uint64(1)).astype(np.int32)
+        meta["react_off"] = react_off[:-1].astype(np.int32)
+        meta["react_cnt"] = react_cnt
+
+        # ---- comments (packed in a secondary pool appended to pool) ----
+        has_c = (h3 % np.uint64(1000)).astype(np.float64) < (
+            self.cfg.comment_rate * 1000
+        )
+        ccnt = np.where(
+            has_c, (h3 >> np.uint64(32)) % np.uint64(
+                self.cfg.max_comments_per_post
+            ) + np.uint64(1), 0
+        ).astype(np.int32)
+        meta["reply_count"] = ccnt
+        com_off = np.zeros(N + 1, dtype=np.int64)
+        np.cumsum(ccnt, out=com_off[1:])
+        C = int(com_off[-1])
+        meta["com_off"] = com_off[:-1].astype(np.int32)
+        meta["com_cnt"] = ccnt
+
+        cmsg = np.repeat(np.arange(N), ccnt)
+        cslot = (np.arange(C) - com_off[cmsg]).astype(np.int64)
+        ch = _splitmix64(h3[cmsg] + cslot.astype(np.uint64) * np.uint64(104729))
+        ctext_idx = (ch % np.uint64(len(COMMENT_TEXTS))).astype(np.int64)
+        clens = np.array([len(b) for b in self._comment_bytes])
+        ctext_len = clens[ctext_idx]
+        # comment pool: text + handle per comment, appended after message pool
+        cblock_len = ctext_len + HANDLE_WIDTH
+        cblock_off = np.zeros(C + 1, dtype=np.int64)
+        np.cumsum(cblock_len, out=cblock_off[1:])
+        cpool = np.zeros(int(cblock_off[-1]), dtype=np.uint8)
+        for k, cb in enumerate(self._comment_bytes):
+            rows = np.nonzero(ctext_idx == k)[0]
+            if len(rows) == 0:
+                continue
+            arr = np.frombuffer(cb, dtype=np.uint8)
+            idx = cblock_off[rows][:, None] + np.arange(len(arr))[None, :]
+            cpool[idx] = arr[None, :]
+        hoff = cblock_off[:-1] + ctext_len
+        if C:
+            cpool[hoff] = ord("u")
+            hv = (ch >> np.uint64(8)) % np.uint64(100_000_000)
+            hv = hv.astype(np.int64)
+            for d in range(8):
+                digit = (hv // (10 ** (7 - d))) % 10
+                cpool[hoff + 1 + d] = (48 + digit).astype(np.uint8)
+        com_text_off = (cblock_off[:-1] + total).astype(np.int32)
+        com_handle_off = (hoff + total).astype(np.int32)
+        if total + int(cblock_off[-1]) >= (1 << 31):
+            raise ValueError("pool exceeds int32 offsets; shrink the batch")
+
+        # comment reactions: none (empty maps) — matches typical thread data
+        com_react_off = np.zeros(C, dtype=np.int32)
+        com_react_cnt = np.zeros(C, dtype=np.int32)
+        com_views = ((ch >> np.uint64(16)) % np.uint64(10_000)).astype(np.int32)
+        com_replies = ((ch >> np.uint64(24)) % np.uint64(50)).astype(np.int32)
+
+        full_pool = np.concatenate([pool, cpool]) if C else pool
+
+        # ---- channel table ----
+        rows = self.channel_rows(channel_ids, P)
+        kpool_parts = []
+        ch_user_off = np.zeros(K, dtype=np.int32)
+        ch_user_len = np.zeros(K, dtype=np.int32)
+        ch_title_off = np.zeros(K, dtype=np.int32)
+        ch_title_len = np.zeros(K, dtype=np.int32)
+        koff = len(full_pool)
+        for c, row in enumerate(rows):
+            ub = row.username.encode()
+            tb2 = row.title.encode()
+            ch_user_off[c] = koff
+            ch_user_len[c] = len(ub)
+            kpool_parts.append(ub)
+            koff += len(ub)
+            ch_title_off[c] = koff
+            ch_title_len[c] = len(tb2)
+            kpool_parts.append(tb2)
+            koff += len(tb2)
+        full_pool = np.concatenate(
+            [full_pool, np.frombuffer(b"".join(kpool_parts), dtype=np.uint8)]
+        )
+
+        t = torch.from_numpy
+        return B.MessageBatch(
+            n=N,
+            chat_id=t(chat_id),
+            msg_id=t(msg_id),
+            text_off=t(text_off),
+            meta={f: t(np.ascontiguousarray(v)) for f, v in meta.items()},
+            text_pool=t(full_pool),
+            entities=t(entities),
+            react_emoji=t(react_emoji),
+            react_count=t(react_count),
+            com_text_off=t(com_text_off),
+            com_text_len=t(ctext_len.astype(np.int32)),
+            com_handle_off=t(com_handle_off),
+            com_handle_len=t(np.full(C, HANDLE_WIDTH, dtype=np.int32)),
+            com_views=t(com_views),
+            com_replies=t(com_replies),
+            com_react_off=t(com_react_off),
+            com_react_cnt=t(com_react_cnt),
+            n_channels=K,
+            ch_chat_id=t(np.array([r.chat_id for r in rows], dtype=np.int64)),
+            ch_member=t(np.array([r.member_count for r in rows], dtype=np.int32)),
+            ch_postcount=t(np.array([r.post_count for r in rows], dtype=np.int32)),
+            ch_totalviews=t(np.array([r.total_views for r in rows], dtype=np.int32)),
+            ch_user_off=t(ch_user_off),
+            ch_user_len=t(ch_user_len),
+            ch_title_off=t(ch_title_off),
+            ch_title_len=t(ch_title_len),
+        )

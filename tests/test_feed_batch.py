@@ -1,0 +1,140 @@
+"""Synthetic feed + packed batch tests: determinism, unpack/golden round trip."""
+import datetime as dt
+import json
+
+import numpy as np
+import pytest
+
+from crawler_amd.feed import FeedConfig, SyntheticFeed
+from crawler_amd.ops import batch as B
+from crawler_amd.ops import golden as G
+from crawler_amd.ops.golden_batch import encode_batch
+
+UTC = dt.timezone.utc
+NOW = dt.datetime(2026, 1, 1, tzinfo=UTC)
+
+
+@pytest.fixture(scope="module")
+def feed():
+    return SyntheticFeed(FeedConfig(seed=42, universe=1000))
+
+
+@pytest.fixture(scope="module")
+def small_batch(feed):
+    return feed.build_batch(np.array([3, 7, 12]), posts_per_channel=50)
+
+
+def test_batch_shapes(small_batch):
+    b = small_batch
+    assert b.n == 150
+    assert b.n_channels == 3
+    assert b.chat_id.shape[0] == 150
+    assert b.text_pool.dtype.__str__() == "torch.uint8"
+
+
+def test_determinism(feed):
+    b1 = feed.build_batch(np.array([3]), posts_per_channel=20)
+    b2 = feed.build_batch(np.array([3]), posts_per_channel=20)
+    assert (b1.text_pool == b2.text_pool).all()
+    assert (b1.msg_id == b2.msg_id).all()
+    for k in b1.meta:
+        assert (b1.meta[k] == b2.meta[k]).all(), k
+
+
+def test_usernames_fixed_width(feed):
+    assert feed.username_of(5) == "c0000000005"
+    assert len(feed.username_of(999999)) == 11
+
+
+def test_unpack_message_valid_utf8(small_batch):
+    for i in range(0, 150, 7):
+        m = B.unpack_message(small_batch, i)
+        if m.text:
+            m.text.text.encode("utf-8")
+        assert m.content_type in B.CONTENT_TYPES
+
+
+def test_extracted_links_are_valid_usernames(small_batch):
+    found = 0
+    for i in range(150):
+        m = B.unpack_message(small_batch, i)
+        for link in G.extract_links_with_source(m):
+            found += 1
+            assert link.name.startswith("c")
+            assert len(link.name) == 11
+            v, _ = G.filter_username(link.name)
+            assert v
+    assert found > 10  # the template mix must produce links
+
+
+def test_mention_entities_resolve(small_batch):
+    saw_mention = False
+    for i in range(150):
+        m = B.unpack_message(small_batch, i)
+        for link in G.extract_links_with_source(m):
+            if link.source_type == "mention":
+                saw_mention = True
+    assert saw_mention
+
+
+def test_text_url_entities_resolve(feed):
+    b = feed.build_batch(np.arange(20), posts_per_channel=40)
+    srcs = set()
+    for i in range(b.n):
+        m = B.unpack_message(b, i)
+        for link in G.extract_links_with_source(m):
+            srcs.add(link.source_type)
+    assert "text_url" in srcs
+    assert "plaintext" in srcs
+
+
+def test_reactions_sorted_by_emoji_table(small_batch):
+    m = small_batch.meta
+    for i in range(150):
+        off, cnt = int(m["react_off"][i]), int(m["react_cnt"][i])
+        idxs = [int(small_batch.react_emoji[r]) for r in range(off, off + cnt)]
+        assert idxs == sorted(idxs)
+
+
+def test_encode_batch_produces_valid_go_jsonl(small_batch):
+    lines, links = encode_batch(small_batch, now=NOW)
+    assert len(lines) == 150
+    n_links = 0
+    for i, line in enumerate(lines):
+        obj = json.loads(line)
+        assert obj["platform_name"] == "Telegram"
+        assert obj["post_uid"].endswith(
+            "-c%010d" % [3, 7, 12][int(small_batch.meta["channel_idx"][i])]
+        )
+        assert obj["post_link"].startswith("https://t.me/c")
+        assert obj["channel_data"]["channel_engagement_data"]["post_count"] == 50
+        n_links += len(links[i])
+    assert n_links > 10
+
+
+def test_comments_present_and_encoded(feed):
+    b = feed.build_batch(np.arange(10), posts_per_channel=200)
+    lines, _ = encode_batch(b, now=NOW)
+    with_comments = [
+        json.loads(l) for l in lines if json.loads(l)["comments_count"] > 0
+    ]
+    assert with_comments, "comment_rate should yield some commented posts"
+    c = with_comments[0]["comments"][0]
+    assert set(c) == {"text", "reactions", "view_count", "reply_count", "handle"}
+
+
+def test_min_post_date_filter(small_batch):
+    lines, _ = encode_batch(
+        small_batch, now=NOW,
+        min_post_date=dt.datetime(2100, 1, 1, tzinfo=UTC),
+    )
+    assert all(l == b"" for l in lines)
+
+
+def test_media_flags_skip_media(small_batch):
+    lines, _ = encode_batch(small_batch, now=NOW, skip_media=True)
+    for line in lines:
+        obj = json.loads(line)
+        assert obj["thumb_url"] == ""  # skip_media drops thumbs
+        if "messageDocument" in obj["post_type"]:
+            assert obj["media_url"].startswith("AgAD")
