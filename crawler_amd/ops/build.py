@@ -1,0 +1,40 @@
+"""In-tree build of the CDNA4 HIP extension (gfx950 only, no fallbacks).
+
+Usage: python -m crawler_amd.ops.build
+Called by __graft_entry__.build() — the driver's build check. hipcc
+cross-compiles on CPU-only hosts; the .so travels to GPU boxes in-tree.
+"""
+from __future__ import annotations
+
+import os
+import subprocess
+import sys
+
+CSRC = os.path.dirname(os.path.abspath(__file__)) + "/csrc"
+SOURCES = ["parse_encode.hip"]
+OUT = os.path.join(CSRC, "libcrawlhip.so")
+
+
+def needs_build() -> bool:
+    if not os.path.exists(OUT):
+        return True
+    out_m = os.path.getmtime(OUT)
+    deps = [os.path.join(CSRC, s) for s in SOURCES]
+    deps += [os.path.join(CSRC, "common.h")]
+    return any(os.path.getmtime(d) > out_m for d in deps)
+
+
+def build(force: bool = False) -> str:
+    if not force and not needs_build():
+        return OUT
+    cmd = [
+        "hipcc", "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
+        "-shared",
+    ] + [os.path.join(CSRC, s) for s in SOURCES] + ["-o", OUT]
+    subprocess.run(cmd, check=True)
+    return OUT
+
+
+if __name__ == "__main__":
+    path = build(force="--force" in sys.argv)
+    print(path)

@@ -1,0 +1,239 @@
+// Device helpers for the CDNA4 (gfx950) crawl hot-path kernels.
+//
+// Wave-cooperative primitives over 64-lane wavefronts: striped memcpy,
+// Go-compatible JSON string escaping, integer/date formatting, UTF-16
+// offset resolution, and the t.me link scanner. Every routine here has a
+// pure-Python oracle in crawler_amd/ops/golden.py (reference semantics
+// cited there against telegramhelper/tdutils.go).
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#define WAVE 64
+#define DEV __device__ __forceinline__
+
+namespace crawl {
+
+DEV int lane_id() { return threadIdx.x & (WAVE - 1); }
+DEV int wave_id() { return threadIdx.x >> 6; }
+
+// ---------- byte classification ----------
+
+DEV bool is_ascii_letter(uint8_t c) {
+  return (c >= 'a' && c <= 'z') || (c >= 'A' && c <= 'Z');
+}
+DEV bool is_word_char(uint8_t c) {  // [a-zA-Z0-9_]
+  return is_ascii_letter(c) || (c >= '0' && c <= '9') || c == '_';
+}
+DEV uint8_t to_lower(uint8_t c) {
+  return (c >= 'A' && c <= 'Z') ? (c + 32) : c;
+}
+
+// UTF-16 code units contributed by the byte at p (0 for continuation bytes).
+DEV int u16_units_of_byte(uint8_t b) {
+  if (b < 0x80) return 1;          // ascii
+  if (b < 0xC0) return 0;          // continuation
+  if (b < 0xF0) return 1;          // 2- or 3-byte leader (BMP)
+  return 2;                        // 4-byte leader -> surrogate pair
+}
+
+// ---------- Go encoding/json escape model ----------
+// Output bytes for input byte s[p] (with lookahead for U+2028/U+2029).
+// 1 = verbatim; 2 = two-char escape; 6 = \u00xx / \u202x; 0 = swallowed
+// (continuation bytes of an escaped U+2028/29 sequence).
+DEV int escape_len_at(const uint8_t* s, int n, int p) {
+  uint8_t c = s[p];
+  if (c == '"' || c == '\\' || c == '\n' || c == '\r' || c == '\t') return 2;
+  if (c < 0x20) return 6;
+  if (c == '<' || c == '>' || c == '&') return 6;
+  if (c == 0xE2 && p + 2 < n && s[p + 1] == 0x80 &&
+      (s[p + 2] == 0xA8 || s[p + 2] == 0xA9))
+    return 6;
+  if (c == 0x80 && p >= 1 && s[p - 1] == 0xE2 && p + 1 < n &&
+      (s[p + 1] == 0xA8 || s[p + 1] == 0xA9))
+    return 0;
+  if ((c == 0xA8 || c == 0xA9) && p >= 2 && s[p - 2] == 0xE2 &&
+      s[p - 1] == 0x80)
+    return 0;
+  return 1;
+}
+
+DEV const char* HEXD() { return "0123456789abcdef"; }
+
+// Write the escape expansion of s[p] to out; returns bytes written.
+DEV int escape_write_at(const uint8_t* s, int n, int p, uint8_t* out) {
+  uint8_t c = s[p];
+  switch (c) {
+    case '"':  out[0] = '\\'; out[1] = '"';  return 2;
+    case '\\': out[0] = '\\'; out[1] = '\\'; return 2;
+    case '\n': out[0] = '\\'; out[1] = 'n';  return 2;
+    case '\r': out[0] = '\\'; out[1] = 'r';  return 2;
+    case '\t': out[0] = '\\'; out[1] = 't';  return 2;
+    default: break;
+  }
+  if (c < 0x20 || c == '<' || c == '>' || c == '&') {
+    out[0] = '\\'; out[1] = 'u'; out[2] = '0'; out[3] = '0';
+    out[4] = HEXD()[c >> 4]; out[5] = HEXD()[c & 15];
+    return 6;
+  }
+  if (c == 0xE2 && p + 2 < n && s[p + 1] == 0x80 &&
+      (s[p + 2] == 0xA8 || s[p + 2] == 0xA9)) {
+    out[0] = '\\'; out[1] = 'u'; out[2] = '2'; out[3] = '0'; out[4] = '2';
+    out[5] = (s[p + 2] == 0xA8) ? '8' : '9';
+    return 6;
+  }
+  if (escape_len_at(s, n, p) == 0) return 0;
+  out[0] = c;
+  return 1;
+}
+
+// ---------- wave primitives ----------
+
+DEV int wave_prefix_excl(int v) {
+  // exclusive prefix sum across the 64-lane wave
+  int lane = lane_id();
+  for (int d = 1; d < WAVE; d <<= 1) {
+    int up = __shfl_up(v, d);
+    if (lane >= d) v += up;
+  }
+  return __shfl_up(v, 1) * (lane > 0 ? 1 : 0);
+}
+
+DEV int wave_sum(int v) {
+  for (int d = WAVE / 2; d > 0; d >>= 1) v += __shfl_down(v, d);
+  return __shfl(v, 0);
+}
+
+// ---------- cursor-based cooperative writer ----------
+// One wave writes one JSON line. The cursor is wave-uniform; every helper
+// leaves all 64 lanes with the same cursor value.
+
+struct Writer {
+  uint8_t* out;   // line base
+  int cur;        // wave-uniform write offset
+
+  DEV void raw_lane0(const char* s, int n) {
+    if (lane_id() == 0)
+      for (int j = 0; j < n; ++j) out[cur + j] = (uint8_t)s[j];
+    cur += n;
+  }
+  DEV void bytes(const uint8_t* s, int n) {  // striped copy, no escaping
+    for (int j = lane_id(); j < n; j += WAVE) out[cur + j] = s[j];
+    cur += n;
+  }
+  DEV void lit(const char* s) {  // compile-time-ish literal (strlen on lane)
+    int n = 0;
+    while (s[n]) ++n;
+    bytes((const uint8_t*)s, n);
+  }
+
+  // Escaped JSON string content (no quotes). Wave-parallel: per-64-byte
+  // stripe, each lane computes its byte's expansion length, wave-scans for
+  // offsets, writes its expansion.
+  DEV void escaped(const uint8_t* s, int n) {
+    int lane = lane_id();
+    int base = cur;
+    for (int start = 0; start < n; start += WAVE) {
+      int p = start + lane;
+      int el = (p < n) ? escape_len_at(s, n, p) : 0;
+      int off = wave_prefix_excl(el);
+      if (p < n && el > 0) {
+        uint8_t tmp[6];
+        int w = escape_write_at(s, n, p, tmp);
+        for (int j = 0; j < w; ++j) out[base + off + j] = tmp[j];
+      }
+      int total = wave_sum(el);
+      base += total;
+    }
+    cur = base;
+  }
+
+  // Escaped length without writing (same loop shape).
+  DEV static int escaped_len(const uint8_t* s, int n) {
+    int lane = lane_id();
+    int total = 0;
+    for (int start = 0; start < n; start += WAVE) {
+      int p = start + lane;
+      int el = (p < n) ? escape_len_at(s, n, p) : 0;
+      total += wave_sum(el);
+    }
+    return total;
+  }
+
+  DEV void quoted(const uint8_t* s, int n) {
+    lit("\"");
+    escaped(s, n);
+    lit("\"");
+  }
+
+  // unsigned/signed integer -> decimal (lane 0 writes; cursor uniform)
+  DEV void u64_dec(uint64_t v) {
+    char buf[20];
+    int n = 0;
+    do { buf[n++] = '0' + (v % 10); v /= 10; } while (v);
+    if (lane_id() == 0)
+      for (int j = 0; j < n; ++j) out[cur + j] = buf[n - 1 - j];
+    cur += n;
+  }
+  DEV void i64_dec(int64_t v) {
+    if (v < 0) { lit("-"); u64_dec((uint64_t)(-v)); }
+    else u64_dec((uint64_t)v);
+  }
+
+  // RFC3339 UTC "YYYY-MM-DDTHH:MM:SSZ" from unix seconds (civil calendar).
+  DEV void rfc3339(int64_t secs) {
+    int64_t days = secs / 86400;
+    int64_t rem = secs % 86400;
+    if (rem < 0) { rem += 86400; days -= 1; }
+    int hh = (int)(rem / 3600), mm = (int)((rem % 3600) / 60),
+        ss = (int)(rem % 60);
+    // civil_from_days (Howard Hinnant's algorithm)
+    int64_t z = days + 719468;
+    int64_t era = (z >= 0 ? z : z - 146096) / 146097;
+    int64_t doe = z - era * 146097;
+    int64_t yoe = (doe - doe / 1460 + doe / 36524 - doe / 146096) / 365;
+    int64_t y = yoe + era * 400;
+    int64_t doy = doe - (365 * yoe + yoe / 4 - yoe / 100);
+    int64_t mp = (5 * doy + 2) / 153;
+    int64_t d = doy - (153 * mp + 2) / 5 + 1;
+    int64_t m = mp + (mp < 10 ? 3 : -9);
+    y += (m <= 2);
+    char buf[20];
+    int yy = (int)y;
+    buf[0] = '0' + (yy / 1000) % 10; buf[1] = '0' + (yy / 100) % 10;
+    buf[2] = '0' + (yy / 10) % 10;   buf[3] = '0' + yy % 10;
+    buf[4] = '-';
+    buf[5] = '0' + ((int)m) / 10; buf[6] = '0' + ((int)m) % 10;
+    buf[7] = '-';
+    buf[8] = '0' + ((int)d) / 10; buf[9] = '0' + ((int)d) % 10;
+    buf[10] = 'T';
+    buf[11] = '0' + hh / 10; buf[12] = '0' + hh % 10; buf[13] = ':';
+    buf[14] = '0' + mm / 10; buf[15] = '0' + mm % 10; buf[16] = ':';
+    buf[17] = '0' + ss / 10; buf[18] = '0' + ss % 10; buf[19] = 'Z';
+    if (lane_id() == 0)
+      for (int j = 0; j < 20; ++j) out[cur + j] = buf[j];
+    cur += 20;
+  }
+};
+
+DEV int u64_dec_len(uint64_t v) {
+  int n = 0;
+  do { ++n; v /= 10; } while (v);
+  return n;
+}
+DEV int i64_dec_len(int64_t v) {
+  return (v < 0) ? 1 + u64_dec_len((uint64_t)(-v)) : u64_dec_len((uint64_t)v);
+}
+
+// ---------- hashing (FNV-1a 64, matches python oracle) ----------
+DEV uint64_t fnv1a64(const uint8_t* s, int n) {
+  uint64_t h = 1469598103934665603ULL;
+  for (int i = 0; i < n; ++i) {
+    h ^= s[i];
+    h *= 1099511628211ULL;
+  }
+  return h;
+}
+
+}  // namespace crawl

@@ -1,0 +1,654 @@
+// CDNA4 (gfx950) parse + JSONL-encode kernels — the per-post hot path.
+//
+// MI355X-native replacement for the reference's per-message hot loop:
+//   - telegramhelper.ParseMessage content switch + field extraction
+//     (tdutils.go:380-732)
+//   - entity walk + UTF-16 offset conversion + t.me regex + username rules
+//     (tdutils.go:55-78, 897-1002; username_filter.go:26-81)
+//   - Go-compatible JSONL emission of model.Post (model/data.go:9-75,
+//     storageproviders.go:275-298)
+//
+// Geometry: 256-thread blocks = 4 waves; ONE WAVE PER MESSAGE, grid-stride.
+// Each message's JSON line is emitted wave-cooperatively (striped copies,
+// wave-scanned escape expansion); scalar fields are formatted by lane 0
+// with a wave-uniform cursor. Two passes share ONE templated emitter:
+//   pass A (WRITE=false): exact line length + link extraction into scratch
+//   pass B (WRITE=true):  byte emission at exclusive-scanned offsets
+// so measure and write can never drift. The oracle is
+// crawler_amd/ops/golden_batch.py (byte-identical output required).
+
+#include "common.h"
+
+namespace crawl {
+
+#define MAX_LINKS 8
+#define LINK_NAME_BYTES 32
+
+struct BatchView {
+  // per-message
+  const long* chat_id;
+  const long* msg_id;
+  const long* text_off;
+  const int* date;
+  const int* content_type;
+  const int* views;
+  const int* forwards;
+  const int* media_album_id;
+  const int* channel_idx;
+  const int* flags;
+  const int* text_len;
+  const int* aux_off;
+  const int* aux_len;
+  const int* ent_off;
+  const int* ent_cnt;
+  const int* react_off;
+  const int* react_cnt;
+  const int* com_off;
+  const int* com_cnt;
+  const int* poster_off;
+  const int* poster_len;
+  // pools / tables
+  const unsigned char* pool;
+  const int* entities;  // [E,5] etype,off16,len16,url_off,url_len
+  const int* react_emoji;
+  const int* react_count;
+  const int* com_text_off;
+  const int* com_text_len;
+  const int* com_handle_off;
+  const int* com_handle_len;
+  const int* com_views;
+  const int* com_replies;
+  const int* com_react_off;
+  const int* com_react_cnt;
+  // channel table
+  const long* ch_chat_id;
+  const int* ch_member;
+  const int* ch_postcount;
+  const int* ch_totalviews;
+  const int* ch_user_off;
+  const int* ch_user_len;
+  const int* ch_title_off;
+  const int* ch_title_len;
+  // emoji vocabulary
+  const unsigned char* emoji_pool;
+  const int* emoji_off;
+  const int* emoji_len;
+  // timestamps (preformatted Go time strings)
+  const unsigned char* created_str;
+  int created_len;
+  const unsigned char* capture_str;
+  int capture_len;
+  // config
+  int n;
+  int skip_media;
+  long min_post_date;  // unix secs; <= filters nothing when INT64_MIN
+  // content-type name table
+  const unsigned char* ctname_pool;
+  const int* ctname_off;
+  const int* ctname_len;
+};
+
+struct LinkOut {
+  unsigned char* name;     // [N, MAX_LINKS, 32]
+  unsigned char* name_len; // [N, MAX_LINKS]
+  unsigned char* src;      // [N, MAX_LINKS] 0=mention 1=text_url 2=url 3=plain
+  int* cnt;                // [N]
+  unsigned long long* hash;// [N, MAX_LINKS] fnv1a64 of name
+};
+
+// ---------- UTF-16 offset resolution (oracle: golden.utf16_offset_to_bytes)
+
+DEV void utf16_to_bytes(const unsigned char* s, int n, int off16, int len16,
+                        int* out_start, int* out_end) {
+  int lane = lane_id();
+  int run = 0;
+  int start = -1, end = -1;
+  const int t1 = off16, t2 = off16 + len16;
+  for (int base = 0; base < n; base += WAVE) {
+    int p = base + lane;
+    int u = (p < n) ? u16_units_of_byte(s[p]) : 0;
+    int pre = run + wave_prefix_excl(u);
+    bool boundary = (p < n) && u > 0;
+    unsigned long long m1 = __ballot(boundary && pre == t1);
+    unsigned long long m2 = __ballot(boundary && pre == t2);
+    if (start < 0 && m1) start = base + __ffsll(m1) - 1;
+    if (end < 0 && m2) end = base + __ffsll(m2) - 1;
+    run += wave_sum(u);
+    if (end >= 0) break;
+  }
+  if (start < 0) { *out_start = 0; *out_end = 0; return; }
+  *out_start = start;
+  *out_end = (end >= 0) ? end : n;
+}
+
+// ---------- t.me link scanning (oracle: golden.CHANNEL_LINK_RE walk) ------
+
+// Reserved t.me path segments >= 5 chars (shorter ones can't match the
+// username group). tdutils.go:27-32.
+DEV bool is_reserved_path(const unsigned char* name, int n) {
+  const char* tbl[10] = {"joinchat", "addlist", "addstickers", "addtheme",
+                         "setlanguage", "share", "proxy", "socks", "login",
+                         "confirm"};
+  const int lens[10] = {8, 7, 11, 8, 11, 5, 5, 5, 5, 7};
+  for (int k = 0; k < 10; ++k) {
+    if (lens[k] != n) continue;
+    bool eq = true;
+    for (int j = 0; j < n; ++j) eq &= (name[j] == (unsigned char)tbl[k][j]);
+    if (eq) return true;
+  }
+  return false;
+}
+
+// Extract username after "t.me/" at byte q (exclusive bound hi). Writes the
+// LOWERCASED name; returns length (5..32) or 0. Runs uniformly on all lanes.
+DEV int extract_name_at(const unsigned char* s, int hi, int q,
+                        unsigned char* name) {
+  if (q >= hi || !is_ascii_letter(s[q])) return 0;
+  int len = 0;
+  while (len < 32 && q + len < hi && is_word_char(s[q + len])) ++len;
+  if (len < 5) return 0;
+  for (int j = 0; j < len; ++j) name[j] = to_lower(s[q + j]);
+  if (is_reserved_path(name, len)) return 0;
+  return len;
+}
+
+// First occurrence of "t.me/" at p in [from, hi-5]; wave-parallel.
+DEV int find_tme(const unsigned char* s, int from, int hi, int lane) {
+  for (int base = from; base + 5 <= hi; base += WAVE) {
+    int p = base + lane;
+    bool m = (p + 5 <= hi) && s[p] == 't' && s[p + 1] == '.' &&
+             s[p + 2] == 'm' && s[p + 3] == 'e' && s[p + 4] == '/';
+    unsigned long long bal = __ballot(m);
+    if (bal) return base + __ffsll(bal) - 1;
+  }
+  return -1;
+}
+
+// usernameRegex.search: first letter in [lo,hi) followed by >=4 word chars.
+DEV int find_username(const unsigned char* s, int lo, int hi, int lane,
+                      unsigned char* name) {
+  for (int base = lo; base < hi; base += WAVE) {
+    int p = base + lane;
+    bool m = false;
+    if (p < hi && p + 5 <= hi && is_ascii_letter(s[p])) {
+      m = is_word_char(s[p + 1]) && is_word_char(s[p + 2]) &&
+          is_word_char(s[p + 3]) && is_word_char(s[p + 4]);
+    }
+    unsigned long long bal = __ballot(m);
+    if (bal) {
+      int q = base + __ffsll(bal) - 1;
+      int len = 0;
+      while (len < 32 && q + len < hi && is_word_char(s[q + len])) ++len;
+      for (int j = 0; j < len; ++j) name[j] = to_lower(s[q + j]);
+      return len;
+    }
+  }
+  return 0;
+}
+
+struct LinkList {
+  unsigned char* names;      // MAX_LINKS * 32 (global scratch)
+  unsigned char* lens;
+  unsigned char* srcs;
+  unsigned long long* hashes;
+  int cnt;
+
+  DEV void add(const unsigned char* name, int len, int src, int lane) {
+    // first-wins dedup by name (tdutils.go:903-907 addIfNew)
+    for (int k = 0; k < cnt; ++k) {
+      if (lens[k] != (unsigned char)len) continue;
+      bool eq = true;
+      for (int j = 0; j < len; ++j) eq &= (names[k * 32 + j] == name[j]);
+      if (eq) return;
+    }
+    if (cnt >= MAX_LINKS) return;
+    // All lanes store identical values (name is wave-uniform): keeps the
+    // next add()'s dedup reads coherent without a cross-lane fence.
+    for (int j = 0; j < len; ++j) names[cnt * 32 + j] = name[j];
+    lens[cnt] = (unsigned char)len;
+    srcs[cnt] = (unsigned char)src;
+    hashes[cnt] = fnv1a64(name, len);
+    ++cnt;
+  }
+};
+
+// Full link extraction for one message; fills the LinkList.
+// Source enum: 0=mention 1=text_url 2=url 3=plaintext (golden SOURCE_*).
+DEV void extract_links(const BatchView& B, int i, LinkList& L, int lane) {
+  int ct = B.content_type[i];
+  if (ct > 6) return;  // not a FormattedText-bearing type (tdutils.go:953-975)
+  const unsigned char* text = B.pool + B.text_off[i];
+  const int tn = B.text_len[i];
+  unsigned char name[32];
+
+  // entity walk, in entity order (tdutils.go:909-941)
+  const int e0 = B.ent_off[i], ec = B.ent_cnt[i];
+  for (int e = e0; e < e0 + ec; ++e) {
+    const int* row = B.entities + (long)e * 5;
+    int etype = row[0];
+    if (etype == 1) {  // text_url: scan the URL attribute
+      const unsigned char* url = B.pool + row[3];
+      int un = row[4];
+      int p = find_tme(url, 0, un, lane);
+      if (p >= 0) {
+        int len = extract_name_at(url, un, p + 5, name);
+        if (len) L.add(name, len, 1, lane);
+      }
+    } else {  // mention (0) / url (2): slice text at utf16 offsets
+      int lo, hi;
+      utf16_to_bytes(text, tn, row[1], row[2], &lo, &hi);
+      if (!(lo < hi && hi <= tn)) continue;
+      if (etype == 0) {
+        int len = find_username(text, lo, hi, lane, name);
+        if (len) L.add(name, len, 0, lane);
+      } else {
+        int p = find_tme(text, lo, hi, lane);
+        if (p >= 0) {
+          int len = extract_name_at(text, hi, p + 5, name);
+          if (len) L.add(name, len, 2, lane);
+        }
+      }
+    }
+  }
+
+  // plaintext scan with non-overlap cursor (FindAllStringSubmatch semantics)
+  int cursor = 0;
+  int from = 0;
+  while (true) {
+    int p = find_tme(text, from, tn, lane);
+    if (p < 0) break;
+    from = p + 1;  // next candidate search position
+    if (p < cursor) continue;
+    int len = extract_name_at(text, tn, p + 5, name);
+    if (len) {
+      L.add(name, len, 3, lane);
+      cursor = p + 5 + len;
+      from = cursor;
+    } else {
+      // invalid name: Go regex has no match AT this t.me; try the next one
+      continue;
+    }
+  }
+}
+
+// ---------- the templated line emitter ----------
+
+template <bool W>
+struct Emit {
+  unsigned char* out;
+  int cur;
+
+  DEV void lit(const char* s) {
+    int n = 0;
+    while (s[n]) ++n;
+    if (W)
+      for (int j = lane_id(); j < n; j += WAVE) out[cur + j] = (unsigned char)s[j];
+    cur += n;
+  }
+  DEV void raw(const unsigned char* s, int n) {
+    if (W)
+      for (int j = lane_id(); j < n; j += WAVE) out[cur + j] = s[j];
+    cur += n;
+  }
+  DEV void esc(const unsigned char* s, int n) {
+    int lane = lane_id();
+    for (int start = 0; start < n; start += WAVE) {
+      int p = start + lane;
+      int el = (p < n) ? escape_len_at(s, n, p) : 0;
+      if (W) {
+        int off = wave_prefix_excl(el);
+        if (p < n && el > 0) {
+          unsigned char tmp[6];
+          int w = escape_write_at(s, n, p, tmp);
+          for (int j = 0; j < w; ++j) out[cur + off + j] = tmp[j];
+        }
+      }
+      cur += wave_sum(el);
+    }
+  }
+  DEV void qesc(const unsigned char* s, int n) { lit("\""); esc(s, n); lit("\""); }
+  DEV void u64(unsigned long long v) {
+    if (W) {
+      char buf[20];
+      int n = 0;
+      unsigned long long x = v;
+      do { buf[n++] = '0' + (x % 10); x /= 10; } while (x);
+      if (lane_id() == 0)
+        for (int j = 0; j < n; ++j) out[cur + j] = buf[n - 1 - j];
+      cur += n;
+    } else {
+      cur += u64_dec_len(v);
+    }
+  }
+  DEV void i64(long long v) {
+    if (v < 0) { lit("-"); u64((unsigned long long)(-v)); }
+    else u64((unsigned long long)v);
+  }
+  DEV void rfc3339(long secs) {
+    if (W) {
+      Writer w{out, cur};
+      w.rfc3339(secs);
+    }
+    cur += 20;
+  }
+};
+
+template <bool W>
+DEV int emit_line(const BatchView& B, int i, unsigned char* out,
+                  LinkList& L) {
+  const int lane = lane_id();
+  Emit<W> e{out, 0};
+  const int c = B.channel_idx[i];
+  const unsigned char* user = B.pool + B.ch_user_off[c];
+  const int user_n = B.ch_user_len[c];
+  const unsigned char* title = B.pool + B.ch_title_off[c];
+  const int title_n = B.ch_title_len[c];
+  const long pub_id = B.msg_id[i] >> 20;
+  const int ct = B.content_type[i];
+  const int ncom = B.com_cnt[i];
+  const bool has_user = user_n > 0;
+
+  // post_link / url (tdutils.go:1005-1031; empty for private channels)
+  auto post_link = [&]() {
+    if (!has_user) { e.lit("\"\""); return; }
+    e.lit("\"https://t.me/");
+    e.esc(user, user_n);
+    e.lit("/");
+    e.i64(pub_id);
+    if (B.media_album_id[i] != 0) e.lit("?single");
+    e.lit("\"");
+  };
+
+  e.lit("{\"post_link\":");
+  post_link();
+  e.lit(",\"channel_id\":\"");
+  e.i64(B.chat_id[i]);
+  e.lit("\",\"post_uid\":\"");
+  e.i64(pub_id);
+  e.lit("-");
+  e.esc(user, user_n);
+  e.lit("\",\"url\":");
+  post_link();
+  e.lit(",\"published_at\":\"");
+  e.rfc3339(B.date[i]);
+  e.lit("\",\"created_at\":\"");
+  e.raw(B.created_str, B.created_len);
+  e.lit("\",\"language_code\":\"\",\"engagement\":");
+  e.i64(B.views[i]);
+  e.lit(",\"view_count\":");
+  e.i64(B.views[i]);
+  e.lit(",\"like_count\":0,\"share_count\":");
+  e.i64(B.forwards[i]);
+  e.lit(",\"comment_count\":");
+  e.i64(ncom);
+  e.lit(",\"crawl_label\":\"\",\"list_ids\":null,\"channel_name\":");
+  e.qesc(title, title_n);
+  e.lit(",\"search_terms\":null,\"search_term_ids\":null,\"project_ids\":null,"
+        "\"exercise_ids\":null,\"label_data\":null,\"labels_metadata\":null,"
+        "\"project_labeled_post_ids\":null,\"labeler_ids\":null,"
+        "\"all_labels\":null,\"label_ids\":null,\"is_ad\":false,"
+        "\"transcript_text\":\"\",\"image_text\":\"\",\"video_length\":null,"
+        "\"is_verified\":null,\"channel_data\":{\"channel_id\":\"");
+  e.i64(B.chat_id[i]);
+  e.lit("\",\"channel_name\":");
+  e.qesc(title, title_n);
+  e.lit(",\"channel_description\":\"\",\"channel_profile_image\":\"\","
+        "\"channel_engagement_data\":{\"follower_count\":");
+  e.i64(B.ch_member[c]);
+  e.lit(",\"following_count\":0,\"like_count\":0,\"post_count\":");
+  e.i64(B.ch_postcount[c]);
+  e.lit(",\"views_count\":");
+  e.i64(B.ch_totalviews[c]);
+  e.lit(",\"comment_count\":0,\"share_count\":0},"
+        "\"channel_url_external\":\"https://t.me/c/");
+  e.esc(user, user_n);
+  e.lit("\",\"channel_url\":\"https://t.me/c/");
+  e.esc(user, user_n);
+  e.lit("\",\"country_code\":\"\",\"published_at\":\"0001-01-01T00:00:00Z\"},"
+        "\"platform_name\":\"Telegram\",\"shared_id\":null,"
+        "\"quoted_id\":null,\"replied_id\":null,\"ai_label\":null,"
+        "\"root_post_id\":null,\"engagement_steps_count\":0,\"ocr_data\":null,"
+        "\"performance_scores\":{\"likes\":null,\"shares\":null,"
+        "\"comments\":null,\"views\":0},\"has_embed_media\":null,"
+        "\"description\":");
+  // description per content switch (tdutils.go:443-587)
+  {
+    const unsigned char* d;
+    int dn;
+    if (ct == 0 || ct == 1 || ct == 2 || ct == 4 || ct == 14) {
+      d = B.pool + B.text_off[i];
+      dn = B.text_len[i];
+    } else if (ct == 3 || ct == 9 || ct == 10 || ct == 11) {
+      d = B.pool + B.aux_off[i];
+      dn = B.aux_len[i];
+    } else {
+      d = nullptr;
+      dn = 0;
+    }
+    e.qesc(d, dn);
+  }
+  e.lit(",\"repost_channel_data\":null,\"post_type\":[\"");
+  e.raw(B.ctname_pool + B.ctname_off[ct], B.ctname_len[ct]);
+  e.lit("\"],\"inner_link\":{},\"post_title\":null,\"media_data\":"
+        "{\"document_name\":\"\"},\"is_reply\":null,\"ad_fields\":null,"
+        "\"likes_count\":0,\"shares_count\":");
+  e.i64(B.forwards[i]);
+  e.lit(",\"comments_count\":");
+  e.i64(ncom);
+  e.lit(",\"views_count\":");
+  e.i64(B.views[i]);
+  e.lit(",\"searchable_text\":\"\",\"all_text\":\"\","
+        "\"contrast_agent_project_ids\":null,\"agent_ids\":null,"
+        "\"segment_ids\":null,\"thumb_url\":\"");
+  // media (fetchAndUploadMedia skip rules, tdutils.go:233-239): GPU path
+  // always runs skip_media (media-on is staged host-side).
+  e.lit("\",\"media_url\":\"");
+  if ((ct == 3 || ct == 8) && (B.flags[i] & 2)) {
+    e.lit("AgAD");
+    e.i64(pub_id);
+    e.lit("v");
+  }
+  e.lit("\",\"comments\":[");
+  {
+    const int c0 = B.com_off[i];
+    for (int k = 0; k < ncom; ++k) {
+      if (k) e.lit(",");
+      const int cc = c0 + k;
+      e.lit("{\"text\":");
+      e.qesc(B.pool + B.com_text_off[cc], B.com_text_len[cc]);
+      e.lit(",\"reactions\":{");
+      const int r0 = B.com_react_off[cc], rc = B.com_react_cnt[cc];
+      for (int r = 0; r < rc; ++r) {
+        if (r) e.lit(",");
+        int em = B.react_emoji[r0 + r];
+        e.lit("\"");
+        e.raw(B.emoji_pool + B.emoji_off[em], B.emoji_len[em]);
+        e.lit("\":");
+        e.i64(B.react_count[r0 + r]);
+      }
+      e.lit("},\"view_count\":");
+      e.i64(B.com_views[cc]);
+      e.lit(",\"reply_count\":");
+      e.i64(B.com_replies[cc]);
+      e.lit(",\"handle\":");
+      e.qesc(B.pool + B.com_handle_off[cc], B.com_handle_len[cc]);
+      e.lit("}");
+    }
+  }
+  e.lit("],\"reactions\":{");
+  {
+    const int r0 = B.react_off[i], rc = B.react_cnt[i];
+    for (int r = 0; r < rc; ++r) {
+      if (r) e.lit(",");
+      int em = B.react_emoji[r0 + r];
+      e.lit("\"");
+      e.raw(B.emoji_pool + B.emoji_off[em], B.emoji_len[em]);
+      e.lit("\":");
+      e.i64(B.react_count[r0 + r]);
+    }
+  }
+  e.lit("},\"outlinks\":[");
+  for (int k = 0; k < L.cnt; ++k) {
+    if (k) e.lit(",");
+    e.lit("\"");
+    e.raw(L.names + k * 32, L.lens[k]);
+    e.lit("\"");
+  }
+  e.lit("],\"capture_time\":\"");
+  e.raw(B.capture_str, B.capture_len);
+  e.lit("\",\"handle\":");
+  e.qesc(B.pool + B.poster_off[i], B.poster_len[i]);
+  e.lit("}\n");
+  return e.cur;
+}
+
+// ---------- kernels ----------
+
+__global__ void __launch_bounds__(256)
+measure_extract_kernel(BatchView B, LinkOut LO, int* line_len) {
+  const int lane = lane_id();
+  const int wave = wave_id();
+  const int waves_per_grid = gridDim.x * 4;
+  for (int i = blockIdx.x * 4 + wave; i < B.n; i += waves_per_grid) {
+    // min_post_date filter (tdutils.go:419-421)
+    if ((long)B.date[i] < B.min_post_date) {
+      if (lane == 0) { line_len[i] = 0; LO.cnt[i] = 0; }
+      continue;
+    }
+    LinkList L{LO.name + (size_t)i * MAX_LINKS * 32,
+               LO.name_len + (size_t)i * MAX_LINKS,
+               LO.src + (size_t)i * MAX_LINKS,
+               LO.hash + (size_t)i * MAX_LINKS, 0};
+    extract_links(B, i, L, lane);
+    int len = emit_line<false>(B, i, nullptr, L);
+    if (lane == 0) {
+      line_len[i] = len;
+      LO.cnt[i] = L.cnt;
+    }
+  }
+}
+
+__global__ void __launch_bounds__(256)
+write_kernel(BatchView B, LinkOut LO, const long* line_off,
+             const int* line_len, unsigned char* out) {
+  const int lane = lane_id();
+  const int wave = wave_id();
+  const int waves_per_grid = gridDim.x * 4;
+  for (int i = blockIdx.x * 4 + wave; i < B.n; i += waves_per_grid) {
+    if (line_len[i] == 0) continue;
+    LinkList L{LO.name + (size_t)i * MAX_LINKS * 32,
+               LO.name_len + (size_t)i * MAX_LINKS,
+               LO.src + (size_t)i * MAX_LINKS,
+               LO.hash + (size_t)i * MAX_LINKS, LO.cnt[i]};
+    emit_line<true>(B, i, out + line_off[i], L);
+  }
+}
+
+}  // namespace crawl
+
+// ---- C ABI (ctypes-friendly; torch-header-free) ----
+//
+// Pointer/scalar array layouts MUST stay in sync with ops/gpu.py
+// (_BATCH_PTR_ORDER / _SCALAR_ORDER). The Python side passes
+// tensor.data_ptr() values in that order.
+
+namespace crawl {
+
+static BatchView make_view(void** p, const long* s) {
+  BatchView B;
+  int k = 0;
+  B.chat_id = (const long*)p[k++];
+  B.msg_id = (const long*)p[k++];
+  B.text_off = (const long*)p[k++];
+  B.date = (const int*)p[k++];
+  B.content_type = (const int*)p[k++];
+  B.views = (const int*)p[k++];
+  B.forwards = (const int*)p[k++];
+  B.media_album_id = (const int*)p[k++];
+  B.channel_idx = (const int*)p[k++];
+  B.flags = (const int*)p[k++];
+  B.text_len = (const int*)p[k++];
+  B.aux_off = (const int*)p[k++];
+  B.aux_len = (const int*)p[k++];
+  B.ent_off = (const int*)p[k++];
+  B.ent_cnt = (const int*)p[k++];
+  B.react_off = (const int*)p[k++];
+  B.react_cnt = (const int*)p[k++];
+  B.com_off = (const int*)p[k++];
+  B.com_cnt = (const int*)p[k++];
+  B.poster_off = (const int*)p[k++];
+  B.poster_len = (const int*)p[k++];
+  B.pool = (const unsigned char*)p[k++];
+  B.entities = (const int*)p[k++];
+  B.react_emoji = (const int*)p[k++];
+  B.react_count = (const int*)p[k++];
+  B.com_text_off = (const int*)p[k++];
+  B.com_text_len = (const int*)p[k++];
+  B.com_handle_off = (const int*)p[k++];
+  B.com_handle_len = (const int*)p[k++];
+  B.com_views = (const int*)p[k++];
+  B.com_replies = (const int*)p[k++];
+  B.com_react_off = (const int*)p[k++];
+  B.com_react_cnt = (const int*)p[k++];
+  B.ch_chat_id = (const long*)p[k++];
+  B.ch_member = (const int*)p[k++];
+  B.ch_postcount = (const int*)p[k++];
+  B.ch_totalviews = (const int*)p[k++];
+  B.ch_user_off = (const int*)p[k++];
+  B.ch_user_len = (const int*)p[k++];
+  B.ch_title_off = (const int*)p[k++];
+  B.ch_title_len = (const int*)p[k++];
+  B.emoji_pool = (const unsigned char*)p[k++];
+  B.emoji_off = (const int*)p[k++];
+  B.emoji_len = (const int*)p[k++];
+  B.created_str = (const unsigned char*)p[k++];
+  B.capture_str = (const unsigned char*)p[k++];
+  B.ctname_pool = (const unsigned char*)p[k++];
+  B.ctname_off = (const int*)p[k++];
+  B.ctname_len = (const int*)p[k++];
+  B.n = (int)s[0];
+  B.skip_media = (int)s[1];
+  B.min_post_date = s[2];
+  B.created_len = (int)s[3];
+  B.capture_len = (int)s[4];
+  return B;
+}
+
+static LinkOut make_links(void** p) {
+  LinkOut L;
+  L.name = (unsigned char*)p[0];
+  L.name_len = (unsigned char*)p[1];
+  L.src = (unsigned char*)p[2];
+  L.cnt = (int*)p[3];
+  L.hash = (unsigned long long*)p[4];
+  return L;
+}
+
+}  // namespace crawl
+
+extern "C" {
+
+int crawl_batch_ptr_count() { return 49; }
+
+int crawl_measure_extract(void** batch_ptrs, const long* scalars,
+                          void** link_ptrs, void* line_len, int grid,
+                          void* stream) {
+  crawl::BatchView B = crawl::make_view(batch_ptrs, scalars);
+  crawl::LinkOut LO = crawl::make_links(link_ptrs);
+  hipLaunchKernelGGL(crawl::measure_extract_kernel, dim3(grid), dim3(256), 0,
+                     (hipStream_t)stream, B, LO, (int*)line_len);
+  return (int)hipGetLastError();
+}
+
+int crawl_write(void** batch_ptrs, const long* scalars, void** link_ptrs,
+                const void* line_off, const void* line_len, void* out,
+                int grid, void* stream) {
+  crawl::BatchView B = crawl::make_view(batch_ptrs, scalars);
+  crawl::LinkOut LO = crawl::make_links(link_ptrs);
+  hipLaunchKernelGGL(crawl::write_kernel, dim3(grid), dim3(256), 0,
+                     (hipStream_t)stream, B, LO, (const long*)line_off,
+                     (const int*)line_len, (unsigned char*)out);
+  return (int)hipGetLastError();
+}
+
+}  // extern "C"

@@ -1,0 +1,245 @@
+"""ctypes driver for the CDNA4 hot-path kernels (ops/csrc/libcrawlhip.so).
+
+The HIP extension is built in-tree by ``__graft_entry__.build()`` (or
+``python -m crawler_amd.ops.build``) with hipcc --offload-arch=gfx950.
+On a GPU host the extension is REQUIRED: :func:`require_lib` raises rather
+than letting callers silently fall back to the (1000x slower) Python path.
+"""
+from __future__ import annotations
+
+import ctypes
+import dataclasses
+import datetime as _dt
+import os
+from typing import List, Optional, Tuple
+
+import torch
+
+from ..models.post import format_go_time
+from . import batch as B
+
+_CSRC = os.path.dirname(os.path.abspath(__file__)) + "/csrc"
+_LIB_PATH = os.path.join(_CSRC, "libcrawlhip.so")
+
+MAX_LINKS = 8
+
+# Must match crawl::make_view() field order in csrc/parse_encode.hip.
+_BATCH_PTR_ORDER = [
+    "chat_id", "msg_id", "text_off",
+    # meta int32 fields, in BatchView order:
+    "m:date", "m:content_type", "m:views", "m:forwards", "m:media_album_id",
+    "m:channel_idx", "m:flags", "m:text_len", "m:aux_off", "m:aux_len",
+    "m:ent_off", "m:ent_cnt", "m:react_off", "m:react_cnt", "m:com_off",
+    "m:com_cnt", "m:poster_off", "m:poster_len",
+    "text_pool", "entities", "react_emoji", "react_count",
+    "com_text_off", "com_text_len", "com_handle_off", "com_handle_len",
+    "com_views", "com_replies", "com_react_off", "com_react_cnt",
+    "ch_chat_id", "ch_member", "ch_postcount", "ch_totalviews",
+    "ch_user_off", "ch_user_len", "ch_title_off", "ch_title_len",
+]
+
+_lib = None
+
+
+def lib_available() -> bool:
+    return os.path.exists(_LIB_PATH)
+
+
+def load_lib():
+    global _lib
+    if _lib is not None:
+        return _lib
+    lib = ctypes.CDLL(_LIB_PATH)
+    lib.crawl_batch_ptr_count.restype = ctypes.c_int
+    lib.crawl_measure_extract.restype = ctypes.c_int
+    lib.crawl_measure_extract.argtypes = [
+        ctypes.POINTER(ctypes.c_void_p), ctypes.POINTER(ctypes.c_long),
+        ctypes.POINTER(ctypes.c_void_p), ctypes.c_void_p, ctypes.c_int,
+        ctypes.c_void_p,
+    ]
+    lib.crawl_write.restype = ctypes.c_int
+    lib.crawl_write.argtypes = [
+        ctypes.POINTER(ctypes.c_void_p), ctypes.POINTER(ctypes.c_long),
+        ctypes.POINTER(ctypes.c_void_p), ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+    ]
+    _lib = lib
+    return lib
+
+
+def require_lib():
+    """On a GPU host, a missing/failed extension is an ERROR, not a fallback."""
+    if not lib_available():
+        raise RuntimeError(
+            "libcrawlhip.so not built — run __graft_entry__.build() "
+            "(hipcc --offload-arch=gfx950) before GPU execution; the Python "
+            "golden path is not a substitute on GPU hosts"
+        )
+    return load_lib()
+
+
+# Emoji / content-type constant tables shipped to the device once.
+def _const_tables(device):
+    epool = b"".join(e.encode("utf-8") for e in B.EMOJI_TABLE)
+    eoff, cur = [], 0
+    elen = []
+    for e in B.EMOJI_TABLE:
+        n = len(e.encode("utf-8"))
+        eoff.append(cur)
+        elen.append(n)
+        cur += n
+    cpool = b"".join(c.encode() for c in B.CONTENT_TYPES)
+    coff, ccur = [], 0
+    clen = []
+    for c in B.CONTENT_TYPES:
+        coff.append(ccur)
+        clen.append(len(c))
+        ccur += len(c)
+    t = lambda x, dt_: torch.tensor(x, dtype=dt_, device=device)
+    return {
+        "emoji_pool": torch.frombuffer(bytearray(epool), dtype=torch.uint8).to(device),
+        "emoji_off": t(eoff, torch.int32),
+        "emoji_len": t(elen, torch.int32),
+        "ctname_pool": torch.frombuffer(bytearray(cpool), dtype=torch.uint8).to(device),
+        "ctname_off": t(coff, torch.int32),
+        "ctname_len": t(clen, torch.int32),
+    }
+
+
+_tables_cache = {}
+
+
+def const_tables(device):
+    key = str(device)
+    if key not in _tables_cache:
+        _tables_cache[key] = _const_tables(device)
+    return _tables_cache[key]
+
+
+@dataclasses.dataclass
+class EncodeResult:
+    """Device-side results of one parse+encode pass."""
+
+    out: torch.Tensor        # uint8[total_bytes] JSONL (concatenated lines)
+    line_off: torch.Tensor   # int64[N] start offset of each line
+    line_len: torch.Tensor   # int32[N] (0 = filtered by min_post_date)
+    link_name: torch.Tensor  # uint8[N, MAX_LINKS, 32]
+    link_len: torch.Tensor   # uint8[N, MAX_LINKS]
+    link_src: torch.Tensor   # uint8[N, MAX_LINKS]
+    link_cnt: torch.Tensor   # int32[N]
+    link_hash: torch.Tensor  # int64[N, MAX_LINKS] (fnv1a64 bits)
+
+
+def _ptr_array(tensors: List[torch.Tensor]):
+    arr = (ctypes.c_void_p * len(tensors))()
+    for k, t in enumerate(tensors):
+        arr[k] = ctypes.c_void_p(t.data_ptr())
+    return arr
+
+
+def parse_encode(
+    batch: B.MessageBatch,
+    now: Optional[_dt.datetime] = None,
+    min_post_date: Optional[_dt.datetime] = None,
+    grid: int = 0,
+    stream: Optional[torch.cuda.Stream] = None,
+) -> EncodeResult:
+    """Run measure+extract then write on the current CUDA device.
+
+    Returns device tensors; the caller D2H-copies ``out[:total]`` (or slices
+    per line) for the host writer. Byte-for-byte equal to
+    golden_batch.encode_batch(batch, now=now, ...).
+    """
+    lib = require_lib()
+    dev = batch.device
+    assert dev.type == "cuda", "parse_encode requires a CUDA (ROCm) batch"
+    n = batch.n
+    tables = const_tables(dev)
+
+    now = now or _dt.datetime.now(_dt.timezone.utc)
+    created = format_go_time(now.replace(microsecond=0)).encode()
+    capture = format_go_time(now).encode()
+    created_t = torch.frombuffer(bytearray(created), dtype=torch.uint8).to(dev)
+    capture_t = torch.frombuffer(bytearray(capture), dtype=torch.uint8).to(dev)
+
+    tensors = []
+    for name in _BATCH_PTR_ORDER:
+        if name.startswith("m:"):
+            tensors.append(batch.meta[name[2:]])
+        else:
+            tensors.append(getattr(batch, name))
+    tensors += [
+        tables["emoji_pool"], tables["emoji_off"], tables["emoji_len"],
+        created_t, capture_t,
+        tables["ctname_pool"], tables["ctname_off"], tables["ctname_len"],
+    ]
+    for t in tensors:
+        assert t.device == dev and t.is_contiguous()
+    batch_ptrs = _ptr_array(tensors)
+    assert len(tensors) == lib.crawl_batch_ptr_count()
+
+    if min_post_date is not None:
+        mpd = int(min_post_date.timestamp())
+    else:
+        mpd = -(1 << 62)
+    scalars = (ctypes.c_long * 5)(n, 1, mpd, len(created), len(capture))
+
+    link_name = torch.empty((n, MAX_LINKS, 32), dtype=torch.uint8, device=dev)
+    link_len = torch.zeros((n, MAX_LINKS), dtype=torch.uint8, device=dev)
+    link_src = torch.zeros((n, MAX_LINKS), dtype=torch.uint8, device=dev)
+    link_cnt = torch.zeros(n, dtype=torch.int32, device=dev)
+    link_hash = torch.zeros((n, MAX_LINKS), dtype=torch.int64, device=dev)
+    line_len = torch.zeros(n, dtype=torch.int32, device=dev)
+    link_ptrs = _ptr_array([link_name, link_len, link_src, link_cnt, link_hash])
+
+    if grid <= 0:
+        grid = min((n + 3) // 4, 8192)
+    stream_ptr = ctypes.c_void_p(
+        torch.cuda.current_stream().cuda_stream if stream is None
+        else stream.cuda_stream
+    )
+
+    rc = lib.crawl_measure_extract(
+        batch_ptrs, scalars, link_ptrs,
+        ctypes.c_void_p(line_len.data_ptr()), grid, stream_ptr,
+    )
+    if rc != 0:
+        raise RuntimeError(f"crawl_measure_extract launch failed: hip error {rc}")
+
+    line_off = torch.zeros(n, dtype=torch.int64, device=dev)
+    torch.cumsum(line_len.to(torch.int64)[:-1], 0, out=line_off[1:])
+    total = int(line_off[-1].item() + line_len[-1].item()) if n else 0
+    out = torch.empty(total, dtype=torch.uint8, device=dev)
+
+    rc = lib.crawl_write(
+        batch_ptrs, scalars, link_ptrs,
+        ctypes.c_void_p(line_off.data_ptr()),
+        ctypes.c_void_p(line_len.data_ptr()),
+        ctypes.c_void_p(out.data_ptr()), grid, stream_ptr,
+    )
+    if rc != 0:
+        raise RuntimeError(f"crawl_write launch failed: hip error {rc}")
+
+    return EncodeResult(
+        out=out, line_off=line_off, line_len=line_len,
+        link_name=link_name, link_len=link_len, link_src=link_src,
+        link_cnt=link_cnt, link_hash=link_hash,
+    )
+
+
+def links_to_python(res: EncodeResult) -> List[List[Tuple[str, str]]]:
+    """Decode the device link output to [(name, source_type), ...] per msg."""
+    srcs = ["mention", "text_url", "url", "plaintext"]
+    name = res.link_name.cpu().numpy()
+    ln = res.link_len.cpu().numpy()
+    src = res.link_src.cpu().numpy()
+    cnt = res.link_cnt.cpu().numpy()
+    out = []
+    for i in range(len(cnt)):
+        row = []
+        for k in range(int(cnt[i])):
+            row.append(
+                (bytes(name[i, k, : ln[i, k]]).decode(), srcs[int(src[i, k])])
+            )
+        out.append(row)
+    return out
