@@ -120,7 +120,12 @@ class MessageBatch:
         kw = {}
         for f in dataclasses.fields(self):
             v = getattr(self, f.name)
-            kw[f.name] = v.to(device) if isinstance(v, torch.Tensor) else v
+            if isinstance(v, torch.Tensor):
+                kw[f.name] = v.to(device)
+            elif isinstance(v, dict):
+                kw[f.name] = {k: t.to(device) for k, t in v.items()}
+            else:
+                kw[f.name] = v
         return MessageBatch(**kw)
 
     @property

@@ -173,8 +173,13 @@ def parse_encode(
         created_t, capture_t,
         tables["ctname_pool"], tables["ctname_off"], tables["ctname_len"],
     ]
-    for t in tensors:
-        assert t.device == dev and t.is_contiguous()
+    names = _BATCH_PTR_ORDER + [
+        "emoji_pool", "emoji_off", "emoji_len", "created", "capture",
+        "ctname_pool", "ctname_off", "ctname_len",
+    ]
+    for nm, t in zip(names, tensors):
+        assert t.device == dev, f"{nm} on {t.device}, batch on {dev}"
+        assert t.is_contiguous(), f"{nm} not contiguous"
     batch_ptrs = _ptr_array(tensors)
     assert len(tensors) == lib.crawl_batch_ptr_count()
 
