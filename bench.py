@@ -1,0 +1,222 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: posts/sec, synthetic Telegram feed, GPU parse+encode.
+
+BASELINE.json config #2 (scaled to N GPUs, weak scaling): per GPU, 1k
+synthetic channels x 10k posts; the timed step runs the full per-post hot
+path on one pre-generated device-resident chunk:
+
+    HIP parse+encode (content switch, UTF-16 entity walk, t.me link
+    extraction, Go-JSON line emission)
+ -> seen-set claim (atomicCAS hash table) + bloom update
+ -> cross-rank discovery exchange (all-gather of newly-claimed hashes,
+    RCCL over xGMI) when world_size > 1
+ -> D2H copy of the JSONL bytes into a pinned host ring (the storage
+    binding boundary; --sink file additionally writes to disk)
+
+Launch: python bench.py --gpus N --steps K --warmup W
+(for N>1 the driver uses torch.distributed.run; RANK/LOCAL_RANK/WORLD_SIZE
+are read from the env). Rank 0 prints ONE json line per the bench contract.
+"""
+import argparse
+import datetime as dt
+import json
+import os
+import statistics
+import sys
+import time
+
+import numpy as np
+import torch
+
+
+def log(msg):
+    if int(os.environ.get("RANK", "0")) == 0:
+        print(f"[bench] {msg}", file=sys.stderr, flush=True)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=6)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--channels", type=int, default=1000,
+                    help="channels per GPU (flagship: 1000)")
+    ap.add_argument("--posts", type=int, default=10000,
+                    help="posts per channel (flagship: 10000)")
+    ap.add_argument("--chunk-channels", type=int, default=125,
+                    help="channels per device chunk (= step granularity)")
+    ap.add_argument("--sink", choices=["pinned", "file", "none"],
+                    default="pinned")
+    ap.add_argument("--sink-dir", default="/tmp/crawl-bench")
+    ap.add_argument("--cpu", action="store_true",
+                    help="debug: run the Python golden path on CPU (tiny)")
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+
+    if args.cpu:
+        return run_cpu_debug(args)
+
+    assert torch.cuda.is_available(), "bench requires a ROCm GPU"
+    torch.cuda.set_device(local_rank)
+    device = torch.device("cuda", local_rank)
+
+    if world > 1:
+        torch.distributed.init_process_group("nccl")
+
+    from crawler_amd.feed import FeedConfig, SyntheticFeed
+    from crawler_amd.ops import gpu
+
+    gpu.require_lib()
+
+    # ---- setup (untimed): generate the per-rank corpus ----
+    n_chunks = max(1, args.channels // args.chunk_channels)
+    chunk_ch = args.chunk_channels
+    posts = args.posts
+    feed = SyntheticFeed(FeedConfig(seed=1234 + rank, universe=1_000_000))
+    chunks = []
+    t_gen = time.time()
+    for c in range(n_chunks):
+        ids = np.arange(c * chunk_ch, (c + 1) * chunk_ch) + rank * args.channels
+        chunks.append(feed.build_batch(ids, posts_per_channel=posts).to(device))
+        log(f"chunk {c + 1}/{n_chunks} generated "
+            f"({(c + 1) * chunk_ch * posts / 1e6:.2f}M posts, "
+            f"{time.time() - t_gen:.0f}s)")
+    chunk_posts = chunk_ch * posts
+
+    now = dt.datetime(2026, 1, 1, tzinfo=dt.timezone.utc)
+    seen = gpu.SeenSet(device)
+
+    # pinned host ring for the JSONL output (the storage boundary)
+    # worst case ~2.6KB/post
+    pinned = torch.empty(int(chunk_posts * 3000), dtype=torch.uint8,
+                         pin_memory=True)
+    sink_f = None
+    if args.sink == "file" and rank == 0:
+        os.makedirs(args.sink_dir, exist_ok=True)
+        sink_f = open(os.path.join(args.sink_dir, f"rank{rank}.jsonl"), "wb")
+
+    total_out_bytes = 0
+    new_discoveries = 0
+
+    def step(si):
+        nonlocal total_out_bytes, new_discoveries
+        chunk = chunks[si % n_chunks]
+        res = gpu.parse_encode(chunk, now=now)
+        new_mask = seen.claim(res)
+        if world > 1:
+            flat = new_mask.flatten().bool()
+            new_hashes = res.link_hash.flatten()[flat]
+            cap = 65536
+            buf = torch.zeros(cap, dtype=torch.int64, device=device)
+            k = min(new_hashes.numel(), cap)
+            buf[:k] = new_hashes[:k]
+            gathered = [torch.empty_like(buf) for _ in range(world)]
+            torch.distributed.all_gather(gathered, buf)
+            for r, g in enumerate(gathered):
+                if r != rank:
+                    gz = g[g != 0]
+                    seen.insert_hashes(gz)
+        nbytes = int(res.line_off[-1].item() + res.line_len[-1].item())
+        pinned[:nbytes].copy_(res.out[:nbytes], non_blocking=True)
+        torch.cuda.synchronize()
+        total_out_bytes += nbytes
+        new_discoveries += seen.new_count()
+        if sink_f is not None:
+            sink_f.write(bytes(pinned[:nbytes].numpy()))
+        return nbytes
+
+    # ---- warmup ----
+    for w in range(args.warmup):
+        step(w)
+    log(f"warmup done ({args.warmup} steps)")
+
+    # ---- timed region ----
+    if world > 1:
+        torch.distributed.barrier()
+    torch.cuda.synchronize()
+    step_times = []
+    t0 = time.perf_counter()
+    for si in range(args.steps):
+        ts = time.perf_counter()
+        step(args.warmup + si)
+        step_times.append((time.perf_counter() - ts) * 1000)
+    torch.cuda.synchronize()
+    if world > 1:
+        torch.distributed.barrier()
+    t1 = time.perf_counter()
+
+    elapsed = t1 - t0
+    if world > 1:
+        e = torch.tensor([elapsed], device=device)
+        torch.distributed.all_reduce(e, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    posts_done = args.steps * chunk_posts
+    value = world * posts_done / elapsed
+
+    if sink_f is not None:
+        sink_f.close()
+
+    if rank == 0:
+        result = {
+            "metric": "posts/sec (whole node) synthetic feed",
+            "value": round(value, 1),
+            "unit": "posts/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed * 1000 / args.steps, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "uint8",
+            "data": "synthetic",
+            "config": {
+                "model": "telegram-crawl 1k-channels x 10k-posts per GPU "
+                         "(BASELINE config #2, weak-scaled)",
+                "global_batch": world * chunk_posts,
+                "seq_len": 0,
+                "parallelism": f"dp{world} (channel-sharded, RCCL discovery "
+                               "all-gather)" if world > 1 else "dp1",
+                "channels_per_gpu": args.channels,
+                "posts_per_channel": args.posts,
+                "chunk_posts": chunk_posts,
+                "jsonl_bytes_per_step": total_out_bytes // max(
+                    1, args.warmup + args.steps),
+                "p50_step_ms": round(statistics.median(step_times), 2),
+                "new_discoveries": new_discoveries,
+                "sink": args.sink,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    if world > 1:
+        torch.distributed.destroy_process_group()
+
+
+def run_cpu_debug(args):
+    """Tiny CPU sanity run of the same pipeline via the golden encoder."""
+    from crawler_amd.feed import FeedConfig, SyntheticFeed
+    from crawler_amd.ops.golden_batch import encode_batch
+
+    feed = SyntheticFeed(FeedConfig(seed=1234, universe=10_000))
+    batch = feed.build_batch(np.arange(4), posts_per_channel=100)
+    now = dt.datetime(2026, 1, 1, tzinfo=dt.timezone.utc)
+    t0 = time.perf_counter()
+    lines, links = encode_batch(batch, now=now)
+    dt_s = time.perf_counter() - t0
+    print(json.dumps({
+        "metric": "posts/sec (cpu golden debug)",
+        "value": round(batch.n / dt_s, 1),
+        "unit": "posts/s",
+        "n_gpus": 0,
+        "bytes": sum(len(l) for l in lines),
+        "links": sum(len(l) for l in links),
+    }))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,116 @@
+// Device-resident channel dedup: open-addressing hash-set claim + bloom.
+//
+// MI355X-native replacement for the reference's discovered-channel set and
+// URL dedup cache (state/datamodels.go:118-162, state/daprstate.go:550-657):
+// the seen-channel set lives in HBM as a power-of-two open-addressing table
+// of 64-bit FNV hashes; discovery claims are atomicCAS inserts, giving the
+// same exactly-once guarantee as the reference's map+mutex (and of the
+// validator's INSERT ... ON CONFLICT claim, daprstate.go:4198-4225).
+// A bloom side-structure (bitwise-OR mergeable) is maintained for cheap
+// cross-GPU union via RCCL all-reduce (BOR/max) over xGMI.
+
+#include "common.h"
+
+namespace crawl {
+
+// Insert each link hash of each message into the table. table size must be
+// a power of two; empty slot = 0 (hash 0 remaps to 1). new_mask[i*maxl+k]=1
+// iff THIS call first-inserted the hash (claim won).
+__global__ void __launch_bounds__(256)
+claim_links_kernel(const unsigned long long* hashes, const int* cnt, int n,
+                   int maxl, unsigned long long* table, long long tmask,
+                   unsigned char* new_mask, unsigned int* n_new) {
+  int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  int stride = gridDim.x * blockDim.x;
+  for (int idx = tid; idx < n * maxl; idx += stride) {
+    int i = idx / maxl, k = idx % maxl;
+    if (k >= cnt[i]) continue;
+    unsigned long long h = hashes[idx];
+    if (h == 0) h = 1;
+    unsigned long long slot = h & (unsigned long long)tmask;
+    bool won = false;
+    for (int probe = 0; probe < 10000; ++probe) {
+      unsigned long long prev = atomicCAS(
+          (unsigned long long*)&table[slot], 0ULL, h);
+      if (prev == 0ULL) { won = true; break; }   // claimed
+      if (prev == h) { won = false; break; }      // already seen
+      slot = (slot + 1) & (unsigned long long)tmask;
+    }
+    new_mask[idx] = won ? 1 : 0;
+    if (won) atomicAdd(n_new, 1u);
+  }
+}
+
+// Mark hashes in a bloom bit array (uint32 words; mergeable by bitwise OR /
+// max across ranks). Two probes per hash (upper/lower 32 bits).
+__global__ void __launch_bounds__(256)
+bloom_update_kernel(const unsigned long long* hashes, const int* cnt, int n,
+                    int maxl, unsigned int* bloom, long long bloom_bits_mask) {
+  int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  int stride = gridDim.x * blockDim.x;
+  for (int idx = tid; idx < n * maxl; idx += stride) {
+    int i = idx / maxl, k = idx % maxl;
+    if (k >= cnt[i]) continue;
+    unsigned long long h = hashes[idx];
+    unsigned long long b1 = h & (unsigned long long)bloom_bits_mask;
+    unsigned long long b2 = (h >> 32) & (unsigned long long)bloom_bits_mask;
+    atomicOr(&bloom[b1 >> 5], 1u << (b1 & 31));
+    atomicOr(&bloom[b2 >> 5], 1u << (b2 & 31));
+  }
+}
+
+// Bulk-insert pre-hashed values (e.g. a merged remote seen-set after an
+// all-gather) without producing claim output.
+__global__ void __launch_bounds__(256)
+insert_hashes_kernel(const unsigned long long* hashes, long n,
+                     unsigned long long* table, long long tmask) {
+  long tid = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  long stride = gridDim.x * (long)blockDim.x;
+  for (long idx = tid; idx < n; idx += stride) {
+    unsigned long long h = hashes[idx];
+    if (h == 0) h = 1;
+    unsigned long long slot = h & (unsigned long long)tmask;
+    for (int probe = 0; probe < 10000; ++probe) {
+      unsigned long long prev = atomicCAS(
+          (unsigned long long*)&table[slot], 0ULL, h);
+      if (prev == 0ULL || prev == h) break;
+      slot = (slot + 1) & (unsigned long long)tmask;
+    }
+  }
+}
+
+}  // namespace crawl
+
+extern "C" {
+
+int crawl_claim_links(const void* hashes, const void* cnt, int n, int maxl,
+                      void* table, long long table_slots, void* new_mask,
+                      void* n_new, int grid, void* stream) {
+  hipLaunchKernelGGL(crawl::claim_links_kernel, dim3(grid), dim3(256), 0,
+                     (hipStream_t)stream,
+                     (const unsigned long long*)hashes, (const int*)cnt, n,
+                     maxl, (unsigned long long*)table, table_slots - 1,
+                     (unsigned char*)new_mask, (unsigned int*)n_new);
+  return (int)hipGetLastError();
+}
+
+int crawl_bloom_update(const void* hashes, const void* cnt, int n, int maxl,
+                       void* bloom, long long bloom_bits, int grid,
+                       void* stream) {
+  hipLaunchKernelGGL(crawl::bloom_update_kernel, dim3(grid), dim3(256), 0,
+                     (hipStream_t)stream,
+                     (const unsigned long long*)hashes, (const int*)cnt, n,
+                     maxl, (unsigned int*)bloom, bloom_bits - 1);
+  return (int)hipGetLastError();
+}
+
+int crawl_insert_hashes(const void* hashes, long n, void* table,
+                        long long table_slots, int grid, void* stream) {
+  hipLaunchKernelGGL(crawl::insert_hashes_kernel, dim3(grid), dim3(256), 0,
+                     (hipStream_t)stream,
+                     (const unsigned long long*)hashes, n,
+                     (unsigned long long*)table, table_slots - 1);
+  return (int)hipGetLastError();
+}
+
+}  // extern "C"

@@ -63,8 +63,28 @@ def load_lib():
         ctypes.POINTER(ctypes.c_void_p), ctypes.c_void_p, ctypes.c_void_p,
         ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
     ]
+    _bind_dedup(lib)
     _lib = lib
     return lib
+
+
+def _bind_dedup(lib):
+    lib.crawl_claim_links.restype = ctypes.c_int
+    lib.crawl_claim_links.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int,
+        ctypes.c_void_p, ctypes.c_longlong, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_int, ctypes.c_void_p,
+    ]
+    lib.crawl_bloom_update.restype = ctypes.c_int
+    lib.crawl_bloom_update.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int,
+        ctypes.c_void_p, ctypes.c_longlong, ctypes.c_int, ctypes.c_void_p,
+    ]
+    lib.crawl_insert_hashes.restype = ctypes.c_int
+    lib.crawl_insert_hashes.argtypes = [
+        ctypes.c_void_p, ctypes.c_long, ctypes.c_void_p, ctypes.c_longlong,
+        ctypes.c_int, ctypes.c_void_p,
+    ]
 
 
 def require_lib():
@@ -230,6 +250,92 @@ def parse_encode(
         link_name=link_name, link_len=link_len, link_src=link_src,
         link_cnt=link_cnt, link_hash=link_hash,
     )
+
+
+class SeenSet:
+    """Device-resident seen-channel set (hash table + bloom).
+
+    Replaces the reference's DiscoveredChannels map / URL dedup cache
+    (state/datamodels.go:118-162, daprstate.go:550-657) with an HBM
+    open-addressing table claimed via atomicCAS (exactly-once discovery)
+    plus a bitwise-OR-mergeable bloom for cross-rank union over RCCL.
+    """
+
+    def __init__(self, device, slots_log2: int = 22, bloom_bits_log2: int = 26):
+        self.device = device
+        self.slots = 1 << slots_log2
+        self.bloom_bits = 1 << bloom_bits_log2
+        self.table = torch.zeros(self.slots, dtype=torch.int64, device=device)
+        self.bloom = torch.zeros(
+            self.bloom_bits // 32, dtype=torch.int32, device=device
+        )
+        self._n_new = torch.zeros(1, dtype=torch.int32, device=device)
+        self.lib = require_lib()
+
+    def _stream(self):
+        return ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+
+    def claim(self, res: EncodeResult) -> torch.Tensor:
+        """Claim every extracted link; returns new_mask uint8[N, MAX_LINKS].
+
+        Also updates the bloom. Reads self.new_count() for the number of
+        first-time discoveries in this batch.
+        """
+        n = res.link_cnt.shape[0]
+        new_mask = torch.zeros(
+            (n, MAX_LINKS), dtype=torch.uint8, device=self.device
+        )
+        self._n_new.zero_()
+        grid = min((n * MAX_LINKS + 255) // 256, 4096)
+        rc = self.lib.crawl_claim_links(
+            ctypes.c_void_p(res.link_hash.data_ptr()),
+            ctypes.c_void_p(res.link_cnt.data_ptr()),
+            n, MAX_LINKS,
+            ctypes.c_void_p(self.table.data_ptr()),
+            ctypes.c_longlong(self.slots),
+            ctypes.c_void_p(new_mask.data_ptr()),
+            ctypes.c_void_p(self._n_new.data_ptr()),
+            grid, self._stream(),
+        )
+        if rc != 0:
+            raise RuntimeError(f"crawl_claim_links failed: hip error {rc}")
+        rc = self.lib.crawl_bloom_update(
+            ctypes.c_void_p(res.link_hash.data_ptr()),
+            ctypes.c_void_p(res.link_cnt.data_ptr()),
+            n, MAX_LINKS,
+            ctypes.c_void_p(self.bloom.data_ptr()),
+            ctypes.c_longlong(self.bloom_bits),
+            grid, self._stream(),
+        )
+        if rc != 0:
+            raise RuntimeError(f"crawl_bloom_update failed: hip error {rc}")
+        return new_mask
+
+    def new_count(self) -> int:
+        return int(self._n_new.item())
+
+    def insert_hashes(self, hashes: torch.Tensor) -> None:
+        """Bulk-insert merged remote hashes (post all-gather)."""
+        n = hashes.numel()
+        if n == 0:
+            return
+        grid = min((n + 255) // 256, 4096)
+        rc = self.lib.crawl_insert_hashes(
+            ctypes.c_void_p(hashes.data_ptr()), ctypes.c_long(n),
+            ctypes.c_void_p(self.table.data_ptr()),
+            ctypes.c_longlong(self.slots), grid, self._stream(),
+        )
+        if rc != 0:
+            raise RuntimeError(f"crawl_insert_hashes failed: hip error {rc}")
+
+
+def fnv1a64(data: bytes) -> int:
+    """Python oracle for the device hash (csrc/common.h fnv1a64)."""
+    h = 1469598103934665603
+    for b in data:
+        h ^= b
+        h = (h * 1099511628211) & 0xFFFFFFFFFFFFFFFF
+    return h
 
 
 def links_to_python(res: EncodeResult) -> List[List[Tuple[str, str]]]:
