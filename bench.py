@@ -80,10 +80,13 @@ def main():
     t_gen = time.time()
     for c in range(n_chunks):
         ids = np.arange(c * chunk_ch, (c + 1) * chunk_ch) + rank * args.channels
-        chunks.append(feed.build_batch(ids, posts_per_channel=posts).to(device))
-        log(f"chunk {c + 1}/{n_chunks} generated "
+        chunks.append(
+            feed.build_batch_device(ids, device, posts_per_channel=posts)
+        )
+        torch.cuda.synchronize()
+        log(f"chunk {c + 1}/{n_chunks} generated on-device "
             f"({(c + 1) * chunk_ch * posts / 1e6:.2f}M posts, "
-            f"{time.time() - t_gen:.0f}s)")
+            f"{time.time() - t_gen:.1f}s)")
     chunk_posts = chunk_ch * posts
 
     now = dt.datetime(2026, 1, 1, tzinfo=dt.timezone.utc)

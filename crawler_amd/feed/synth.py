@@ -283,6 +283,234 @@ class SyntheticFeed:
                                   for t in self.templates])
         self._comment_bytes = [c.encode("utf-8") for c in COMMENT_TEXTS]
 
+    # ---- device-side generation (csrc/feedgen.hip) ----
+
+    def _device_tables(self, device):
+        """Upload the compiled template tables once per device."""
+        if not hasattr(self, "_dev_tables"):
+            self._dev_tables = {}
+        key = str(device)
+        if key in self._dev_tables:
+            return self._dev_tables[key]
+        import torch as T
+
+        def cat_u8(parts):
+            blob = b"".join(parts)
+            return T.frombuffer(bytearray(blob or b"\0"), dtype=T.uint8).to(device)
+
+        def offs(parts):
+            off, cur = [], 0
+            for p in parts:
+                off.append(cur)
+                cur += len(p)
+            return off
+
+        t32 = lambda x: T.tensor(x, dtype=T.int32, device=device)
+        tmpl_pool = [t.pool for t in self.templates]
+        aux_pool = [a for a in self._aux]
+        ent_rows = []
+        ent_off, ent_cnt = [], []
+        for te in self._ents:
+            ent_off.append(len(ent_rows))
+            ent_cnt.append(len(te))
+            ent_rows.extend(int(v) for row in te for v in row)
+        slot_flat, slot_off, slot_cnt = [], [], []
+        for t in self.templates:
+            slot_off.append(len(slot_flat))
+            slot_cnt.append(len(t.slot_positions))
+            slot_flat.extend(t.slot_positions)
+        tables = {
+            "pool": cat_u8(tmpl_pool),
+            "pool_off": t32(offs(tmpl_pool)),
+            "pool_len": t32([len(p) for p in tmpl_pool]),
+            "text_len": t32([t.text_len for t in self.templates]),
+            "ctype": t32([int(v) for v in self._ctype]),
+            "tflags": t32([int(v) for v in self._flags]),
+            "aux": cat_u8(aux_pool),
+            "aux_off": t32(offs(aux_pool)),
+            "aux_len": t32([len(a) for a in aux_pool]),
+            "ents": t32(ent_rows or [0]),
+            "ent_off": t32(ent_off),
+            "ent_cnt": t32(ent_cnt),
+            "slots": t32(slot_flat or [0]),
+            "slot_off": t32(slot_off),
+            "slot_cnt": t32(slot_cnt),
+            "cdf": T.tensor(self._cdf, dtype=T.float64, device=device),
+            "ctext": cat_u8(self._comment_bytes),
+            "ctext_off": t32(offs(self._comment_bytes)),
+            "ctext_len": t32([len(b) for b in self._comment_bytes]),
+        }
+        self._dev_tables[key] = tables
+        return tables
+
+    _TBL_ORDER = [
+        "pool", "pool_off", "pool_len", "text_len", "ctype", "tflags",
+        "aux", "aux_off", "aux_len", "ents", "ent_off", "ent_cnt",
+        "slots", "slot_off", "slot_cnt", "cdf", "ctext", "ctext_off",
+        "ctext_len",
+    ]
+
+    def build_batch_device(self, channel_ids: np.ndarray, device,
+                           posts_per_channel: Optional[int] = None
+                           ) -> B.MessageBatch:
+        """GPU-side batch generation, bit-identical to build_batch()."""
+        import ctypes
+        import torch as T
+
+        from ..ops import gpu as gpu_mod
+
+        lib = gpu_mod.require_lib()
+        cfg = self.cfg
+        P = posts_per_channel or cfg.posts_per_channel
+        K = len(channel_ids)
+        N = K * P
+        permille = cfg.comment_rate * 1000
+        assert permille == int(permille), (
+            "comment_rate must be an integer number of permille for the "
+            "device generator"
+        )
+        tables = self._device_tables(device)
+        tbl_ptrs = (ctypes.c_void_p * len(self._TBL_ORDER))()
+        for k, name in enumerate(self._TBL_ORDER):
+            tbl_ptrs[k] = ctypes.c_void_p(tables[name].data_ptr())
+        scalars = (ctypes.c_long * 10)(
+            len(self.templates), len(self._comment_bytes), cfg.seed,
+            cfg.universe, cfg.base_date, cfg.date_step, int(permille),
+            cfg.max_comments_per_post, K, P,
+        )
+        cids_t = T.tensor(np.asarray(channel_ids, dtype=np.int64),
+                          device=device)
+        stream = ctypes.c_void_p(T.cuda.current_stream().cuda_stream)
+
+        z64 = lambda n: T.zeros(n, dtype=T.int64, device=device)
+        z32 = lambda n: T.zeros(n, dtype=T.int32, device=device)
+        chat_id, msg_id, block_len = z64(N), z64(N), z64(N)
+        m = {f: z32(N) for f in B._META_FIELDS_I32}
+        tidx = z32(N)
+        out_ptrs = (ctypes.c_void_p * 18)(
+            *[ctypes.c_void_p(x.data_ptr()) for x in [
+                chat_id, msg_id, m["date"], m["content_type"], m["views"],
+                m["forwards"], m["media_album_id"], m["channel_idx"],
+                m["flags"], m["text_len"], m["aux_len"], m["ent_cnt"],
+                m["react_cnt"], m["com_cnt"], m["poster_len"],
+                m["reply_count"], block_len, tidx,
+            ]]
+        )
+        grid = min((N + 255) // 256, 8192)
+        rc = lib.crawl_feed_meta(tbl_ptrs, scalars,
+                                 ctypes.c_void_p(cids_t.data_ptr()),
+                                 out_ptrs, grid, stream)
+        if rc != 0:
+            raise RuntimeError(f"crawl_feed_meta failed: hip {rc}")
+
+        def excl_cumsum(x64):
+            out = T.zeros(x64.numel() + 1, dtype=T.int64, device=device)
+            T.cumsum(x64, 0, out=out[1:])
+            return out
+
+        block_off = excl_cumsum(block_len)
+        ent_off_l = excl_cumsum(m["ent_cnt"].to(T.int64))
+        react_off_l = excl_cumsum(m["react_cnt"].to(T.int64))
+        com_off_l = excl_cumsum(m["com_cnt"].to(T.int64))
+        msg_total = int(block_off[-1].item())
+        E = int(ent_off_l[-1].item())
+        R = int(react_off_l[-1].item())
+        C = int(com_off_l[-1].item())
+        cblock = max(len(b) for b in self._comment_bytes) + HANDLE_WIDTH
+
+        # channel table bytes (host-small)
+        rows = self.channel_rows(np.asarray(channel_ids), P)
+        ch_parts = []
+        ch_user_off = np.zeros(K, dtype=np.int32)
+        ch_user_len = np.zeros(K, dtype=np.int32)
+        ch_title_off = np.zeros(K, dtype=np.int32)
+        ch_title_len = np.zeros(K, dtype=np.int32)
+        koff = msg_total + C * cblock
+        for c, row in enumerate(rows):
+            ub, tb2 = row.username.encode(), row.title.encode()
+            ch_user_off[c] = koff; ch_user_len[c] = len(ub)
+            ch_parts.append(ub); koff += len(ub)
+            ch_title_off[c] = koff; ch_title_len[c] = len(tb2)
+            ch_parts.append(tb2); koff += len(tb2)
+        ch_blob = b"".join(ch_parts)
+        total_pool = msg_total + C * cblock + len(ch_blob)
+        if total_pool >= (1 << 31):
+            raise ValueError("pool exceeds int32 offsets; shrink the batch")
+        pool = T.zeros(total_pool, dtype=T.uint8, device=device)
+        pool[msg_total + C * cblock:] = T.frombuffer(
+            bytearray(ch_blob), dtype=T.uint8
+        ).to(device)
+
+        text_off = z64(N)
+        entities = z32(max(E, 1) * 5).view(-1, 5)[:E] if E else T.zeros(
+            (0, 5), dtype=T.int32, device=device)
+        ent_store = entities if E else z32(5).view(1, 5)
+        react_emoji, react_count = z32(max(R, 1)), z32(max(R, 1))
+        fill_ptrs = (ctypes.c_void_p * 9)(
+            *[ctypes.c_void_p(x.data_ptr()) for x in [
+                pool, text_off, m["aux_off"], m["poster_off"], m["ent_off"],
+                m["react_off"], ent_store, react_emoji, react_count,
+            ]]
+        )
+        rc = lib.crawl_feed_fill(
+            tbl_ptrs, scalars, ctypes.c_void_p(cids_t.data_ptr()),
+            ctypes.c_void_p(tidx.data_ptr()),
+            ctypes.c_void_p(block_off.data_ptr()),
+            ctypes.c_void_p(ent_off_l.data_ptr()),
+            ctypes.c_void_p(react_off_l.data_ptr()),
+            fill_ptrs, min((N + 3) // 4, 8192), stream,
+        )
+        if rc != 0:
+            raise RuntimeError(f"crawl_feed_fill failed: hip {rc}")
+
+        com = {f: z32(max(C, 1)) for f in [
+            "text_off", "text_len", "handle_off", "handle_len", "views",
+            "replies", "react_off", "react_cnt"]}
+        if C:
+            cmsg_of = T.repeat_interleave(
+                T.arange(N, dtype=T.int64, device=device),
+                m["com_cnt"].to(T.int64))
+            com_ptrs = (ctypes.c_void_p * 9)(
+                *[ctypes.c_void_p(x.data_ptr()) for x in [
+                    pool, com["text_off"], com["text_len"],
+                    com["handle_off"], com["handle_len"], com["views"],
+                    com["replies"], com["react_off"], com["react_cnt"],
+                ]]
+            )
+            rc = lib.crawl_feed_comments(
+                tbl_ptrs, scalars, ctypes.c_void_p(cids_t.data_ptr()),
+                ctypes.c_void_p(com_off_l.data_ptr()), cblock,
+                ctypes.c_void_p(cmsg_of.data_ptr()), msg_total, C,
+                com_ptrs, min((C + 255) // 256, 8192), stream,
+            )
+            if rc != 0:
+                raise RuntimeError(f"crawl_feed_comments failed: hip {rc}")
+
+        m["com_off"] = com_off_l[:-1].to(T.int32)
+        t = lambda arr, dt_: T.tensor(arr, dtype=dt_, device=device)
+        return B.MessageBatch(
+            n=N, chat_id=chat_id, msg_id=msg_id, text_off=text_off,
+            meta=m, text_pool=pool,
+            entities=entities if E else T.zeros((0, 5), dtype=T.int32,
+                                                device=device),
+            react_emoji=react_emoji[:R], react_count=react_count[:R],
+            com_text_off=com["text_off"][:C], com_text_len=com["text_len"][:C],
+            com_handle_off=com["handle_off"][:C],
+            com_handle_len=com["handle_len"][:C],
+            com_views=com["views"][:C], com_replies=com["replies"][:C],
+            com_react_off=com["react_off"][:C],
+            com_react_cnt=com["react_cnt"][:C],
+            n_channels=K,
+            ch_chat_id=t(np.array([r.chat_id for r in rows]), T.int64),
+            ch_member=t(np.array([r.member_count for r in rows]), T.int32),
+            ch_postcount=t(np.array([r.post_count for r in rows]), T.int32),
+            ch_totalviews=t(np.array([r.total_views for r in rows]), T.int32),
+            ch_user_off=t(ch_user_off, T.int32),
+            ch_user_len=t(ch_user_len, T.int32),
+            ch_title_off=t(ch_title_off, T.int32),
+            ch_title_len=t(ch_title_len, T.int32),
+        )
+
     # ---- channel metadata ----
 
     @staticmethod
@@ -466,10 +694,12 @@ class SyntheticFeed:
         ctext_idx = (ch % np.uint64(len(COMMENT_TEXTS))).astype(np.int64)
         clens = np.array([len(b) for b in self._comment_bytes])
         ctext_len = clens[ctext_idx]
-        # comment pool: text + handle per comment, appended after message pool
-        cblock_len = ctext_len + HANDLE_WIDTH
-        cblock_off = np.zeros(C + 1, dtype=np.int64)
-        np.cumsum(cblock_len, out=cblock_off[1:])
+        # comment pool: FIXED-WIDTH blocks (max text len + handle) appended
+        # after the message pool — fixed stride keeps the device generator
+        # (csrc/feedgen.hip) cumsum-free for comments; unused tail bytes of a
+        # block are zero and never referenced.
+        cblock = int(max(len(b) for b in self._comment_bytes)) + HANDLE_WIDTH
+        cblock_off = np.arange(C + 1, dtype=np.int64) * cblock
         cpool = np.zeros(int(cblock_off[-1]), dtype=np.uint8)
         for k, cb in enumerate(self._comment_bytes):
             rows = np.nonzero(ctext_idx == k)[0]

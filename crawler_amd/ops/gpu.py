@@ -64,6 +64,7 @@ def load_lib():
         ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
     ]
     _bind_dedup(lib)
+    _bind_feedgen(lib)
     _lib = lib
     return lib
 
@@ -83,6 +84,29 @@ def _bind_dedup(lib):
     lib.crawl_insert_hashes.restype = ctypes.c_int
     lib.crawl_insert_hashes.argtypes = [
         ctypes.c_void_p, ctypes.c_long, ctypes.c_void_p, ctypes.c_longlong,
+        ctypes.c_int, ctypes.c_void_p,
+    ]
+
+
+def _bind_feedgen(lib):
+    lib.crawl_feed_meta.restype = ctypes.c_int
+    lib.crawl_feed_meta.argtypes = [
+        ctypes.POINTER(ctypes.c_void_p), ctypes.POINTER(ctypes.c_long),
+        ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p), ctypes.c_int,
+        ctypes.c_void_p,
+    ]
+    lib.crawl_feed_fill.restype = ctypes.c_int
+    lib.crawl_feed_fill.argtypes = [
+        ctypes.POINTER(ctypes.c_void_p), ctypes.POINTER(ctypes.c_long),
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.POINTER(ctypes.c_void_p), ctypes.c_int,
+        ctypes.c_void_p,
+    ]
+    lib.crawl_feed_comments.restype = ctypes.c_int
+    lib.crawl_feed_comments.argtypes = [
+        ctypes.POINTER(ctypes.c_void_p), ctypes.POINTER(ctypes.c_long),
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_long, ctypes.c_void_p,
+        ctypes.c_long, ctypes.c_long, ctypes.POINTER(ctypes.c_void_p),
         ctypes.c_int, ctypes.c_void_p,
     ]
 
