@@ -310,9 +310,11 @@ class SyntheticFeed:
         aux_pool = [a for a in self._aux]
         ent_rows = []
         ent_off, ent_cnt = [], []
+        nrows = 0
         for te in self._ents:
-            ent_off.append(len(ent_rows))
+            ent_off.append(nrows)  # ROW offset (kernel indexes rows * 5)
             ent_cnt.append(len(te))
+            nrows += len(te)
             ent_rows.extend(int(v) for row in te for v in row)
         slot_flat, slot_off, slot_cnt = [], [], []
         for t in self.templates:
