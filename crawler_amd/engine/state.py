@@ -1,0 +1,844 @@
+"""State management: layer map, page/message status, checkpoint/resume,
+media cache, random-walk store.
+
+Parity targets (reference file:line):
+- Page / Message / EdgeRecord / PendingEdge* / CrawlMetadata / State models:
+  state/datamodels.go:41-214
+- BaseStateManager in-mem maps, AddLayer URL dedup + MaxPages deadend
+  replacement budget: state/base.go:219-322
+- DiscoveredChannels O(1) random pick: state/datamodels.go:118-162
+- LocalStateManager file layout (state.json / metadata.json /
+  media-cache.json under basePath/crawlID; posts ->
+  crawlID/channel/posts/posts.jsonl; media -> crawlID/media/channel/):
+  state/storageproviders.go:74-647
+- FindIncompleteCrawl resume rules (never resume completed; requires a
+  non-empty layer map): state/base.go:466-516, daprstate.go:1703-2199
+- RandomWalkStore: the Postgres tables (page_buffer, edge_records,
+  seed_channels, invalid_channels, pending_edge_batches, pending_edges)
+  re-hosted as in-process structures with the same claim semantics
+  (FOR UPDATE SKIP LOCKED -> lock + claimed-set; attempt_count poison
+  detection): sql/random-walk-schema.sql, sql/validator-schema.sql,
+  state/daprstate.go:3150-4391. The per-name exactly-once discovery set
+  itself lives on the GPU (ops/gpu.SeenSet) in GPU execution.
+"""
+from __future__ import annotations
+
+import dataclasses
+import datetime as _dt
+import json
+import os
+import random
+import threading
+import uuid
+from typing import Dict, List, Optional, Tuple
+
+from ..models.post import Post
+
+UTC = _dt.timezone.utc
+
+
+def _now() -> _dt.datetime:
+    return _dt.datetime.now(UTC)
+
+
+@dataclasses.dataclass
+class PageMessage:
+    """state/datamodels.go:67-73."""
+
+    chat_id: int = 0
+    message_id: int = 0
+    status: str = "unfetched"
+    page_id: str = ""
+    platform: str = ""
+
+    def to_dict(self):
+        return {
+            "chatId": self.chat_id, "messageId": self.message_id,
+            "status": self.status, "pageId": self.page_id,
+            "platform": self.platform,
+        }
+
+    @classmethod
+    def from_dict(cls, d):
+        return cls(
+            chat_id=d.get("chatId", 0), message_id=d.get("messageId", 0),
+            status=d.get("status", "unfetched"),
+            page_id=d.get("pageId", ""), platform=d.get("platform", ""),
+        )
+
+
+@dataclasses.dataclass
+class Page:
+    """state/datamodels.go:41-64. Status machine: unfetched -> processing ->
+    fetched | error | deadend (crawl/runner.go, dapr/standalone.go)."""
+
+    id: str = ""
+    url: str = ""
+    depth: int = 0
+    status: str = "unfetched"
+    error: str = ""
+    timestamp: Optional[_dt.datetime] = None
+    platform: str = ""
+    parent_id: str = ""
+    messages: List[PageMessage] = dataclasses.field(default_factory=list)
+    connection_id: str = ""
+    sequence_id: str = ""
+    crawl_id: str = ""
+
+    def to_dict(self):
+        return {
+            "id": self.id, "url": self.url, "depth": self.depth,
+            "status": self.status, "error": self.error,
+            "timestamp": self.timestamp.isoformat() if self.timestamp else None,
+            "platform": self.platform, "parentId": self.parent_id,
+            "messages": [m.to_dict() for m in self.messages],
+            "LastConnectionID": self.connection_id,
+            "sequenceId": self.sequence_id, "crawlId": self.crawl_id,
+        }
+
+    @classmethod
+    def from_dict(cls, d):
+        ts = d.get("timestamp")
+        return cls(
+            id=d.get("id", ""), url=d.get("url", ""),
+            depth=d.get("depth", 0), status=d.get("status", "unfetched"),
+            error=d.get("error", ""),
+            timestamp=_dt.datetime.fromisoformat(ts) if ts else None,
+            platform=d.get("platform", ""), parent_id=d.get("parentId", ""),
+            messages=[PageMessage.from_dict(m)
+                      for m in d.get("messages", [])],
+            connection_id=d.get("LastConnectionID", ""),
+            sequence_id=d.get("sequenceId", ""),
+            crawl_id=d.get("crawlId", ""),
+        )
+
+
+@dataclasses.dataclass
+class EdgeRecord:
+    """state/datamodels.go:76-84."""
+
+    destination_channel: str = ""
+    discovery_time: Optional[_dt.datetime] = None
+    source_channel: str = ""
+    walkback: bool = False
+    skipped: bool = False
+    sequence_id: str = ""
+    crawl_id: str = ""
+
+
+@dataclasses.dataclass
+class PendingEdgeBatch:
+    """state/datamodels.go:89-98. open -> closed -> processing -> completed."""
+
+    batch_id: str = ""
+    crawl_id: str = ""
+    source_channel: str = ""
+    source_page_id: str = ""
+    source_depth: int = 0
+    sequence_id: str = ""
+    status: str = "open"
+    attempt_count: int = 0
+
+
+@dataclasses.dataclass
+class PendingEdge:
+    """state/datamodels.go:101-113."""
+
+    pending_id: int = 0
+    batch_id: str = ""
+    crawl_id: str = ""
+    destination_channel: str = ""
+    source_channel: str = ""
+    sequence_id: str = ""
+    discovery_time: Optional[_dt.datetime] = None
+    source_type: str = ""
+    validation_status: str = "pending"
+    validation_reason: str = ""
+
+
+@dataclasses.dataclass
+class CrawlMetadata:
+    """state/datamodels.go:172-183."""
+
+    crawl_id: str = ""
+    execution_id: str = ""
+    start_time: Optional[_dt.datetime] = None
+    end_time: Optional[_dt.datetime] = None
+    status: str = "running"
+    previous_crawl_id: List[str] = dataclasses.field(default_factory=list)
+    platform: str = ""
+    target_channels: List[str] = dataclasses.field(default_factory=list)
+    messages_count: int = 0
+    errors_count: int = 0
+
+    def to_dict(self):
+        return {
+            "crawlId": self.crawl_id, "executionId": self.execution_id,
+            "startTime": self.start_time.isoformat() if self.start_time else None,
+            "endTime": self.end_time.isoformat() if self.end_time else None,
+            "status": self.status, "previousCrawlId": self.previous_crawl_id,
+            "platform": self.platform,
+            "targetChannels": self.target_channels,
+            "messagesCount": self.messages_count,
+            "errorsCount": self.errors_count,
+        }
+
+    @classmethod
+    def from_dict(cls, d):
+        def ts(k):
+            v = d.get(k)
+            return _dt.datetime.fromisoformat(v) if v else None
+        return cls(
+            crawl_id=d.get("crawlId", ""),
+            execution_id=d.get("executionId", ""),
+            start_time=ts("startTime"), end_time=ts("endTime"),
+            status=d.get("status", "running"),
+            previous_crawl_id=d.get("previousCrawlId") or [],
+            platform=d.get("platform", ""),
+            target_channels=d.get("targetChannels") or [],
+            messages_count=d.get("messagesCount", 0),
+            errors_count=d.get("errorsCount", 0),
+        )
+
+
+class DiscoveredChannels:
+    """Map + key slice for O(1) random pick (state/datamodels.go:118-162)."""
+
+    def __init__(self):
+        self._items: Dict[str, bool] = {}
+        self._keys: List[str] = []
+        self._lock = threading.RLock()
+
+    def add(self, name: str) -> bool:
+        with self._lock:
+            if name in self._items:
+                return False
+            self._items[name] = True
+            self._keys.append(name)
+            return True
+
+    def contains(self, name: str) -> bool:
+        with self._lock:
+            return name in self._items
+
+    def random(self, rng: Optional[random.Random] = None) -> Optional[str]:
+        with self._lock:
+            if not self._keys:
+                return None
+            r = rng or random
+            return self._keys[r.randrange(len(self._keys))]
+
+    def __len__(self):
+        with self._lock:
+            return len(self._keys)
+
+    def keys(self) -> List[str]:
+        with self._lock:
+            return list(self._keys)
+
+
+class BaseStateManager:
+    """In-memory layer map / page map core (state/base.go:15-552)."""
+
+    def __init__(self, config):
+        self.config = config
+        self.pages: Dict[str, Page] = {}
+        self.layer_map: Dict[int, List[str]] = {}
+        self.metadata = CrawlMetadata(
+            crawl_id=getattr(config, "crawl_id", ""),
+            execution_id=getattr(config, "crawl_id", ""),
+            start_time=_now(),
+            platform=getattr(config, "platform", "telegram"),
+        )
+        self.discovered = DiscoveredChannels()
+        self._lock = threading.RLock()
+        self._chat_id_cache: Dict[str, int] = {}
+        self._seed_channels: Dict[str, bool] = {}
+
+    # ---- layer management ----
+
+    def initialize(self, seed_urls: List[str]) -> None:
+        pages = [
+            Page(id=str(uuid.uuid4()), url=u, depth=0, status="unfetched",
+                 timestamp=_now())
+            for u in seed_urls
+        ]
+        self.add_layer(pages)
+        self.metadata.target_channels = list(seed_urls)
+
+    def add_layer(self, pages: List[Page]) -> List[str]:
+        """URL-dedup + MaxPages deadend-replacement budget
+        (state/base.go:219-322). Returns added page IDs."""
+        if not pages:
+            return []
+        with self._lock:
+            total = len(self.pages)
+            deadends = sum(1 for p in self.pages.values()
+                           if p.status == "deadend")
+            max_pages = getattr(self.config, "max_pages", 0) or 0
+            max_reached = max_pages > 0 and total >= max_pages
+            replacements = deadends
+            existing_urls = {p.url: pid for pid, p in self.pages.items()}
+            depth = pages[0].depth
+            self.layer_map.setdefault(depth, [])
+            added = []
+            for p in pages:
+                if p.url in existing_urls:
+                    continue
+                if max_reached:
+                    if replacements <= 0:
+                        continue
+                    replacements -= 1
+                if not p.id:
+                    p.id = str(uuid.uuid4())
+                if p.timestamp is None:
+                    p.timestamp = _now()
+                self.pages[p.id] = p
+                existing_urls[p.url] = p.id
+                self.layer_map[depth].append(p.id)
+                added.append(p.id)
+            return added
+
+    def get_page(self, page_id: str) -> Page:
+        with self._lock:
+            if page_id not in self.pages:
+                raise KeyError(f"page not found: {page_id}")
+            return self.pages[page_id]
+
+    def update_page(self, page: Page) -> None:
+        with self._lock:
+            self.pages[page.id] = page
+
+    def update_message(self, page_id: str, chat_id: int, message_id: int,
+                       status: str) -> None:
+        with self._lock:
+            page = self.pages.get(page_id)
+            if page is None:
+                raise KeyError(f"page not found: {page_id}")
+            for m in page.messages:
+                if m.chat_id == chat_id and m.message_id == message_id:
+                    m.status = status
+                    return
+            page.messages.append(PageMessage(
+                chat_id=chat_id, message_id=message_id, status=status,
+                page_id=page_id,
+            ))
+
+    def get_layer_by_depth(self, depth: int) -> List[Page]:
+        with self._lock:
+            return [self.pages[i] for i in self.layer_map.get(depth, [])]
+
+    def get_max_depth(self) -> int:
+        with self._lock:
+            return max(self.layer_map.keys(), default=-1)
+
+    # ---- discovered channels / seed cache ----
+
+    def is_discovered_channel(self, name: str) -> bool:
+        return self.discovered.contains(name)
+
+    def add_discovered_channel(self, name: str) -> None:
+        self.discovered.add(name)
+
+    def get_random_discovered_channel(self, rng=None) -> Optional[str]:
+        return self.discovered.random(rng)
+
+    def upsert_seed_channel_chat_id(self, username: str, chat_id: int):
+        with self._lock:
+            self._chat_id_cache[username] = chat_id
+            self._seed_channels[username] = True
+
+    def get_cached_chat_id(self, username: str) -> Tuple[int, bool]:
+        with self._lock:
+            v = self._chat_id_cache.get(username)
+            return (v, True) if v is not None else (0, False)
+
+    def is_seed_channel(self, username: str) -> bool:
+        with self._lock:
+            return username in self._seed_channels
+
+
+class LocalStateManager(BaseStateManager):
+    """File-backed state manager (state/storageproviders.go:74-647).
+
+    Layout under base_path/crawl_id/:
+      state.json, metadata.json, media-cache.json, progress.json (legacy
+      format documented in README "Directory Structure"),
+      <channel>/posts/posts.jsonl, media/<channel>/<file>
+    """
+
+    def __init__(self, config, base_path: Optional[str] = None):
+        super().__init__(config)
+        self.base_path = base_path or getattr(
+            config, "storage_root", "/tmp/crawl"
+        )
+        self.media_cache: Dict[str, dict] = {}
+        self._post_files = {}
+
+    # paths (storageproviders.go:634-647)
+    def _crawl_dir(self):
+        return os.path.join(self.base_path, self.metadata.crawl_id)
+
+    def _state_path(self):
+        return os.path.join(self._crawl_dir(), "state.json")
+
+    def _metadata_path(self):
+        return os.path.join(self._crawl_dir(), "metadata.json")
+
+    def _media_cache_path(self):
+        return os.path.join(self._crawl_dir(), "media-cache.json")
+
+    def _progress_path(self):
+        return os.path.join(self._crawl_dir(), "progress.json")
+
+    def _posts_path(self, channel: str):
+        return os.path.join(self._crawl_dir(), channel, "posts",
+                            "posts.jsonl")
+
+    # ---- persistence ----
+
+    def save_state(self) -> None:
+        os.makedirs(self._crawl_dir(), exist_ok=True)
+        with self._lock:
+            state = {
+                "layers": [
+                    {
+                        "depth": depth,
+                        "pages": [self.pages[i].to_dict()
+                                  for i in ids],
+                    }
+                    for depth, ids in sorted(self.layer_map.items())
+                ],
+                "metadata": self.metadata.to_dict(),
+                "lastUpdated": _now().isoformat(),
+            }
+        tmp = self._state_path() + ".tmp"
+        with open(tmp, "w") as f:
+            json.dump(state, f)
+        os.replace(tmp, self._state_path())
+        with open(self._metadata_path() + ".tmp", "w") as f:
+            json.dump(self.metadata.to_dict(), f)
+        os.replace(self._metadata_path() + ".tmp", self._metadata_path())
+        self._save_progress()
+
+    def _save_progress(self) -> None:
+        """Legacy progress.json checkpoint (README "Directory Structure",
+        docs/architecture.md:166-177): per-depth completion cursor."""
+        with self._lock:
+            layers = []
+            for depth, ids in sorted(self.layer_map.items()):
+                done = sum(
+                    1 for i in ids
+                    if self.pages[i].status in
+                    ("fetched", "error", "deadend")
+                )
+                layers.append({
+                    "depth": depth, "total": len(ids), "completed": done,
+                })
+            progress = {
+                "crawlId": self.metadata.crawl_id,
+                "executionId": self.metadata.execution_id,
+                "status": self.metadata.status,
+                "layers": layers,
+                "updatedAt": _now().isoformat(),
+            }
+        with open(self._progress_path() + ".tmp", "w") as f:
+            json.dump(progress, f)
+        os.replace(self._progress_path() + ".tmp", self._progress_path())
+
+    def load_state(self) -> bool:
+        if not os.path.exists(self._state_path()):
+            return False
+        with open(self._state_path()) as f:
+            state = json.load(f)
+        with self._lock:
+            self.pages.clear()
+            self.layer_map.clear()
+            for layer in state.get("layers", []):
+                ids = []
+                for pd in layer.get("pages", []):
+                    page = Page.from_dict(pd)
+                    self.pages[page.id] = page
+                    ids.append(page.id)
+                self.layer_map[layer.get("depth", 0)] = ids
+            self.metadata = CrawlMetadata.from_dict(
+                state.get("metadata", {})
+            )
+        if os.path.exists(self._media_cache_path()):
+            try:
+                with open(self._media_cache_path()) as f:
+                    self.media_cache = json.load(f)
+            except (OSError, json.JSONDecodeError):
+                self.media_cache = {}
+        return True
+
+    def find_incomplete_crawl(self, crawl_id: str) -> Tuple[str, bool]:
+        """Resume rules (state/base.go:466-516): a crawl is resumable iff
+        its state exists, has a non-empty layer map, and is not completed.
+        Returns (execution_id, exists)."""
+        path = os.path.join(self.base_path, crawl_id, "state.json")
+        if not os.path.exists(path):
+            return "", False
+        try:
+            with open(path) as f:
+                state = json.load(f)
+        except (OSError, json.JSONDecodeError):
+            return "", False
+        meta = state.get("metadata", {})
+        if meta.get("status") == "completed":
+            return "", False
+        if not state.get("layers"):
+            return "", False
+        return meta.get("executionId") or crawl_id, True
+
+    def get_previous_crawls(self) -> List[str]:
+        if not os.path.isdir(self.base_path):
+            return []
+        return sorted(
+            d for d in os.listdir(self.base_path)
+            if os.path.isdir(os.path.join(self.base_path, d))
+            and d != self.metadata.crawl_id
+        )
+
+    def update_crawl_metadata(self, crawl_id: str, md: dict) -> None:
+        with self._lock:
+            if "status" in md:
+                self.metadata.status = md["status"]
+            if "endTime" in md:
+                v = md["endTime"]
+                self.metadata.end_time = (
+                    v if isinstance(v, _dt.datetime)
+                    else _dt.datetime.fromisoformat(v)
+                )
+            if "messagesCount" in md:
+                self.metadata.messages_count = md["messagesCount"]
+            if "errorsCount" in md:
+                self.metadata.errors_count = md["errorsCount"]
+
+    # ---- posts / files ----
+
+    def store_post(self, channel: str, post: Post) -> None:
+        path = self._posts_path(channel)
+        f = self._post_files.get(path)
+        if f is None:
+            os.makedirs(os.path.dirname(path), exist_ok=True)
+            f = open(path, "ab")
+            self._post_files[path] = f
+        f.write(post.to_jsonl().encode("utf-8"))
+
+    def store_post_lines(self, channel: str, data: bytes) -> None:
+        """Bulk JSONL append (the GPU path hands whole encoded blocks)."""
+        path = self._posts_path(channel)
+        f = self._post_files.get(path)
+        if f is None:
+            os.makedirs(os.path.dirname(path), exist_ok=True)
+            f = open(path, "ab")
+            self._post_files[path] = f
+        f.write(data)
+
+    def store_file(self, channel: str, source_path: str,
+                   file_name: str) -> Tuple[str, str]:
+        dst_dir = os.path.join(self._crawl_dir(), "media", channel)
+        os.makedirs(dst_dir, exist_ok=True)
+        dst = os.path.join(dst_dir, file_name)
+        os.replace(source_path, dst)
+        return dst, file_name
+
+    # ---- media cache ----
+
+    def has_processed_media(self, media_id: str) -> bool:
+        return media_id in self.media_cache
+
+    def mark_media_as_processed(self, media_id: str) -> None:
+        self.media_cache[media_id] = {
+            "id": media_id, "firstSeen": _now().isoformat(),
+        }
+
+    def close(self) -> None:
+        for f in self._post_files.values():
+            f.close()
+        self._post_files.clear()
+        os.makedirs(self._crawl_dir(), exist_ok=True)
+        with open(self._media_cache_path() + ".tmp", "w") as f:
+            json.dump(self.media_cache, f)
+        os.replace(self._media_cache_path() + ".tmp",
+                   self._media_cache_path())
+
+
+class RandomWalkStore:
+    """In-process re-host of the random-walk / validator Postgres tables.
+
+    Claim operations reproduce `FOR UPDATE SKIP LOCKED` + attempt_count
+    semantics (state/daprstate.go:3944-4391) with a lock + status fields:
+    a claim transitions rows atomically so concurrent claimers never see
+    the same row. Poison detection: attempt_count >= 3 parks the batch.
+    """
+
+    MAX_ATTEMPTS = 3
+
+    def __init__(self):
+        self._lock = threading.RLock()
+        self.page_buffer: Dict[str, Page] = {}       # page_id -> Page
+        self.edge_records: List[EdgeRecord] = []
+        self.seed_channels: Dict[str, dict] = {}     # username -> row
+        self.invalid_channels: Dict[str, _dt.datetime] = {}
+        self.pending_batches: Dict[str, PendingEdgeBatch] = {}
+        self.pending_edges: Dict[int, PendingEdge] = {}
+        self.discovered_channels: Dict[str, dict] = {}
+        self.access_events: List[dict] = []
+        self.source_type_stats: Dict[Tuple[str, str], int] = {}
+        self._next_edge_id = 1
+
+    # ---- page_buffer (sql/random-walk-schema.sql page_buffer) ----
+
+    def add_page(self, page: Page) -> None:
+        with self._lock:
+            if not page.id:
+                page.id = str(uuid.uuid4())
+            self.page_buffer[page.id] = page
+
+    def get_pages(self, limit: int) -> List[Page]:
+        with self._lock:
+            return list(self.page_buffer.values())[:limit]
+
+    def delete_pages(self, page_ids: List[str]) -> None:
+        with self._lock:
+            for pid in page_ids:
+                self.page_buffer.pop(pid, None)
+
+    def buffer_size(self) -> int:
+        with self._lock:
+            return len(self.page_buffer)
+
+    # ---- edge_records ----
+
+    def save_edge_records(self, edges: List[EdgeRecord]) -> None:
+        with self._lock:
+            for e in edges:
+                if e.discovery_time is None:
+                    e.discovery_time = _now()
+                self.edge_records.append(e)
+
+    def get_random_skipped_edge(self, exclude: set,
+                                rng=None) -> Optional[EdgeRecord]:
+        """Promote a skipped edge (daprstate.go:3206-3277 semantics)."""
+        r = rng or random
+        with self._lock:
+            candidates = [
+                e for e in self.edge_records
+                if e.skipped and e.destination_channel not in exclude
+            ]
+            return r.choice(candidates) if candidates else None
+
+    # ---- seed_channels ----
+
+    def upsert_seed_channel(self, username: str, chat_id: int = 0) -> None:
+        with self._lock:
+            row = self.seed_channels.setdefault(username, {
+                "username": username, "chat_id": 0,
+                "last_crawled_at": None, "invalidated_at": None,
+            })
+            if chat_id:
+                row["chat_id"] = chat_id
+
+    def mark_channel_crawled(self, username: str, chat_id: int) -> None:
+        with self._lock:
+            self.upsert_seed_channel(username, chat_id)
+            self.seed_channels[username]["last_crawled_at"] = _now()
+
+    def get_channel_last_crawled(self, username: str) -> Optional[_dt.datetime]:
+        with self._lock:
+            row = self.seed_channels.get(username)
+            return row["last_crawled_at"] if row else None
+
+    def mark_seed_channel_invalid(self, username: str) -> None:
+        with self._lock:
+            row = self.seed_channels.get(username)
+            if row:
+                row["invalidated_at"] = _now()
+
+    def load_seed_channels(self, ttl_days: int = 30) -> List[dict]:
+        """Rows whose invalidation is absent or older than the TTL
+        (daprstate.go:3327-3429)."""
+        cutoff = _now() - _dt.timedelta(days=ttl_days)
+        with self._lock:
+            return [
+                dict(row) for row in self.seed_channels.values()
+                if row["invalidated_at"] is None
+                or row["invalidated_at"] < cutoff
+            ]
+
+    # ---- invalid_channels (30-day TTL cache) ----
+
+    def mark_invalid_channel(self, username: str) -> None:
+        with self._lock:
+            self.invalid_channels[username] = _now()
+
+    def is_invalid_channel(self, username: str, ttl_days: int = 30) -> bool:
+        with self._lock:
+            t = self.invalid_channels.get(username)
+            if t is None:
+                return False
+            return t > _now() - _dt.timedelta(days=ttl_days)
+
+    # ---- pending edge batches / edges (tandem validator) ----
+
+    def open_batch(self, crawl_id: str, source_channel: str,
+                   source_page_id: str, source_depth: int,
+                   sequence_id: str) -> str:
+        with self._lock:
+            bid = str(uuid.uuid4())
+            self.pending_batches[bid] = PendingEdgeBatch(
+                batch_id=bid, crawl_id=crawl_id,
+                source_channel=source_channel,
+                source_page_id=source_page_id, source_depth=source_depth,
+                sequence_id=sequence_id, status="open",
+            )
+            return bid
+
+    def insert_pending_edge(self, batch_id: str, crawl_id: str,
+                            destination: str, source: str,
+                            sequence_id: str, source_type: str) -> int:
+        with self._lock:
+            eid = self._next_edge_id
+            self._next_edge_id += 1
+            self.pending_edges[eid] = PendingEdge(
+                pending_id=eid, batch_id=batch_id, crawl_id=crawl_id,
+                destination_channel=destination, source_channel=source,
+                sequence_id=sequence_id, discovery_time=_now(),
+                source_type=source_type, validation_status="pending",
+            )
+            return eid
+
+    def close_batch(self, batch_id: str) -> None:
+        with self._lock:
+            b = self.pending_batches.get(batch_id)
+            if b and b.status == "open":
+                b.status = "closed"
+
+    def claim_pending_edges(self, limit: int) -> List[PendingEdge]:
+        """SKIP LOCKED claim (daprstate.go:3944-3996)."""
+        with self._lock:
+            claimed = []
+            for e in self.pending_edges.values():
+                if e.validation_status == "pending":
+                    e.validation_status = "validating"
+                    claimed.append(dataclasses.replace(e))
+                    if len(claimed) >= limit:
+                        break
+            return claimed
+
+    def update_pending_edges(self, updates: List[Tuple[int, str, str]]):
+        with self._lock:
+            for (eid, status, reason) in updates:
+                e = self.pending_edges.get(eid)
+                if e:
+                    e.validation_status = status
+                    e.validation_reason = reason
+
+    def claim_walkback_batch(self) -> Optional[PendingEdgeBatch]:
+        """Closed batch with no pending/validating edges
+        (daprstate.go:4016-4033)."""
+        with self._lock:
+            for b in self.pending_batches.values():
+                if b.status != "closed":
+                    continue
+                if b.attempt_count >= self.MAX_ATTEMPTS:
+                    continue
+                busy = any(
+                    e.batch_id == b.batch_id
+                    and e.validation_status in ("pending", "validating")
+                    for e in self.pending_edges.values()
+                )
+                if busy:
+                    continue
+                b.status = "processing"
+                b.attempt_count += 1
+                return dataclasses.replace(b)
+            return None
+
+    def complete_batch(self, batch_id: str) -> None:
+        with self._lock:
+            b = self.pending_batches.get(batch_id)
+            if b:
+                b.status = "completed"
+
+    def count_incomplete_batches(self, crawl_id: str) -> int:
+        with self._lock:
+            return sum(
+                1 for b in self.pending_batches.values()
+                if b.crawl_id == crawl_id
+                and b.status in ("open", "closed", "processing")
+            )
+
+    def edges_of_batch(self, batch_id: str) -> List[PendingEdge]:
+        with self._lock:
+            return [dataclasses.replace(e)
+                    for e in self.pending_edges.values()
+                    if e.batch_id == batch_id]
+
+    # ---- stale / orphan recovery (daprstate.go:4264-4391) ----
+
+    def recover_stale_claims(self) -> Tuple[int, int]:
+        """Reset 'validating' edges to 'pending' and 'processing' batches to
+        'closed' (attempt_count retained for poison detection)."""
+        with self._lock:
+            ne = nb = 0
+            for e in self.pending_edges.values():
+                if e.validation_status == "validating":
+                    e.validation_status = "pending"
+                    ne += 1
+            for b in self.pending_batches.values():
+                if b.status == "processing":
+                    b.status = "closed"
+                    nb += 1
+            return ne, nb
+
+    def delete_orphan_edges(self) -> int:
+        with self._lock:
+            valid_batches = set(self.pending_batches)
+            orphans = [eid for eid, e in self.pending_edges.items()
+                       if e.batch_id not in valid_batches]
+            for eid in orphans:
+                del self.pending_edges[eid]
+            return len(orphans)
+
+    # ---- discovered_channels (validator exactly-once claim CTE) ----
+
+    def claim_discovered_channel(self, username: str, crawl_id: str) -> bool:
+        """INSERT ... ON CONFLICT DO NOTHING claim (daprstate.go:4198-4225):
+        True iff this call discovered the channel first."""
+        with self._lock:
+            if username in self.discovered_channels:
+                return False
+            self.discovered_channels[username] = {
+                "username": username, "crawl_id": crawl_id,
+                "discovered_at": _now(),
+            }
+            return True
+
+    # ---- stats / access events ----
+
+    def flush_batch_stats(self, crawl_id: str, stats: Dict[str, int]):
+        with self._lock:
+            for stype, cnt in stats.items():
+                key = (crawl_id, stype)
+                self.source_type_stats[key] = (
+                    self.source_type_stats.get(key, 0) + cnt
+                )
+
+    def insert_access_event(self, event_type: str, detail: str = ""):
+        with self._lock:
+            self.access_events.append({
+                "type": event_type, "detail": detail, "at": _now(),
+            })
+
+
+class StateManagerFactory:
+    """state/statefactory.go:20-52 — local is the only backend here; the
+    Dapr/Redis/Postgres stack is replaced by LocalStateManager +
+    RandomWalkStore + the GPU-resident seen-set."""
+
+    @staticmethod
+    def create(config, base_path: Optional[str] = None) -> LocalStateManager:
+        return LocalStateManager(config, base_path=base_path)
