@@ -619,16 +619,53 @@ class RandomWalkStore:
                     e.discovery_time = _now()
                 self.edge_records.append(e)
 
-    def get_random_skipped_edge(self, exclude: set,
-                                rng=None) -> Optional[EdgeRecord]:
-        """Promote a skipped edge (daprstate.go:3206-3277 semantics)."""
+    def get_random_skipped_edge(self, exclude: set, rng=None,
+                                sequence_id: Optional[str] = None,
+                                source_channel: Optional[str] = None
+                                ) -> Optional[EdgeRecord]:
+        """Random skipped edge, optionally filtered to the same sequence +
+        source (daprstate.go:3206-3277 semantics)."""
         r = rng or random
         with self._lock:
             candidates = [
                 e for e in self.edge_records
                 if e.skipped and e.destination_channel not in exclude
+                and (sequence_id is None or e.sequence_id == sequence_id)
+                and (source_channel is None
+                     or e.source_channel == source_channel)
             ]
             return r.choice(candidates) if candidates else None
+
+    def get_edge_record(self, sequence_id: str,
+                        destination: str) -> Optional[EdgeRecord]:
+        """Non-skipped edge arriving at destination in this chain
+        (daprstate.go GetEdgeRecord semantics)."""
+        with self._lock:
+            for e in self.edge_records:
+                if (e.sequence_id == sequence_id
+                        and e.destination_channel == destination
+                        and not e.skipped):
+                    return e
+            return None
+
+    def delete_edge_record(self, sequence_id: str, destination: str) -> None:
+        with self._lock:
+            self.edge_records = [
+                e for e in self.edge_records
+                if not (e.sequence_id == sequence_id
+                        and e.destination_channel == destination
+                        and not e.skipped)
+            ]
+
+    def promote_edge(self, sequence_id: str, destination: str) -> None:
+        """Flip a skipped edge to followed (daprstate.go PromoteEdge)."""
+        with self._lock:
+            for e in self.edge_records:
+                if (e.sequence_id == sequence_id
+                        and e.destination_channel == destination
+                        and e.skipped):
+                    e.skipped = False
+                    return
 
     # ---- seed_channels ----
 
