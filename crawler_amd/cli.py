@@ -1,0 +1,289 @@
+"""CLI entry point — flag-compatible with the reference (main.go:751-812).
+
+Precedence mirrors viper (main.go:231-261): flags > env (CRAWLER_*,
+dots/dashes -> underscores) > config yaml (./config.yaml, ~/.crawler,
+/etc/crawler). Mode dispatch (main.go:586-643): standalone,
+dapr-standalone (local runtime stand-in), orchestrator, worker.
+"""
+from __future__ import annotations
+
+import argparse
+import os
+import sys
+from typing import List, Optional
+
+from .config import (
+    CrawlerConfig,
+    config_from_env,
+    generate_crawl_id,
+    parse_time_ago,
+    validate_sampling_method,
+)
+from .models.null_handler import NullValidator
+
+
+def build_parser() -> argparse.ArgumentParser:
+    p = argparse.ArgumentParser(
+        prog="crawler_amd",
+        description="MI355X-native distributed crawl/ingest engine",
+    )
+    a = p.add_argument
+    # Global flags (main.go:753-788)
+    a("--config", default="", help="config file (default ./config.yaml)")
+    a("--log-level", default="debug")
+    a("--dapr", action="store_true", help="compat flag (local runtime)")
+    a("--dapr-mode", default="job", choices=["job", "standalone"])
+    a("--dapr-port", type=int, default=6481)
+    a("--concurrency", type=int, default=1)
+    a("--timeout", type=int, default=30)
+    a("--user-agent", default="Mozilla/5.0 Crawler")
+    a("--output", default="json")
+    a("--storage-root", default="/tmp/crawl")
+    a("--min-post-date", default="")
+    a("--time-ago", default="")
+    a("--max-crawl-duration", default="")
+    a("--date-between", default="")
+    a("--sample-size", type=int, default=0)
+    a("--tdlib-database-url", default="")
+    a("--tdlib-database-urls", default="")
+    a("--min-users", type=int, default=100)
+    a("--crawl-id", default="")
+    a("--crawl-label", default="")
+    a("--max-comments", type=int, default=-1)
+    a("--max-depth", type=int, default=-1)
+    a("--max-posts", type=int, default=-1)
+    a("--max-pages", type=int, default=108000)
+    a("--tdlib-verbosity", type=int, default=1)
+    a("--skip-media", action="store_true")
+    a("--youtube-api-key", default="")
+    a("--platform", default="telegram")
+    a("--sampling", default="channel")
+    a("--seed-size", type=int, default=0)
+    a("--walkback-rate", type=int, default=15)
+    a("--min-channel-videos", type=int, default=10)
+    a("--null-config", default="{}")
+    a("--exit-on-complete", action="store_true")
+    # Validator / tandem flags (main.go:790-795)
+    a("--tandem-crawl", action="store_true")
+    a("--validate-only", action="store_true")
+    a("--validator-request-rate", type=float, default=6.0)
+    a("--validator-request-jitter-ms", type=int, default=200)
+    a("--validator-claim-batch-size", type=int, default=10)
+    a("--validator-timeout", default="")
+    # Combine-files flags (main.go:798-803)
+    a("--combine-files", action="store_true")
+    a("--combine-watch-dir", default="/tmp/watch-files")
+    a("--combine-temp-dir", default="/tmp/temp-files")
+    a("--combine-write-dir", default="/tmp/combine-write")
+    a("--combine-trigger-size", type=int, default=170)
+    a("--combine-hard-cap", type=int, default=200)
+    # Distributed mode flags (main.go:806-808)
+    a("--mode", default="",
+      choices=["", "standalone", "dapr-standalone", "orchestrator",
+               "worker"])
+    a("--worker-id", default="")
+    # Standalone flags (main.go:810-813)
+    a("--urls", default="")
+    a("--url-file", default="")
+    a("--url-file-url", default="")
+    a("--generate-code", action="store_true")
+    # MI355X-native knobs
+    a("--gpu", action="store_true",
+      help="run the per-post hot path on the GPU (HIP kernels)")
+    a("--synthetic-seed", type=int, default=1234)
+    a("--synthetic-universe", type=int, default=1_000_000)
+    a("--synthetic-posts", type=int, default=1000)
+    a("--disable-rate-limits", action="store_true")
+    a("--pool-size", type=int, default=2)
+    return p
+
+
+def _load_yaml_config(path: str) -> dict:
+    candidates = [path] if path else [
+        "./config.yaml",
+        os.path.expanduser("~/.crawler/config.yaml"),
+        "/etc/crawler/config.yaml",
+    ]
+    for c in candidates:
+        if c and os.path.exists(c):
+            import yaml
+
+            with open(c) as f:
+                return yaml.safe_load(f) or {}
+    return {}
+
+
+def parse_config(argv: Optional[List[str]] = None) -> CrawlerConfig:
+    parser = build_parser()
+    args = parser.parse_args(argv)
+    yaml_cfg = _load_yaml_config(args.config)
+
+    def pick(flag_val, default, yaml_key):
+        # flags > env (applied below) > yaml > defaults
+        if flag_val != default:
+            return flag_val
+        return yaml_cfg.get(yaml_key, flag_val)
+
+    cfg = CrawlerConfig(
+        concurrency=pick(args.concurrency, 1, "concurrency"),
+        timeout=pick(args.timeout, 30, "timeout"),
+        user_agent=args.user_agent,
+        output_format=args.output,
+        storage_root=pick(args.storage_root, "/tmp/crawl", "storage-root"),
+        min_users=pick(args.min_users, 100, "min-users"),
+        crawl_id=args.crawl_id or generate_crawl_id(),
+        crawl_label=args.crawl_label,
+        max_comments=args.max_comments,
+        max_depth=pick(args.max_depth, -1, "max-depth"),
+        max_posts=args.max_posts,
+        max_pages=args.max_pages,
+        skip_media_download=args.skip_media,
+        platform=pick(args.platform, "telegram", "platform"),
+        sampling_method=pick(args.sampling, "channel", "sampling"),
+        seed_size=args.seed_size,
+        walkback_rate=args.walkback_rate,
+        min_channel_videos=args.min_channel_videos,
+        null_config=args.null_config,
+        exit_on_complete=args.exit_on_complete,
+        tandem_crawl=args.tandem_crawl,
+        validate_only=args.validate_only,
+        validator_request_rate=args.validator_request_rate,
+        validator_request_jitter_ms=args.validator_request_jitter_ms,
+        validator_claim_batch_size=args.validator_claim_batch_size,
+        combine_files=args.combine_files,
+        combine_watch_dir=args.combine_watch_dir,
+        combine_temp_dir=args.combine_temp_dir,
+        combine_write_dir=args.combine_write_dir,
+        combine_trigger_size=args.combine_trigger_size,
+        combine_hard_cap=args.combine_hard_cap,
+        youtube_api_key=args.youtube_api_key,
+        synthetic_seed=args.synthetic_seed,
+        disable_rate_limits=args.disable_rate_limits,
+    )
+    if args.min_post_date:
+        import datetime as dt
+
+        cfg.min_post_date = dt.datetime.strptime(
+            args.min_post_date, "%Y-%m-%d"
+        ).replace(tzinfo=dt.timezone.utc)
+    if args.time_ago:
+        import datetime as dt
+
+        cfg.post_recency = parse_time_ago(args.time_ago).replace(
+            tzinfo=dt.timezone.utc
+        )
+    if args.date_between:
+        import datetime as dt
+
+        lo, hi = args.date_between.split(",")
+        cfg.date_between_min = dt.datetime.strptime(
+            lo, "%Y-%m-%d").replace(tzinfo=dt.timezone.utc)
+        cfg.date_between_max = dt.datetime.strptime(
+            hi, "%Y-%m-%d").replace(tzinfo=dt.timezone.utc)
+        cfg.sample_size = args.sample_size
+    if args.validator_timeout:
+        cfg.validator_timeout_s = _parse_duration(args.validator_timeout)
+    if args.max_crawl_duration:
+        cfg.max_crawl_duration_s = _parse_duration(args.max_crawl_duration)
+    cfg = config_from_env(cfg)
+    cfg.null_validator = NullValidator(cfg.platform, cfg.null_config)
+    # stash non-config CLI values for the dispatcher
+    cfg._cli = args  # type: ignore[attr-defined]
+    return cfg
+
+
+def _parse_duration(s: str) -> float:
+    """Go-style durations: '48h', '24h30m', '90s'."""
+    import re
+
+    total = 0.0
+    for num, unit in re.findall(r"([\d.]+)([hms])", s):
+        total += float(num) * {"h": 3600, "m": 60, "s": 1}[unit]
+    if total == 0:
+        raise ValueError(f"invalid duration: {s}")
+    return total
+
+
+def resolve_urls(args) -> List[str]:
+    urls: List[str] = []
+    if args.urls:
+        urls += [u.strip() for u in args.urls.split(",") if u.strip()]
+    if args.url_file:
+        with open(args.url_file) as f:
+            urls += [l.strip() for l in f if l.strip()]
+    return urls
+
+
+def main(argv: Optional[List[str]] = None) -> int:
+    cfg = parse_config(argv)
+    args = cfg._cli  # type: ignore[attr-defined]
+    urls = resolve_urls(args)
+    validate_sampling_method(
+        cfg.platform, cfg.sampling_method, url_list=urls,
+        url_file=args.url_file, url_file_url=args.url_file_url,
+        mode=args.mode, seed_size=cfg.seed_size, crawl_id=cfg.crawl_id,
+    )
+
+    mode = args.mode or ("standalone" if urls or cfg.seed_size else "")
+    if not mode:
+        print("no mode and no URLs given; nothing to do", file=sys.stderr)
+        return 2
+
+    from .feed import FeedConfig, SyntheticFeed
+    from .feed.client import ConnectionPool
+    from .engine import LocalStateManager, RandomWalkStore
+    from .engine.runner import RandomWalkRunner, StandaloneRunner
+
+    feed = SyntheticFeed(FeedConfig(
+        seed=args.synthetic_seed, universe=args.synthetic_universe,
+        posts_per_channel=args.synthetic_posts,
+    ))
+
+    if mode == "orchestrator":
+        from .parallel.orchestrator import orchestrator_main
+
+        return orchestrator_main(cfg, urls)
+    if mode == "worker":
+        from .parallel.worker import worker_main
+
+        return worker_main(cfg, feed)
+
+    if cfg.platform == "youtube":
+        from .youtube.runner import run_youtube
+
+        stats = run_youtube(cfg, urls)
+        print(f"youtube crawl complete: {stats}", file=sys.stderr)
+        return 0
+
+    pool = ConnectionPool(
+        feed, args.pool_size, cfg.rate_limit,
+        posts_per_channel=args.synthetic_posts,
+        disable_rate_limits=cfg.disable_rate_limits,
+    )
+    sm = LocalStateManager(cfg)
+
+    if cfg.validate_only:
+        from .engine.validator import run_validation_loop
+
+        rw = RandomWalkStore()
+        run_validation_loop(cfg, sm, rw)
+        return 0
+
+    if cfg.sampling_method == "random-walk":
+        rw = RandomWalkStore()
+        runner = RandomWalkRunner(cfg, sm, rw, pool)
+        if cfg.seed_size and not urls:
+            urls = [feed.username_of(i) for i in range(cfg.seed_size)]
+        runner.seed(urls)
+        stats = runner.run(
+            max_seconds=cfg.max_crawl_duration_s or None
+        )
+    else:
+        runner = StandaloneRunner(cfg, sm, pool)
+        stats = runner.run(urls)
+    print(f"crawl complete: {stats}", file=sys.stderr)
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())
