@@ -1,0 +1,219 @@
+"""Distributed layer tests: queues, contracts, orchestrator+worker rounds,
+reassignment, plus a real multi-process gloo all-gather (world_size 2).
+
+Mirrors reference distributed/{messages,pubsub,integration}_test.go and
+orchestrator/worker tests."""
+import os
+import threading
+import time
+
+import pytest
+
+from crawler_amd.config import CrawlerConfig
+from crawler_amd.engine import LocalStateManager
+from crawler_amd.feed import FeedConfig, SyntheticFeed
+from crawler_amd.feed.client import ConnectionPool
+from crawler_amd.parallel import messages as M
+from crawler_amd.parallel.orchestrator import Orchestrator
+from crawler_amd.parallel.queue import Heartbeats, InMemoryStore, StoreQueue
+from crawler_amd.parallel.worker import Worker
+
+
+# ---------- contracts ----------
+
+def test_work_item_roundtrip_and_validation():
+    item = M.WorkItem(id="i1", url="chan", depth=1, crawl_id="c1",
+                      trace_id=M.new_trace_id())
+    item.validate()
+    got = M.WorkItem.from_json(item.to_json())
+    assert got == item
+    with pytest.raises(ValueError):
+        M.WorkItem(url="chan", crawl_id="c1").validate()
+    with pytest.raises(ValueError):
+        M.WorkItem(id="x", crawl_id="c1").validate()
+
+
+def test_trace_ids_unique():
+    assert M.new_trace_id() != M.new_trace_id()
+
+
+# ---------- store queue ----------
+
+def test_queue_fifo_and_exactly_once():
+    store = InMemoryStore()
+    q = StoreQueue(store, "t1")
+    for i in range(5):
+        q.publish(f"m{i}")
+    assert q.size() == 5
+    got = [q.claim() for _ in range(5)]
+    assert got == [f"m{i}" for i in range(5)]
+    assert q.claim(timeout_s=0.0) is None
+
+
+def test_queue_concurrent_claims_no_duplicates():
+    store = InMemoryStore()
+    q = StoreQueue(store, "t2")
+    for i in range(200):
+        q.publish(str(i))
+    claimed = []
+    lock = threading.Lock()
+
+    def worker():
+        while True:
+            v = q.claim(timeout_s=0.0)
+            if v is None:
+                return
+            with lock:
+                claimed.append(v)
+
+    threads = [threading.Thread(target=worker) for _ in range(8)]
+    [t.start() for t in threads]
+    [t.join() for t in threads]
+    assert sorted(map(int, claimed)) == list(range(200))
+
+
+def test_heartbeats_offline_detection():
+    store = InMemoryStore()
+    hb = Heartbeats(store)
+    hb.register("w1")
+    hb.register("w2")
+    hb.beat("w1")
+    hb.beat("w2")
+    assert hb.offline_workers(timeout_s=10) == []
+    # age w2's beat artificially
+    ts, st = hb.last_seen("w2")
+    store.set("hb/w2", f"{ts - 600}|active")
+    assert hb.offline_workers(timeout_s=300) == ["w2"]
+
+
+# ---------- orchestrator + worker integration (in-proc, like the
+# reference's integration_test.go which shares a mock state manager) ----
+
+def mk_dist_env(tmp_path, sampling="snowball", max_depth=1, n_workers=2):
+    cfg = CrawlerConfig(
+        crawl_id="d1", storage_root=str(tmp_path), min_users=1,
+        sampling_method=sampling, max_depth=max_depth,
+        disable_rate_limits=True,
+    )
+    feed = SyntheticFeed(FeedConfig(seed=21, universe=200,
+                                    posts_per_channel=40))
+    store = InMemoryStore()
+    sm = LocalStateManager(cfg)
+    orch = Orchestrator(cfg, sm, store, worker_timeout_s=300)
+    workers = []
+    for i in range(n_workers):
+        pool = ConnectionPool(feed, 1, cfg.rate_limit,
+                              posts_per_channel=40,
+                              disable_rate_limits=True)
+        wcfg = CrawlerConfig(
+            crawl_id="d1", storage_root=str(tmp_path / f"w{i}"),
+            min_users=1, sampling_method=sampling, max_depth=max_depth,
+            disable_rate_limits=True,
+        )
+        workers.append(Worker(f"w{i}", wcfg, pool, store))
+    return cfg, orch, workers, store
+
+
+def drive(orch, workers, rounds=200):
+    for _ in range(rounds):
+        orch.distribute()
+        for w in workers:
+            w.run_once(timeout_s=0.0)
+        orch.pump_results()
+        if orch.done:
+            break
+    return orch
+
+
+def test_orchestrator_worker_snowball_crawl(tmp_path):
+    cfg, orch, workers, store = mk_dist_env(tmp_path)
+    orch.sm.initialize(["c0000000001"])
+    drive(orch, workers)
+    assert orch.done
+    assert orch.stats["results"] >= 1
+    # depth advanced into discovered pages
+    assert orch.sm.get_max_depth() >= 1
+    layer1 = orch.sm.get_layer_by_depth(1)
+    assert layer1 and all(
+        p.status in ("fetched", "deadend", "error") for p in layer1
+    )
+    # both workers processed something
+    assert sum(w.processed for w in workers) == orch.stats["results"]
+
+
+def test_work_split_across_workers(tmp_path):
+    cfg, orch, workers, store = mk_dist_env(tmp_path, max_depth=1)
+    orch.sm.initialize(["c%010d" % i for i in range(8)])
+    drive(orch, workers)
+    assert orch.done
+    assert all(w.processed > 0 for w in workers)
+
+
+def test_reassignment_after_timeout(tmp_path):
+    cfg, orch, workers, store = mk_dist_env(tmp_path, sampling="channel",
+                                            n_workers=1)
+    t = {"now": 0.0}
+    orch.clock = lambda: t["now"]
+    orch.worker_timeout_s = 100
+    orch.sm.initialize(["c0000000001"])
+    orch.distribute()
+    assert orch.work_q.size() == 1
+    # worker claims the item but "dies" (never publishes a result)
+    raw = orch.work_q.claim()
+    assert raw is not None
+    t["now"] = 200.0
+    n = orch.check_worker_health()
+    assert n == 1
+    item = M.WorkItem.from_json(orch.work_q.claim())
+    assert item.retry_count == 1
+    assert item.priority == M.PRIORITY_HIGH
+
+
+def test_error_pages_marked(tmp_path):
+    cfg, orch, workers, store = mk_dist_env(tmp_path, sampling="channel")
+    orch.sm.initialize(["c0000009999"])  # outside universe -> 400
+    drive(orch, workers)
+    page = orch.sm.get_layer_by_depth(0)[0]
+    assert page.status == "error"
+
+
+def test_poison_pill_stops_workers(tmp_path):
+    cfg, orch, workers, store = mk_dist_env(tmp_path, n_workers=1)
+    orch.broadcast_stop(1)
+    assert workers[0].run_once(timeout_s=0.0) is False
+
+
+# ---------- real multi-process gloo collective (world_size 2) ----------
+
+def _gloo_worker(rank, world, port, results_dir):
+    import torch
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    # simulate the per-step discovery exchange of bench.py: all-gather
+    # fixed-size hash buffers and union
+    local = torch.zeros(8, dtype=torch.int64)
+    local[: 2 + rank] = torch.arange(1, 3 + rank) + rank * 100
+    gathered = [torch.empty_like(local) for _ in range(world)]
+    dist.all_gather(gathered, local)
+    union = set()
+    for g in gathered:
+        union |= {int(x) for x in g[g != 0]}
+    with open(os.path.join(results_dir, f"r{rank}.txt"), "w") as f:
+        f.write(",".join(map(str, sorted(union))))
+    dist.destroy_process_group()
+
+
+def test_gloo_discovery_allgather_two_procs(tmp_path):
+    import torch.multiprocessing as mp
+
+    port = 29712
+    ctx = mp.spawn(
+        _gloo_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True
+    )
+    r0 = (tmp_path / "r0.txt").read_text()
+    r1 = (tmp_path / "r1.txt").read_text()
+    assert r0 == r1
+    assert "101" in r0 and "1" in r0
