@@ -1,0 +1,161 @@
+"""YouTube subsystem tests (reference crawler/youtube/*_test.go coverage:
+sampling dispatch, prefix validity rule, channel gate, quota economics,
+duration parsing, video->Post conversion)."""
+import json
+import random
+
+import pytest
+
+from crawler_amd.config import CrawlerConfig
+from crawler_amd.youtube import (
+    QuotaExceeded,
+    SyntheticYouTubeClient,
+    SyntheticYouTubeIndex,
+    convert_video_to_post,
+    parse_iso8601_duration,
+)
+from crawler_amd.youtube.convert import extract_urls
+from crawler_amd.youtube.runner import run_youtube
+
+
+@pytest.fixture
+def client():
+    return SyntheticYouTubeClient(
+        SyntheticYouTubeIndex(seed=5, universe_channels=500),
+        rng=random.Random(1),
+    )
+
+
+# ---------- validity rule (youtube_client.go:1266-1281) ----------
+
+@pytest.mark.parametrize("vid,prefix,ok", [
+    ("abcde-12345", "abcde", True),
+    ("abcde-1234", "abcde", False),     # len != 11
+    ("abcdef12345", "abcde", False),    # no dash at 5
+    ("xbcde-12345", "abcde", False),    # wrong prefix
+])
+def test_prefix_validity_rule(vid, prefix, ok):
+    assert SyntheticYouTubeClient.is_valid_sample(vid, prefix) == ok
+
+
+def test_random_prefix_shape():
+    rng = random.Random(3)
+    p = SyntheticYouTubeClient.generate_random_prefix(rng)
+    assert len(p) == 5 and p.isalpha() and p.islower()
+
+
+# ---------- duration parsing ----------
+
+@pytest.mark.parametrize("s,secs", [
+    ("PT5M30S", 330),
+    ("PT1H2M3S", 3723),
+    ("P1DT1S", 86401),
+    ("PT45S", 45),
+])
+def test_iso8601_duration(s, secs):
+    assert parse_iso8601_duration(s) == secs
+
+
+def test_url_extraction_trims_punctuation():
+    urls = extract_urls("see https://a.example/x, and (https://b.example/y).")
+    assert urls == ["https://a.example/x", "https://b.example/y"]
+
+
+# ---------- quota ----------
+
+def test_quota_accounting(client):
+    q0 = client.quota_used
+    client.search("abcde")
+    assert client.quota_used == q0 + 100
+    client.list_videos(["abcde-12345"])
+    assert client.quota_used == q0 + 101
+
+
+def test_quota_exceeded_stops_sampling():
+    c = SyntheticYouTubeClient(
+        SyntheticYouTubeIndex(seed=5, universe_channels=100),
+        daily_quota=150, rng=random.Random(1),
+    )
+    with pytest.raises(QuotaExceeded):
+        c.search("aaaaa")
+        c.search("bbbbb")
+
+
+# ---------- random sampling ----------
+
+def test_random_sampling_respects_gate_and_rule(client):
+    videos = client.get_random_videos(limit=20, max_searches=40)
+    assert videos
+    for v in videos:
+        assert len(v.id) == 11 and v.id[5] == "-"
+        ch = client.get_channel_info(v.channel_id)
+        assert ch.video_count > client.min_channel_videos
+    assert client.stats["invalid_ids"] > 0  # noise ids got filtered
+    assert client.stats["gate_rejected"] >= 0
+
+
+# ---------- snowball ----------
+
+def test_snowball_discovers_channels(client):
+    seed = client.index.channel_id_of(7)
+    client.index.channel(7)  # register in reverse map
+    videos = client.get_snowball_videos([seed], limit=30, max_depth=2)
+    assert videos
+    chans = {v.channel_id for v in videos}
+    assert len(chans) >= 1
+
+
+# ---------- conversion ----------
+
+def test_convert_video_to_post_fields(client):
+    vid = client.index.video("abcde-12345")
+    ch = client.get_channel_info(vid.channel_id)
+    post = convert_video_to_post(vid, ch, crawl_label="yt-test")
+    obj = json.loads(post.to_json())
+    assert obj["platform_name"] == "youtube"
+    assert obj["post_uid"] == "abcde-12345"
+    assert obj["post_link"] == "https://www.youtube.com/watch?v=abcde-12345"
+    assert obj["post_type"] == ["video"]
+    assert obj["crawl_label"] == "yt-test"
+    assert obj["engagement"] == (
+        vid.like_count + vid.comment_count + vid.view_count // 100
+    )
+    assert obj["performance_scores"]["likes"] == vid.like_count
+    assert obj["has_embed_media"] is True
+    assert obj["post_title"] == vid.title
+    assert obj["media_data"]["document_name"].endswith(".mp4")
+    if vid.duration == "P0D":
+        assert obj["video_length"] is None
+    assert obj["outlinks"]  # description contains URLs
+
+
+def test_p0d_duration_null(client):
+    vid = client.index.video("abcde-12345")
+    vid.duration = "P0D"
+    post = convert_video_to_post(vid, None)
+    assert post.video_length is None
+
+
+# ---------- runner ----------
+
+def test_run_youtube_random_mode(tmp_path):
+    cfg = CrawlerConfig(crawl_id="yt1", storage_root=str(tmp_path),
+                        platform="youtube", sampling_method="random",
+                        max_posts=10)
+    stats = run_youtube(cfg, [])
+    assert stats["posts"] >= 1
+    assert stats["quota_used"] > 0
+    # posts landed in per-channel JSONL files
+    crawl_dir = tmp_path / "yt1"
+    jsonls = list(crawl_dir.rglob("posts.jsonl"))
+    assert jsonls
+    obj = json.loads(jsonls[0].read_bytes().splitlines()[0])
+    assert obj["platform_name"] == "youtube"
+
+
+def test_run_youtube_channel_mode(tmp_path):
+    cfg = CrawlerConfig(crawl_id="yt2", storage_root=str(tmp_path),
+                        platform="youtube", sampling_method="channel",
+                        max_posts=20)
+    stats = run_youtube(cfg, ["3"])
+    assert stats["videos"] >= 1
