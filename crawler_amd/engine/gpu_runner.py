@@ -1,0 +1,195 @@
+"""GPU crawl engine: whole-layer batch execution of the per-post hot path.
+
+This is the MI355X-native execution mode of the crawl engine (BASELINE
+configs #2/#3): instead of per-message host parsing, a BFS layer's
+channels are generated/fetched as one packed device batch and pushed
+through the HIP kernels:
+
+    feed (device feedgen or host batch)  ->  parse+encode (JSONL bytes)
+      ->  SeenSet.claim (exactly-once discovery, atomicCAS)
+      ->  host spill: per-channel JSONL slices appended to the same
+          posts.jsonl layout the CPU path writes
+      ->  newly-claimed link names (small D2H) become the next layer
+
+Equivalence contract: for the same channels the JSONL bytes are identical
+to the CPU pipeline's Post.to_jsonl() output (enforced by the byte-exact
+kernel tests) and the discovered-channel set matches golden extraction.
+
+Cross-rank: ranks shard the layer's channels; newly-claimed hashes are
+all-gathered and merged into each rank's SeenSet (bench.py does the same
+per step) so a channel is crawled by exactly one rank per crawl.
+"""
+from __future__ import annotations
+
+import datetime as _dt
+from typing import List, Optional, Tuple
+
+import numpy as np
+import torch
+
+from ..feed.synth import SyntheticFeed
+from .state import LocalStateManager, Page
+
+
+class GpuCrawlEngine:
+    def __init__(self, cfg, sm: LocalStateManager, feed: SyntheticFeed,
+                 device="cuda:0", posts_per_channel: Optional[int] = None,
+                 chunk_channels: int = 256, use_device_gen: bool = True):
+        from ..ops import gpu as gpu_mod
+
+        self.gpu = gpu_mod
+        gpu_mod.require_lib()
+        self.cfg = cfg
+        self.sm = sm
+        self.feed = feed
+        self.device = torch.device(device)
+        self.ppc = posts_per_channel or feed.cfg.posts_per_channel
+        self.chunk_channels = chunk_channels
+        self.seen = gpu_mod.SeenSet(self.device)
+        self.stats = {"pages": 0, "posts": 0, "jsonl_bytes": 0,
+                      "discovered": 0, "deadends": 0}
+
+    # ---- helpers ----
+
+    def _cid_of(self, username: str) -> Optional[int]:
+        if username.startswith("c") and username[1:].isdigit():
+            cid = int(username[1:])
+            if cid < self.feed.cfg.universe:
+                return cid
+        return None
+
+    def process_channels(self, usernames: List[str],
+                         now: Optional[_dt.datetime] = None
+                         ) -> Tuple[List[str], int]:
+        """Process a list of channels; returns (newly discovered names,
+        posts stored)."""
+        now = now or _dt.datetime.now(_dt.timezone.utc)
+        discovered: List[str] = []
+        posts_total = 0
+        valid = [(u, self._cid_of(u)) for u in usernames]
+        bad = [u for u, c in valid if c is None]
+        ok = [(u, c) for u, c in valid if c is not None]
+        for u in bad:
+            self.stats["deadends"] += 1
+        for i in range(0, len(ok), self.chunk_channels):
+            chunk = ok[i:i + self.chunk_channels]
+            cids = np.array([c for _u, c in chunk], dtype=np.int64)
+            batch = self.feed.build_batch_device(
+                cids, self.device, posts_per_channel=self.ppc
+            )
+            res = self.gpu.parse_encode(
+                batch, now=now, min_post_date=self.cfg.min_post_date
+            )
+            new_mask = self.seen.claim(res)
+            torch.cuda.synchronize()
+
+            # host spill: per-channel JSONL slices. Messages are grouped
+            # by channel (K x P layout), so channel k owns lines
+            # [k*P, (k+1)*P) -> bytes [line_off[kP], line_off[(k+1)P-1]+len).
+            out_host = torch.empty_like(res.out, device="cpu",
+                                        pin_memory=True)
+            out_host.copy_(res.out, non_blocking=True)
+            line_off = res.line_off.cpu().numpy()
+            line_len = res.line_len.cpu().numpy()
+            torch.cuda.synchronize()
+            buf = out_host.numpy()
+            P = self.ppc
+            for k, (uname, _cid) in enumerate(chunk):
+                lo = int(line_off[k * P])
+                last = (k + 1) * P - 1
+                hi = int(line_off[last] + line_len[last])
+                if hi > lo:
+                    self.sm.store_post_lines(uname, buf[lo:hi].tobytes())
+                n_lines = int((line_len[k * P:(k + 1) * P] > 0).sum())
+                posts_total += n_lines
+                self.stats["pages"] += 1
+            self.stats["jsonl_bytes"] += int(buf.shape[0])
+
+            # newly-claimed names (small: first-discovery rows only)
+            nz = new_mask.nonzero()
+            if nz.numel():
+                rows = nz[:, 0]
+                cols = nz[:, 1]
+                names = res.link_name[rows, cols].cpu().numpy()
+                lens = res.link_len[rows, cols].cpu().numpy()
+                for b, ln in zip(names, lens):
+                    discovered.append(bytes(b[:ln]).decode())
+        self.stats["posts"] += posts_total
+        self.stats["discovered"] += len(discovered)
+        return discovered, posts_total
+
+    # ---- BFS crawl (snowball / channel) ----
+
+    def run(self, seed_urls: List[str], comm=None) -> dict:
+        """comm: optional torch.distributed process group for multi-rank
+        discovery exchange (rank-sharded layers)."""
+        import torch.distributed as dist
+
+        sm = self.sm
+        sm.initialize(seed_urls)
+        depth = 0
+        while True:
+            layer = [p for p in sm.get_layer_by_depth(depth)
+                     if p.status == "unfetched"]
+            if not layer:
+                break
+            names = [p.url for p in layer]
+            if comm is not None:
+                rank = dist.get_rank()
+                world = dist.get_world_size()
+                mine = names[rank::world]
+            else:
+                mine = names
+            discovered, _ = self.process_channels(mine)
+            for p in layer:
+                p.status = "fetched"
+                sm.update_page(p)
+            if comm is not None:
+                # exchange discoveries (names, fixed-width padded)
+                discovered = self._allgather_names(discovered)
+            if (self.cfg.sampling_method == "snowball"
+                    and (self.cfg.max_depth < 0
+                         or depth < self.cfg.max_depth)
+                    and discovered):
+                pages = [Page(url=n, depth=depth + 1, status="unfetched")
+                         for n in sorted(set(discovered))]
+                sm.add_layer(pages)
+            sm.save_state()
+            if self.cfg.sampling_method == "channel":
+                break
+            if self.cfg.max_depth >= 0 and depth >= self.cfg.max_depth:
+                break
+            depth += 1
+        sm.update_crawl_metadata(sm.metadata.crawl_id,
+                                 {"status": "completed"})
+        sm.save_state()
+        sm.close()
+        return dict(self.stats)
+
+    @staticmethod
+    def _allgather_names(names: List[str], width: int = 32) -> List[str]:
+        import torch.distributed as dist
+
+        world = dist.get_world_size()
+        n = len(names)
+        counts = torch.tensor([n], dtype=torch.int64)
+        all_counts = [torch.zeros(1, dtype=torch.int64)
+                      for _ in range(world)]
+        dist.all_gather(all_counts, counts)
+        max_n = max(int(c.item()) for c in all_counts)
+        buf = torch.zeros(max_n, width, dtype=torch.uint8)
+        for i, name in enumerate(names):
+            b = name.encode()[:width]
+            buf[i, :len(b)] = torch.frombuffer(bytearray(b),
+                                               dtype=torch.uint8)
+        gathered = [torch.empty_like(buf) for _ in range(world)]
+        dist.all_gather(gathered, buf)
+        out = []
+        for r, g in enumerate(gathered):
+            cnt = int(all_counts[r].item())
+            arr = g.numpy()
+            for i in range(cnt):
+                row = arr[i]
+                ln = int((row != 0).sum())
+                out.append(bytes(row[:ln]).decode())
+        return out
