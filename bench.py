@@ -92,10 +92,17 @@ def main():
     now = dt.datetime(2026, 1, 1, tzinfo=dt.timezone.utc)
     seen = gpu.SeenSet(device)
 
-    # pinned host ring for the JSONL output (the storage boundary)
-    # worst case ~2.6KB/post
-    pinned = torch.empty(int(chunk_posts * 3000), dtype=torch.uint8,
-                         pin_memory=True)
+    # Double-buffered pinned host ring for the JSONL output (the storage
+    # boundary): D2H of step k overlaps the kernels of step k+1 on a
+    # separate copy stream. Worst case ~2.6KB/post.
+    pinned = [
+        torch.empty(int(chunk_posts * 3000), dtype=torch.uint8,
+                    pin_memory=True)
+        for _ in range(2)
+    ]
+    compute_stream = torch.cuda.Stream()
+    copy_stream = torch.cuda.Stream()
+    inflight = [None, None]  # keep res tensors alive while copying
     sink_f = None
     if args.sink == "file" and rank == 0:
         os.makedirs(args.sink_dir, exist_ok=True)
@@ -107,28 +114,41 @@ def main():
     def step(si):
         nonlocal total_out_bytes, new_discoveries
         chunk = chunks[si % n_chunks]
-        res = gpu.parse_encode(chunk, now=now)
-        new_mask = seen.claim(res)
-        if world > 1:
-            flat = new_mask.flatten().bool()
-            new_hashes = res.link_hash.flatten()[flat]
-            cap = 65536
-            buf = torch.zeros(cap, dtype=torch.int64, device=device)
-            k = min(new_hashes.numel(), cap)
-            buf[:k] = new_hashes[:k]
-            gathered = [torch.empty_like(buf) for _ in range(world)]
-            torch.distributed.all_gather(gathered, buf)
-            for r, g in enumerate(gathered):
-                if r != rank:
-                    gz = g[g != 0]
-                    seen.insert_hashes(gz)
-        nbytes = int(res.line_off[-1].item() + res.line_len[-1].item())
-        pinned[:nbytes].copy_(res.out[:nbytes], non_blocking=True)
-        torch.cuda.synchronize()
+        slot = si % 2
+        with torch.cuda.stream(compute_stream):
+            res = gpu.parse_encode(chunk, now=now)
+            new_mask = seen.claim(res)
+            if world > 1:
+                flat = new_mask.flatten().bool()
+                new_hashes = res.link_hash.flatten()[flat]
+                cap = 65536
+                buf = torch.zeros(cap, dtype=torch.int64, device=device)
+                k = min(new_hashes.numel(), cap)
+                buf[:k] = new_hashes[:k]
+                gathered = [torch.empty_like(buf) for _ in range(world)]
+                torch.distributed.all_gather(gathered, buf)
+                for r, g in enumerate(gathered):
+                    if r != rank:
+                        seen.insert_hashes(g[g != 0])
+            nd = seen.new_count()  # syncs within the compute stream
+        nbytes = res.out.numel()
+        ev = torch.cuda.Event()
+        ev.record(compute_stream)
+        copy_stream.wait_event(ev)
+        # the previous copy into this pinned slot must be done before reuse
+        if inflight[slot] is not None:
+            inflight[slot][1].synchronize()
+        with torch.cuda.stream(copy_stream):
+            pinned[slot][:nbytes].copy_(res.out, non_blocking=True)
+            res.out.record_stream(copy_stream)
+        done = torch.cuda.Event()
+        done.record(copy_stream)
+        inflight[slot] = (res.out, done)
         total_out_bytes += nbytes
-        new_discoveries += seen.new_count()
+        new_discoveries += nd
         if sink_f is not None:
-            sink_f.write(bytes(pinned[:nbytes].numpy()))
+            done.synchronize()
+            sink_f.write(bytes(pinned[slot][:nbytes].numpy()))
         return nbytes
 
     # ---- warmup ----

@@ -544,6 +544,79 @@ write_kernel(BatchView B, LinkOut LO, const long* line_off,
   }
 }
 
+// Single-pass variant: emit straight into per-message scratch slots (the
+// host supplies a sound stride bound), then a vectorized compaction gathers
+// the final contiguous JSONL. Replaces measure+write (saves the whole
+// measuring pass; compaction is pure memcpy-rate).
+
+__global__ void __launch_bounds__(256)
+write_scratch_kernel(BatchView B, LinkOut LO, unsigned char* scratch,
+                     long stride, int* line_len, int* overflow) {
+  const int lane = lane_id();
+  const int wave = wave_id();
+  const int waves_per_grid = gridDim.x * 4;
+  for (int i = blockIdx.x * 4 + wave; i < B.n; i += waves_per_grid) {
+    if ((long)B.date[i] < B.min_post_date) {
+      if (lane == 0) { line_len[i] = 0; LO.cnt[i] = 0; }
+      continue;
+    }
+    LinkList L{LO.name + (size_t)i * MAX_LINKS * 32,
+               LO.name_len + (size_t)i * MAX_LINKS,
+               LO.src + (size_t)i * MAX_LINKS,
+               LO.hash + (size_t)i * MAX_LINKS, 0};
+    extract_links(B, i, L, lane);
+    int len = emit_line<true>(B, i, scratch + (size_t)i * stride, L);
+    if (lane == 0) {
+      line_len[i] = len;
+      LO.cnt[i] = L.cnt;
+      if (len > stride) atomicMax(overflow, len);  // host asserts 0
+    }
+  }
+}
+
+// Wave-cooperative aligned-dst copy: head bytes to 4B alignment, then
+// funnel-shifted dword stores (src may be misaligned), byte tail.
+DEV void copy_line(const unsigned char* src, unsigned char* dst, int n,
+                   int lane) {
+  int h = (int)((4 - ((unsigned long)(size_t)dst & 3)) & 3);
+  if (h > n) h = n;
+  if (lane == 0)
+    for (int j = 0; j < h; ++j) dst[j] = src[j];
+  const unsigned char* s = src + h;
+  unsigned char* d = dst + h;
+  int rem = n - h;
+  int nw = rem >> 2;
+  int sh = (int)((unsigned long)(size_t)s & 3);
+  const unsigned int* sw = (const unsigned int*)(s - sh);
+  for (int k = lane; k < nw; k += WAVE) {
+    unsigned int w0 = sw[k];
+    unsigned int v;
+    if (sh) {
+      unsigned int w1 = sw[k + 1];
+      v = (unsigned int)((((unsigned long long)w1 << 32) | w0) >> (8 * sh));
+    } else {
+      v = w0;
+    }
+    ((unsigned int*)d)[k] = v;
+  }
+  if (lane == 0)
+    for (int j = h + (nw << 2); j < n; ++j) dst[j] = src[j];
+}
+
+__global__ void __launch_bounds__(256)
+compact_kernel(const unsigned char* scratch, long stride,
+               const long* line_off, const int* line_len,
+               unsigned char* out, int n) {
+  const int lane = lane_id();
+  const int wave = wave_id();
+  const int waves_per_grid = gridDim.x * 4;
+  for (int i = blockIdx.x * 4 + wave; i < n; i += waves_per_grid) {
+    int len = line_len[i];
+    if (len == 0) continue;
+    copy_line(scratch + (size_t)i * stride, out + line_off[i], len, lane);
+  }
+}
+
 }  // namespace crawl
 
 // ---- C ABI (ctypes-friendly; torch-header-free) ----
@@ -648,6 +721,28 @@ int crawl_write(void** batch_ptrs, const long* scalars, void** link_ptrs,
   hipLaunchKernelGGL(crawl::write_kernel, dim3(grid), dim3(256), 0,
                      (hipStream_t)stream, B, LO, (const long*)line_off,
                      (const int*)line_len, (unsigned char*)out);
+  return (int)hipGetLastError();
+}
+
+int crawl_write_scratch(void** batch_ptrs, const long* scalars,
+                        void** link_ptrs, void* scratch, long stride,
+                        void* line_len, void* overflow, int grid,
+                        void* stream) {
+  crawl::BatchView B = crawl::make_view(batch_ptrs, scalars);
+  crawl::LinkOut LO = crawl::make_links(link_ptrs);
+  hipLaunchKernelGGL(crawl::write_scratch_kernel, dim3(grid), dim3(256), 0,
+                     (hipStream_t)stream, B, LO, (unsigned char*)scratch,
+                     stride, (int*)line_len, (int*)overflow);
+  return (int)hipGetLastError();
+}
+
+int crawl_compact(const void* scratch, long stride, const void* line_off,
+                  const void* line_len, void* out, int n, int grid,
+                  void* stream) {
+  hipLaunchKernelGGL(crawl::compact_kernel, dim3(grid), dim3(256), 0,
+                     (hipStream_t)stream, (const unsigned char*)scratch,
+                     stride, (const long*)line_off, (const int*)line_len,
+                     (unsigned char*)out, n);
   return (int)hipGetLastError();
 }
 

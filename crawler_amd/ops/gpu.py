@@ -65,8 +65,23 @@ def load_lib():
     ]
     _bind_dedup(lib)
     _bind_feedgen(lib)
+    _bind_v2(lib)
     _lib = lib
     return lib
+
+
+def _bind_v2(lib):
+    lib.crawl_write_scratch.restype = ctypes.c_int
+    lib.crawl_write_scratch.argtypes = [
+        ctypes.POINTER(ctypes.c_void_p), ctypes.POINTER(ctypes.c_long),
+        ctypes.POINTER(ctypes.c_void_p), ctypes.c_void_p, ctypes.c_long,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+    ]
+    lib.crawl_compact.restype = ctypes.c_int
+    lib.crawl_compact.argtypes = [
+        ctypes.c_void_p, ctypes.c_long, ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
+    ]
 
 
 def _bind_dedup(lib):
@@ -181,18 +196,52 @@ def _ptr_array(tensors: List[torch.Tensor]):
     return arr
 
 
+def _stride_bound(batch: B.MessageBatch) -> int:
+    """Sound per-message upper bound on the JSONL line length.
+
+    Worst-case expansion of any escaped byte is 6x; the constant skeleton,
+    numeric fields and timestamps fit in the 2200-byte base. Comment text /
+    handle / reaction contributions are segment-summed per message."""
+    m = batch.meta
+    dev = batch.device
+    i64 = lambda t: t.to(torch.int64)
+    c = m["channel_idx"].to(torch.int64)
+    bound = (
+        2200
+        + 6 * i64(m["text_len"]) + 6 * i64(m["aux_len"])
+        + 12 * i64(batch.ch_title_len[c])
+        + 36 * i64(batch.ch_user_len[c])
+        + 6 * i64(m["poster_len"])
+        + 45 * i64(m["react_cnt"])
+        + 170 * i64(m["com_cnt"])
+        + 35 * MAX_LINKS
+    )
+    ccnt = i64(m["com_cnt"])
+    total_c = int(ccnt.sum().item())
+    if total_c:
+        cmsg = torch.repeat_interleave(
+            torch.arange(batch.n, dtype=torch.int64, device=dev), ccnt
+        )
+        extra = (6 * (i64(batch.com_text_len) + i64(batch.com_handle_len))
+                 + 45 * i64(batch.com_react_cnt))
+        bound = bound.index_add(0, cmsg, extra)
+    return int(bound.max().item())
+
+
 def parse_encode(
     batch: B.MessageBatch,
     now: Optional[_dt.datetime] = None,
     min_post_date: Optional[_dt.datetime] = None,
     grid: int = 0,
     stream: Optional[torch.cuda.Stream] = None,
+    single_pass: bool = True,
 ) -> EncodeResult:
-    """Run measure+extract then write on the current CUDA device.
+    """Parse + JSONL-encode a batch on the current CUDA device.
 
-    Returns device tensors; the caller D2H-copies ``out[:total]`` (or slices
-    per line) for the host writer. Byte-for-byte equal to
-    golden_batch.encode_batch(batch, now=now, ...).
+    Default path (single_pass): one emit pass into strided scratch, then a
+    funnel-shift vectorized compaction — no separate measuring pass.
+    Returns device tensors; the caller D2H-copies ``out`` for the host
+    writer. Byte-for-byte equal to golden_batch.encode_batch(...).
     """
     lib = require_lib()
     dev = batch.device
@@ -248,26 +297,57 @@ def parse_encode(
         else stream.cuda_stream
     )
 
-    rc = lib.crawl_measure_extract(
-        batch_ptrs, scalars, link_ptrs,
-        ctypes.c_void_p(line_len.data_ptr()), grid, stream_ptr,
-    )
-    if rc != 0:
-        raise RuntimeError(f"crawl_measure_extract launch failed: hip error {rc}")
-
-    line_off = torch.zeros(n, dtype=torch.int64, device=dev)
-    torch.cumsum(line_len.to(torch.int64)[:-1], 0, out=line_off[1:])
-    total = int(line_off[-1].item() + line_len[-1].item()) if n else 0
-    out = torch.empty(total, dtype=torch.uint8, device=dev)
-
-    rc = lib.crawl_write(
-        batch_ptrs, scalars, link_ptrs,
-        ctypes.c_void_p(line_off.data_ptr()),
-        ctypes.c_void_p(line_len.data_ptr()),
-        ctypes.c_void_p(out.data_ptr()), grid, stream_ptr,
-    )
-    if rc != 0:
-        raise RuntimeError(f"crawl_write launch failed: hip error {rc}")
+    if single_pass:
+        stride = (_stride_bound(batch) + 31) & ~15
+        scratch = torch.empty(n * stride, dtype=torch.uint8, device=dev)
+        overflow = torch.zeros(1, dtype=torch.int32, device=dev)
+        rc = lib.crawl_write_scratch(
+            batch_ptrs, scalars, link_ptrs,
+            ctypes.c_void_p(scratch.data_ptr()), stride,
+            ctypes.c_void_p(line_len.data_ptr()),
+            ctypes.c_void_p(overflow.data_ptr()), grid, stream_ptr,
+        )
+        if rc != 0:
+            raise RuntimeError(f"crawl_write_scratch failed: hip error {rc}")
+        line_off = torch.zeros(n, dtype=torch.int64, device=dev)
+        torch.cumsum(line_len.to(torch.int64)[:-1], 0, out=line_off[1:])
+        total = int(line_off[-1].item() + line_len[-1].item()) if n else 0
+        ovf = int(overflow.item())
+        if ovf:
+            raise RuntimeError(
+                f"line overflowed scratch stride ({ovf} > {stride}); "
+                "stride bound is unsound for this batch"
+            )
+        out = torch.empty(total, dtype=torch.uint8, device=dev)
+        rc = lib.crawl_compact(
+            ctypes.c_void_p(scratch.data_ptr()), stride,
+            ctypes.c_void_p(line_off.data_ptr()),
+            ctypes.c_void_p(line_len.data_ptr()),
+            ctypes.c_void_p(out.data_ptr()), n, grid, stream_ptr,
+        )
+        if rc != 0:
+            raise RuntimeError(f"crawl_compact failed: hip error {rc}")
+    else:
+        rc = lib.crawl_measure_extract(
+            batch_ptrs, scalars, link_ptrs,
+            ctypes.c_void_p(line_len.data_ptr()), grid, stream_ptr,
+        )
+        if rc != 0:
+            raise RuntimeError(
+                f"crawl_measure_extract launch failed: hip error {rc}"
+            )
+        line_off = torch.zeros(n, dtype=torch.int64, device=dev)
+        torch.cumsum(line_len.to(torch.int64)[:-1], 0, out=line_off[1:])
+        total = int(line_off[-1].item() + line_len[-1].item()) if n else 0
+        out = torch.empty(total, dtype=torch.uint8, device=dev)
+        rc = lib.crawl_write(
+            batch_ptrs, scalars, link_ptrs,
+            ctypes.c_void_p(line_off.data_ptr()),
+            ctypes.c_void_p(line_len.data_ptr()),
+            ctypes.c_void_p(out.data_ptr()), grid, stream_ptr,
+        )
+        if rc != 0:
+            raise RuntimeError(f"crawl_write launch failed: hip error {rc}")
 
     return EncodeResult(
         out=out, line_off=line_off, line_len=line_len,
