@@ -116,7 +116,10 @@ def main():
         chunk = chunks[si % n_chunks]
         slot = si % 2
         with torch.cuda.stream(compute_stream):
-            res = gpu.parse_encode(chunk, now=now)
+            res = gpu.parse_encode(
+                chunk, now=now,
+                single_pass=os.environ.get("CRAWL_SINGLE_PASS", "") == "1",
+            )
             new_mask = seen.claim(res)
             if world > 1:
                 flat = new_mask.flatten().bool()

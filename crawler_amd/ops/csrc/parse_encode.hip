@@ -544,6 +544,39 @@ write_kernel(BatchView B, LinkOut LO, const long* line_off,
   }
 }
 
+// LDS-staged write: emit each line into a per-wave LDS buffer (fast byte
+// stores, no vmem latency on the hot emission path), then stream it to its
+// exact global offset with the funnel-shift dword copy. Lines larger than
+// the LDS budget (rare) fall back to direct global emission.
+
+#define LDS_LINE_BYTES (12 * 1024)
+
+DEV void copy_line(const unsigned char* src, unsigned char* dst, int n,
+                   int lane);
+
+__global__ void __launch_bounds__(256)
+write_lds_kernel(BatchView B, LinkOut LO, const long* line_off,
+                 const int* line_len, unsigned char* out) {
+  __shared__ unsigned char lbuf[4][LDS_LINE_BYTES];
+  const int lane = lane_id();
+  const int wave = wave_id();
+  const int waves_per_grid = gridDim.x * 4;
+  for (int i = blockIdx.x * 4 + wave; i < B.n; i += waves_per_grid) {
+    const int len = line_len[i];
+    if (len == 0) continue;
+    LinkList L{LO.name + (size_t)i * MAX_LINKS * 32,
+               LO.name_len + (size_t)i * MAX_LINKS,
+               LO.src + (size_t)i * MAX_LINKS,
+               LO.hash + (size_t)i * MAX_LINKS, LO.cnt[i]};
+    if (len > LDS_LINE_BYTES) {
+      emit_line<true>(B, i, out + line_off[i], L);
+      continue;
+    }
+    emit_line<true>(B, i, &lbuf[wave][0], L);
+    copy_line(&lbuf[wave][0], out + line_off[i], len, lane);
+  }
+}
+
 // Single-pass variant: emit straight into per-message scratch slots (the
 // host supplies a sound stride bound), then a vectorized compaction gathers
 // the final contiguous JSONL. Replaces measure+write (saves the whole
@@ -733,6 +766,17 @@ int crawl_write_scratch(void** batch_ptrs, const long* scalars,
   hipLaunchKernelGGL(crawl::write_scratch_kernel, dim3(grid), dim3(256), 0,
                      (hipStream_t)stream, B, LO, (unsigned char*)scratch,
                      stride, (int*)line_len, (int*)overflow);
+  return (int)hipGetLastError();
+}
+
+int crawl_write_lds(void** batch_ptrs, const long* scalars, void** link_ptrs,
+                    const void* line_off, const void* line_len, void* out,
+                    int grid, void* stream) {
+  crawl::BatchView B = crawl::make_view(batch_ptrs, scalars);
+  crawl::LinkOut LO = crawl::make_links(link_ptrs);
+  hipLaunchKernelGGL(crawl::write_lds_kernel, dim3(grid), dim3(256), 0,
+                     (hipStream_t)stream, B, LO, (const long*)line_off,
+                     (const int*)line_len, (unsigned char*)out);
   return (int)hipGetLastError();
 }
 
