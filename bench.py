@@ -101,7 +101,10 @@ def main():
         for _ in range(2)
     ]
     compute_stream = torch.cuda.Stream()
-    copy_stream = torch.cuda.Stream()
+    # two copy streams: MI355X has multiple SDMA engines; splitting the
+    # 2.5GB D2H in half across streams uses two of them
+    copy_streams = [torch.cuda.Stream(), torch.cuda.Stream()]
+    n_copy = 2 if os.environ.get("CRAWL_ONE_SDMA", "") != "1" else 1
     inflight = [None, None]  # keep res tensors alive while copying
     sink_f = None
     if args.sink == "file" and rank == 0:
@@ -137,20 +140,31 @@ def main():
         nbytes = res.out.numel()
         ev = torch.cuda.Event()
         ev.record(compute_stream)
-        copy_stream.wait_event(ev)
         # the previous copy into this pinned slot must be done before reuse
         if inflight[slot] is not None:
-            inflight[slot][1].synchronize()
-        with torch.cuda.stream(copy_stream):
-            pinned[slot][:nbytes].copy_(res.out, non_blocking=True)
-            res.out.record_stream(copy_stream)
-        done = torch.cuda.Event()
-        done.record(copy_stream)
-        inflight[slot] = (res.out, done)
+            for d in inflight[slot][1]:
+                d.synchronize()
+        dones = []
+        half = (nbytes // n_copy + 15) & ~15
+        for ci in range(n_copy):
+            cs = copy_streams[ci]
+            cs.wait_event(ev)
+            lo = ci * half
+            hi = min(nbytes, lo + half)
+            if lo >= hi:
+                continue
+            with torch.cuda.stream(cs):
+                pinned[slot][lo:hi].copy_(res.out[lo:hi], non_blocking=True)
+                res.out.record_stream(cs)
+            d = torch.cuda.Event()
+            d.record(cs)
+            dones.append(d)
+        inflight[slot] = (res.out, dones)
         total_out_bytes += nbytes
         new_discoveries += nd
         if sink_f is not None:
-            done.synchronize()
+            for d in dones:
+                d.synchronize()
             sink_f.write(bytes(pinned[slot][:nbytes].numpy()))
         return nbytes
 

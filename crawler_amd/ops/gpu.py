@@ -348,8 +348,9 @@ def parse_encode(
         torch.cumsum(line_len.to(torch.int64)[:-1], 0, out=line_off[1:])
         total = int(line_off[-1].item() + line_len[-1].item()) if n else 0
         out = torch.empty(total, dtype=torch.uint8, device=dev)
-        writer = (lib.crawl_write if os.environ.get("CRAWL_PLAIN_WRITE")
-                  else lib.crawl_write_lds)
+        writer = (lib.crawl_write_lds if os.environ.get("CRAWL_LDS_WRITE")
+                  else lib.crawl_write)  # LDS staging measured 2x slower
+                  # (48KB static shared tanks occupancy; see profiles/)
         rc = writer(
             batch_ptrs, scalars, link_ptrs,
             ctypes.c_void_p(line_off.data_ptr()),
