@@ -544,6 +544,47 @@ class LocalStateManager(BaseStateManager):
         os.replace(source_path, dst)
         return dst, file_name
 
+    # ---- page export (daprstate.go:2906-3012) ----
+
+    export_chunk_size_bytes = 100 * 1024 * 1024  # daprstate.go:2909
+
+    def export_pages_to_binding(self, crawl_id: str) -> List[str]:
+        """Chunked JSONL export of all pages to
+        analysis/channels/channel-pages-<crawlID>-partN.jsonl
+        (100 MB raw per chunk; daprstate.go:2906-3012)."""
+        out_dir = os.path.join(self.base_path, "analysis", "channels")
+        os.makedirs(out_dir, exist_ok=True)
+        paths: List[str] = []
+        part = 0
+        buf: List[bytes] = []
+        size = 0
+
+        def flush():
+            nonlocal part, buf, size
+            if not buf:
+                return
+            path = os.path.join(
+                out_dir, f"channel-pages-{crawl_id}-part{part}.jsonl"
+            )
+            with open(path, "wb") as f:
+                f.writelines(buf)
+            paths.append(path)
+            part += 1
+            buf = []
+            size = 0
+
+        with self._lock:
+            pages = [self.pages[i] for ids in self.layer_map.values()
+                     for i in ids]
+        for p in pages:
+            line = (json.dumps(p.to_dict()) + "\n").encode()
+            if size + len(line) > self.export_chunk_size_bytes and buf:
+                flush()
+            buf.append(line)
+            size += len(line)
+        flush()
+        return paths
+
     # ---- media cache ----
 
     def has_processed_media(self, media_id: str) -> bool:
