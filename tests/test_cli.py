@@ -1,0 +1,85 @@
+"""CLI surface + end-to-end standalone runs (BASELINE config #1: two mock
+channels, CPU, --skip-media, local JSONL)."""
+import json
+
+import pytest
+
+from crawler_amd.cli import build_parser, main, parse_config
+
+
+def test_parser_accepts_reference_flag_surface():
+    p = build_parser()
+    args = p.parse_args([
+        "--mode", "standalone", "--urls", "a,b", "--max-depth", "2",
+        "--concurrency", "4", "--skip-media", "--platform", "telegram",
+        "--sampling", "snowball", "--min-users", "50", "--max-posts", "10",
+        "--crawl-id", "c123", "--walkback-rate", "20", "--tandem-crawl",
+        "--validator-request-rate", "6", "--combine-files",
+        "--combine-trigger-size", "170", "--time-ago", "30d",
+        "--storage-root", "/tmp/x", "--exit-on-complete",
+    ])
+    assert args.sampling == "snowball"
+    assert args.walkback_rate == 20
+
+
+def test_parse_config_time_ago_and_dates():
+    cfg = parse_config([
+        "--urls", "abcde", "--time-ago", "30d", "--crawl-id", "t",
+    ])
+    assert cfg.post_recency is not None
+    cfg = parse_config([
+        "--urls", "abcde", "--date-between", "2024-01-01,2024-02-01",
+        "--sample-size", "5", "--crawl-id", "t",
+    ])
+    assert cfg.date_between_min.year == 2024
+    assert cfg.sample_size == 5
+
+
+def test_invalid_sampling_combo_raises():
+    with pytest.raises(ValueError):
+        main(["--mode", "standalone", "--platform", "telegram",
+              "--sampling", "random", "--urls", "abcde"])
+
+
+def test_e2e_standalone_two_mock_channels(tmp_path):
+    """BASELINE config #1: standalone, 2 mock channels, CPU, skip-media."""
+    rc = main([
+        "--mode", "standalone", "--urls", "c0000000001,c0000000002",
+        "--skip-media", "--storage-root", str(tmp_path),
+        "--crawl-id", "fix1", "--synthetic-universe", "100",
+        "--synthetic-posts", "20", "--disable-rate-limits",
+        "--min-users", "1",
+    ])
+    assert rc == 0
+    crawl = tmp_path / "fix1"
+    assert (crawl / "progress.json").exists()
+    prog = json.loads((crawl / "progress.json").read_text())
+    assert prog["layers"][0]["total"] == 2
+    assert prog["layers"][0]["completed"] == 2
+    for ch in ("c0000000001", "c0000000002"):
+        lines = (crawl / ch / "posts" / "posts.jsonl").read_bytes()
+        assert lines.count(b"\n") == 20
+        obj = json.loads(lines.splitlines()[0])
+        assert obj["platform_name"] == "Telegram"
+        assert obj["thumb_url"] == ""  # skip-media
+
+
+def test_e2e_random_walk_seed_size(tmp_path):
+    rc = main([
+        "--mode", "standalone", "--sampling", "random-walk",
+        "--seed-size", "3", "--storage-root", str(tmp_path),
+        "--crawl-id", "rw1", "--synthetic-universe", "200",
+        "--synthetic-posts", "15", "--disable-rate-limits",
+        "--min-users", "1", "--max-pages", "6",
+        "--max-crawl-duration", "30s",
+    ])
+    assert rc == 0
+
+
+def test_e2e_youtube_random(tmp_path):
+    rc = main([
+        "--mode", "standalone", "--platform", "youtube",
+        "--sampling", "random", "--storage-root", str(tmp_path),
+        "--crawl-id", "yt9", "--max-posts", "5",
+    ])
+    assert rc == 0
