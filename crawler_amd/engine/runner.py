@@ -52,6 +52,10 @@ class StandaloneRunner:
         if not resumed:
             sm.initialize(seed_urls)
             sm.save_state()
+        # cross-crawl URL dedup (daprstate.go:550-623); random-walk mode
+        # deliberately skips it (duplicate visits allowed)
+        if self.cfg.sampling_method != "random-walk":
+            sm.load_url_dedup_cache()
 
         depth = 0
         max_depth = self.cfg.max_depth

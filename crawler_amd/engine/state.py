@@ -279,11 +279,16 @@ class BaseStateManager:
             max_reached = max_pages > 0 and total >= max_pages
             replacements = deadends
             existing_urls = {p.url: pid for pid, p in self.pages.items()}
+            url_dedup = getattr(self, "url_dedup", {})
             depth = pages[0].depth
             self.layer_map.setdefault(depth, [])
             added = []
             for p in pages:
                 if p.url in existing_urls:
+                    continue
+                if p.url in url_dedup and depth > 0:
+                    # crawled by a previous crawl (cross-crawl URL cache);
+                    # seeds (depth 0) are always admitted
                     continue
                 if max_reached:
                     if replacements <= 0:
@@ -374,6 +379,31 @@ class LocalStateManager(BaseStateManager):
         )
         self.media_cache: Dict[str, dict] = {}
         self._post_files = {}
+        self.url_dedup: Dict[str, str] = {}
+
+    def load_url_dedup_cache(self) -> int:
+        """Cross-crawl URL dedup: url -> "crawlID:pageID" loaded from all
+        previous crawls' state files (daprstate.go:550-623). Skipped
+        entirely for random-walk (duplicate visits are allowed there,
+        daprstate.go:642-657 — callers must not invoke this in that mode).
+        """
+        cache: Dict[str, str] = {}
+        for prev in self.get_previous_crawls():
+            path = os.path.join(self.base_path, prev, "state.json")
+            if not os.path.exists(path):
+                continue
+            try:
+                with open(path) as f:
+                    state = json.load(f)
+            except (OSError, json.JSONDecodeError):
+                continue
+            for layer in state.get("layers", []):
+                for pd in layer.get("pages", []):
+                    url = pd.get("url")
+                    if url and url not in cache:
+                        cache[url] = f"{prev}:{pd.get('id', '')}"
+        self.url_dedup = cache
+        return len(cache)
 
     # paths (storageproviders.go:634-647)
     def _crawl_dir(self):
