@@ -130,24 +130,35 @@ class MediaEngine:
         return path
 
     def _spill_gpu(self, media_id: str, size: int, path: str):
-        """Stage in HBM, spill via the pinned ring in slot-sized pieces."""
+        """Stage in HBM, spill via the pinned ring in slot-sized pieces.
+
+        Issue and drain are PIPELINED: when the ring is exhausted the
+        oldest in-flight piece is drained before issuing the next (a
+        blocking get() with all slots in flight would deadlock)."""
+        import collections
+
         blob = device_blob(media_id, size, self.device, self.seed)
-        pieces = []
-        with torch.cuda.stream(self._stream):
-            for off in range(0, size, self._slot_bytes):
-                n = min(self._slot_bytes, size - off)
-                buf = self._ring_free.get()
+        pending = collections.deque()  # (buf, n, ev, dst_off)
+        data = bytearray(size)
+
+        def drain_one():
+            buf, n, ev, dst = pending.popleft()
+            ev.synchronize()
+            data[dst:dst + n] = bytes(buf[:n].numpy())
+            self._ring_free.put(buf)
+
+        for off in range(0, size, self._slot_bytes):
+            n = min(self._slot_bytes, size - off)
+            while self._ring_free.empty() and pending:
+                drain_one()
+            buf = self._ring_free.get()
+            with torch.cuda.stream(self._stream):
                 buf[:n].copy_(blob[off:off + n], non_blocking=True)
                 ev = torch.cuda.Event()
                 ev.record(self._stream)
-                pieces.append((buf, n, ev))
-        data = bytearray(size)
-        off = 0
-        for (buf, n, ev) in pieces:
-            ev.synchronize()
-            data[off:off + n] = bytes(buf[:n].numpy())
-            off += n
-            self._ring_free.put(buf)
+            pending.append((buf, n, ev, off))
+        while pending:
+            drain_one()
         self._q.put((path, bytes(data), None))
 
     def flush(self):
