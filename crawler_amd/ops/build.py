@@ -11,7 +11,7 @@ import subprocess
 import sys
 
 CSRC = os.path.dirname(os.path.abspath(__file__)) + "/csrc"
-SOURCES = ["parse_encode.hip", "dedup.hip", "feedgen.hip"]
+SOURCES = ["parse_encode.hip", "dedup.hip", "feedgen.hip", "yt_encode.hip"]
 OUT = os.path.join(CSRC, "libcrawlhip.so")
 
 

@@ -66,6 +66,7 @@ def load_lib():
     _bind_dedup(lib)
     _bind_feedgen(lib)
     _bind_v2(lib)
+    _bind_yt(lib)
     _lib = lib
     return lib
 
@@ -88,6 +89,78 @@ def _bind_v2(lib):
         ctypes.c_void_p, ctypes.c_long, ctypes.c_void_p, ctypes.c_void_p,
         ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
     ]
+
+
+def _bind_yt(lib):
+    lib.crawl_yt_ptr_count.restype = ctypes.c_int
+    lib.crawl_yt_measure.restype = ctypes.c_int
+    lib.crawl_yt_measure.argtypes = [
+        ctypes.POINTER(ctypes.c_void_p), ctypes.POINTER(ctypes.c_long),
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+    ]
+    lib.crawl_yt_write.restype = ctypes.c_int
+    lib.crawl_yt_write.argtypes = [
+        ctypes.POINTER(ctypes.c_void_p), ctypes.POINTER(ctypes.c_long),
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+    ]
+
+
+# Must match crawl::yt_make_view() in csrc/yt_encode.hip.
+_YT_PTR_ORDER = [
+    "vid_off", "channel_idx", "published", "views", "likes", "comments",
+    "duration_s", "lang", "title_off", "title_len", "desc_off", "desc_len",
+    "pool", "ch_id_off", "ch_title_off", "ch_title_len", "ch_desc_off",
+    "ch_desc_len", "ch_subs", "ch_videos", "ch_views", "ch_country_off",
+    "ch_country_len", "ch_published",
+]
+
+
+def yt_parse_encode(batch, now: Optional[_dt.datetime] = None,
+                    grid: int = 0) -> Tuple[torch.Tensor, torch.Tensor,
+                                            torch.Tensor]:
+    """YouTube batch -> (out bytes, line_off, line_len); byte-identical to
+    youtube.batch.encode_yt_batch."""
+    lib = require_lib()
+    dev = batch.device
+    assert dev.type == "cuda"
+    n = batch.n
+    now = now or _dt.datetime.now(_dt.timezone.utc)
+    created = format_go_time(now.replace(microsecond=0)).encode()
+    capture = format_go_time(now).encode()
+    to_dev = lambda b: torch.frombuffer(bytearray(b or b"\0"),
+                                        dtype=torch.uint8).to(dev)
+    label_t = to_dev(batch.crawl_label.encode())
+    from ..youtube.batch import LANGS
+
+    lang_t = to_dev("".join(LANGS).encode())
+    created_t, capture_t = to_dev(created), to_dev(capture)
+    tensors = [getattr(batch, name) for name in _YT_PTR_ORDER]
+    tensors += [label_t, lang_t, created_t, capture_t]
+    ptrs = _ptr_array(tensors)
+    assert len(tensors) == lib.crawl_yt_ptr_count()
+    scalars = (ctypes.c_long * 4)(
+        n, len(batch.crawl_label.encode()), len(created), len(capture)
+    )
+    if grid <= 0:
+        grid = min((n + 3) // 4, 8192)
+    stream_ptr = ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+    line_len = torch.zeros(n, dtype=torch.int32, device=dev)
+    rc = lib.crawl_yt_measure(ptrs, scalars,
+                              ctypes.c_void_p(line_len.data_ptr()),
+                              grid, stream_ptr)
+    if rc != 0:
+        raise RuntimeError(f"crawl_yt_measure failed: hip {rc}")
+    line_off = torch.zeros(n, dtype=torch.int64, device=dev)
+    torch.cumsum(line_len.to(torch.int64)[:-1], 0, out=line_off[1:])
+    total = int(line_off[-1].item() + line_len[-1].item()) if n else 0
+    out = torch.empty(total, dtype=torch.uint8, device=dev)
+    rc = lib.crawl_yt_write(ptrs, scalars,
+                            ctypes.c_void_p(line_off.data_ptr()),
+                            ctypes.c_void_p(out.data_ptr()),
+                            grid, stream_ptr)
+    if rc != 0:
+        raise RuntimeError(f"crawl_yt_write failed: hip {rc}")
+    return out, line_off, line_len
 
 
 def _bind_dedup(lib):

@@ -137,8 +137,10 @@ def convert_video_to_post(
         searchable_text=f"{video.title} {video.description}",
         all_text=f"{video.title} {video.description}",
         thumb_url=thumb,
-        comments=[],
-        reactions={},
+        # the reference's convertVideoToPost never sets Comments/Reactions
+        # (nil slices -> null); comment COUNTS come from the API stats
+        comments=None,
+        reactions=None,
         outlinks=extract_urls(video.description),
         capture_time=now,
         handle=channel_name,

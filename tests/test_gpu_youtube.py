@@ -1,0 +1,45 @@
+"""GPU YouTube encoder byte-exactness tests vs the CPU oracle."""
+import datetime as dt
+
+import pytest
+import torch
+
+from crawler_amd.youtube.batch import build_corpus, encode_yt_batch
+from crawler_amd.youtube.synth import SyntheticYouTubeIndex
+
+pytestmark = pytest.mark.gpu
+
+NOW = dt.datetime(2026, 3, 4, 5, 6, 7, tzinfo=dt.timezone.utc)
+
+
+def test_yt_bytes_identical():
+    from crawler_amd.ops import gpu
+
+    idx = SyntheticYouTubeIndex(seed=5, universe_channels=3000)
+    batch = build_corpus(idx, 500, crawl_label="yt-bench")
+    golden = encode_yt_batch(batch, now=NOW)
+    out, line_off, line_len = gpu.yt_parse_encode(batch.to("cuda:0"),
+                                                  now=NOW)
+    torch.cuda.synchronize()
+    got = bytes(out.cpu().numpy())
+    expect = b"".join(golden)
+    if got != expect:
+        offs = line_off.cpu().numpy()
+        lens = line_len.cpu().numpy()
+        for i, gl in enumerate(golden):
+            dev_line = got[offs[i]: offs[i] + lens[i]]
+            assert dev_line == gl, (
+                f"video {i}:\nGPU: {dev_line[:400]!r}\nCPU: {gl[:400]!r}"
+            )
+    assert got == expect
+
+
+def test_yt_p0d_and_empty_label():
+    from crawler_amd.ops import gpu
+
+    idx = SyntheticYouTubeIndex(seed=9, universe_channels=100)
+    batch = build_corpus(idx, 64)
+    golden = encode_yt_batch(batch, now=NOW)
+    out, _, _ = gpu.yt_parse_encode(batch.to("cuda:0"), now=NOW)
+    torch.cuda.synchronize()
+    assert bytes(out.cpu().numpy()) == b"".join(golden)
