@@ -48,6 +48,14 @@ def main():
     ap.add_argument("--sink", choices=["pinned", "file", "none"],
                     default="pinned")
     ap.add_argument("--sink-dir", default="/tmp/crawl-bench")
+    ap.add_argument("--platform", choices=["telegram", "youtube", "mixed"],
+                    default="telegram",
+                    help="mixed = even ranks telegram, odd ranks youtube "
+                         "(BASELINE config #4 dispatch)")
+    ap.add_argument("--videos", type=int, default=400_000,
+                    help="videos per GPU for the youtube platform")
+    ap.add_argument("--chunk-videos", type=int, default=100_000)
+    ap.add_argument("--max-comments", type=int, default=1000)
     ap.add_argument("--cpu", action="store_true",
                     help="debug: run the Python golden path on CPU (tiny)")
     args = ap.parse_args()
@@ -72,22 +80,45 @@ def main():
     gpu.require_lib()
 
     # ---- setup (untimed): generate the per-rank corpus ----
-    n_chunks = max(1, args.channels // args.chunk_channels)
-    chunk_ch = args.chunk_channels
-    posts = args.posts
-    feed = SyntheticFeed(FeedConfig(seed=1234 + rank, universe=1_000_000))
+    my_platform = args.platform
+    if args.platform == "mixed":
+        my_platform = "telegram" if rank % 2 == 0 else "youtube"
+
     chunks = []
     t_gen = time.time()
-    for c in range(n_chunks):
-        ids = np.arange(c * chunk_ch, (c + 1) * chunk_ch) + rank * args.channels
-        chunks.append(
-            feed.build_batch_device(ids, device, posts_per_channel=posts)
-        )
-        torch.cuda.synchronize()
-        log(f"chunk {c + 1}/{n_chunks} generated on-device "
-            f"({(c + 1) * chunk_ch * posts / 1e6:.2f}M posts, "
-            f"{time.time() - t_gen:.1f}s)")
-    chunk_posts = chunk_ch * posts
+    if my_platform == "telegram":
+        n_chunks = max(1, args.channels // args.chunk_channels)
+        chunk_ch = args.chunk_channels
+        posts = args.posts
+        feed = SyntheticFeed(FeedConfig(seed=1234 + rank,
+                                        universe=1_000_000))
+        for c in range(n_chunks):
+            ids = (np.arange(c * chunk_ch, (c + 1) * chunk_ch)
+                   + rank * args.channels)
+            chunks.append(
+                feed.build_batch_device(ids, device, posts_per_channel=posts)
+            )
+            torch.cuda.synchronize()
+            log(f"chunk {c + 1}/{n_chunks} generated on-device "
+                f"({(c + 1) * chunk_ch * posts / 1e6:.2f}M posts, "
+                f"{time.time() - t_gen:.1f}s)")
+        chunk_posts = chunk_ch * posts
+    else:
+        from crawler_amd.youtube.batch import build_corpus
+        from crawler_amd.youtube.synth import SyntheticYouTubeIndex
+
+        idx = SyntheticYouTubeIndex(seed=99 + rank,
+                                    universe_channels=1_000_000)
+        n_chunks = max(1, args.videos // args.chunk_videos)
+        for c in range(n_chunks):
+            chunks.append(
+                build_corpus(idx, args.chunk_videos,
+                             crawl_label="yt-bench").to(device)
+            )
+            log(f"yt chunk {c + 1}/{n_chunks} "
+                f"({args.chunk_videos} videos, "
+                f"{time.time() - t_gen:.1f}s)")
+        chunk_posts = args.chunk_videos
 
     now = dt.datetime(2026, 1, 1, tzinfo=dt.timezone.utc)
     seen = gpu.SeenSet(device)
@@ -119,25 +150,41 @@ def main():
         chunk = chunks[si % n_chunks]
         slot = si % 2
         with torch.cuda.stream(compute_stream):
-            res = gpu.parse_encode(
-                chunk, now=now,
-                single_pass=os.environ.get("CRAWL_SINGLE_PASS", "") == "1",
-            )
-            new_mask = seen.claim(res)
-            if world > 1:
-                flat = new_mask.flatten().bool()
-                new_hashes = res.link_hash.flatten()[flat]
-                cap = 65536
-                buf = torch.zeros(cap, dtype=torch.int64, device=device)
-                k = min(new_hashes.numel(), cap)
-                buf[:k] = new_hashes[:k]
-                gathered = [torch.empty_like(buf) for _ in range(world)]
-                torch.distributed.all_gather(gathered, buf)
-                for r, g in enumerate(gathered):
-                    if r != rank:
-                        seen.insert_hashes(g[g != 0])
-            nd = seen.new_count()  # syncs within the compute stream
-        nbytes = res.out.numel()
+            if my_platform == "telegram":
+                res = gpu.parse_encode(
+                    chunk, now=now,
+                    single_pass=os.environ.get("CRAWL_SINGLE_PASS", "") == "1",
+                )
+                out_t = res.out
+                new_mask = seen.claim(res)
+                if world > 1:
+                    flat = new_mask.flatten().bool()
+                    new_hashes = res.link_hash.flatten()[flat]
+                    cap = 65536
+                    buf = torch.zeros(cap, dtype=torch.int64, device=device)
+                    k = min(new_hashes.numel(), cap)
+                    buf[:k] = new_hashes[:k]
+                    gathered = [torch.empty_like(buf) for _ in range(world)]
+                    torch.distributed.all_gather(gathered, buf)
+                    for r, g in enumerate(gathered):
+                        if r != rank:
+                            seen.insert_hashes(g[g != 0])
+                nd = seen.new_count()  # syncs within the compute stream
+            else:
+                out_t, _off, _len = gpu.yt_parse_encode(chunk, now=now)
+                if world > 1:
+                    # mixed dispatch still exchanges (empty) discovery
+                    # buffers so collectives stay symmetric across ranks
+                    buf = torch.zeros(65536, dtype=torch.int64,
+                                      device=device)
+                    gathered = [torch.empty_like(buf)
+                                for _ in range(world)]
+                    torch.distributed.all_gather(gathered, buf)
+                    for r, g in enumerate(gathered):
+                        if r != rank:
+                            seen.insert_hashes(g[g != 0])
+                nd = 0
+        nbytes = out_t.numel()
         ev = torch.cuda.Event()
         ev.record(compute_stream)
         # the previous copy into this pinned slot must be done before reuse
@@ -154,12 +201,12 @@ def main():
             if lo >= hi:
                 continue
             with torch.cuda.stream(cs):
-                pinned[slot][lo:hi].copy_(res.out[lo:hi], non_blocking=True)
-                res.out.record_stream(cs)
+                pinned[slot][lo:hi].copy_(out_t[lo:hi], non_blocking=True)
+                out_t.record_stream(cs)
             d = torch.cuda.Event()
             d.record(cs)
             dones.append(d)
-        inflight[slot] = (res.out, dones)
+        inflight[slot] = (out_t, dones)
         total_out_bytes += nbytes
         new_discoveries += nd
         if sink_f is not None:
@@ -215,10 +262,13 @@ def main():
             "dtype": "uint8",
             "data": "synthetic",
             "config": {
-                "model": "telegram-crawl 1k-channels x 10k-posts per GPU "
-                         "(BASELINE config #2, weak-scaled)",
+                "model": ("telegram-crawl 1k-channels x 10k-posts per GPU "
+                          "(BASELINE config #2, weak-scaled)"
+                          if args.platform == "telegram" else
+                          f"{args.platform} crawl (BASELINE config #4 dispatch)"),
                 "global_batch": world * chunk_posts,
                 "seq_len": 0,
+                "platform": args.platform,
                 "parallelism": f"dp{world} (channel-sharded, RCCL discovery "
                                "all-gather)" if world > 1 else "dp1",
                 "channels_per_gpu": args.channels,
