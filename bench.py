@@ -220,6 +220,28 @@ def main():
         step(w)
     log(f"warmup done ({args.warmup} steps)")
 
+    # ---- p50 channel latency (BASELINE secondary metric): one channel's
+    # batch through the full pipeline (parse+encode -> claim -> D2H) ----
+    p50_channel_ms = None
+    if my_platform == "telegram":
+        lat = []
+        one = feed.build_batch_device(
+            np.array([999_990 + rank]), device, posts_per_channel=args.posts
+        )
+        lat_pin = torch.empty(args.posts * 3000, dtype=torch.uint8,
+                              pin_memory=True)
+        for _ in range(11):
+            torch.cuda.synchronize()
+            t = time.perf_counter()
+            r1 = gpu.parse_encode(one, now=now)
+            seen.claim(r1)
+            lat_pin[: r1.out.numel()].copy_(r1.out, non_blocking=True)
+            torch.cuda.synchronize()
+            lat.append((time.perf_counter() - t) * 1000)
+        p50_channel_ms = round(statistics.median(lat), 3)
+        log(f"p50 channel latency: {p50_channel_ms} ms "
+            f"({args.posts} posts/channel)")
+
     # ---- timed region ----
     if world > 1:
         torch.distributed.barrier()
@@ -277,6 +299,7 @@ def main():
                 "jsonl_bytes_per_step": total_out_bytes // max(
                     1, args.warmup + args.steps),
                 "p50_step_ms": round(statistics.median(step_times), 2),
+                "p50_channel_latency_ms": p50_channel_ms,
                 "new_discoveries": new_discoveries,
                 "sink": args.sink,
             },
