@@ -215,13 +215,38 @@ class RandomWalkRunner:
         cap = max_pages or self.cfg.max_pages
         max_workers = max(1, self.cfg.concurrency)
         in_flight = set()
+        blocked_since = None  # tandem circuit breaker state
         while self.stats["pages"] < cap:
             if deadline and time.monotonic() > deadline:
                 break
             pages = [p for p in self.rw.get_pages(max_workers)
                      if p.id not in in_flight]
             if not pages:
-                break  # synthetic buffer never refills asynchronously here
+                if not self.cfg.tandem_crawl:
+                    break
+                # Tandem: an empty buffer with incomplete batches means we
+                # are waiting on the validator; a circuit breaker aborts
+                # after validator_timeout of zero progress
+                # (dapr/standalone.go:837-867).
+                incomplete = self.rw.count_incomplete_batches(
+                    self.cfg.crawl_id
+                )
+                if incomplete == 0:
+                    break  # crawl complete
+                now_t = time.monotonic()
+                if blocked_since is None:
+                    blocked_since = now_t
+                timeout = self.cfg.validator_timeout_s
+                if timeout and now_t - blocked_since > timeout:
+                    self.stats["circuit_breaker"] = 1
+                    raise E.PoolExhausted(
+                        f"validator made no progress for {timeout}s with "
+                        f"{incomplete} incomplete batches — aborting "
+                        "(circuit breaker)"
+                    )
+                time.sleep(self.poll_interval or 0.01)
+                continue
+            blocked_since = None
             for page in pages:
                 if self.stats["pages"] >= cap:
                     break
