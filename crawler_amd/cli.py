@@ -224,6 +224,16 @@ def main(argv: Optional[List[str]] = None) -> int:
         mode=args.mode, seed_size=cfg.seed_size, crawl_id=cfg.crawl_id,
     )
 
+    if args.generate_code:
+        # TDLib auth-code generator stand-in (standalone/runner.go:77-192):
+        # the synthetic engine needs no live auth; emit a deterministic
+        # pairing code so operator scripts keep working.
+        import hashlib
+
+        code = hashlib.sha256(cfg.crawl_id.encode()).hexdigest()[:6]
+        print(f"synthetic-tdlib auth code: {code}")
+        return 0
+
     mode = args.mode or ("standalone" if urls or cfg.seed_size else "")
     if not mode:
         print("no mode and no URLs given; nothing to do", file=sys.stderr)

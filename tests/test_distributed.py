@@ -217,3 +217,30 @@ def test_gloo_discovery_allgather_two_procs(tmp_path):
     r1 = (tmp_path / "r1.txt").read_text()
     assert r0 == r1
     assert "101" in r0 and "1" in r0
+
+
+def _names_worker(rank, world, port, results_dir):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from crawler_amd.engine.gpu_runner import GpuCrawlEngine
+
+    mine = [f"rank{rank}chan{i}" for i in range(2 + rank)]
+    merged = GpuCrawlEngine._allgather_names(mine)
+    with open(os.path.join(results_dir, f"n{rank}.txt"), "w") as f:
+        f.write(",".join(sorted(merged)))
+    dist.destroy_process_group()
+
+
+def test_gpu_engine_name_exchange_two_procs(tmp_path):
+    """The multi-rank discovered-name exchange used by the GPU snowball
+    engine, run on gloo CPU (the RCCL path shares this code)."""
+    import torch.multiprocessing as mp
+
+    mp.spawn(_names_worker, args=(2, 29713, str(tmp_path)), nprocs=2,
+             join=True)
+    r0 = (tmp_path / "n0.txt").read_text()
+    assert r0 == (tmp_path / "n1.txt").read_text()
+    assert "rank0chan0" in r0 and "rank1chan2" in r0
