@@ -279,6 +279,16 @@ def main(argv: Optional[List[str]] = None) -> int:
         run_validation_loop(cfg, sm, rw)
         return 0
 
+    if args.gpu and cfg.sampling_method in ("channel", "snowball"):
+        # MI355X execution mode: whole layers through the HIP kernels
+        from .engine.gpu_runner import GpuCrawlEngine
+
+        engine = GpuCrawlEngine(cfg, sm, feed,
+                                posts_per_channel=args.synthetic_posts)
+        stats = engine.run(urls)
+        print(f"gpu crawl complete: {stats}", file=sys.stderr)
+        return 0
+
     if cfg.sampling_method == "random-walk":
         rw = RandomWalkStore()
         runner = RandomWalkRunner(cfg, sm, rw, pool)

@@ -76,3 +76,26 @@ def test_exactly_once_across_layers(tmp_path):
     for d in range(sm2.get_max_depth() + 1):
         seen_urls += [p.url for p in sm2.get_layer_by_depth(d)]
     assert len(seen_urls) == len(set(seen_urls))
+
+
+def test_cli_gpu_mode_end_to_end(tmp_path):
+    """--gpu standalone snowball through the CLI: JSONL files + completed
+    progress.json, hot path on the HIP kernels."""
+    from crawler_amd.cli import main
+
+    rc = main([
+        "--mode", "standalone", "--gpu", "--sampling", "snowball",
+        "--urls", "c0000000001", "--max-depth", "1",
+        "--storage-root", str(tmp_path), "--crawl-id", "gcli1",
+        "--synthetic-universe", "300", "--synthetic-posts", "32",
+        "--min-users", "1", "--skip-media",
+    ])
+    assert rc == 0
+    import json
+
+    prog = json.loads((tmp_path / "gcli1" / "progress.json").read_text())
+    assert prog["status"] == "completed"
+    jsonls = list((tmp_path / "gcli1").rglob("posts.jsonl"))
+    assert len(jsonls) >= 2  # seed + discovered channels
+    obj = json.loads(jsonls[0].read_bytes().splitlines()[0])
+    assert obj["platform_name"] == "Telegram"
