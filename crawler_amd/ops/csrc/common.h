@@ -105,117 +105,12 @@ DEV int wave_sum(int v) {
   return __shfl(v, 0);
 }
 
-// ---------- cursor-based cooperative writer ----------
-// One wave writes one JSON line. The cursor is wave-uniform; every helper
-// leaves all 64 lanes with the same cursor value.
-
-struct Writer {
-  uint8_t* out;   // line base
-  int cur;        // wave-uniform write offset
-
-  DEV void raw_lane0(const char* s, int n) {
-    if (lane_id() == 0)
-      for (int j = 0; j < n; ++j) out[cur + j] = (uint8_t)s[j];
-    cur += n;
-  }
-  DEV void bytes(const uint8_t* s, int n) {  // striped copy, no escaping
-    for (int j = lane_id(); j < n; j += WAVE) out[cur + j] = s[j];
-    cur += n;
-  }
-  DEV void lit(const char* s) {  // compile-time-ish literal (strlen on lane)
-    int n = 0;
-    while (s[n]) ++n;
-    bytes((const uint8_t*)s, n);
-  }
-
-  // Escaped JSON string content (no quotes). Wave-parallel: per-64-byte
-  // stripe, each lane computes its byte's expansion length, wave-scans for
-  // offsets, writes its expansion.
-  DEV void escaped(const uint8_t* s, int n) {
-    int lane = lane_id();
-    int base = cur;
-    for (int start = 0; start < n; start += WAVE) {
-      int p = start + lane;
-      int el = (p < n) ? escape_len_at(s, n, p) : 0;
-      int off = wave_prefix_excl(el);
-      if (p < n && el > 0) {
-        uint8_t tmp[6];
-        int w = escape_write_at(s, n, p, tmp);
-        for (int j = 0; j < w; ++j) out[base + off + j] = tmp[j];
-      }
-      int total = wave_sum(el);
-      base += total;
-    }
-    cur = base;
-  }
-
-  // Escaped length without writing (same loop shape).
-  DEV static int escaped_len(const uint8_t* s, int n) {
-    int lane = lane_id();
-    int total = 0;
-    for (int start = 0; start < n; start += WAVE) {
-      int p = start + lane;
-      int el = (p < n) ? escape_len_at(s, n, p) : 0;
-      total += wave_sum(el);
-    }
-    return total;
-  }
-
-  DEV void quoted(const uint8_t* s, int n) {
-    lit("\"");
-    escaped(s, n);
-    lit("\"");
-  }
-
-  // unsigned/signed integer -> decimal (lane 0 writes; cursor uniform)
-  DEV void u64_dec(uint64_t v) {
-    char buf[20];
-    int n = 0;
-    do { buf[n++] = '0' + (v % 10); v /= 10; } while (v);
-    if (lane_id() == 0)
-      for (int j = 0; j < n; ++j) out[cur + j] = buf[n - 1 - j];
-    cur += n;
-  }
-  DEV void i64_dec(int64_t v) {
-    if (v < 0) { lit("-"); u64_dec((uint64_t)(-v)); }
-    else u64_dec((uint64_t)v);
-  }
-
-  // RFC3339 UTC "YYYY-MM-DDTHH:MM:SSZ" from unix seconds (civil calendar).
-  DEV void rfc3339(int64_t secs) {
-    int64_t days = secs / 86400;
-    int64_t rem = secs % 86400;
-    if (rem < 0) { rem += 86400; days -= 1; }
-    int hh = (int)(rem / 3600), mm = (int)((rem % 3600) / 60),
-        ss = (int)(rem % 60);
-    // civil_from_days (Howard Hinnant's algorithm)
-    int64_t z = days + 719468;
-    int64_t era = (z >= 0 ? z : z - 146096) / 146097;
-    int64_t doe = z - era * 146097;
-    int64_t yoe = (doe - doe / 1460 + doe / 36524 - doe / 146096) / 365;
-    int64_t y = yoe + era * 400;
-    int64_t doy = doe - (365 * yoe + yoe / 4 - yoe / 100);
-    int64_t mp = (5 * doy + 2) / 153;
-    int64_t d = doy - (153 * mp + 2) / 5 + 1;
-    int64_t m = mp + (mp < 10 ? 3 : -9);
-    y += (m <= 2);
-    char buf[20];
-    int yy = (int)y;
-    buf[0] = '0' + (yy / 1000) % 10; buf[1] = '0' + (yy / 100) % 10;
-    buf[2] = '0' + (yy / 10) % 10;   buf[3] = '0' + yy % 10;
-    buf[4] = '-';
-    buf[5] = '0' + ((int)m) / 10; buf[6] = '0' + ((int)m) % 10;
-    buf[7] = '-';
-    buf[8] = '0' + ((int)d) / 10; buf[9] = '0' + ((int)d) % 10;
-    buf[10] = 'T';
-    buf[11] = '0' + hh / 10; buf[12] = '0' + hh % 10; buf[13] = ':';
-    buf[14] = '0' + mm / 10; buf[15] = '0' + mm % 10; buf[16] = ':';
-    buf[17] = '0' + ss / 10; buf[18] = '0' + ss % 10; buf[19] = 'Z';
-    if (lane_id() == 0)
-      for (int j = 0; j < 20; ++j) out[cur + j] = buf[j];
-    cur += 20;
-  }
-};
+// ---------- cursor-based cooperative JSON emitter ----------
+// One wave emits one JSON line; the cursor is wave-uniform (every helper
+// leaves all 64 lanes with the same cursor). The W=false instantiation
+// measures (stores compiled out; constant segments fold to `cur += K`).
+// Shared by the Telegram (parse_encode.hip) and YouTube (yt_encode.hip)
+// emitters; LIT(e, "...") supplies compile-time literal lengths.
 
 DEV int u64_dec_len(uint64_t v) {
   int n = 0;
@@ -225,6 +120,104 @@ DEV int u64_dec_len(uint64_t v) {
 DEV int i64_dec_len(int64_t v) {
   return (v < 0) ? 1 + u64_dec_len((uint64_t)(-v)) : u64_dec_len((uint64_t)v);
 }
+
+DEV void rfc3339_write(uint8_t* out, int cur, long long secs) {
+  long long days = secs / 86400;
+  long long rem = secs % 86400;
+  if (rem < 0) { rem += 86400; days -= 1; }
+  int hh = (int)(rem / 3600), mm = (int)((rem % 3600) / 60),
+      ss = (int)(rem % 60);
+  // civil_from_days (Howard Hinnant's algorithm)
+  long long z = days + 719468;
+  long long era = (z >= 0 ? z : z - 146096) / 146097;
+  long long doe = z - era * 146097;
+  long long yoe = (doe - doe / 1460 + doe / 36524 - doe / 146096) / 365;
+  long long y = yoe + era * 400;
+  long long doy = doe - (365 * yoe + yoe / 4 - yoe / 100);
+  long long mp = (5 * doy + 2) / 153;
+  long long d = doy - (153 * mp + 2) / 5 + 1;
+  long long m = mp + (mp < 10 ? 3 : -9);
+  y += (m <= 2);
+  char buf[20];
+  int yy = (int)y;
+  buf[0] = '0' + (yy / 1000) % 10; buf[1] = '0' + (yy / 100) % 10;
+  buf[2] = '0' + (yy / 10) % 10;   buf[3] = '0' + yy % 10;
+  buf[4] = '-';
+  buf[5] = '0' + ((int)m) / 10; buf[6] = '0' + ((int)m) % 10;
+  buf[7] = '-';
+  buf[8] = '0' + ((int)d) / 10; buf[9] = '0' + ((int)d) % 10;
+  buf[10] = 'T';
+  buf[11] = '0' + hh / 10; buf[12] = '0' + hh % 10; buf[13] = ':';
+  buf[14] = '0' + mm / 10; buf[15] = '0' + mm % 10; buf[16] = ':';
+  buf[17] = '0' + ss / 10; buf[18] = '0' + ss % 10; buf[19] = 'Z';
+  if (lane_id() == 0)
+    for (int j = 0; j < 20; ++j) out[cur + j] = (uint8_t)buf[j];
+}
+
+#define LIT(e, s) (e).raw((const unsigned char*)(s), (int)sizeof(s) - 1)
+
+template <bool W>
+struct JsonEmit {
+  uint8_t* out;
+  int cur;
+
+  DEV void raw(const uint8_t* s, int n) {
+    if (W)
+      for (int j = lane_id(); j < n; j += WAVE) out[cur + j] = s[j];
+    cur += n;
+  }
+
+  // Escaped JSON string content (no quotes). Clean 64-byte stripes (the
+  // common case) take an identity-copy fast path with no scans.
+  DEV void esc(const uint8_t* s, int n) {
+    int lane = lane_id();
+    for (int start = 0; start < n; start += WAVE) {
+      int p = start + lane;
+      int span = n - start;
+      if (span > WAVE) span = WAVE;
+      int el = (p < n) ? escape_len_at(s, n, p) : 0;
+      unsigned long long dirty = __ballot(p < n && el != 1);
+      if (dirty == 0) {
+        if (W && p < n) out[cur + p - start] = s[p];
+        cur += span;
+        continue;
+      }
+      if (W) {
+        int off = wave_prefix_excl(el);
+        if (p < n && el > 0) {
+          uint8_t tmp[6];
+          int w = escape_write_at(s, n, p, tmp);
+          for (int j = 0; j < w; ++j) out[cur + off + j] = tmp[j];
+        }
+      }
+      cur += wave_sum(el);
+    }
+  }
+
+  DEV void u64(unsigned long long v) {
+    if (W) {
+      char buf[20];
+      int n = 0;
+      unsigned long long x = v;
+      do { buf[n++] = '0' + (x % 10); x /= 10; } while (x);
+      if (lane_id() == 0)
+        for (int j = 0; j < n; ++j) out[cur + j] = (uint8_t)buf[n - 1 - j];
+      cur += n;
+    } else {
+      cur += u64_dec_len(v);
+    }
+  }
+  DEV void i64(long long v) {
+    if (v < 0) { LIT(*this, "-"); u64((unsigned long long)(-v)); }
+    else u64((unsigned long long)v);
+  }
+
+  // RFC3339 UTC "YYYY-MM-DDTHH:MM:SSZ" from unix seconds (fixed 20 bytes).
+  DEV void rfc3339(long long secs) {
+    if (W) rfc3339_write(out, cur, secs);
+    cur += 20;
+  }
+};
 
 // ---------- hashing (FNV-1a 64, matches python oracle) ----------
 DEV uint64_t fnv1a64(const uint8_t* s, int n) {

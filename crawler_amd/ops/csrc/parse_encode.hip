@@ -272,77 +272,15 @@ DEV void extract_links(const BatchView& B, int i, LinkList& L, int lane) {
 }
 
 // ---------- the templated line emitter ----------
-
-// Compile-time literal emission: sizeof-based length means the measuring
-// instantiation folds every constant segment into `cur += K` with no loads.
-#define LIT(e, s) (e).raw((const unsigned char*)(s), (int)sizeof(s) - 1)
+// JsonEmit<W> (common.h) provides raw/esc/u64/i64/rfc3339 and the LIT
+// macro; Emit adds the quoted-escape shorthand used below.
 
 template <bool W>
-struct Emit {
-  unsigned char* out;
-  int cur;
-
-  DEV void lit(const char* s) {
-    int n = 0;
-    while (s[n]) ++n;
-    if (W)
-      for (int j = lane_id(); j < n; j += WAVE) out[cur + j] = (unsigned char)s[j];
-    cur += n;
-  }
-  DEV void raw(const unsigned char* s, int n) {
-    if (W)
-      for (int j = lane_id(); j < n; j += WAVE) out[cur + j] = s[j];
-    cur += n;
-  }
-  DEV void esc(const unsigned char* s, int n) {
-    int lane = lane_id();
-    for (int start = 0; start < n; start += WAVE) {
-      int p = start + lane;
-      int span = n - start;
-      if (span > WAVE) span = WAVE;
-      int el = (p < n) ? escape_len_at(s, n, p) : 0;
-      // Clean-stripe fast path: no escapes => identity copy, no scans.
-      unsigned long long dirty = __ballot(p < n && el != 1);
-      if (dirty == 0) {
-        if (W && p < n) out[cur + p - start] = s[p];
-        cur += span;
-        continue;
-      }
-      if (W) {
-        int off = wave_prefix_excl(el);
-        if (p < n && el > 0) {
-          unsigned char tmp[6];
-          int w = escape_write_at(s, n, p, tmp);
-          for (int j = 0; j < w; ++j) out[cur + off + j] = tmp[j];
-        }
-      }
-      cur += wave_sum(el);
-    }
-  }
-  DEV void qesc(const unsigned char* s, int n) { lit("\""); esc(s, n); lit("\""); }
-  DEV void u64(unsigned long long v) {
-    if (W) {
-      char buf[20];
-      int n = 0;
-      unsigned long long x = v;
-      do { buf[n++] = '0' + (x % 10); x /= 10; } while (x);
-      if (lane_id() == 0)
-        for (int j = 0; j < n; ++j) out[cur + j] = buf[n - 1 - j];
-      cur += n;
-    } else {
-      cur += u64_dec_len(v);
-    }
-  }
-  DEV void i64(long long v) {
-    if (v < 0) { lit("-"); u64((unsigned long long)(-v)); }
-    else u64((unsigned long long)v);
-  }
-  DEV void rfc3339(long secs) {
-    if (W) {
-      Writer w{out, cur};
-      w.rfc3339(secs);
-    }
-    cur += 20;
+struct Emit : JsonEmit<W> {
+  DEV void qesc(const unsigned char* s, int n) {
+    LIT(*this, "\"");
+    this->esc(s, n);
+    LIT(*this, "\"");
   }
 };
 
@@ -350,7 +288,9 @@ template <bool W>
 DEV int emit_line(const BatchView& B, int i, unsigned char* out,
                   LinkList& L) {
   const int lane = lane_id();
-  Emit<W> e{out, 0};
+  Emit<W> e{};
+  e.out = out;
+  e.cur = 0;
   const int c = B.channel_idx[i];
   const unsigned char* user = B.pool + B.ch_user_off[c];
   const int user_n = B.ch_user_len[c];

@@ -99,68 +99,13 @@ DEV int extract_urls(const unsigned char* s, int n, int* starts,
   return cnt;
 }
 
-#define LIT(e, s) (e).raw((const unsigned char*)(s), (int)sizeof(s) - 1)
+// YEmit extends the shared JsonEmit (common.h) with the sanitized
+// filename writer (runs of non-[A-Za-z0-9._-] -> '_', cap 80;
+// convert.sanitize_filename).
 
 template <bool W>
-struct YEmit {
-  unsigned char* out;
-  int cur;
-
-  DEV void raw(const unsigned char* s, int n) {
-    if (W)
-      for (int j = lane_id(); j < n; j += WAVE) out[cur + j] = s[j];
-    cur += n;
-  }
-  DEV void esc(const unsigned char* s, int n) {
-    int lane = lane_id();
-    for (int start = 0; start < n; start += WAVE) {
-      int p = start + lane;
-      int span = n - start;
-      if (span > WAVE) span = WAVE;
-      int el = (p < n) ? escape_len_at(s, n, p) : 0;
-      unsigned long long dirty = __ballot(p < n && el != 1);
-      if (dirty == 0) {
-        if (W && p < n) out[cur + p - start] = s[p];
-        cur += span;
-        continue;
-      }
-      if (W) {
-        int off = wave_prefix_excl(el);
-        if (p < n && el > 0) {
-          unsigned char tmp[6];
-          int w = escape_write_at(s, n, p, tmp);
-          for (int j = 0; j < w; ++j) out[cur + off + j] = tmp[j];
-        }
-      }
-      cur += wave_sum(el);
-    }
-  }
-  DEV void u64(unsigned long long v) {
-    if (W) {
-      char buf[20];
-      int n = 0;
-      unsigned long long x = v;
-      do { buf[n++] = '0' + (x % 10); x /= 10; } while (x);
-      if (lane_id() == 0)
-        for (int j = 0; j < n; ++j) out[cur + j] = buf[n - 1 - j];
-      cur += n;
-    } else {
-      cur += u64_dec_len(v);
-    }
-  }
-  DEV void i64(long long v) {
-    if (v < 0) { LIT(*this, "-"); u64((unsigned long long)(-v)); }
-    else u64((unsigned long long)v);
-  }
-  DEV void rfc(long secs) {
-    if (W) {
-      Writer w{out, cur};
-      w.rfc3339(secs);
-    }
-    cur += 20;
-  }
-  // sanitized filename: runs of non-[A-Za-z0-9._-] -> '_', cap 80 chars
-  // (convert.sanitize_filename). Serial on all lanes (titles are short).
+struct YEmit : JsonEmit<W> {
+  DEV void rfc(long secs) { this->rfc3339(secs); }
   DEV void sanitized(const unsigned char* s, int n) {
     int o = 0;
     int p = 0;
@@ -168,11 +113,11 @@ struct YEmit {
       unsigned char c = s[p];
       bool ok = is_word_char(c) || c == '.' || c == '-';
       if (ok) {
-        if (W && lane_id() == 0) out[cur + o] = c;
+        if (W && lane_id() == 0) this->out[this->cur + o] = c;
         ++o;
         ++p;
       } else {
-        if (W && lane_id() == 0) out[cur + o] = '_';
+        if (W && lane_id() == 0) this->out[this->cur + o] = '_';
         ++o;
         while (p < n) {
           unsigned char d = s[p];
@@ -181,13 +126,15 @@ struct YEmit {
         }
       }
     }
-    cur += o;
+    this->cur += o;
   }
 };
 
 template <bool W>
 DEV int emit_yt(const YtView& B, int i, unsigned char* out) {
-  YEmit<W> e{out, 0};
+  YEmit<W> e{};
+  e.out = out;
+  e.cur = 0;
   const int c = B.channel_idx[i];
   const unsigned char* vid = B.pool + B.vid_off[i];
   const unsigned char* cid = B.pool + B.ch_id_off[c];
