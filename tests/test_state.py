@@ -248,3 +248,30 @@ def test_skipped_edge_promotion_excludes():
     e = rw.get_random_skipped_edge({"d1"})
     assert e is not None and e.destination_channel == "d2"
     assert rw.get_random_skipped_edge({"d1", "d2"}) is None
+
+
+def test_concurrent_edge_claims_stress():
+    """Race exercise (reference concurrent_test.go pattern):
+    8 threads claiming pending edges never double-claim."""
+    import threading
+
+    rw = RandomWalkStore()
+    bid = rw.open_batch("c1", "src", "p1", 0, "seq")
+    for i in range(400):
+        rw.insert_pending_edge(bid, "c1", f"d{i}", "src", "seq", "url")
+    claimed = []
+    lock = threading.Lock()
+
+    def worker():
+        while True:
+            got = rw.claim_pending_edges(7)
+            if not got:
+                return
+            with lock:
+                claimed.extend(e.pending_id for e in got)
+
+    ts = [threading.Thread(target=worker) for _ in range(8)]
+    [t.start() for t in ts]
+    [t.join() for t in ts]
+    assert len(claimed) == 400
+    assert len(set(claimed)) == 400
