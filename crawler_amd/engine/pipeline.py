@@ -20,15 +20,13 @@ from __future__ import annotations
 
 import dataclasses
 import random
-import time
-import uuid
-from typing import Callable, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 from ..feed.client import TelegramAPIError
 from ..models.post import ChannelData, EngagementData
 from ..ops import golden as G
 from . import errors as E
-from .state import EdgeRecord, Page, PageMessage
+from .state import Page
 
 
 @dataclasses.dataclass

@@ -25,7 +25,7 @@ import dataclasses
 import random
 import threading
 import time
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 import numpy as np
 

@@ -26,7 +26,7 @@ data itself (reference: telegramhelper/rate_limiter.go, crawl/runner.go:55-104).
 from __future__ import annotations
 
 import dataclasses
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 import numpy as np
 import torch

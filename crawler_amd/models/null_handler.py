@@ -17,7 +17,7 @@ from __future__ import annotations
 import dataclasses
 import datetime as _dt
 import json
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 from . import post as _post
 

@@ -16,7 +16,7 @@ from __future__ import annotations
 import datetime as _dt
 import time
 import uuid
-from typing import Dict, Optional
+from typing import Dict
 
 from ..engine.state import LocalStateManager, Page
 from . import messages as M
