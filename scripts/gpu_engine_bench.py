@@ -1,0 +1,63 @@
+#!/usr/bin/env python3
+"""End-to-end GPU crawl ENGINE measurement (not the bench.py microbench):
+a real snowball crawl through GpuCrawlEngine — device feed generation,
+HIP parse+encode, seen-set claims, per-channel JSONL files on disk,
+state checkpoints — reporting whole-crawl posts/sec including the host
+filesystem writes. This is the config #3 shape on one GPU.
+
+Usage: python scripts/gpu_engine_bench.py [--seeds 200 --posts 2000
+       --max-depth 2 --storage /tmp/gpu-crawl]
+"""
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+from crawler_amd.config import CrawlerConfig  # noqa: E402
+from crawler_amd.engine import LocalStateManager  # noqa: E402
+from crawler_amd.engine.gpu_runner import GpuCrawlEngine  # noqa: E402
+from crawler_amd.feed import FeedConfig, SyntheticFeed  # noqa: E402
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--seeds", type=int, default=200)
+    ap.add_argument("--posts", type=int, default=2000)
+    ap.add_argument("--max-depth", type=int, default=2)
+    ap.add_argument("--universe", type=int, default=1_000_000)
+    ap.add_argument("--storage", default="/tmp/gpu-crawl")
+    args = ap.parse_args()
+
+    cfg = CrawlerConfig(
+        crawl_id="engine-bench", storage_root=args.storage,
+        sampling_method="snowball", max_depth=args.max_depth, min_users=1,
+        skip_media_download=True,
+    )
+    feed = SyntheticFeed(FeedConfig(seed=2026, universe=args.universe,
+                                    posts_per_channel=args.posts))
+    sm = LocalStateManager(cfg)
+    eng = GpuCrawlEngine(cfg, sm, feed, posts_per_channel=args.posts,
+                         chunk_channels=512)
+    seeds = [feed.username_of(i) for i in range(args.seeds)]
+    t0 = time.perf_counter()
+    stats = eng.run(seeds)
+    elapsed = time.perf_counter() - t0
+    out = {
+        "metric": "engine posts/sec (full crawl incl. disk JSONL)",
+        "value": round(stats["posts"] / elapsed, 1),
+        "elapsed_s": round(elapsed, 2),
+        "pages": stats["pages"],
+        "posts": stats["posts"],
+        "jsonl_gb": round(stats["jsonl_bytes"] / 1e9, 2),
+        "discovered": stats["discovered"],
+        "deadends": stats["deadends"],
+        "max_depth": args.max_depth,
+    }
+    print(json.dumps(out))
+
+
+if __name__ == "__main__":
+    main()
