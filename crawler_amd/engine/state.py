@@ -378,7 +378,12 @@ class LocalStateManager(BaseStateManager):
             config, "storage_root", "/tmp/crawl"
         )
         self.media_cache: Dict[str, dict] = {}
-        self._post_files = {}
+        # LRU-capped append-handle cache: crawls touch thousands of
+        # channels; keeping every posts.jsonl open exhausts the fd limit
+        self._post_files: "collections.OrderedDict" = (
+            __import__("collections").OrderedDict()
+        )
+        self.max_open_post_files = 256
         self.url_dedup: Dict[str, str] = {}
 
     def load_url_dedup_cache(self) -> int:
@@ -547,24 +552,26 @@ class LocalStateManager(BaseStateManager):
 
     # ---- posts / files ----
 
-    def store_post(self, channel: str, post: Post) -> None:
+    def _post_file(self, channel: str):
         path = self._posts_path(channel)
         f = self._post_files.get(path)
-        if f is None:
-            os.makedirs(os.path.dirname(path), exist_ok=True)
-            f = open(path, "ab")
-            self._post_files[path] = f
-        f.write(post.to_jsonl().encode("utf-8"))
+        if f is not None:
+            self._post_files.move_to_end(path)
+            return f
+        os.makedirs(os.path.dirname(path), exist_ok=True)
+        f = open(path, "ab")
+        self._post_files[path] = f
+        while len(self._post_files) > self.max_open_post_files:
+            _old_path, old_f = self._post_files.popitem(last=False)
+            old_f.close()
+        return f
+
+    def store_post(self, channel: str, post: Post) -> None:
+        self._post_file(channel).write(post.to_jsonl().encode("utf-8"))
 
     def store_post_lines(self, channel: str, data: bytes) -> None:
         """Bulk JSONL append (the GPU path hands whole encoded blocks)."""
-        path = self._posts_path(channel)
-        f = self._post_files.get(path)
-        if f is None:
-            os.makedirs(os.path.dirname(path), exist_ok=True)
-            f = open(path, "ab")
-            self._post_files[path] = f
-        f.write(data)
+        self._post_file(channel).write(data)
 
     def store_file(self, channel: str, source_path: str,
                    file_name: str) -> Tuple[str, str]:
