@@ -29,11 +29,13 @@ def main():
     ap.add_argument("--max-depth", type=int, default=2)
     ap.add_argument("--universe", type=int, default=1_000_000)
     ap.add_argument("--storage", default="/tmp/gpu-crawl")
+    ap.add_argument("--max-pages", type=int, default=1500)
     args = ap.parse_args()
 
     cfg = CrawlerConfig(
         crawl_id="engine-bench", storage_root=args.storage,
         sampling_method="snowball", max_depth=args.max_depth, min_users=1,
+        max_pages=args.max_pages,
         skip_media_download=True,
     )
     feed = SyntheticFeed(FeedConfig(seed=2026, universe=args.universe,
