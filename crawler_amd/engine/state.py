@@ -276,7 +276,6 @@ class BaseStateManager:
             deadends = sum(1 for p in self.pages.values()
                            if p.status == "deadend")
             max_pages = getattr(self.config, "max_pages", 0) or 0
-            max_reached = max_pages > 0 and total >= max_pages
             replacements = deadends
             existing_urls = {p.url: pid for pid, p in self.pages.items()}
             url_dedup = getattr(self, "url_dedup", {})
@@ -290,7 +289,11 @@ class BaseStateManager:
                     # crawled by a previous crawl (cross-crawl URL cache);
                     # seeds (depth 0) are always admitted
                     continue
-                if max_reached:
+                # MaxPages budget. The reference checks the cap only at
+                # layer START (base.go:240-244), so one layer can overshoot
+                # it arbitrarily; we enforce it continuously — the stronger
+                # guarantee, same replacement semantics at the cap.
+                if max_pages > 0 and total + len(added) >= max_pages:
                     if replacements <= 0:
                         continue
                     replacements -= 1
