@@ -92,14 +92,15 @@ class GpuCrawlEngine:
             line_off = res.line_off.cpu().numpy()
             line_len = res.line_len.cpu().numpy()
             torch.cuda.synchronize()
-            buf = out_host.numpy()
+            buf = memoryview(out_host.numpy())
             P = self.ppc
             for k, (uname, _cid) in enumerate(chunk):
                 lo = int(line_off[k * P])
                 last = (k + 1) * P - 1
                 hi = int(line_off[last] + line_len[last])
                 if hi > lo:
-                    self.sm.store_post_lines(uname, buf[lo:hi].tobytes())
+                    # zero-copy slice: file.write accepts the memoryview
+                    self.sm.store_post_lines(uname, buf[lo:hi])
                 n_lines = int((line_len[k * P:(k + 1) * P] > 0).sum())
                 posts_total += n_lines
                 self.stats["pages"] += 1
