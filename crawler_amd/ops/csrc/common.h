@@ -113,9 +113,28 @@ DEV int wave_sum(int v) {
 // emitters; LIT(e, "...") supplies compile-time literal lengths.
 
 DEV int u64_dec_len(uint64_t v) {
-  int n = 0;
-  do { ++n; v /= 10; } while (v);
-  return n;
+  // comparison ladder (v is wave-uniform at every call site: scalar
+  // branches, no divide chain)
+  if (v < 10ULL) return 1;
+  if (v < 100ULL) return 2;
+  if (v < 1000ULL) return 3;
+  if (v < 10000ULL) return 4;
+  if (v < 100000ULL) return 5;
+  if (v < 1000000ULL) return 6;
+  if (v < 10000000ULL) return 7;
+  if (v < 100000000ULL) return 8;
+  if (v < 1000000000ULL) return 9;
+  if (v < 10000000000ULL) return 10;
+  if (v < 100000000000ULL) return 11;
+  if (v < 1000000000000ULL) return 12;
+  if (v < 10000000000000ULL) return 13;
+  if (v < 100000000000000ULL) return 14;
+  if (v < 1000000000000000ULL) return 15;
+  if (v < 10000000000000000ULL) return 16;
+  if (v < 100000000000000000ULL) return 17;
+  if (v < 1000000000000000000ULL) return 18;
+  if (v < 10000000000000000000ULL) return 19;
+  return 20;
 }
 DEV int i64_dec_len(int64_t v) {
   return (v < 0) ? 1 + u64_dec_len((uint64_t)(-v)) : u64_dec_len((uint64_t)v);
@@ -150,8 +169,8 @@ DEV void rfc3339_write(uint8_t* out, int cur, long long secs) {
   buf[11] = '0' + hh / 10; buf[12] = '0' + hh % 10; buf[13] = ':';
   buf[14] = '0' + mm / 10; buf[15] = '0' + mm % 10; buf[16] = ':';
   buf[17] = '0' + ss / 10; buf[18] = '0' + ss % 10; buf[19] = 'Z';
-  if (lane_id() == 0)
-    for (int j = 0; j < 20; ++j) out[cur + j] = (uint8_t)buf[j];
+  int lane = lane_id();
+  if (lane < 20) out[cur + lane] = (uint8_t)buf[lane];
 }
 
 #define LIT(e, s) (e).raw((const unsigned char*)(s), (int)sizeof(s) - 1)
@@ -194,18 +213,29 @@ struct JsonEmit {
     }
   }
 
+  // Wave-parallel decimal emission: lane j computes digit j-from-the-right
+  // ((v / 10^j) % 10) so a 10-digit number costs ~1 division per LANE in
+  // parallel instead of a serial divide chain on lane 0 (the emitters are
+  // VALU-instruction-bound; profiles/r01_pmc_counters.csv).
   DEV void u64(unsigned long long v) {
+    int n = u64_dec_len(v);
     if (W) {
-      char buf[20];
-      int n = 0;
-      unsigned long long x = v;
-      do { buf[n++] = '0' + (x % 10); x /= 10; } while (x);
-      if (lane_id() == 0)
-        for (int j = 0; j < n; ++j) out[cur + j] = (uint8_t)buf[n - 1 - j];
-      cur += n;
-    } else {
-      cur += u64_dec_len(v);
+      int lane = lane_id();
+      if (lane < n) {
+        unsigned d;
+        if (v <= 0xFFFFFFFFULL) {
+          unsigned p = 1;
+          for (int k = 0; k < lane; ++k) p *= 10u;   // lane<10: cheap
+          d = ((unsigned)v / p) % 10u;
+        } else {
+          unsigned long long p = 1;
+          for (int k = 0; k < lane; ++k) p *= 10ULL;
+          d = (unsigned)((v / p) % 10ULL);
+        }
+        out[cur + n - 1 - lane] = (uint8_t)('0' + d);
+      }
     }
+    cur += n;
   }
   DEV void i64(long long v) {
     if (v < 0) { LIT(*this, "-"); u64((unsigned long long)(-v)); }
