@@ -126,11 +126,16 @@ def main():
     # Double-buffered pinned host ring for the JSONL output (the storage
     # boundary): D2H of step k overlaps the kernels of step k+1 on a
     # separate copy stream. Worst case ~2.6KB/post.
-    pinned = [
-        torch.empty(int(chunk_posts * 3000), dtype=torch.uint8,
-                    pin_memory=True)
-        for _ in range(2)
-    ]
+    def alloc_host(nbytes):
+        # 8 ranks x 2 buffers of pinned memory can brush against host
+        # lockable-memory limits; fall back to pageable rather than fail
+        try:
+            return torch.empty(nbytes, dtype=torch.uint8, pin_memory=True)
+        except RuntimeError:
+            log("pinned allocation failed; using pageable host memory")
+            return torch.empty(nbytes, dtype=torch.uint8)
+
+    pinned = [alloc_host(int(chunk_posts * 2600)) for _ in range(2)]
     compute_stream = torch.cuda.Stream()
     # two copy streams: MI355X has multiple SDMA engines; splitting the
     # 2.5GB D2H in half across streams uses two of them
@@ -185,6 +190,9 @@ def main():
                             seen.insert_hashes(g[g != 0])
                 nd = 0
         nbytes = out_t.numel()
+        assert nbytes <= pinned[slot].numel(), (
+            f"host ring too small: {nbytes} > {pinned[slot].numel()}"
+        )
         ev = torch.cuda.Event()
         ev.record(compute_stream)
         # the previous copy into this pinned slot must be done before reuse
@@ -228,8 +236,7 @@ def main():
         one = feed.build_batch_device(
             np.array([999_990 + rank]), device, posts_per_channel=args.posts
         )
-        lat_pin = torch.empty(args.posts * 3000, dtype=torch.uint8,
-                              pin_memory=True)
+        lat_pin = alloc_host(args.posts * 3000)
         for _ in range(11):
             torch.cuda.synchronize()
             t = time.perf_counter()
