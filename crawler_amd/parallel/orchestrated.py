@@ -1,0 +1,170 @@
+"""Orchestrated multi-rank crawl: work-queue fan-out + collective discovery.
+
+BASELINE config #3 ("orchestrator + 8 workers on RCCL, snowball, seen-set
+all-reduce over xGMI") in its MI355X-native shape:
+
+- control plane: rank 0 publishes each BFS layer's channels as chunked
+  WorkItems on a TCPStore queue (the Dapr work-queue equivalent,
+  orchestrator.go:182-277); ALL ranks — including rank 0 — claim chunks
+  dynamically, so load balances even when channels vary in cost;
+- data plane: every rank runs the per-chunk hot path (on the GPU via
+  GpuCrawlEngine.process_channels) and the newly-claimed discovery names
+  are all-gathered per layer (RCCL over xGMI on GPU groups; gloo in CPU
+  tests) — the seen-set union of SURVEY §2.6;
+- results flow back on a results queue for rank 0's bookkeeping
+  (orchestrator.go:315-383); rank 0 owns the persisted state/checkpoints,
+  every rank writes its own JSONL shard;
+- layer synchronization is depth-barriered exactly like the reference's
+  depth-synchronous distribution; every rank derives the next layer
+  deterministically from the same gathered name set.
+
+Heartbeats remain for observability; a rank that dies mid-layer parts the
+collective (RCCL semantics), which the launcher surfaces — reassignment
+at the work-queue level only helps before the layer barrier, matching the
+reference's at-least-once republish model.
+"""
+from __future__ import annotations
+
+import datetime as _dt
+import uuid
+from typing import Callable, List, Optional, Tuple
+
+from ..engine.state import LocalStateManager, Page
+from . import messages as M
+from .queue import Heartbeats, StoreQueue
+
+
+class OrchestratedCrawl:
+    def __init__(self, cfg, sm: LocalStateManager, store, rank: int,
+                 world: int,
+                 process_fn: Callable[[List[str]], Tuple[List[str], int]],
+                 chunk_channels: int = 64, dist=None):
+        """process_fn(names) -> (discovered_names, posts_stored).
+        `dist` is torch.distributed (injected so CPU tests can pass gloo
+        and unit tests can fake it)."""
+        self.cfg = cfg
+        self.sm = sm
+        self.rank = rank
+        self.world = world
+        self.process_fn = process_fn
+        self.chunk = chunk_channels
+        if dist is None:
+            import torch.distributed as dist  # noqa: PLC0415
+
+        self.dist = dist
+        self.work_q = StoreQueue(store, M.TOPIC_WORK_QUEUE)
+        self.result_q = StoreQueue(store, M.TOPIC_RESULTS)
+        self.heartbeats = Heartbeats(store)
+        self.stats = {"chunks": 0, "pages": 0, "posts": 0,
+                      "discovered": 0, "layers": 0}
+
+    # ---- name exchange (fixed-width padded all-gather) ----
+
+    def _allgather_names(self, names: List[str], width: int = 32
+                         ) -> List[str]:
+        import torch
+
+        dist = self.dist
+        counts = torch.tensor([len(names)], dtype=torch.int64)
+        all_counts = [torch.zeros(1, dtype=torch.int64)
+                      for _ in range(self.world)]
+        dist.all_gather(all_counts, counts)
+        max_n = max(int(c.item()) for c in all_counts) or 1
+        buf = torch.zeros(max_n, width, dtype=torch.uint8)
+        for i, name in enumerate(names):
+            b = name.encode()[:width]
+            buf[i, :len(b)] = torch.frombuffer(bytearray(b),
+                                               dtype=torch.uint8)
+        gathered = [torch.empty_like(buf) for _ in range(self.world)]
+        dist.all_gather(gathered, buf)
+        out: List[str] = []
+        for r, g in enumerate(gathered):
+            cnt = int(all_counts[r].item())
+            arr = g.numpy()
+            for i in range(cnt):
+                row = arr[i]
+                ln = int((row != 0).sum())
+                out.append(bytes(row[:ln]).decode())
+        return out
+
+    # ---- one layer ----
+
+    def _run_layer(self, names: List[str], depth: int) -> List[str]:
+        dist = self.dist
+        self.heartbeats.register(f"rank{self.rank}")
+        self.heartbeats.beat(f"rank{self.rank}", M.WORKER_ACTIVE)
+        if self.rank == 0:
+            for i in range(0, len(names), self.chunk):
+                item = M.WorkItem(
+                    id=str(uuid.uuid4()), url=f"chunk-{depth}-{i}",
+                    depth=depth, crawl_id=self.sm.metadata.crawl_id,
+                    trace_id=M.new_trace_id(),
+                    config={"channels": names[i:i + self.chunk]},
+                )
+                item.validate()
+                self.work_q.publish(item.to_json())
+        dist.barrier()  # queue fully published before claims start
+
+        my_discovered: List[str] = []
+        while True:
+            raw = self.work_q.claim(timeout_s=0.0)
+            if raw is None:
+                break
+            item = M.WorkItem.from_json(raw)
+            chans = item.config["channels"]
+            discovered, posts = self.process_fn(chans)
+            my_discovered.extend(discovered)
+            self.stats["chunks"] += 1
+            self.stats["pages"] += len(chans)
+            self.stats["posts"] += posts
+            self.result_q.publish(M.WorkResult(
+                work_item_id=item.id, worker_id=f"rank{self.rank}",
+                posts_stored=posts, discovered=discovered,
+                trace_id=item.trace_id,
+            ).to_json())
+        self.heartbeats.beat(f"rank{self.rank}", M.WORKER_IDLE)
+        return self._allgather_names(my_discovered)
+
+    # ---- the crawl ----
+
+    def run(self, seed_urls: List[str]) -> dict:
+        sm = self.sm
+        sm.initialize(seed_urls)
+        depth = 0
+        while True:
+            layer = [p for p in sm.get_layer_by_depth(depth)
+                     if p.status == "unfetched"]
+            if not layer:
+                break
+            names = [p.url for p in layer]
+            all_discovered = self._run_layer(names, depth)
+            self.stats["layers"] += 1
+            self.stats["discovered"] += len(all_discovered)
+            for p in layer:
+                p.status = "fetched"
+                sm.update_page(p)
+            if self.rank == 0:
+                # drain results for bookkeeping (page counts / errors)
+                self.result_q.drain()
+            if (self.cfg.sampling_method == "snowball"
+                    and (self.cfg.max_depth < 0
+                         or depth < self.cfg.max_depth)
+                    and all_discovered):
+                # every rank derives the SAME next layer (sorted set ->
+                # deterministic add_layer admission order)
+                pages = [Page(url=n, depth=depth + 1, status="unfetched")
+                         for n in sorted(set(all_discovered))]
+                sm.add_layer(pages)
+            if self.rank == 0:
+                sm.save_state()
+            if self.cfg.sampling_method == "channel":
+                break
+            if self.cfg.max_depth >= 0 and depth >= self.cfg.max_depth:
+                break
+            depth += 1
+        sm.update_crawl_metadata(sm.metadata.crawl_id,
+                                 {"status": "completed"})
+        if self.rank == 0:
+            sm.save_state()
+        sm.close()
+        return dict(self.stats)

@@ -21,18 +21,28 @@ from typing import List, Optional
 
 
 class StoreQueue:
-    """One named topic over a TCPStore-like object (set/get/add/check)."""
+    """One named topic over a TCPStore-like object (set/get/add/check).
+
+    Publish order: a slot number is drawn from `pub`, the payload is
+    written, THEN `tail` is bumped — so `tail` only ever covers readable
+    slots (safe with several concurrent publishers). A claimer that races
+    another claimer past `tail` STASHES its ticket instead of blocking:
+    the ticket is honored on a later claim() once more elements are
+    published, so no element is lost and no claimer ever blocks on a slot
+    that may never fill (which would deadlock depth-barriered layers)."""
 
     def __init__(self, store, topic: str):
         self.store = store
         self.topic = topic
+        self._pub_key = f"{topic}/pub"
         self._tail_key = f"{topic}/tail"
         self._claim_key = f"{topic}/claim"
+        self._stash: Optional[int] = None
 
     def publish(self, payload: str) -> int:
-        tail = self.store.add(self._tail_key, 1)
-        idx = tail - 1
+        idx = self.store.add(self._pub_key, 1) - 1
         self.store.set(f"{self.topic}/{idx}", payload)
+        self.store.add(self._tail_key, 1)
         return idx
 
     def _counters(self):
@@ -49,14 +59,20 @@ class StoreQueue:
         """Claim the next element; None if empty past the timeout."""
         deadline = time.monotonic() + timeout_s
         while True:
-            claim, tail = self._counters()
-            if claim < tail:
-                idx = self.store.add(self._claim_key, 1) - 1
-                if idx < tail:
+            _, tail = self._counters()
+            if self._stash is not None:
+                if self._stash < tail:
+                    idx = self._stash
+                    self._stash = None
                     return self._get_blocking(idx)
-                # raced past tail: a publisher will fill it eventually —
-                # the slot is ours, wait for its payload
-                return self._get_blocking(idx)
+            else:
+                claim, tail = self._counters()
+                if claim < tail:
+                    idx = self.store.add(self._claim_key, 1) - 1
+                    if idx < tail:
+                        return self._get_blocking(idx)
+                    # raced past tail: hold the ticket for the future
+                    self._stash = idx
             if time.monotonic() >= deadline:
                 return None
             time.sleep(poll_s)
@@ -83,6 +99,18 @@ class StoreQueue:
         return out
 
 
+def _try_get(store, key: str):
+    """Non-blocking store read: TCPStore.get WAITS for missing keys (up to
+    its timeout), so probe with check() first where supported."""
+    try:
+        if hasattr(store, "check") and not store.check([key]):
+            return None
+        v = store.get(key)
+        return v.decode() if isinstance(v, (bytes, bytearray)) else v
+    except Exception:
+        return None
+
+
 class Heartbeats:
     """Per-worker heartbeat keys (worker/worker.go:234-252)."""
 
@@ -102,24 +130,18 @@ class Heartbeats:
             self.store.set(f"{self.prefix}/members", ",".join(members))
 
     def _members(self) -> List[str]:
-        try:
-            v = self.store.get(f"{self.prefix}/members")
-            s = v.decode() if isinstance(v, (bytes, bytearray)) else v
-            return [m for m in s.split(",") if m]
-        except Exception:
-            return []
+        s = _try_get(self.store, f"{self.prefix}/members")
+        return [m for m in s.split(",") if m] if s else []
 
     def workers(self) -> List[str]:
         return self._members()
 
     def last_seen(self, worker_id: str):
-        try:
-            v = self.store.get(f"{self.prefix}/{worker_id}")
-            s = v.decode() if isinstance(v, (bytes, bytearray)) else v
-            ts, status = s.split("|", 1)
-            return float(ts), status
-        except Exception:
+        s = _try_get(self.store, f"{self.prefix}/{worker_id}")
+        if not s:
             return None, "unknown"
+        ts, status = s.split("|", 1)
+        return float(ts), status
 
     def offline_workers(self, timeout_s: float) -> List[str]:
         """Workers whose last beat is older than timeout_s
@@ -158,3 +180,7 @@ class InMemoryStore:
             cur += n
             self._d[k] = str(cur).encode()
             return cur
+
+    def check(self, keys) -> bool:
+        with self._lock:
+            return all(k in self._d for k in keys)
