@@ -191,3 +191,32 @@ class SyntheticYouTubeClient:
                            limit: int = 50) -> List[YouTubeVideo]:
         ids = self.index.channel_uploads(channel_id, limit)
         return self.list_videos(ids)
+
+
+class YouTubeClientPool:
+    """Rotating client pool (ytWorker, dapr/standalone.go:1245-1272):
+    each client retires after ~50 +/- 10 uses and is replaced by a fresh
+    one, spreading quota pressure the way the reference rotates workers."""
+
+    def __init__(self, make_client, retire_at: int = 50,
+                 retire_jitter: int = 10, rng=None):
+        self.make_client = make_client
+        self.retire_at = retire_at
+        self.retire_jitter = retire_jitter
+        self.rng = rng or random.Random(0)
+        self.retired = 0
+        self._fresh()
+
+    def _fresh(self):
+        self.client = self.make_client()
+        self.uses = 0
+        self.limit = self.retire_at + self.rng.randint(
+            -self.retire_jitter, self.retire_jitter
+        )
+
+    def get(self) -> SyntheticYouTubeClient:
+        if self.uses >= self.limit:
+            self.retired += 1
+            self._fresh()
+        self.uses += 1
+        return self.client

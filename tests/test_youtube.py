@@ -159,3 +159,26 @@ def test_run_youtube_channel_mode(tmp_path):
                         max_posts=20)
     stats = run_youtube(cfg, ["3"])
     assert stats["videos"] >= 1
+
+
+def test_client_pool_rotation():
+    """ytWorker rotation (dapr/standalone.go:1245-1272): retire ~50+-10."""
+    import random as _r
+
+    from crawler_amd.youtube.client import YouTubeClientPool
+
+    made = []
+
+    def mk():
+        c = SyntheticYouTubeClient(
+            SyntheticYouTubeIndex(seed=1, universe_channels=10)
+        )
+        made.append(c)
+        return c
+
+    pool = YouTubeClientPool(mk, retire_at=50, retire_jitter=10,
+                             rng=_r.Random(5))
+    for _ in range(200):
+        pool.get()
+    assert pool.retired >= 2
+    assert 3 <= len(made) <= 6  # ~200/50 rotations
