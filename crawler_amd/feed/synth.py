@@ -48,7 +48,6 @@ class Template:
     text: str                      # with one "{U}" per link slot (11-wide when filled)
     entity_specs: List[Tuple[str, int]] = dataclasses.field(default_factory=list)
     # entity_specs: (etype, slot_index) — offsets computed after slot fill
-    url_slots: int = 0             # text_url URL-tail slots ("https://t.me/" + user)
     aux: str = ""                  # emoji / poll question / prize / doc name
     has_thumb: bool = False
     has_video: bool = False
@@ -71,7 +70,6 @@ _PH = "c" + "0" * 10
 def _compile(t: Template) -> Template:
     """Fill slots with the placeholder, compute byte/UTF-16 entity offsets."""
     parts = t.text.split("{U}")
-    n_slots = len(parts) - 1
     text = _PH.join(parts)
     tb = text.encode("utf-8")
 
@@ -92,7 +90,6 @@ def _compile(t: Template) -> Template:
     entities = []
     url_tail = b""
     url_off_base = len(tb)
-    url_k = 0
     for (etype, slot) in t.entity_specs:
         # text_url URLs live in the tail, not at a text slot
         spos = (slot_pos[slot] - 1) if etype != "text_url" else -1
@@ -122,7 +119,6 @@ def _compile(t: Template) -> Template:
                              _utf16_len(anchor),
                              url_off_base + len(url_tail), len(url)))
             url_tail += url.encode("utf-8")
-            url_k += 1
         else:
             raise ValueError(etype)
 

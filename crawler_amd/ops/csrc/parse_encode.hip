@@ -22,7 +22,6 @@
 namespace crawl {
 
 #define MAX_LINKS 8
-#define LINK_NAME_BYTES 32
 
 struct BatchView {
   // per-message
