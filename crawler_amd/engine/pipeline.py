@@ -35,6 +35,7 @@ class ChannelResult:
     status: str = "fetched"      # fetched | deadend | error
     error: str = ""
     posts_stored: int = 0
+    skipped_fetched: int = 0
     discovered: List[str] = dataclasses.field(default_factory=list)
     discovered_links: List[G.DiscoveredLink] = dataclasses.field(
         default_factory=list
@@ -186,11 +187,31 @@ def run_for_channel(client, page: Page, sm, cfg, rw=None, seen=None,
         return result
 
     # -- per-message processing (processAllMessagesWithProcessor) --
+    # resampleMarker / addNewMessages (crawl/runner.go:1572-1697): on a
+    # re-crawl, messages already "fetched" are skipped (no duplicate
+    # posts); tracked messages absent from the new fetch are marked
+    # "deleted"; everything else (re)processes. (The GPU engine path
+    # processes a channel atomically in one batch, so its resume
+    # granularity is the channel — see gpu_runner.py.)
     username = sg["active_usernames"][0] if sg["active_usernames"] else ""
+    tracked = {}
+    if page.messages:
+        tracked = {(pm.chat_id, pm.message_id): pm for pm in page.messages}
+        discovered_keys = {(m.chat_id, m.msg_id) for m in messages}
+        for key, pm in tracked.items():
+            if pm.status == "fetched":
+                continue
+            pm.status = "resample" if key in discovered_keys else "deleted"
     all_links: dict = {}
     for m in messages:
+        pm = tracked.get((m.chat_id, m.msg_id))
+        if pm is not None and pm.status == "fetched":
+            result.skipped_fetched += 1
+            continue
         if page.id and page.id in getattr(sm, "pages", {}):
             sm.update_message(page.id, m.chat_id, m.msg_id, "fetched")
+        elif pm is not None:
+            pm.status = "fetched"
         comments = None
         if m.reply_count > 0:
             comments = client.get_message_comments(

@@ -204,3 +204,32 @@ def test_walk_stats_and_seed_marking(tmp_path):
     runner.seed(["c0000000005"])
     runner.run(max_pages=3)
     assert rw.get_channel_last_crawled("c0000000005") is not None
+
+
+def test_resample_marker_skips_fetched_and_marks_deleted(tmp_path):
+    """resampleMarker semantics (crawl/runner.go:1572-1633): re-crawling a
+    page skips already-fetched messages (no duplicate posts) and marks
+    tracked-but-vanished messages deleted."""
+    from crawler_amd.engine.state import PageMessage
+
+    cfg, feed, pool, sm = mk_env(tmp_path, posts=20)
+    page = Page(id="pr1", url="c0000000003", depth=0)
+    sm.add_layer([page])
+    res1 = run_for_channel_with_pool(pool, page, sm, cfg)
+    assert res1.posts_stored == 20
+    page = sm.get_page("pr1")
+    assert len(page.messages) == 20
+    assert all(m.status == "fetched" for m in page.messages)
+
+    # track a phantom message that the next fetch won't contain
+    page.messages.append(PageMessage(chat_id=page.messages[0].chat_id,
+                                     message_id=999 << 20, status="unfetched",
+                                     page_id="pr1"))
+    res2 = run_for_channel_with_pool(pool, page, sm, cfg)
+    assert res2.posts_stored == 0          # nothing re-stored
+    assert res2.skipped_fetched == 20
+    phantom = [m for m in page.messages if m.message_id == 999 << 20][0]
+    assert phantom.status == "deleted"
+    sm.close()
+    path = tmp_path / "t1" / "c0000000003" / "posts" / "posts.jsonl"
+    assert path.read_bytes().count(b"\n") == 20  # no duplicates on disk
