@@ -272,6 +272,29 @@ def main(argv: Optional[List[str]] = None) -> int:
     )
     sm = LocalStateManager(cfg)
 
+    chunker = None
+    if cfg.combine_files:
+        # CombineFiles mode (dapr/standalone.go:253-271): posts go through
+        # the temp->watch combiner; combined files land under the crawl's
+        # combined/ dir (the upload-binding mock)
+        import shutil
+
+        from .engine.chunker import Chunker
+
+        combined_dir = os.path.join(cfg.storage_root, cfg.crawl_id,
+                                    "combined")
+        os.makedirs(combined_dir, exist_ok=True)
+        chunker = Chunker(
+            cfg.combine_temp_dir, cfg.combine_watch_dir,
+            cfg.combine_write_dir,
+            upload=lambda p: shutil.copy(p, combined_dir),
+            trigger_bytes=cfg.combine_trigger_size * 1024 * 1024,
+            hard_cap_bytes=cfg.combine_hard_cap * 1024 * 1024,
+        )
+        chunker.verify_cleanup()  # crash recovery at startup
+        chunker.start()
+        sm.attach_chunker(chunker)
+
     if cfg.validate_only:
         from .engine.validator import run_validation_loop
 
@@ -301,6 +324,8 @@ def main(argv: Optional[List[str]] = None) -> int:
     else:
         runner = StandaloneRunner(cfg, sm, pool)
         stats = runner.run(urls)
+    if chunker is not None:
+        chunker.stop()
     print(f"crawl complete: {stats}", file=sys.stderr)
     return 0
 

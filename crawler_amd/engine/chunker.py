@@ -161,6 +161,9 @@ class Chunker:
         self._stop.set()
         if self._thread:
             self._thread.join(timeout=10)
+        # final sweep: pick up files the poll loop had not seen yet, then
+        # flush the partial batch
+        self.scan_once()
         self.flush()
 
     def verify_cleanup(self) -> dict:

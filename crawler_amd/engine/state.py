@@ -555,6 +555,22 @@ class LocalStateManager(BaseStateManager):
 
     # ---- posts / files ----
 
+    # CombineFiles mode: when a chunker sink is attached, post JSONL goes
+    # through the temp->watch protocol instead of direct per-channel
+    # appends (reference daprstate.go:1106-1248: CombineFiles writes temp
+    # files for the chunker rather than the storage binding).
+    post_sink = None  # Chunker or None
+    _sink_seq = 0
+
+    def attach_chunker(self, chunker) -> None:
+        self.post_sink = chunker
+
+    def _sink_write(self, channel: str, data) -> None:
+        self._sink_seq += 1
+        name = (f"{self.metadata.crawl_id}-{channel}-"
+                f"{self._sink_seq:08d}.jsonl")
+        self.post_sink.write_temp_then_watch(name, bytes(data))
+
     def _post_file(self, channel: str):
         path = self._posts_path(channel)
         f = self._post_files.get(path)
@@ -570,10 +586,17 @@ class LocalStateManager(BaseStateManager):
         return f
 
     def store_post(self, channel: str, post: Post) -> None:
-        self._post_file(channel).write(post.to_jsonl().encode("utf-8"))
+        data = post.to_jsonl().encode("utf-8")
+        if self.post_sink is not None:
+            self._sink_write(channel, data)
+            return
+        self._post_file(channel).write(data)
 
-    def store_post_lines(self, channel: str, data: bytes) -> None:
+    def store_post_lines(self, channel: str, data) -> None:
         """Bulk JSONL append (the GPU path hands whole encoded blocks)."""
+        if self.post_sink is not None:
+            self._sink_write(channel, data)
+            return
         self._post_file(channel).write(data)
 
     def store_file(self, channel: str, source_path: str,

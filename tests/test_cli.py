@@ -83,3 +83,27 @@ def test_e2e_youtube_random(tmp_path):
         "--crawl-id", "yt9", "--max-posts", "5",
     ])
     assert rc == 0
+
+
+def test_e2e_combine_files_mode(tmp_path):
+    """--combine-files routes post JSONL through the chunker
+    (temp->watch->combined), landing combined files under the crawl."""
+    rc = main([
+        "--mode", "standalone", "--urls", "c0000000001,c0000000002",
+        "--skip-media", "--storage-root", str(tmp_path),
+        "--crawl-id", "comb1", "--synthetic-universe", "100",
+        "--synthetic-posts", "15", "--disable-rate-limits",
+        "--min-users", "1", "--combine-files",
+        "--combine-temp-dir", str(tmp_path / "tempd"),
+        "--combine-watch-dir", str(tmp_path / "watchd"),
+        "--combine-write-dir", str(tmp_path / "writed"),
+        "--combine-trigger-size", "1", "--combine-hard-cap", "2",
+    ])
+    assert rc == 0
+    combined = list((tmp_path / "comb1" / "combined").glob("*.jsonl"))
+    assert combined, "combined files must be uploaded"
+    lines = b"".join(p.read_bytes() for p in combined).splitlines()
+    assert len(lines) == 30  # 2 channels x 15 posts, all combined
+    json.loads(lines[0])
+    # per-channel direct appends did not happen in combine mode
+    assert not (tmp_path / "comb1" / "c0000000001").exists()
