@@ -211,6 +211,17 @@ def resolve_urls(args) -> List[str]:
     if args.url_file:
         with open(args.url_file) as f:
             urls += [l.strip() for l in f if l.strip()]
+    if args.url_file_url:
+        # DownloadURLFile (common/utils.go:115-187); this deployment has
+        # no egress, so only file:// URLs resolve
+        if args.url_file_url.startswith("file://"):
+            with open(args.url_file_url[len("file://"):]) as f:
+                urls += [l.strip() for l in f if l.strip()]
+        else:
+            raise ValueError(
+                "remote --url-file-url requires network egress; use "
+                "file:// or --url-file"
+            )
     return urls
 
 
@@ -265,8 +276,10 @@ def main(argv: Optional[List[str]] = None) -> int:
         print(f"youtube crawl complete: {stats}", file=sys.stderr)
         return 0
 
+    db_urls = [u for u in args.tdlib_database_urls.split(",") if u]
+    pool_size = len(db_urls) if db_urls else args.pool_size
     pool = ConnectionPool(
-        feed, args.pool_size, cfg.rate_limit,
+        feed, pool_size, cfg.rate_limit,
         posts_per_channel=args.synthetic_posts,
         disable_rate_limits=cfg.disable_rate_limits,
     )

@@ -107,3 +107,49 @@ def test_e2e_combine_files_mode(tmp_path):
     json.loads(lines[0])
     # per-channel direct appends did not happen in combine mode
     assert not (tmp_path / "comb1" / "c0000000001").exists()
+
+
+def test_yaml_config_precedence(tmp_path):
+    """viper precedence (main.go:231-261): flags > yaml > defaults."""
+    cfgfile = tmp_path / "config.yaml"
+    cfgfile.write_text(
+        "concurrency: 7\nmin-users: 3\nsampling: snowball\n"
+    )
+    cfg = parse_config(["--config", str(cfgfile), "--urls", "abcde",
+                        "--min-users", "9", "--crawl-id", "y1"])
+    assert cfg.concurrency == 7          # yaml wins over default
+    assert cfg.min_users == 9            # flag wins over yaml
+    assert cfg.sampling_method == "snowball"
+
+
+def test_url_file_and_file_url(tmp_path):
+    f1 = tmp_path / "urls.txt"
+    f1.write_text("chan_a\nchan_b\n")
+    from crawler_amd.cli import build_parser, resolve_urls
+
+    args = build_parser().parse_args(["--url-file", str(f1)])
+    assert resolve_urls(args) == ["chan_a", "chan_b"]
+    args2 = build_parser().parse_args(
+        ["--url-file-url", f"file://{f1}"]
+    )
+    assert resolve_urls(args2) == ["chan_a", "chan_b"]
+    args3 = build_parser().parse_args(
+        ["--url-file-url", "https://example.com/u.txt"]
+    )
+    import pytest as _pt
+
+    with _pt.raises(ValueError):
+        resolve_urls(args3)
+
+
+def test_pool_sized_by_tdlib_database_urls(tmp_path):
+    """PreloadConnections: one pooled session per DB archive URL
+    (connection_pool.go:97-149)."""
+    rc = main([
+        "--mode", "standalone", "--urls", "c0000000001",
+        "--storage-root", str(tmp_path), "--crawl-id", "pz1",
+        "--synthetic-universe", "100", "--synthetic-posts", "5",
+        "--disable-rate-limits", "--min-users", "1",
+        "--tdlib-database-urls", "u1,u2,u3",
+    ])
+    assert rc == 0
