@@ -121,13 +121,21 @@ class GpuCrawlEngine:
 
     # ---- BFS crawl (snowball / channel) ----
 
-    def run(self, seed_urls: List[str], comm=None) -> dict:
+    def run(self, seed_urls: List[str], comm=None,
+            resume: bool = True) -> dict:
         """comm: optional torch.distributed process group for multi-rank
-        discovery exchange (rank-sharded layers)."""
+        discovery exchange (rank-sharded layers). Resume follows the
+        standalone rules (never resume completed; state.json required)."""
         import torch.distributed as dist
 
         sm = self.sm
-        sm.initialize(seed_urls)
+        resumed = False
+        if resume:
+            _exec, ok = sm.find_incomplete_crawl(sm.metadata.crawl_id)
+            if ok and sm.load_state():
+                resumed = True
+        if not resumed:
+            sm.initialize(seed_urls)
         depth = 0
         while True:
             layer = [p for p in sm.get_layer_by_depth(depth)

@@ -99,3 +99,31 @@ def test_cli_gpu_mode_end_to_end(tmp_path):
     assert len(jsonls) >= 2  # seed + discovered channels
     obj = json.loads(jsonls[0].read_bytes().splitlines()[0])
     assert obj["platform_name"] == "Telegram"
+
+
+def test_gpu_engine_resume_skips_fetched(tmp_path):
+    """Resume rules on the GPU path: fetched pages of an incomplete crawl
+    are not re-processed; completed crawls start fresh."""
+    from crawler_amd.engine.state import Page
+
+    cfg, feed, sm, eng = mk_engine(tmp_path, sampling_method="channel")
+    # craft an incomplete crawl: one fetched, one unfetched
+    sm.initialize(["c%010d" % 5, "c%010d" % 6])
+    p0 = sm.get_layer_by_depth(0)[0]
+    p0.status = "fetched"
+    sm.update_page(p0)
+    sm.save_state()
+
+    cfg2 = cfg
+    sm2 = LocalStateManager(cfg2)
+    from crawler_amd.engine.gpu_runner import GpuCrawlEngine
+
+    eng2 = GpuCrawlEngine(cfg2, sm2, feed, posts_per_channel=64)
+    stats = eng2.run(["ignored_seed_list"])  # resume path ignores seeds
+    assert stats["pages"] == 1  # only the unfetched page ran
+    sm3 = LocalStateManager(cfg2)
+    sm3.load_state()
+    assert sm3.metadata.status == "completed"
+    assert {p.url for p in sm3.get_layer_by_depth(0)} == {
+        "c%010d" % 5, "c%010d" % 6
+    }
