@@ -25,9 +25,8 @@ reference's at-least-once republish model.
 """
 from __future__ import annotations
 
-import datetime as _dt
 import uuid
-from typing import Callable, List, Optional, Tuple
+from typing import Callable, List, Tuple
 
 from ..engine.state import LocalStateManager, Page
 from . import messages as M
