@@ -11,7 +11,8 @@ import subprocess
 import sys
 
 CSRC = os.path.dirname(os.path.abspath(__file__)) + "/csrc"
-SOURCES = ["parse_encode.hip", "dedup.hip", "feedgen.hip", "yt_encode.hip"]
+SOURCES = ["parse_encode.hip", "dedup.hip", "feedgen.hip", "yt_encode.hip",
+           "aggregate.hip", "htmlclass.hip"]
 OUT = os.path.join(CSRC, "libcrawlhip.so")
 
 

@@ -67,6 +67,7 @@ def load_lib():
     _bind_feedgen(lib)
     _bind_v2(lib)
     _bind_yt(lib)
+    _bind_aux(lib)
     _lib = lib
     return lib
 
@@ -161,6 +162,98 @@ def yt_parse_encode(batch, now: Optional[_dt.datetime] = None,
     if rc != 0:
         raise RuntimeError(f"crawl_yt_write failed: hip {rc}")
     return out, line_off, line_len
+
+
+
+
+def _bind_aux(lib):
+    lib.crawl_channel_stats.restype = ctypes.c_int
+    lib.crawl_channel_stats.argtypes = [ctypes.c_void_p] * 4 + [
+        ctypes.c_int, ctypes.c_int] + [ctypes.c_void_p] * 5 + [
+        ctypes.c_void_p]
+    lib.crawl_html_classify.restype = ctypes.c_int
+    lib.crawl_html_classify.argtypes = [
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
+        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+    ]
+
+
+def channel_stats(batch: B.MessageBatch, line_len=None):
+    """Segmented per-channel engagement aggregation (SURVEY §2.6:
+    GetTotalChannelViews etc. -> one workgroup per channel segment).
+    Returns dict of device tensors: views/forwards/replies int64[K],
+    posts int32[K], totals int64[4] (views,forwards,replies,posts)."""
+    lib = require_lib()
+    dev = batch.device
+    K = batch.n_channels
+    P = batch.n // K
+    assert K * P == batch.n, "batch must be K x P channel-grouped"
+    out = {
+        "views": torch.zeros(K, dtype=torch.int64, device=dev),
+        "forwards": torch.zeros(K, dtype=torch.int64, device=dev),
+        "replies": torch.zeros(K, dtype=torch.int64, device=dev),
+        "posts": torch.zeros(K, dtype=torch.int32, device=dev),
+        "totals": torch.zeros(4, dtype=torch.int64, device=dev),
+    }
+    m = batch.meta
+    rc = lib.crawl_channel_stats(
+        ctypes.c_void_p(m["views"].data_ptr()),
+        ctypes.c_void_p(m["forwards"].data_ptr()),
+        ctypes.c_void_p(m["reply_count"].data_ptr()),
+        ctypes.c_void_p(line_len.data_ptr()) if line_len is not None
+        else None,
+        P, K,
+        ctypes.c_void_p(out["views"].data_ptr()),
+        ctypes.c_void_p(out["forwards"].data_ptr()),
+        ctypes.c_void_p(out["replies"].data_ptr()),
+        ctypes.c_void_p(out["posts"].data_ptr()),
+        ctypes.c_void_p(out["totals"].data_ptr()),
+        ctypes.c_void_p(torch.cuda.current_stream().cuda_stream),
+    )
+    if rc != 0:
+        raise RuntimeError(f"crawl_channel_stats failed: hip {rc}")
+    return out
+
+
+_HTML_STATUS = ["valid", "not_channel", "invalid"]
+_HTML_REASON = ["", "not_supergroup", "username_not_occupied", "not_found",
+                "unrecognized"]
+
+
+def html_classify(docs, device="cuda:0"):
+    """Batch t.me HTML classification on GPU (oracle:
+    engine.htmlvalidator.parse_channel_html). docs: list[bytes].
+    Returns list of (status, reason) strings."""
+    lib = require_lib()
+    dev = torch.device(device)
+    n = len(docs)
+    blob = b"".join(docs) or b"\0"
+    pool = torch.frombuffer(bytearray(blob), dtype=torch.uint8).to(dev)
+    offs, cur = [], 0
+    lens = []
+    for d in docs:
+        offs.append(cur)
+        lens.append(len(d))
+        cur += len(d)
+    doc_off = torch.tensor(offs, dtype=torch.int64, device=dev)
+    doc_len = torch.tensor(lens, dtype=torch.int32, device=dev)
+    status = torch.zeros(n, dtype=torch.int32, device=dev)
+    reason = torch.zeros(n, dtype=torch.int32, device=dev)
+    rc = lib.crawl_html_classify(
+        ctypes.c_void_p(pool.data_ptr()),
+        ctypes.c_void_p(doc_off.data_ptr()),
+        ctypes.c_void_p(doc_len.data_ptr()), n,
+        ctypes.c_void_p(status.data_ptr()),
+        ctypes.c_void_p(reason.data_ptr()),
+        min(max(1, (n + 3) // 4), 8192),
+        ctypes.c_void_p(torch.cuda.current_stream().cuda_stream),
+    )
+    if rc != 0:
+        raise RuntimeError(f"crawl_html_classify failed: hip {rc}")
+    st = status.cpu().numpy()
+    rs = reason.cpu().numpy()
+    return [(_HTML_STATUS[int(st[i])], _HTML_REASON[int(rs[i])])
+            for i in range(n)]
 
 
 def _bind_dedup(lib):
