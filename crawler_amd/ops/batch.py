@@ -386,3 +386,17 @@ def channel_row(batch: MessageBatch, c: int) -> ChannelRow:
         post_count=int(batch.ch_postcount[c]),
         total_views=int(batch.ch_totalviews[c]),
     )
+
+def select_rows(batch: MessageBatch, idx: torch.Tensor) -> MessageBatch:
+    """Row-subset view of a batch (pools shared; per-message tensors
+    gathered). Offsets reference the shared pool, so any row order is
+    valid — used by the sampling path (date-between + sample-size)."""
+    idx = idx.to(batch.chat_id.device, torch.long)
+    return dataclasses.replace(
+        batch,
+        n=int(idx.numel()),
+        chat_id=batch.chat_id[idx].contiguous(),
+        msg_id=batch.msg_id[idx].contiguous(),
+        text_off=batch.text_off[idx].contiguous(),
+        meta={k: v[idx].contiguous() for k, v in batch.meta.items()},
+    )

@@ -171,6 +171,11 @@ def _bind_aux(lib):
     lib.crawl_channel_stats.argtypes = [ctypes.c_void_p] * 4 + [
         ctypes.c_int, ctypes.c_int] + [ctypes.c_void_p] * 5 + [
         ctypes.c_void_p]
+    lib.crawl_reservoir_sample.restype = ctypes.c_int
+    lib.crawl_reservoir_sample.argtypes = [
+        ctypes.c_long, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
+    ]
     lib.crawl_html_classify.restype = ctypes.c_int
     lib.crawl_html_classify.argtypes = [
         ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
@@ -212,6 +217,45 @@ def channel_stats(batch: B.MessageBatch, line_len=None):
     )
     if rc != 0:
         raise RuntimeError(f"crawl_channel_stats failed: hip {rc}")
+    return out
+
+
+def reservoir_sample(batch: B.MessageBatch, k: int, seed: int = 0):
+    """Per-channel uniform sample without replacement (SURVEY §2.6
+    Fisher-Yates row). Returns int32[K, k] global row indices (device)."""
+    lib = require_lib()
+    dev = batch.device
+    K = batch.n_channels
+    P = batch.n // K
+    out = torch.zeros((K, k), dtype=torch.int32, device=dev)
+    rc = lib.crawl_reservoir_sample(
+        seed, P, K, k, ctypes.c_void_p(out.data_ptr()),
+        min(max(1, (K + 3) // 4), 8192),
+        ctypes.c_void_p(torch.cuda.current_stream().cuda_stream),
+    )
+    if rc != 0:
+        raise RuntimeError(f"crawl_reservoir_sample failed: hip {rc}")
+    return out
+
+
+def reservoir_sample_oracle(P: int, K: int, k: int, seed: int = 0):
+    """Python replay of the device reservoir (same splitmix stream)."""
+    def sm64(x):
+        z = (x + 0x9E3779B97F4A7C15) & (2**64 - 1)
+        z = ((z ^ (z >> 30)) * 0xBF58476D1CE4E5B9) & (2**64 - 1)
+        z = ((z ^ (z >> 27)) * 0x94D049BB133111EB) & (2**64 - 1)
+        return z ^ (z >> 31)
+
+    out = []
+    for c in range(K):
+        slot = list(range(c * P, c * P + k))
+        for i in range(k, P):
+            r = sm64((seed ^ ((c * 0x9E3779B1 + i) & (2**64 - 1)))
+                     & (2**64 - 1))
+            j = r % (i + 1)
+            if j < k:
+                slot[j] = c * P + i
+        out.append(slot)
     return out
 
 

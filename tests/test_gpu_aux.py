@@ -64,3 +64,37 @@ def test_html_classify_fixtures_and_mock():
             f"doc {i}: GPU {got[i]} != oracle "
             f"{(oracle.status, oracle.reason)}"
         )
+
+
+def test_reservoir_sample_matches_oracle_and_encodes():
+    from crawler_amd.ops import batch as B
+    from crawler_amd.ops import gpu
+    from crawler_amd.ops.golden_batch import encode_batch
+    import datetime as dt
+
+    NOW = dt.datetime(2026, 1, 1, tzinfo=dt.timezone.utc)
+    feed = SyntheticFeed(FeedConfig(seed=10, universe=400))
+    K, P, k = 6, 200, 25
+    dev_batch = feed.build_batch_device(np.arange(K),
+                                        torch.device("cuda:0"),
+                                        posts_per_channel=P)
+    idx = gpu.reservoir_sample(dev_batch, k, seed=123)
+    torch.cuda.synchronize()
+    got = idx.cpu().numpy().tolist()
+    expect = gpu.reservoir_sample_oracle(P, K, k, seed=123)
+    assert got == expect
+    # uniform-without-replacement invariants
+    for c in range(K):
+        rows = got[c]
+        assert len(set(rows)) == k
+        assert all(c * P <= r < (c + 1) * P for r in rows)
+
+    # the sampled subset flows through the full encode path byte-exactly
+    sub_idx = idx.flatten()
+    sub = B.select_rows(dev_batch, sub_idx)
+    res = gpu.parse_encode(sub, now=NOW)
+    torch.cuda.synchronize()
+    cpu_batch = feed.build_batch(np.arange(K), posts_per_channel=P)
+    cpu_sub = B.select_rows(cpu_batch, sub_idx.cpu())
+    golden_lines, _ = encode_batch(cpu_sub, now=NOW)
+    assert bytes(res.out.cpu().numpy()) == b"".join(golden_lines)
