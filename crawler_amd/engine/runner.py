@@ -75,6 +75,8 @@ class StandaloneRunner:
             depth += 1
         sm.update_crawl_metadata(sm.metadata.crawl_id, {
             "status": "completed",
+            "messagesCount": self.stats["posts"],
+            "errorsCount": self.stats["errors"],
         })
         sm.save_state()
         sm.close()

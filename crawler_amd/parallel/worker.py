@@ -62,6 +62,16 @@ class Worker:
             return True
         return "timeout" in msg.lower() or "connection" in msg.lower()
 
+    def _merged_cfg(self, item: M.WorkItem):
+        """WorkItemConfig overrides the worker's base config
+        (worker.go:302-381 applies the orchestrator-sent config)."""
+        import dataclasses as _dc
+
+        cfg = self.cfg
+        overrides = {k: v for k, v in (item.config or {}).items()
+                     if hasattr(cfg, k) and v not in (None, "", -1)}
+        return _dc.replace(cfg, **overrides) if overrides else cfg
+
     def process_item(self, item: M.WorkItem) -> M.WorkResult:
         """ProcessWorkItem (worker.go:302-381)."""
         t0 = time.perf_counter()
@@ -69,7 +79,7 @@ class Worker:
                     parent_id=item.parent_id)
         try:
             res = run_for_channel_with_pool(
-                self.pool, page, self.sm, self.cfg
+                self.pool, page, self.sm, self._merged_cfg(item)
             )
         except Exception as err:  # classified below
             status = (M.STATUS_RETRY if self.should_retry_error(err)

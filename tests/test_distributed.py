@@ -244,3 +244,16 @@ def test_gpu_engine_name_exchange_two_procs(tmp_path):
     r0 = (tmp_path / "n0.txt").read_text()
     assert r0 == (tmp_path / "n1.txt").read_text()
     assert "rank0chan0" in r0 and "rank1chan2" in r0
+
+
+def test_work_item_config_propagates_to_worker(tmp_path):
+    """WorkItemConfig overrides the worker's base config
+    (worker.go:302-381): the orchestrator's max_posts cap applies."""
+    cfg, orch, workers, store = mk_dist_env(tmp_path, sampling="channel",
+                                            n_workers=1)
+    w = workers[0]
+    item = M.WorkItem(id="wi1", url="c0000000001", depth=0, crawl_id="d1",
+                      config={"max_posts": 5, "min_users": 1})
+    res = w.process_item(item)
+    assert res.status == M.STATUS_SUCCESS
+    assert res.posts_stored == 5  # capped by the item config, not base cfg
