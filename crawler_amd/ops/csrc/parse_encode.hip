@@ -456,7 +456,7 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
 
 // ---------- kernels ----------
 
-__global__ void __launch_bounds__(256, 4)
+__global__ void __launch_bounds__(256, 6)
 measure_extract_kernel(BatchView B, LinkOut LO, int* line_len) {
   const int lane = lane_id();
   const int wave = wave_id();
