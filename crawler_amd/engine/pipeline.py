@@ -25,8 +25,11 @@ from typing import List, Optional, Tuple
 from ..feed.client import TelegramAPIError
 from ..models.post import ChannelData, EngagementData
 from ..ops import golden as G
+from ..utils.logging import get_logger
 from . import errors as E
 from .state import Page
+
+_log = get_logger("pipeline")
 
 
 @dataclasses.dataclass
@@ -36,6 +39,7 @@ class ChannelResult:
     error: str = ""
     posts_stored: int = 0
     skipped_fetched: int = 0
+    parse_errors: int = 0
     discovered: List[str] = dataclasses.field(default_factory=list)
     discovered_links: List[G.DiscoveredLink] = dataclasses.field(
         default_factory=list
@@ -212,29 +216,38 @@ def run_for_channel(client, page: Page, sm, cfg, rw=None, seen=None,
             sm.update_message(page.id, m.chat_id, m.msg_id, "fetched")
         elif pm is not None:
             pm.status = "fetched"
-        comments = None
-        if m.reply_count > 0:
-            comments = client.get_message_comments(
-                m.chat_id, m.msg_id, cfg.max_comments
+        # Panic containment at message granularity (tdutils.go:395-405
+        # recover()): one malformed message never aborts the channel.
+        try:
+            comments = None
+            if m.reply_count > 0:
+                comments = client.get_message_comments(
+                    m.chat_id, m.msg_id, cfg.max_comments
+                )
+            post = G.parse_message(
+                m,
+                channel_username=username or channel,
+                chat_title=info.title,
+                member_count=sg["member_count"],
+                post_count=info.message_count,
+                total_views=info.total_views,
+                comments=comments,
+                min_post_date=cfg.min_post_date,
+                skip_media=cfg.skip_media_download,
+                now=now,
             )
-        post = G.parse_message(
-            m,
-            channel_username=username or channel,
-            chat_title=info.title,
-            member_count=sg["member_count"],
-            post_count=info.message_count,
-            total_views=info.total_views,
-            comments=comments,
-            min_post_date=cfg.min_post_date,
-            skip_media=cfg.skip_media_download,
-            now=now,
-        )
-        if post is not None:
-            sm.store_post(channel, post)
-            result.posts_stored += 1
-        for link in G.extract_links_with_source(m):
-            if link.name not in all_links:
-                all_links[link.name] = link
+            if post is not None:
+                sm.store_post(channel, post)
+                result.posts_stored += 1
+            for link in G.extract_links_with_source(m):
+                if link.name not in all_links:
+                    all_links[link.name] = link
+        except TelegramAPIError:
+            raise  # API faults keep their channel-level semantics
+        except Exception as perr:
+            result.parse_errors += 1
+            _log.warn("message-parse-recovered", channel=channel,
+                      msg_id=m.msg_id, error=str(perr))
     result.discovered_links = list(all_links.values())
 
     # -- outlink handling per sampling mode --

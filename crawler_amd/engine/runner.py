@@ -108,6 +108,14 @@ class StandaloneRunner:
                 if self.pool.empty():
                     raise
                 return
+            except Exception as err:  # panic containment per page
+                # (recoverFromPanic, crawl/runner.go: one bad channel
+                # never kills the layer)
+                page.status = "error"
+                page.error = f"recovered: {err}"
+                sm.update_page(page)
+                self.stats["errors"] += 1
+                return
             page.status = res.status
             page.error = res.error
             sm.update_page(page)
@@ -202,6 +210,9 @@ class RandomWalkRunner:
             except E.WalkbackExhausted:
                 self.stats["walkback_exhausted"] += 1
             return True  # delete page (standalone.go:912-921)
+        except Exception:  # panic containment: drop the page, keep walking
+            self.stats["errors"] += 1
+            return True
         self.stats["pages"] += 1
         self.stats["posts"] += res.posts_stored
         if res.status == "error":

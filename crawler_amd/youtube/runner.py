@@ -56,10 +56,15 @@ def run_youtube(cfg, urls: List[str], client=None, sm=None,
             ch = client.get_channel_info(v.channel_id)
         except QuotaExceeded:
             pass
-        return convert_video_to_post(v, ch, crawl_label=cfg.crawl_label)
+        # panic containment per video (youtube panic_test.go analog):
+        # one malformed video never kills the conversion pool
+        try:
+            return convert_video_to_post(v, ch, crawl_label=cfg.crawl_label)
+        except Exception:
+            return None
 
     with cf.ThreadPoolExecutor(max_workers=10) as ex:
-        posts = list(ex.map(conv, videos))
+        posts = [p for p in ex.map(conv, videos) if p is not None]
     for p in posts:
         sm.store_post(p.channel_id, p)
     sm.save_state()
