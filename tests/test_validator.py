@@ -188,3 +188,164 @@ def test_recover_runs_at_startup(tmp_path):
     v.recover()
     assert all(e.validation_status == "pending"
                for e in rw.pending_edges.values())
+
+
+# ---------- scripted-validate_fn matrix (validator.go:194-310 branches) ----
+
+class Script:
+    """validate_fn whose outcome per call is scripted; records calls."""
+
+    def __init__(self, outcomes):
+        self.outcomes = list(outcomes)
+        self.calls = 0
+
+    def __call__(self, username):
+        self.calls += 1
+        out = self.outcomes.pop(0) if self.outcomes else "valid"
+        if out == "blocked":
+            raise ValidationHTTPError("blocked", 429)
+        if out == "transient":
+            raise ValidationHTTPError("transient", 503)
+        if out == "not_channel":
+            return ChannelValidationResult("not_channel", "not_supergroup")
+        if out == "invalid":
+            return ChannelValidationResult("invalid", "not_found")
+        return ChannelValidationResult("valid")
+
+
+def mk_scripted(tmp_path, outcomes, **cfg_kw):
+    cfg_kw.setdefault("walkback_rate", 0)
+    cfg = CrawlerConfig(crawl_id="v1", storage_root=str(tmp_path), **cfg_kw)
+    sm = LocalStateManager(cfg)
+    rw = RandomWalkStore()
+    script = Script(outcomes)
+    v = TandemValidator(cfg, sm, rw, validate_fn=script,
+                        rng=random.Random(9), probe_interval=0.0)
+    return cfg, sm, rw, v, script
+
+
+def test_transient_error_leaves_edge_pending(tmp_path):
+    cfg, sm, rw, v, script = mk_scripted(tmp_path, ["transient", "valid"])
+    bid = seed_batch(rw, ["aaaa1", "bbbb2"])
+    v.pump_edges()
+    edges = {e.destination_channel: e.validation_status
+             for e in rw.edges_of_batch(bid)}
+    assert edges["aaaa1"] == "pending"   # left for a later claim round
+    assert edges["bbbb2"] == "valid"
+    assert not v.blocked and v.consecutive_blocked == 0
+    # next round re-claims ONLY the pending edge
+    v.pump_edges()
+    assert script.calls == 3
+
+
+def test_blocked_counter_resets_on_success(tmp_path):
+    """4 blocked, 1 ok, 4 blocked: never crosses the threshold of 5
+    CONSECUTIVE blocked outcomes (validator.go:34-38)."""
+    outcomes = ["blocked"] * 4 + ["valid"] + ["blocked"] * 4
+    cfg, sm, rw, v, script = mk_scripted(tmp_path, outcomes)
+    seed_batch(rw, ["n%04d" % i for i in range(9)])
+    while v.pump_edges():
+        pass
+    assert not v.blocked
+    assert v.stats["blocked_events"] == 0
+
+
+def test_rate_limiter_skipped_on_fast_paths(tmp_path):
+    """The HTTP rate limiter is consumed only for real HTTP validations —
+    cached-invalid and duplicate fast paths bypass it
+    (validator rate limiter, telegramhelper/validator_rate_limiter.go)."""
+    cfg, sm, rw, v, script = mk_scripted(tmp_path, ["valid"])
+
+    class CountingLimiter:
+        n = 0
+
+        def acquire(self):
+            self.n += 1
+
+    v.rate_limiter = CountingLimiter()
+    rw.mark_invalid_channel("badcha")
+    sm.add_discovered_channel("dupcha")
+    rw.discovered_channels["dupcha"] = {"username": "dupcha"}
+    seed_batch(rw, ["badcha", "dupcha", "newcha"])
+    v.pump_edges()
+    assert v.rate_limiter.n == 1         # only "newcha" hit HTTP
+    assert script.calls == 1
+
+
+def test_probe_gating_and_unblock_event(tmp_path):
+    outcomes = ["blocked"] * 5 + ["blocked", "valid"]
+    cfg, sm, rw, v, script = mk_scripted(tmp_path, outcomes)
+    t = {"now": 0.0}
+    v.clock = lambda: t["now"]
+    v.probe_interval = 300.0
+    seed_batch(rw, ["n%04d" % i for i in range(5)])
+    while v.pump_edges():
+        pass
+    assert v.blocked
+    # within the probe interval: no probe, no work
+    t["now"] += 100
+    assert v.pump_edges() == 0
+    assert script.calls == 5
+    # past the interval: probe runs but is still blocked
+    t["now"] += 300
+    assert v.pump_edges() == 0
+    assert script.calls == 6
+    # next interval: probe succeeds -> unblocked, edges get re-validated
+    t["now"] += 300
+    v.pump_edges()
+    assert not v.blocked
+    assert any(e["type"] == "ip_unblocked" for e in rw.access_events)
+
+
+def test_walkback_rate_forces_walkback_with_valid_channels(tmp_path):
+    cfg, sm, rw, v, script = mk_scripted(tmp_path, ["valid", "valid"],
+                                         walkback_rate=100)
+    sm.add_discovered_channel("backstop")
+    bid = seed_batch(rw, ["ccc33", "ddd44"])
+    while v.pump_edges():
+        pass
+    assert v.pump_walkback()
+    pages = rw.get_pages(10)
+    assert len(pages) == 1 and pages[0].url == "backstop"
+    # both valid channels became skipped edges; walkback edge primary
+    recs = [e for e in rw.edge_records if e.sequence_id == "seq1"]
+    assert sum(1 for e in recs if e.walkback) == 1
+    assert {e.destination_channel for e in recs if e.skipped} == \
+        {"ccc33", "ddd44"}
+
+
+def test_batch_stats_flushed_per_source_type(tmp_path):
+    cfg, sm, rw, v, script = mk_scripted(tmp_path, ["valid"] * 3)
+    sm.add_discovered_channel("backstop")
+    bid = rw.open_batch("v1", "srcchan1", "p1", 0, "seqS")
+    rw.insert_pending_edge(bid, "v1", "eee55", "srcchan1", "seqS",
+                           "mention")
+    rw.insert_pending_edge(bid, "v1", "fff66", "srcchan1", "seqS",
+                           "mention")
+    rw.insert_pending_edge(bid, "v1", "ggg77", "srcchan1", "seqS",
+                           "forward")
+    rw.close_batch(bid)
+    while v.pump_edges():
+        pass
+    assert v.pump_walkback()
+    assert rw.source_type_stats[("v1", "mention")] == 2
+    assert rw.source_type_stats[("v1", "forward")] == 1
+
+
+def test_walkback_exhausted_leaves_batch_for_recovery(tmp_path):
+    """No discovered channel to walk back to: the claimed batch is NOT
+    completed; stale recovery re-opens it (validator.go:360-487 +
+    daprstate.go:4264-4354)."""
+    cfg, sm, rw, v, script = mk_scripted(tmp_path, [], walkback_rate=100)
+    # an empty closed batch: no valid channels -> forced walkback
+    bid = rw.open_batch("v1", "srcchan1", "p1", 0, "seqE")
+    rw.close_batch(bid)
+    assert not v.pump_walkback()        # WalkbackExhausted inside
+    b = rw.pending_batches[bid]
+    assert b.status == "processing"     # stuck claim
+    rw.recover_stale_claims()
+    assert rw.pending_batches[bid].status == "closed"
+    # once a backstop exists the retried batch completes
+    sm.add_discovered_channel("backstop")
+    assert v.pump_walkback()
+    assert rw.pending_batches[bid].status == "completed"
