@@ -1,0 +1,57 @@
+"""Sampling-method validation matrix (reference
+common/sampling_validation.go:19-66 + main_test.go coverage)."""
+import pytest
+
+from crawler_amd.config import validate_sampling_method
+
+
+def ok(**kw):
+    validate_sampling_method(**kw)
+
+
+def bad(match, **kw):
+    with pytest.raises(ValueError, match=match):
+        validate_sampling_method(**kw)
+
+
+def test_platform_method_support_matrix():
+    ok(platform="telegram", sampling_method="channel", url_list=["a"])
+    ok(platform="telegram", sampling_method="snowball", url_list=["a"])
+    ok(platform="telegram", sampling_method="random-walk", seed_size=5)
+    ok(platform="youtube", sampling_method="channel", url_list=["a"])
+    ok(platform="youtube", sampling_method="random")
+    ok(platform="youtube", sampling_method="snowball", url_list=["a"])
+    bad("not supported", platform="telegram", sampling_method="random")
+    bad("not supported", platform="youtube",
+        sampling_method="random-walk")
+    bad("unsupported platform", platform="tiktok",
+        sampling_method="channel")
+
+
+def test_random_walk_xor_seed_sources():
+    # neither
+    bad("not both or neither", platform="telegram",
+        sampling_method="random-walk")
+    # both
+    bad("not both or neither", platform="telegram",
+        sampling_method="random-walk", url_list=["a"], seed_size=5)
+    # url-file counts as a source
+    ok(platform="telegram", sampling_method="random-walk",
+       url_file="/tmp/urls.txt")
+    ok(platform="telegram", sampling_method="random-walk",
+       url_file_url="file:///tmp/urls.txt")
+
+
+def test_random_walk_crawl_id_length():
+    ok(platform="telegram", sampling_method="random-walk", seed_size=1,
+       crawl_id="x" * 32)
+    bad("32 characters", platform="telegram",
+        sampling_method="random-walk", seed_size=1, crawl_id="x" * 33)
+
+
+def test_urls_required_unless_dapr_job():
+    bad("requires URLs", platform="telegram", sampling_method="channel")
+    bad("requires URLs", platform="telegram", sampling_method="snowball")
+    ok(platform="telegram", sampling_method="channel", mode="dapr-job")
+    # random never needs URLs
+    ok(platform="youtube", sampling_method="random", mode="")
