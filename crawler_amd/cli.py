@@ -32,6 +32,8 @@ def build_parser() -> argparse.ArgumentParser:
     a("--config", default="", help="config file (default ./config.yaml)")
     a("--log-level", default="debug")
     a("--dapr", action="store_true", help="compat flag (local runtime)")
+    a("--debug-port", type=int, default=0,
+      help="debug/metrics HTTP server port (pprof-:6060 analog; 0=off)")
     a("--dapr-mode", default="job", choices=["job", "standalone"])
     a("--dapr-port", type=int, default=6481)
     a("--concurrency", type=int, default=1)
@@ -249,6 +251,10 @@ def main(argv: Optional[List[str]] = None) -> int:
     if not mode:
         print("no mode and no URLs given; nothing to do", file=sys.stderr)
         return 2
+
+    from .utils.debugserver import maybe_start
+
+    dbg = maybe_start(args.debug_port)
 
     from .feed import FeedConfig, SyntheticFeed
     from .feed.client import ConnectionPool

@@ -58,3 +58,41 @@ def test_tagged_logger_json_fields():
     lg2 = TaggedLogger("t2", level="error", stream=buf2)
     lg2.info("hidden")
     assert buf2.getvalue() == ""
+
+
+def test_debug_server_endpoints():
+    """pprof-:6060 analog (reference main.go pprof server)."""
+    import json
+    import urllib.request
+
+    from crawler_amd.utils.debugserver import DebugServer
+    from crawler_amd.utils.metrics import MetricsRegistry
+
+    reg = MetricsRegistry()
+    reg.posts.inc(7)
+    srv = DebugServer(0, metrics=reg).start()
+    try:
+        base = f"http://127.0.0.1:{srv.port}"
+        assert urllib.request.urlopen(base + "/healthz").read() == b"ok\n"
+        snap = json.loads(urllib.request.urlopen(base + "/metrics").read())
+        assert snap["posts"] == 7
+        stacks = urllib.request.urlopen(base + "/debug/stacks").read()
+        assert b"thread" in stacks and b"MainThread" in stacks
+        v = json.loads(
+            urllib.request.urlopen(base + "/debug/vars").read())
+        assert v["pid"] > 0 and "uptime_s" in v
+        import urllib.error
+        try:
+            urllib.request.urlopen(base + "/nope")
+            assert False, "expected 404"
+        except urllib.error.HTTPError as e:
+            assert e.code == 404
+    finally:
+        srv.stop()
+
+
+def test_debug_server_maybe_start_disabled():
+    from crawler_amd.utils.debugserver import maybe_start
+
+    assert maybe_start(0) is None
+    assert maybe_start(None) is None
