@@ -127,3 +127,39 @@ def test_gpu_engine_resume_skips_fetched(tmp_path):
     assert {p.url for p in sm3.get_layer_by_depth(0)} == {
         "c%010d" % 5, "c%010d" % 6
     }
+
+
+def test_gpu_engine_through_chunker_sink(tmp_path):
+    """GPU engine + combine-files chunker integration: per-channel JSONL
+    slices flow through the temp->watch protocol and come out combined
+    with nothing lost (daprstate.go:1106-1248 CombineFiles + chunk/)."""
+    from crawler_amd.engine.chunker import Chunker
+
+    cfg, feed, sm, eng = mk_engine(tmp_path, sampling_method="channel",
+                                   max_depth=0)
+    uploads = []
+    ch = Chunker(
+        str(tmp_path / "ctemp"), str(tmp_path / "cwatch"),
+        str(tmp_path / "cwrite"),
+        upload=lambda p: uploads.append(open(p, "rb").read()),
+        trigger_bytes=1 << 20, hard_cap_bytes=4 << 20,
+        batch_timeout_s=600,
+    )
+    sm.attach_chunker(ch)
+    names = ["c%010d" % i for i in range(1, 9)]
+    eng.process_channels(names, now=NOW)
+    ch.scan_once()
+    ch.flush()
+    combined = b"".join(uploads)
+    # oracle: same channels through the CPU batch encoder
+    total = 0
+    for n in names:
+        cid = int(n[1:])
+        batch = feed.build_batch(np.array([cid]), posts_per_channel=64)
+        lines, _ = encode_batch(batch, now=NOW)
+        total += sum(len(l) for l in lines)
+    assert len(combined) == total
+    assert combined.count(b"\n") == 8 * 64
+    # no leftovers anywhere in the pipeline dirs
+    assert list((tmp_path / "cwatch").iterdir()) == []
+    assert list((tmp_path / "ctemp").iterdir()) == []
