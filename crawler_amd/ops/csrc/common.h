@@ -186,11 +186,64 @@ struct JsonEmit {
     cur += n;
   }
 
-  // Escaped JSON string content (no quotes). Clean 64-byte stripes (the
-  // common case) take an identity-copy fast path with no scans.
+  // SWAR probe: does dword w contain any byte that Go-JSON escapes
+  // (<0x20, ", \, <, >, &) or an 0xE2 (possible U+2028/29 leader)?
+  // Classic haszero trick: (x-0x01010101) & ~x & 0x80808080 != 0 iff a
+  // zero byte exists in x.
+  static DEV bool swar_special(unsigned w) {
+    unsigned hz;
+    unsigned nonascii = w & 0x80808080u;
+    // hasless(w, 0x20): exact for the "any byte" question (borrow
+    // propagation can only add false positives, which just take the
+    // exact path)
+    unsigned lt20 = (w - 0x20202020u) & ~w & 0x80808080u;
+    if (lt20) return true;
+    hz = ((w ^ 0x22222222u) - 0x01010101u) & ~(w ^ 0x22222222u);   // "
+    hz |= ((w ^ 0x5C5C5C5Cu) - 0x01010101u) & ~(w ^ 0x5C5C5C5Cu);  // \ .
+    hz |= ((w ^ 0x3C3C3C3Cu) - 0x01010101u) & ~(w ^ 0x3C3C3C3Cu);  // <
+    hz |= ((w ^ 0x3E3E3E3Eu) - 0x01010101u) & ~(w ^ 0x3E3E3E3Eu);  // >
+    hz |= ((w ^ 0x26262626u) - 0x01010101u) & ~(w ^ 0x26262626u);  // &
+    if (hz & 0x80808080u) return true;
+    if (nonascii) {  // only 0xE2 needs the slow path among high bytes
+      unsigned e2 = ((w ^ 0xE2E2E2E2u) - 0x01010101u) & ~(w ^ 0xE2E2E2E2u);
+      if (e2 & 0x80808080u) return true;
+    }
+    return false;
+  }
+
+  // Escaped JSON string content (no quotes). The measure pass (!W) scans
+  // clean 256-byte blocks 4 bytes/lane with a SWAR special-byte probe
+  // (the emitters are VALU-bound, not bandwidth-bound — see
+  // profiles/r01_pmc_counters.csv); dirty or boundary-risk blocks and the
+  // write pass fall back to the exact 64-byte stripe path below.
   DEV void esc(const uint8_t* s, int n) {
     int lane = lane_id();
-    for (int start = 0; start < n; start += WAVE) {
+    int start = 0;
+    while (start < n) {
+      if (!W && start + 4 * WAVE <= n) {
+        // a U+2028/29 leader just before the block would swallow bytes
+        // INSIDE it; route those rare blocks to the exact path
+        bool risk = (start >= 1 && s[start - 1] == 0xE2) ||
+                    (start >= 2 && s[start - 2] == 0xE2 &&
+                     s[start - 1] == 0x80);
+        if (!risk) {
+          unsigned w;
+          __builtin_memcpy(&w, s + start + 4 * lane, 4);
+          if (__ballot(swar_special(w)) == 0) {
+            cur += 4 * WAVE;   // every byte emits verbatim
+            start += 4 * WAVE;
+            continue;
+          }
+        }
+      }
+      esc_stripe(s, n, start, lane);
+      start += WAVE;
+    }
+  }
+
+  // One exact 64-byte stripe of the escape walk (the original path).
+  DEV void esc_stripe(const uint8_t* s, int n, int start, int lane) {
+    {
       int p = start + lane;
       int span = n - start;
       if (span > WAVE) span = WAVE;
@@ -199,7 +252,7 @@ struct JsonEmit {
       if (dirty == 0) {
         if (W && p < n) out[cur + p - start] = s[p];
         cur += span;
-        continue;
+        return;
       }
       if (W) {
         int off = wave_prefix_excl(el);

@@ -153,3 +153,58 @@ def test_larger_batch_bytes_identical(gpu_mod, feed):
     golden_lines, _, res = _roundtrip(gpu_mod, batch)
     out = bytes(res.out.cpu().numpy())
     assert out == b"".join(golden_lines)
+
+
+def test_swar_block_boundary_escapes(gpu_mod):
+    """Adversarial placement of escape bytes around the measure pass's
+    256-byte SWAR blocks (common.h esc fast path): U+2028/29 leaders at
+    block edges, specials at every lane-word position, long clean runs."""
+    U2028 = " "
+    U2029 = " "
+    texts = []
+    # E2-leader at byte offsets straddling the 256-byte block boundary.
+    # description starts at the raw text start, so byte index within the
+    # field == byte index within the python string's utf-8 encoding.
+    for lead in (253, 254, 255, 256, 257):
+        t = "a" * lead + U2028 + "b" * 300
+        texts.append(t)
+        texts.append("a" * lead + U2029 + "b" * 300)
+    # special byte at each position of the first lane word
+    for pos in (0, 1, 2, 3, 63, 64, 127, 255, 256, 511):
+        t = ("x" * pos) + '"' + ("y" * 400)
+        texts.append(t)
+    # a control byte deep in an otherwise clean 1KB run
+    texts.append("c" * 700 + "\x07" + "d" * 300)
+    # fully clean 1KB (pure fast path), clean multiple-of-256
+    texts.append("e" * 1024)
+    texts.append("f" * 512)
+    # dirty first block then long clean tail (fast path must re-engage)
+    texts.append("<&>" + "g" * 900)
+    # non-ascii high bytes that are NOT E2 sequences (must stay clean-ish)
+    texts.append("п" * 400)   # 0xD0 0xBF pairs
+    # E2 that is NOT a U+2028/29 (e.g. '…' U+2026 = E2 80 A6)
+    texts.append("h" * 250 + "…" + "i" * 300)
+
+    msgs = []
+    for k, t in enumerate(texts):
+        msgs.append(G.SynthMessage(
+            chat_id=-100555, msg_id=(k + 1) << 20,
+            date=1_700_100_000 + k, content_type="messageText",
+            text=G.FormattedText(text=t),
+            views=k, forwards=0, poster_handle=f"user{k:04d}",
+        ))
+    ch = [B.ChannelRow(chat_id=-100555, username="swarchan1",
+                       title="SWAR", member_count=3,
+                       post_count=len(msgs), total_views=9)]
+    batch = B.pack(msgs, ch, [0] * len(msgs))
+    from crawler_amd.ops import gpu
+
+    golden_lines, _ = encode_batch(batch, now=NOW)
+    res = gpu.parse_encode(batch.to("cuda:0"), now=NOW)
+    torch.cuda.synchronize()
+    out = bytes(res.out.cpu().numpy())
+    offs = res.line_off.cpu().numpy()
+    lens = res.line_len.cpu().numpy()
+    for i, gl in enumerate(golden_lines):
+        dev_line = out[offs[i]: offs[i] + lens[i]]
+        assert dev_line == gl, f"swar case {i}: {texts[i][:50]!r}"
