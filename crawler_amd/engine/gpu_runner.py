@@ -99,6 +99,9 @@ class GpuCrawlEngine:
                 last = (k + 1) * P - 1
                 hi = int(line_off[last] + line_len[last])
                 if hi > lo:
+                    # truncate-then-write = exactly-once per channel even
+                    # when a crash forces the layer to re-process
+                    self.sm.truncate_posts(uname)
                     # zero-copy slice: file.write accepts the memoryview
                     self.sm.store_post_lines(uname, buf[lo:hi])
                 n_lines = int((line_len[k * P:(k + 1) * P] > 0).sum())

@@ -599,6 +599,22 @@ class LocalStateManager(BaseStateManager):
             return
         self._post_file(channel).write(data)
 
+    def truncate_posts(self, channel: str) -> None:
+        """Idempotent re-crawl support for the channel-atomic GPU path: a
+        crash between a channel's JSONL write and the layer's save_state
+        leaves the page 'unfetched', so resume re-processes it — truncate
+        before the re-write so posts are exactly-once at channel
+        granularity (the CPU path gets the same via per-message
+        update_message, crawl/runner.go:1572-1633)."""
+        if self.post_sink is not None:
+            return  # combine-files mode writes unique temp names
+        path = self._posts_path(channel)
+        f = self._post_files.pop(path, None)
+        if f is not None:
+            f.close()
+        if os.path.exists(path):
+            open(path, "wb").close()
+
     def store_file(self, channel: str, source_path: str,
                    file_name: str) -> Tuple[str, str]:
         dst_dir = os.path.join(self._crawl_dir(), "media", channel)
