@@ -174,9 +174,6 @@ def test_crash_resume_does_not_duplicate_posts(tmp_path):
     name = "c%010d" % 21
     eng.process_channels([name], now=NOW)
     path = tmp_path / "g1" / name / "posts" / "posts.jsonl"
-    sm._post_files[str(path)].flush() if str(path) in sm._post_files \
-        else None
-    first = None
     # crash simulation: process the same channel again (as a layer
     # re-run after restart would)
     eng.process_channels([name], now=NOW)
