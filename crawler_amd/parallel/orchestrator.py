@@ -46,7 +46,9 @@ class Orchestrator:
         self.page_of_item: Dict[str, str] = {}
         self.done = False
         self.stats = {"distributed": 0, "results": 0, "reassigned": 0,
-                      "errors": 0}
+                      "errors": 0, "retried": 0}
+
+    MAX_RETRIES = 3  # matches RandomWalkStore.MAX_ATTEMPTS poison cap
 
     # ---- distribution (distributeWork, :182-277) ----
 
@@ -92,13 +94,26 @@ class Orchestrator:
             n += 1
             self.stats["results"] += 1
             entry = self.in_flight.pop(res.work_item_id, None)
+            if res.status == M.STATUS_RETRY and entry is not None:
+                # transient failure (FLOOD_WAIT / timeout): republish at
+                # high priority up to MAX_RETRIES, page stays 'processing'
+                # (the reference gets this via pub/sub NACK redelivery,
+                # worker/worker.go:436 + orchestrator reassignment)
+                item = entry[0]
+                if item.retry_count < self.MAX_RETRIES:
+                    item.retry_count += 1
+                    item.priority = M.PRIORITY_HIGH
+                    self.work_q.publish(item.to_json())
+                    self.in_flight[item.id] = (item, self.clock())
+                    self.stats["retried"] += 1
+                    continue
             page_id = self.page_of_item.pop(res.work_item_id, None)
             if page_id is not None:
                 page = self.sm.get_page(page_id)
                 page.status = res.page_status
                 page.error = res.error
                 self.sm.update_page(page)
-            if res.status == M.STATUS_ERROR:
+            if res.status in (M.STATUS_ERROR, M.STATUS_RETRY):
                 self.stats["errors"] += 1
             if (res.discovered
                     and self.cfg.sampling_method == "snowball"

@@ -257,3 +257,74 @@ def test_work_item_config_propagates_to_worker(tmp_path):
     res = w.process_item(item)
     assert res.status == M.STATUS_SUCCESS
     assert res.posts_stored == 5  # capped by the item config, not base cfg
+
+
+def test_retry_results_republished_then_succeed(tmp_path):
+    """A transient (FLOOD_WAIT-class) failure republishes the item at
+    high priority; the page completes on the retry, not as an error."""
+    cfg, orch, workers, store = mk_dist_env(tmp_path, n_workers=1,
+                                            sampling="channel")
+    orch.sm.initialize(["c0000000001"])
+    w = workers[0]
+    fail_once = {"left": 1}
+    real = w.process_item
+
+    def flaky(item):
+        if fail_once["left"] > 0:
+            fail_once["left"] -= 1
+            return M.WorkResult(
+                work_item_id=item.id, worker_id=w.worker_id,
+                status=M.STATUS_RETRY, error="[429] FLOOD_WAIT_5",
+                page_status="error", trace_id=item.trace_id,
+            )
+        return real(item)
+
+    w.process_item = flaky
+    drive(orch, workers)
+    assert orch.done
+    assert orch.stats["retried"] == 1
+    assert orch.stats["errors"] == 0
+    page = [p for p in orch.sm.pages.values()][0]
+    assert page.status == "fetched"
+
+
+def test_retry_exhaustion_marks_error(tmp_path):
+    cfg, orch, workers, store = mk_dist_env(tmp_path, n_workers=1,
+                                            sampling="channel")
+    orch.sm.initialize(["c0000000001"])
+    w = workers[0]
+
+    def always_retry(item):
+        return M.WorkResult(
+            work_item_id=item.id, worker_id=w.worker_id,
+            status=M.STATUS_RETRY, error="[429] FLOOD_WAIT_5",
+            page_status="error", trace_id=item.trace_id,
+        )
+
+    w.process_item = always_retry
+    drive(orch, workers)
+    assert orch.done
+    assert orch.stats["retried"] == orch.MAX_RETRIES
+    assert orch.stats["errors"] == 1
+    page = [p for p in orch.sm.pages.values()][0]
+    assert page.status == "error"
+
+
+def test_depth_advance_waits_for_processing_pages(tmp_path):
+    """_maybe_advance must not advance or complete while any page of the
+    current layer is processing (orchestrator.go:315-383 barrier)."""
+    cfg, orch, workers, store = mk_dist_env(tmp_path, n_workers=1)
+    orch.sm.initialize(["c0000000001", "c0000000002"])
+    orch.distribute()
+    # worker processes only ONE of the two queued items
+    workers[0].run_once(timeout_s=0.0)
+    orch.pump_results()
+    assert not orch.done
+    assert orch.current_depth == 0
+    # finish the second
+    workers[0].run_once(timeout_s=0.0)
+    orch.pump_results()
+    # both done; layer 0 complete -> advance (snowball found channels)
+    assert orch.current_depth in (0, 1)  # advances iff a next layer exists
+    if orch.current_depth == 0:
+        assert orch.done
