@@ -94,6 +94,7 @@ class GpuCrawlEngine:
             torch.cuda.synchronize()
             buf = memoryview(out_host.numpy())
             P = self.ppc
+            items = []
             for k, (uname, _cid) in enumerate(chunk):
                 lo = int(line_off[k * P])
                 last = (k + 1) * P - 1
@@ -102,11 +103,14 @@ class GpuCrawlEngine:
                     # truncate-then-write = exactly-once per channel even
                     # when a crash forces the layer to re-process
                     self.sm.truncate_posts(uname)
-                    # zero-copy slice: file.write accepts the memoryview
-                    self.sm.store_post_lines(uname, buf[lo:hi])
+                    items.append((uname, lo, hi))
                 n_lines = int((line_len[k * P:(k + 1) * P] > 0).sum())
                 posts_total += n_lines
                 self.stats["pages"] += 1
+            # one fan-out call for the whole chunk: the native sink
+            # (crawler_amd/native) appends all channels in parallel with
+            # the GIL released; pure-Python fallback writes sequentially
+            self.sm.store_post_lines_batch(items, buf)
             self.stats["jsonl_bytes"] += int(buf.shape[0])
 
             # newly-claimed names (small: first-discovery rows only)

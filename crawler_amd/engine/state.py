@@ -599,6 +599,63 @@ class LocalStateManager(BaseStateManager):
             return
         self._post_file(channel).write(data)
 
+    _native_sink = None
+    _native_tried = False
+
+    def _get_native_sink(self):
+        """crawler_amd.native.fanout_native thread-pool writer (the
+        runtime-native spill path); None when the extension is absent or
+        disabled (CRAWL_NO_NATIVE_SINK=1)."""
+        if self._native_tried:
+            return self._native_sink
+        self._native_tried = True
+        if os.environ.get("CRAWL_NO_NATIVE_SINK", "") == "1":
+            return None
+        try:
+            from ..native import load
+
+            self._native_sink = load().FanoutSink(
+                8, self.max_open_post_files
+            )
+        except ImportError:
+            self._native_sink = None
+        return self._native_sink
+
+    def store_post_lines_batch(self, items, buffer) -> None:
+        """Fan out one encoded host buffer to many channels' JSONL files:
+        items = [(channel, lo, hi)]. Uses the native thread-pool sink
+        when built (parallel write(2), GIL released); falls back to the
+        per-channel Python path otherwise. Combine-files mode keeps the
+        chunker protocol."""
+        if self.post_sink is not None:
+            mv = memoryview(buffer)
+            for (channel, lo, hi) in items:
+                if hi > lo:
+                    self._sink_write(channel, mv[lo:hi])
+            return
+        sink = self._get_native_sink()
+        if sink is None:
+            mv = memoryview(buffer)
+            for (channel, lo, hi) in items:
+                if hi > lo:
+                    self.store_post_lines(channel, mv[lo:hi])
+            return
+        paths, los, his = [], [], []
+        for (channel, lo, hi) in items:
+            if hi <= lo:
+                continue
+            path = self._posts_path(channel)
+            # never interleave a buffered Python handle with the O_APPEND
+            # fd: flush+drop the Python handle first
+            f = self._post_files.pop(path, None)
+            if f is not None:
+                f.close()
+            paths.append(path)
+            los.append(lo)
+            his.append(hi)
+        if paths:
+            sink.write_batch(paths, buffer, los, his)
+
     def truncate_posts(self, channel: str) -> None:
         """Idempotent re-crawl support for the channel-atomic GPU path: a
         crash between a channel's JSONL write and the layer's save_state
@@ -678,6 +735,10 @@ class LocalStateManager(BaseStateManager):
         for f in self._post_files.values():
             f.close()
         self._post_files.clear()
+        if self._native_sink is not None:
+            self._native_sink.close()
+            self._native_sink = None
+            self._native_tried = False
         os.makedirs(self._crawl_dir(), exist_ok=True)
         with open(self._media_cache_path() + ".tmp", "w") as f:
             json.dump(self.media_cache, f)

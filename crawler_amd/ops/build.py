@@ -36,6 +36,39 @@ def build(force: bool = False) -> str:
     return OUT
 
 
+NATIVE_DIR = os.path.join(
+    os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "native"
+)
+NATIVE_SRC = os.path.join(NATIVE_DIR, "fanout_sink.cc")
+
+
+def _native_out() -> str:
+    import sysconfig
+
+    suffix = sysconfig.get_config_var("EXT_SUFFIX") or ".so"
+    return os.path.join(NATIVE_DIR, "fanout_native" + suffix)
+
+
+def build_native(force: bool = False) -> str:
+    """Plain-C++ runtime extension (no GPU dependency): g++ + pybind11."""
+    out = _native_out()
+    if (not force and os.path.exists(out)
+            and os.path.getmtime(out) >= os.path.getmtime(NATIVE_SRC)):
+        return out
+    import pybind11
+    import sysconfig
+
+    py_inc = sysconfig.get_paths()["include"]
+    cmd = [
+        "g++", "-O2", "-std=c++17", "-fPIC", "-shared", "-pthread",
+        f"-I{py_inc}", f"-I{pybind11.get_include()}",
+        NATIVE_SRC, "-o", out,
+    ]
+    subprocess.run(cmd, check=True)
+    return out
+
+
 if __name__ == "__main__":
     path = build(force="--force" in sys.argv)
     print(path)
+    print(build_native(force="--force" in sys.argv))
