@@ -46,6 +46,7 @@ class GpuCrawlEngine:
         self.ppc = posts_per_channel or feed.cfg.posts_per_channel
         self.chunk_channels = chunk_channels
         self.seen = gpu_mod.SeenSet(self.device)
+        self._spill_inflight = []  # host buffers of unfinished writes
         self.stats = {"pages": 0, "posts": 0, "jsonl_bytes": 0,
                       "discovered": 0, "deadends": 0}
 
@@ -109,8 +110,14 @@ class GpuCrawlEngine:
                 self.stats["pages"] += 1
             # one fan-out call for the whole chunk: the native sink
             # (crawler_amd/native) appends all channels in parallel with
-            # the GIL released; pure-Python fallback writes sequentially
-            self.sm.store_post_lines_batch(items, buf)
+            # the GIL released. nowait + 1-deep pipeline: chunk i's disk
+            # writes overlap chunk i+1's feed-gen and kernels; the
+            # buffer stays referenced until the next drain.
+            if self._spill_inflight:
+                self.sm.drain_post_writes()
+                self._spill_inflight.clear()
+            self.sm.store_post_lines_batch(items, buf, nowait=True)
+            self._spill_inflight.append(out_host)
             self.stats["jsonl_bytes"] += int(buf.shape[0])
 
             # newly-claimed names (small: first-discovery rows only)
@@ -124,6 +131,9 @@ class GpuCrawlEngine:
                     discovered.append(bytes(b[:ln]).decode())
         self.stats["posts"] += posts_total
         self.stats["discovered"] += len(discovered)
+        # barrier: all spill writes down before the layer's save_state
+        self.sm.drain_post_writes()
+        self._spill_inflight.clear()
         return discovered, posts_total
 
     # ---- BFS crawl (snowball / channel) ----

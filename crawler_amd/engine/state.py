@@ -621,12 +621,14 @@ class LocalStateManager(BaseStateManager):
             self._native_sink = None
         return self._native_sink
 
-    def store_post_lines_batch(self, items, buffer) -> None:
+    def store_post_lines_batch(self, items, buffer,
+                               nowait: bool = False) -> None:
         """Fan out one encoded host buffer to many channels' JSONL files:
         items = [(channel, lo, hi)]. Uses the native thread-pool sink
         when built (parallel write(2), GIL released); falls back to the
         per-channel Python path otherwise. Combine-files mode keeps the
-        chunker protocol."""
+        chunker protocol. nowait=True (native only) returns before the
+        writes land — hold `buffer` alive until drain_post_writes()."""
         if self.post_sink is not None:
             mv = memoryview(buffer)
             for (channel, lo, hi) in items:
@@ -653,8 +655,18 @@ class LocalStateManager(BaseStateManager):
             paths.append(path)
             los.append(lo)
             his.append(hi)
-        if paths:
+        if not paths:
+            return
+        if nowait:
+            # caller MUST keep `buffer` alive until drain_post_writes()
+            sink.write_batch_nowait(paths, buffer, los, his)
+        else:
             sink.write_batch(paths, buffer, los, his)
+
+    def drain_post_writes(self) -> None:
+        """Barrier for store_post_lines_batch(..., nowait=True) writes."""
+        if self._native_sink is not None:
+            self._native_sink.drain()
 
     def truncate_posts(self, channel: str) -> None:
         """Idempotent re-crawl support for the channel-atomic GPU path: a

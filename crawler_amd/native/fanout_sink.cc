@@ -106,6 +106,54 @@ class FanoutSink {
           std::to_string(errors_.load() - errors_before));
   }
 
+  // Fire-and-forget variant: enqueue and return immediately. The CALLER
+  // must keep `buf` alive until drain() returns (the GPU engine holds
+  // the pinned host buffers of in-flight chunks and drains every few
+  // chunks, overlapping disk writes with the next chunk's kernels).
+  void write_batch_nowait(const std::vector<std::string>& paths,
+                          py::buffer buf, const std::vector<size_t>& lo,
+                          const std::vector<size_t>& hi) {
+    if (paths.size() != lo.size() || paths.size() != hi.size())
+      throw std::invalid_argument("paths/lo/hi length mismatch");
+    py::buffer_info info = buf.request();
+    if (info.ndim != 1 || info.itemsize != 1)
+      throw std::invalid_argument("buffer must be 1-D bytes");
+    const char* base = static_cast<const char*>(info.ptr);
+    size_t n_bytes = static_cast<size_t>(info.size);
+    std::vector<Task> tasks;
+    tasks.reserve(paths.size());
+    for (size_t i = 0; i < paths.size(); ++i) {
+      if (hi[i] < lo[i] || hi[i] > n_bytes)
+        throw std::out_of_range("slice outside buffer");
+      if (hi[i] == lo[i]) continue;
+      int fd = fd_for(paths[i]);
+      tasks.push_back(Task{fd, base + lo[i], hi[i] - lo[i]});
+    }
+    {
+      std::unique_lock<std::mutex> lk(mu_);
+      for (auto& t : tasks) q_.push(t);
+      pending_ += tasks.size();
+    }
+    cv_.notify_all();
+    // NOTE: no eviction here — fds stay open until the next drain()
+  }
+
+  // Wait for all queued writes; then evict down to the fd cap. Raises if
+  // any write since the last drain failed.
+  void drain() {
+    size_t errs;
+    {
+      py::gil_scoped_release rel;
+      std::unique_lock<std::mutex> lk(mu_);
+      done_cv_.wait(lk, [this] { return pending_ == 0; });
+      errs = errors_.exchange(0);
+    }
+    evict_to_cap();
+    if (errs != 0)
+      throw std::runtime_error("fanout sink: write(2) failures: " +
+                               std::to_string(errs));
+  }
+
   void flush() {
     py::gil_scoped_release rel;
     std::unique_lock<std::mutex> lk(mu_);
@@ -221,6 +269,8 @@ PYBIND11_MODULE(fanout_native, m) {
       .def(py::init<int, int>(), py::arg("n_threads") = 4,
            py::arg("max_open") = 256)
       .def("write_batch", &FanoutSink::write_batch)
+      .def("write_batch_nowait", &FanoutSink::write_batch_nowait)
+      .def("drain", &FanoutSink::drain)
       .def("flush", &FanoutSink::flush)
       .def("close", &FanoutSink::close)
       .def_property_readonly("bytes_written", &FanoutSink::bytes_written)

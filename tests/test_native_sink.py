@@ -123,3 +123,31 @@ def test_bad_slice_raises(tmp_path):
         sink.write_batch([str(tmp_path / "x")], memoryview(buf),
                          [0], [99])
     sink.close()
+
+
+def test_nowait_then_drain(tmp_path):
+    buf, bounds = mk_buffer(n_channels=6)
+    channels = [f"chan{c}" for c in range(6)]
+    cfg = CrawlerConfig(crawl_id="n5", storage_root=str(tmp_path))
+    sm = LocalStateManager(cfg)
+    items = [(ch, lo, hi) for ch, (lo, hi) in zip(channels, bounds)]
+    sm.store_post_lines_batch(items, memoryview(buf), nowait=True)
+    sm.drain_post_writes()
+    sm.store_post_lines_batch(items, memoryview(buf), nowait=True)
+    sm.drain_post_writes()
+    sm.close()
+    for ch, (lo, hi) in zip(channels, bounds):
+        data = (tmp_path / "n5" / ch / "posts" / "posts.jsonl").read_bytes()
+        assert data == bytes(buf[lo:hi]) * 2
+
+
+def test_open_failure_raises_immediately(tmp_path):
+    """Opening a path that is a DIRECTORY fails at enqueue time (both
+    write_batch and write_batch_nowait open fds synchronously)."""
+    sink = native.FanoutSink(2, 4)
+    buf = np.frombuffer(b"x" * 10, dtype=np.uint8)
+    ro_dir = tmp_path / "ro"
+    ro_dir.mkdir()
+    with pytest.raises(Exception):
+        sink.write_batch_nowait([str(ro_dir)], memoryview(buf), [0], [10])
+    sink.close()
