@@ -120,15 +120,22 @@ class GpuCrawlEngine:
             self._spill_inflight.append(out_host)
             self.stats["jsonl_bytes"] += int(buf.shape[0])
 
-            # newly-claimed names (small: first-discovery rows only)
+            # newly-claimed names (first-discovery rows only). One bulk
+            # D2H + ONE ascii decode of the packed name block, then
+            # string slices — a per-name bytes().decode() loop costs
+            # seconds at ~1M discoveries per crawl.
             nz = new_mask.nonzero()
             if nz.numel():
                 rows = nz[:, 0]
                 cols = nz[:, 1]
                 names = res.link_name[rows, cols].cpu().numpy()
                 lens = res.link_len[rows, cols].cpu().numpy()
-                for b, ln in zip(names, lens):
-                    discovered.append(bytes(b[:ln]).decode())
+                w = names.shape[1]
+                blob = names.tobytes().decode("ascii", "replace")
+                discovered.extend(
+                    blob[i * w:i * w + ln]
+                    for i, ln in enumerate(lens.tolist())
+                )
         self.stats["posts"] += posts_total
         self.stats["discovered"] += len(discovered)
         # barrier: all spill writes down before the layer's save_state
