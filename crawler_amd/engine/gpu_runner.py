@@ -184,8 +184,30 @@ class GpuCrawlEngine:
                     and (self.cfg.max_depth < 0
                          or depth < self.cfg.max_depth)
                     and discovered):
+                # Pre-apply add_layer's own skip rules (URL dedup +
+                # MaxPages/deadend budget) BEFORE constructing Page
+                # objects: a dense crawl discovers ~1M names per layer
+                # and the cap keeps ~1k — building 1M dataclasses to
+                # throw them away costs seconds.
+                max_pages = getattr(self.cfg, "max_pages", 0) or 0
+                budget = None
+                if max_pages > 0:
+                    total = len(sm.pages)
+                    deadends = sum(1 for p in sm.pages.values()
+                                   if p.status == "deadend")
+                    budget = max(0, max_pages - total) + deadends
+                existing = {p.url for p in sm.pages.values()}
+                url_dedup = getattr(sm, "url_dedup", {})
+                cand = []
+                for n in sorted(set(discovered)):
+                    if budget is not None and len(cand) >= budget:
+                        break
+                    if n in existing or n in url_dedup:
+                        continue
+                    existing.add(n)
+                    cand.append(n)
                 pages = [Page(url=n, depth=depth + 1, status="unfetched")
-                         for n in sorted(set(discovered))]
+                         for n in cand]
                 sm.add_layer(pages)
             sm.save_state()
             if self.cfg.sampling_method == "channel":
