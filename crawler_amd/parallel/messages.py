@@ -113,6 +113,9 @@ class WorkResult:
     page_status: str = "fetched"   # fetched|error|deadend
     posts_stored: int = 0
     discovered: List[str] = dataclasses.field(default_factory=list)
+    # count-only form for paths whose next layer travels out-of-band
+    # (OrchestratedCrawl all-gathers names; ~1M names as JSON is waste)
+    discovered_count: int = 0
     duration_ms: float = 0.0
     trace_id: str = ""
 

@@ -79,11 +79,15 @@ class OrchestratedCrawl:
         out: List[str] = []
         for r, g in enumerate(gathered):
             cnt = int(all_counts[r].item())
-            arr = g.numpy()
-            for i in range(cnt):
-                row = arr[i]
-                ln = int((row != 0).sum())
-                out.append(bytes(row[:ln]).decode())
+            arr = g.numpy()[:cnt]
+            if not cnt:
+                continue
+            lens = (arr != 0).sum(axis=1)
+            # one decode of the packed block, then slices (a per-row
+            # bytes().decode() loop costs seconds at ~1M names)
+            blob = arr.tobytes().decode("ascii", "replace")
+            out.extend(blob[i * width:i * width + int(l)]
+                       for i, l in enumerate(lens))
         return out
 
     # ---- one layer ----
@@ -116,9 +120,11 @@ class OrchestratedCrawl:
             self.stats["chunks"] += 1
             self.stats["pages"] += len(chans)
             self.stats["posts"] += posts
+            # bookkeeping only: the next layer comes from the all-gather,
+            # so publish the COUNT, not ~1M names as JSON
             self.result_q.publish(M.WorkResult(
                 work_item_id=item.id, worker_id=f"rank{self.rank}",
-                posts_stored=posts, discovered=discovered,
+                posts_stored=posts, discovered_count=len(discovered),
                 trace_id=item.trace_id,
             ).to_json())
         self.heartbeats.beat(f"rank{self.rank}", M.WORKER_IDLE)
@@ -150,9 +156,29 @@ class OrchestratedCrawl:
                          or depth < self.cfg.max_depth)
                     and all_discovered):
                 # every rank derives the SAME next layer (sorted set ->
-                # deterministic add_layer admission order)
+                # deterministic admission order). Pre-apply add_layer's
+                # URL-dedup + MaxPages budget before building Page
+                # objects (same rule as gpu_runner.run) — deterministic,
+                # so all ranks still agree.
+                max_pages = getattr(self.cfg, "max_pages", 0) or 0
+                budget = None
+                if max_pages > 0:
+                    total = len(sm.pages)
+                    deadends = sum(1 for p in sm.pages.values()
+                                   if p.status == "deadend")
+                    budget = max(0, max_pages - total) + deadends
+                existing = {p.url for p in sm.pages.values()}
+                url_dedup = getattr(sm, "url_dedup", {})
+                cand = []
+                for n in sorted(set(all_discovered)):
+                    if budget is not None and len(cand) >= budget:
+                        break
+                    if n in existing or n in url_dedup:
+                        continue
+                    existing.add(n)
+                    cand.append(n)
                 pages = [Page(url=n, depth=depth + 1, status="unfetched")
-                         for n in sorted(set(all_discovered))]
+                         for n in cand]
                 sm.add_layer(pages)
             if self.rank == 0:
                 sm.save_state()
