@@ -47,6 +47,8 @@ class GpuCrawlEngine:
         self.chunk_channels = chunk_channels
         self.seen = gpu_mod.SeenSet(self.device)
         self._spill_inflight = []  # host buffers of unfinished writes
+        import collections
+        self.timings = collections.defaultdict(float)  # phase seconds
         self.stats = {"pages": 0, "posts": 0, "jsonl_bytes": 0,
                       "discovered": 0, "deadends": 0}
 
@@ -67,6 +69,8 @@ class GpuCrawlEngine:
         now = now or _dt.datetime.now(_dt.timezone.utc)
         discovered: List[str] = []
         posts_total = 0
+        t = self.timings
+        import time as _time
         valid = [(u, self._cid_of(u)) for u in usernames]
         bad = [u for u, c in valid if c is None]
         ok = [(u, c) for u, c in valid if c is not None]
@@ -75,6 +79,7 @@ class GpuCrawlEngine:
         for i in range(0, len(ok), self.chunk_channels):
             chunk = ok[i:i + self.chunk_channels]
             cids = np.array([c for _u, c in chunk], dtype=np.int64)
+            t0 = _time.perf_counter()
             batch = self.feed.build_batch_device(
                 cids, self.device, posts_per_channel=self.ppc
             )
@@ -83,6 +88,7 @@ class GpuCrawlEngine:
             )
             new_mask = self.seen.claim(res)
             torch.cuda.synchronize()
+            t["gen+kernels"] += _time.perf_counter() - t0; t0 = _time.perf_counter()
 
             # host spill: per-channel JSONL slices. Messages are grouped
             # by channel (K x P layout), so channel k owns lines
@@ -93,6 +99,7 @@ class GpuCrawlEngine:
             line_off = res.line_off.cpu().numpy()
             line_len = res.line_len.cpu().numpy()
             torch.cuda.synchronize()
+            t["d2h"] += _time.perf_counter() - t0; t0 = _time.perf_counter()
             buf = memoryview(out_host.numpy())
             P = self.ppc
             items = []
@@ -119,6 +126,7 @@ class GpuCrawlEngine:
             self.sm.store_post_lines_batch(items, buf, nowait=True)
             self._spill_inflight.append(out_host)
             self.stats["jsonl_bytes"] += int(buf.shape[0])
+            t["spill"] += _time.perf_counter() - t0; t0 = _time.perf_counter()
 
             # newly-claimed names (first-discovery rows only). One bulk
             # D2H + ONE ascii decode of the packed name block, then
@@ -136,6 +144,7 @@ class GpuCrawlEngine:
                     blob[i * w:i * w + ln]
                     for i, ln in enumerate(lens.tolist())
                 )
+            t["names"] += _time.perf_counter() - t0
         self.stats["posts"] += posts_total
         self.stats["discovered"] += len(discovered)
         # barrier: all spill writes down before the layer's save_state
@@ -173,7 +182,9 @@ class GpuCrawlEngine:
                 mine = names[rank::world]
             else:
                 mine = names
+            import time as _time
             discovered, _ = self.process_channels(mine)
+            t0 = _time.perf_counter()
             for p in layer:
                 p.status = "fetched"
                 sm.update_page(p)
@@ -209,7 +220,10 @@ class GpuCrawlEngine:
                 pages = [Page(url=n, depth=depth + 1, status="unfetched")
                          for n in cand]
                 sm.add_layer(pages)
+            self.timings["layer-build"] += _time.perf_counter() - t0
+            t0 = _time.perf_counter()
             sm.save_state()
+            self.timings["save_state"] += _time.perf_counter() - t0
             if self.cfg.sampling_method == "channel":
                 break
             if self.cfg.max_depth >= 0 and depth >= self.cfg.max_depth:

@@ -58,6 +58,8 @@ def main():
         "deadends": stats["deadends"],
         "max_depth": args.max_depth,
     }
+    if getattr(eng, "timings", None):
+        out["phase_s"] = {k: round(v, 3) for k, v in eng.timings.items()}
     print(json.dumps(out))
 
 
