@@ -46,7 +46,8 @@ class GpuCrawlEngine:
         self.ppc = posts_per_channel or feed.cfg.posts_per_channel
         self.chunk_channels = chunk_channels
         self.seen = gpu_mod.SeenSet(self.device)
-        self._spill_inflight = []  # host buffers of unfinished writes
+        self._pin_ring = [None, None]      # reusable pinned host slots
+        self._spill_tickets = [None, None]  # sink ticket per ring slot
         import collections
         self.timings = collections.defaultdict(float)  # phase seconds
         self.stats = {"pages": 0, "posts": 0, "jsonl_bytes": 0,
@@ -76,7 +77,7 @@ class GpuCrawlEngine:
         ok = [(u, c) for u, c in valid if c is not None]
         for u in bad:
             self.stats["deadends"] += 1
-        for i in range(0, len(ok), self.chunk_channels):
+        for ci, i in enumerate(range(0, len(ok), self.chunk_channels)):
             chunk = ok[i:i + self.chunk_channels]
             cids = np.array([c for _u, c in chunk], dtype=np.int64)
             t0 = _time.perf_counter()
@@ -93,8 +94,21 @@ class GpuCrawlEngine:
             # host spill: per-channel JSONL slices. Messages are grouped
             # by channel (K x P layout), so channel k owns lines
             # [k*P, (k+1)*P) -> bytes [line_off[kP], line_off[(k+1)P-1]+len).
-            out_host = torch.empty_like(res.out, device="cpu",
-                                        pin_memory=True)
+            # The pinned HOST RING (2 slots, allocated once) replaces a
+            # per-chunk 1GB pin_memory allocation (which cost ~0.1s/chunk
+            # — the d2h phase was allocation-bound, not copy-bound); the
+            # slot's previous SINK TICKET is awaited before the D2H may
+            # overwrite it, making the spill pipeline 2 chunks deep.
+            slot = ci % 2
+            self.sm.wait_post_write(self._spill_tickets[slot])
+            self._spill_tickets[slot] = None
+            need = int(res.out.numel())
+            ring = self._pin_ring
+            if ring[slot] is None or ring[slot].numel() < need:
+                ring[slot] = torch.empty(need + need // 8,
+                                         dtype=torch.uint8,
+                                         pin_memory=True)
+            out_host = ring[slot][:need]
             out_host.copy_(res.out, non_blocking=True)
             line_off = res.line_off.cpu().numpy()
             line_len = res.line_len.cpu().numpy()
@@ -117,14 +131,12 @@ class GpuCrawlEngine:
                 self.stats["pages"] += 1
             # one fan-out call for the whole chunk: the native sink
             # (crawler_amd/native) appends all channels in parallel with
-            # the GIL released. nowait + 1-deep pipeline: chunk i's disk
-            # writes overlap chunk i+1's feed-gen and kernels; the
-            # buffer stays referenced until the next drain.
-            if self._spill_inflight:
-                self.sm.drain_post_writes()
-                self._spill_inflight.clear()
-            self.sm.store_post_lines_batch(items, buf, nowait=True)
-            self._spill_inflight.append(out_host)
+            # the GIL released; the ticket is awaited when this ring slot
+            # comes around again (2-deep: chunk i's disk writes run under
+            # chunks i+1 AND i+2's GPU work).
+            self._spill_tickets[slot] = self.sm.store_post_lines_batch(
+                items, buf, ticket=True
+            )
             self.stats["jsonl_bytes"] += int(buf.shape[0])
             t["spill"] += _time.perf_counter() - t0; t0 = _time.perf_counter()
 
@@ -148,8 +160,10 @@ class GpuCrawlEngine:
         self.stats["posts"] += posts_total
         self.stats["discovered"] += len(discovered)
         # barrier: all spill writes down before the layer's save_state
+        for slot in (0, 1):
+            self.sm.wait_post_write(self._spill_tickets[slot])
+            self._spill_tickets[slot] = None
         self.sm.drain_post_writes()
-        self._spill_inflight.clear()
         return discovered, posts_total
 
     # ---- BFS crawl (snowball / channel) ----

@@ -621,8 +621,8 @@ class LocalStateManager(BaseStateManager):
             self._native_sink = None
         return self._native_sink
 
-    def store_post_lines_batch(self, items, buffer,
-                               nowait: bool = False) -> None:
+    def store_post_lines_batch(self, items, buffer, nowait: bool = False,
+                               ticket: bool = False):
         """Fan out one encoded host buffer to many channels' JSONL files:
         items = [(channel, lo, hi)]. Uses the native thread-pool sink
         when built (parallel write(2), GIL released); falls back to the
@@ -656,12 +656,22 @@ class LocalStateManager(BaseStateManager):
             los.append(lo)
             his.append(hi)
         if not paths:
-            return
+            return None
+        if ticket:
+            # caller MUST keep `buffer` alive until wait_post_write()
+            return sink.write_batch_ticket(paths, buffer, los, his)
         if nowait:
             # caller MUST keep `buffer` alive until drain_post_writes()
             sink.write_batch_nowait(paths, buffer, los, his)
         else:
             sink.write_batch(paths, buffer, los, his)
+        return None
+
+    def wait_post_write(self, ticket) -> None:
+        """Barrier for ONE store_post_lines_batch(..., ticket=True) batch
+        (None tickets — sync fallback paths — are a no-op)."""
+        if ticket is not None and self._native_sink is not None:
+            self._native_sink.wait_ticket(ticket)
 
     def drain_post_writes(self) -> None:
         """Barrier for store_post_lines_batch(..., nowait=True) writes."""

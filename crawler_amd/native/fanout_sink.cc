@@ -50,6 +50,7 @@ struct Task {
   int fd;
   const char* data;
   size_t len;
+  uint64_t ticket;  // 0 = untracked (write_batch/write_batch_nowait)
 };
 
 class FanoutSink {
@@ -136,6 +137,61 @@ class FanoutSink {
     }
     cv_.notify_all();
     // NOTE: no eviction here — fds stay open until the next drain()
+  }
+
+  // Ticketed variant: like write_batch_nowait but returns a ticket that
+  // wait_ticket() blocks on — lets a caller pipeline SEVERAL batches and
+  // wait for the oldest only (the GPU engine runs the spill 2-deep).
+  uint64_t write_batch_ticket(const std::vector<std::string>& paths,
+                              py::buffer buf,
+                              const std::vector<size_t>& lo,
+                              const std::vector<size_t>& hi) {
+    if (paths.size() != lo.size() || paths.size() != hi.size())
+      throw std::invalid_argument("paths/lo/hi length mismatch");
+    py::buffer_info info = buf.request();
+    if (info.ndim != 1 || info.itemsize != 1)
+      throw std::invalid_argument("buffer must be 1-D bytes");
+    const char* base = static_cast<const char*>(info.ptr);
+    size_t n_bytes = static_cast<size_t>(info.size);
+    uint64_t ticket = ++ticket_seq_;
+    std::vector<Task> tasks;
+    tasks.reserve(paths.size());
+    for (size_t i = 0; i < paths.size(); ++i) {
+      if (hi[i] < lo[i] || hi[i] > n_bytes)
+        throw std::out_of_range("slice outside buffer");
+      if (hi[i] == lo[i]) continue;
+      int fd = fd_for(paths[i]);
+      tasks.push_back(Task{fd, base + lo[i], hi[i] - lo[i], ticket});
+    }
+    {
+      std::unique_lock<std::mutex> lk(mu_);
+      ticket_pending_[ticket] = tasks.size();
+      ticket_errors_[ticket] = 0;
+      for (auto& t : tasks) q_.push(t);
+      pending_ += tasks.size();
+    }
+    cv_.notify_all();
+    return ticket;
+  }
+
+  void wait_ticket(uint64_t ticket) {
+    size_t errs;
+    {
+      py::gil_scoped_release rel;
+      std::unique_lock<std::mutex> lk(mu_);
+      done_cv_.wait(lk, [this, ticket] {
+        auto it = ticket_pending_.find(ticket);
+        return it == ticket_pending_.end() || it->second == 0;
+      });
+      errs = ticket_errors_[ticket];
+      ticket_pending_.erase(ticket);
+      ticket_errors_.erase(ticket);
+    }
+    // NOTE: no eviction — other tickets may still be in flight; the
+    // periodic drain() (layer barrier) evicts.
+    if (errs != 0)
+      throw std::runtime_error("fanout sink: write(2) failures: " +
+                               std::to_string(errs));
   }
 
   // Wait for all queued writes; then evict down to the fd cap. Raises if
@@ -243,7 +299,14 @@ class FanoutSink {
         bytes_.fetch_add(t.len);
       {
         std::unique_lock<std::mutex> lk(mu_);
-        if (--pending_ == 0) done_cv_.notify_all();
+        if (t.ticket != 0) {
+          auto it = ticket_pending_.find(t.ticket);
+          if (it != ticket_pending_.end() && it->second > 0) --it->second;
+          if (!ok) ++ticket_errors_[t.ticket];
+        }
+        if (--pending_ == 0 ||
+            (t.ticket != 0 && ticket_pending_[t.ticket] == 0))
+          done_cv_.notify_all();
       }
     }
   }
@@ -251,6 +314,9 @@ class FanoutSink {
   int max_open_;
   bool stop_;
   size_t pending_;
+  std::atomic<uint64_t> ticket_seq_{0};
+  std::unordered_map<uint64_t, size_t> ticket_pending_;   // under mu_
+  std::unordered_map<uint64_t, size_t> ticket_errors_;    // under mu_
   std::atomic<size_t> errors_;
   std::atomic<size_t> bytes_;
   std::unordered_map<std::string, int> fds_;
@@ -270,6 +336,8 @@ PYBIND11_MODULE(fanout_native, m) {
            py::arg("max_open") = 256)
       .def("write_batch", &FanoutSink::write_batch)
       .def("write_batch_nowait", &FanoutSink::write_batch_nowait)
+      .def("write_batch_ticket", &FanoutSink::write_batch_ticket)
+      .def("wait_ticket", &FanoutSink::wait_ticket)
       .def("drain", &FanoutSink::drain)
       .def("flush", &FanoutSink::flush)
       .def("close", &FanoutSink::close)

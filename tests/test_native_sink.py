@@ -151,3 +151,40 @@ def test_open_failure_raises_immediately(tmp_path):
     with pytest.raises(Exception):
         sink.write_batch_nowait([str(ro_dir)], memoryview(buf), [0], [10])
     sink.close()
+
+
+def test_ticketed_batches_independent_waits(tmp_path):
+    sink = native.FanoutSink(3, 16)
+    bufs = [np.frombuffer(bytes([65 + k]) * 1000, dtype=np.uint8)
+            for k in range(4)]
+    tickets = []
+    for k in range(4):
+        tickets.append(sink.write_batch_ticket(
+            [str(tmp_path / f"t{k}.jsonl")], memoryview(bufs[k]),
+            [0], [1000]))
+    # waits may happen out of order
+    sink.wait_ticket(tickets[2])
+    sink.wait_ticket(tickets[0])
+    sink.wait_ticket(tickets[3])
+    sink.wait_ticket(tickets[1])
+    sink.close()
+    for k in range(4):
+        assert (tmp_path / f"t{k}.jsonl").read_bytes() == \
+            bytes([65 + k]) * 1000
+
+
+def test_state_ticket_api_roundtrip(tmp_path):
+    buf, bounds = mk_buffer(n_channels=3)
+    channels = ["chanA", "chanB", "chanC"]
+    cfg = CrawlerConfig(crawl_id="n6", storage_root=str(tmp_path))
+    sm = LocalStateManager(cfg)
+    items = [(ch, lo, hi) for ch, (lo, hi) in zip(channels, bounds)]
+    t1 = sm.store_post_lines_batch(items, memoryview(buf), ticket=True)
+    t2 = sm.store_post_lines_batch(items, memoryview(buf), ticket=True)
+    sm.wait_post_write(t1)
+    sm.wait_post_write(t2)
+    sm.wait_post_write(None)  # no-op
+    sm.close()
+    for ch, (lo, hi) in zip(channels, bounds):
+        data = (tmp_path / "n6" / ch / "posts" / "posts.jsonl").read_bytes()
+        assert data == bytes(buf[lo:hi]) * 2
