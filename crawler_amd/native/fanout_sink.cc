@@ -47,7 +47,7 @@ void make_parent_dirs(const std::string& path) {
 }
 
 struct Task {
-  int fd;
+  std::string path;  // resolved to an fd by the WORKER (parallel opens)
   const char* data;
   size_t len;
   uint64_t ticket;  // 0 = untracked (write_batch/write_batch_nowait)
@@ -86,14 +86,13 @@ class FanoutSink {
       if (hi[i] < lo[i] || hi[i] > n_bytes)
         throw std::out_of_range("slice outside buffer");
       if (hi[i] == lo[i]) continue;
-      int fd = fd_for(paths[i]);
-      tasks.push_back(Task{fd, base + lo[i], hi[i] - lo[i]});
+      tasks.push_back(Task{paths[i], base + lo[i], hi[i] - lo[i], 0});
     }
     {
       py::gil_scoped_release rel;
       {
         std::unique_lock<std::mutex> lk(mu_);
-        for (auto& t : tasks) q_.push(t);
+        for (auto& t : tasks) q_.push(std::move(t));
         pending_ += tasks.size();
       }
       cv_.notify_all();
@@ -127,12 +126,11 @@ class FanoutSink {
       if (hi[i] < lo[i] || hi[i] > n_bytes)
         throw std::out_of_range("slice outside buffer");
       if (hi[i] == lo[i]) continue;
-      int fd = fd_for(paths[i]);
-      tasks.push_back(Task{fd, base + lo[i], hi[i] - lo[i]});
+      tasks.push_back(Task{paths[i], base + lo[i], hi[i] - lo[i], 0});
     }
     {
       std::unique_lock<std::mutex> lk(mu_);
-      for (auto& t : tasks) q_.push(t);
+      for (auto& t : tasks) q_.push(std::move(t));
       pending_ += tasks.size();
     }
     cv_.notify_all();
@@ -160,14 +158,13 @@ class FanoutSink {
       if (hi[i] < lo[i] || hi[i] > n_bytes)
         throw std::out_of_range("slice outside buffer");
       if (hi[i] == lo[i]) continue;
-      int fd = fd_for(paths[i]);
-      tasks.push_back(Task{fd, base + lo[i], hi[i] - lo[i], ticket});
+      tasks.push_back(Task{paths[i], base + lo[i], hi[i] - lo[i], ticket});
     }
     {
       std::unique_lock<std::mutex> lk(mu_);
       ticket_pending_[ticket] = tasks.size();
       ticket_errors_[ticket] = 0;
-      for (auto& t : tasks) q_.push(t);
+      for (auto& t : tasks) q_.push(std::move(t));
       pending_ += tasks.size();
     }
     cv_.notify_all();
@@ -214,6 +211,7 @@ class FanoutSink {
     py::gil_scoped_release rel;
     std::unique_lock<std::mutex> lk(mu_);
     done_cv_.wait(lk, [this] { return pending_ == 0; });
+    std::unique_lock<std::mutex> flk(fd_mu_);
     for (auto& kv : fds_) ::fsync(kv.second);
   }
 
@@ -227,26 +225,40 @@ class FanoutSink {
     cv_.notify_all();
     for (auto& t : workers_) t.join();
     workers_.clear();
+    std::unique_lock<std::mutex> flk(fd_mu_);
     for (auto& kv : fds_) ::close(kv.second);
     fds_.clear();
     lru_.clear();
   }
 
   size_t bytes_written() const { return bytes_.load(); }
-  size_t open_files() const { return fds_.size(); }
+  size_t open_files() {
+    std::unique_lock<std::mutex> lk(fd_mu_);
+    return fds_.size();
+  }
 
  private:
+  // Worker-side open with its own lock: 1500 O_CREAT opens + mkdirs per
+  // chunk are metadata-heavy; resolving in the pool parallelizes them
+  // instead of serializing the enqueue thread. Returns -1 on failure.
   int fd_for(const std::string& path) {
-    auto it = fds_.find(path);
-    if (it != fds_.end()) return it->second;
+    {
+      std::unique_lock<std::mutex> lk(fd_mu_);
+      auto it = fds_.find(path);
+      if (it != fds_.end()) return it->second;
+    }
     int fd = ::open(path.c_str(), O_WRONLY | O_CREAT | O_APPEND, 0644);
     if (fd < 0 && errno == ENOENT) {
       make_parent_dirs(path);
       fd = ::open(path.c_str(), O_WRONLY | O_CREAT | O_APPEND, 0644);
     }
-    if (fd < 0)
-      throw std::runtime_error("open failed: " + path + ": " +
-                               std::strerror(errno));
+    if (fd < 0) return -1;
+    std::unique_lock<std::mutex> lk(fd_mu_);
+    auto it = fds_.find(path);
+    if (it != fds_.end()) {  // raced: another worker opened it first
+      ::close(fd);
+      return it->second;
+    }
     fds_[path] = fd;
     lru_.push_back(path);
     return fd;
@@ -255,6 +267,7 @@ class FanoutSink {
   // Eviction runs ONLY between batches (no task may reference an fd):
   // a same-batch eviction could close an fd a queued write still holds.
   void evict_to_cap() {
+    std::unique_lock<std::mutex> lk(fd_mu_);
     while (fds_.size() > static_cast<size_t>(max_open_) &&
            !lru_.empty()) {
       std::string old = lru_.front();          // oldest first; O_APPEND
@@ -280,11 +293,12 @@ class FanoutSink {
         t = q_.front();
         q_.pop();
       }
+      int fd = fd_for(t.path);
       const char* p = t.data;
       size_t left = t.len;
-      bool ok = true;
-      while (left > 0) {
-        ssize_t w = ::write(t.fd, p, left);
+      bool ok = (fd >= 0);
+      while (ok && left > 0) {
+        ssize_t w = ::write(fd, p, left);
         if (w < 0) {
           if (errno == EINTR) continue;
           ok = false;
@@ -319,6 +333,7 @@ class FanoutSink {
   std::unordered_map<uint64_t, size_t> ticket_errors_;    // under mu_
   std::atomic<size_t> errors_;
   std::atomic<size_t> bytes_;
+  std::mutex fd_mu_;  // guards fds_/lru_ (workers open in parallel)
   std::unordered_map<std::string, int> fds_;
   std::vector<std::string> lru_;
   std::queue<Task> q_;

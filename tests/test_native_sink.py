@@ -141,15 +141,28 @@ def test_nowait_then_drain(tmp_path):
         assert data == bytes(buf[lo:hi]) * 2
 
 
-def test_open_failure_raises_immediately(tmp_path):
-    """Opening a path that is a DIRECTORY fails at enqueue time (both
-    write_batch and write_batch_nowait open fds synchronously)."""
+def test_open_failure_surfaces_at_barrier(tmp_path):
+    """Opens happen in the WORKER pool (parallel metadata ops), so a
+    path that cannot be opened (here: a directory) surfaces as a write
+    failure at the drain/wait barrier, not at enqueue."""
     sink = native.FanoutSink(2, 4)
     buf = np.frombuffer(b"x" * 10, dtype=np.uint8)
     ro_dir = tmp_path / "ro"
     ro_dir.mkdir()
+    sink.write_batch_nowait([str(ro_dir)], memoryview(buf), [0], [10])
     with pytest.raises(Exception):
-        sink.write_batch_nowait([str(ro_dir)], memoryview(buf), [0], [10])
+        sink.drain()
+    sink.close()
+
+
+def test_ticket_open_failure_surfaces_at_wait(tmp_path):
+    sink = native.FanoutSink(2, 4)
+    buf = np.frombuffer(b"x" * 10, dtype=np.uint8)
+    ro_dir = tmp_path / "ro2"
+    ro_dir.mkdir()
+    t = sink.write_batch_ticket([str(ro_dir)], memoryview(buf), [0], [10])
+    with pytest.raises(Exception):
+        sink.wait_ticket(t)
     sink.close()
 
 
