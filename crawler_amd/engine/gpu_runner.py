@@ -63,12 +63,16 @@ class GpuCrawlEngine:
         return None
 
     def process_channels(self, usernames: List[str],
-                         now: Optional[_dt.datetime] = None
-                         ) -> Tuple[List[str], int]:
+                         now: Optional[_dt.datetime] = None,
+                         as_arrays: bool = False):
         """Process a list of channels; returns (newly discovered names,
-        posts stored)."""
+        posts stored). as_arrays=True returns the names as ONE
+        zero-padded uint8[M, 32] array instead of python strings — the
+        BFS loop sorts/dedups those at numpy speed (857k string objects
+        per layer cost ~0.5s to build and sort)."""
         now = now or _dt.datetime.now(_dt.timezone.utc)
         discovered: List[str] = []
+        discovered_arrays = []
         posts_total = 0
         t = self.timings
         import time as _time
@@ -157,12 +161,24 @@ class GpuCrawlEngine:
                 names = res.link_name[rows, cols].cpu().numpy()
                 lens = res.link_len[rows, cols].cpu().numpy()
                 w = names.shape[1]
-                blob = names.tobytes().decode("ascii", "replace")
-                discovered.extend(
-                    blob[i * w:i * w + ln]
-                    for i, ln in enumerate(lens.tolist())
-                )
+                # link_name rows carry garbage past len: zero-pad so the
+                # rows sort/compare as fixed-width byte strings
+                pad_mask = np.arange(w, dtype=np.uint8)[None, :] < \
+                    lens[:, None]
+                padded = np.where(pad_mask, names, 0)
+                if as_arrays:
+                    discovered_arrays.append(padded)
+                else:
+                    blob = padded.tobytes().decode("ascii", "replace")
+                    discovered.extend(
+                        blob[i * w:i * w + ln]
+                        for i, ln in enumerate(lens.tolist())
+                    )
             t["names"] += _time.perf_counter() - t0
+        if as_arrays:
+            discovered = (np.concatenate(discovered_arrays)
+                          if discovered_arrays
+                          else np.zeros((0, 32), dtype=np.uint8))
         self.stats["posts"] += posts_total
         self.stats["discovered"] += len(discovered)
         # barrier: all spill writes down before the layer's save_state
@@ -203,18 +219,28 @@ class GpuCrawlEngine:
             else:
                 mine = names
             import time as _time
-            discovered, _ = self.process_channels(mine)
+            discovered, _ = self.process_channels(mine, as_arrays=True)
             t0 = _time.perf_counter()
             for p in layer:
                 p.status = "fetched"
                 sm.update_page(p)
             if comm is not None:
-                # exchange discoveries (names, fixed-width padded)
-                discovered = self._allgather_names(discovered)
+                # exchange discoveries (names, fixed-width padded):
+                # materialize strings only for the cross-rank hop
+                w = discovered.shape[1] if len(discovered) else 32
+                blob = discovered.tobytes().decode("ascii", "replace")
+                as_strs = [blob[i * w:i * w + 32].rstrip("\x00")
+                           for i in range(len(discovered))]
+                names_x = self._allgather_names(as_strs)
+                discovered = np.zeros((len(names_x), 32), dtype=np.uint8)
+                for i, nm in enumerate(names_x):
+                    b = nm.encode()[:32]
+                    discovered[i, :len(b)] = np.frombuffer(
+                        b, dtype=np.uint8)
             if (self.cfg.sampling_method == "snowball"
                     and (self.cfg.max_depth < 0
                          or depth < self.cfg.max_depth)
-                    and discovered):
+                    and len(discovered)):
                 # Pre-apply add_layer's own skip rules (URL dedup +
                 # MaxPages/deadend budget) BEFORE constructing Page
                 # objects: a dense crawl discovers ~1M names per layer
@@ -229,10 +255,19 @@ class GpuCrawlEngine:
                     budget = max(0, max_pages - total) + deadends
                 existing = {p.url for p in sm.pages.values()}
                 url_dedup = getattr(sm, "url_dedup", {})
+                # numpy sorted-unique over fixed-width byte rows (NUL
+                # padding sorts exactly like the shorter string), then
+                # decode ONLY as many rows as the budget admits — the
+                # same lexicographic admission order as sorted(set(...))
+                # without 857k python string objects
+                uniq = np.unique(
+                    np.ascontiguousarray(discovered).view("S32").ravel()
+                )
                 cand = []
-                for n in sorted(set(discovered)):
+                for b in uniq:
                     if budget is not None and len(cand) >= budget:
                         break
+                    n = b.decode("ascii", "replace")
                     if n in existing or n in url_dedup:
                         continue
                     existing.add(n)
