@@ -307,10 +307,10 @@ class FanoutSink {
         p += w;
         left -= static_cast<size_t>(w);
       }
-      if (!ok)
-        errors_.fetch_add(1);
-      else
+      if (ok)
         bytes_.fetch_add(t.len);
+      else if (t.ticket == 0)
+        errors_.fetch_add(1);  // ticketed failures live in ticket_errors_
       {
         std::unique_lock<std::mutex> lk(mu_);
         if (t.ticket != 0) {
