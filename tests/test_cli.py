@@ -153,3 +153,42 @@ def test_pool_sized_by_tdlib_database_urls(tmp_path):
         "--tdlib-database-urls", "u1,u2,u3",
     ])
     assert rc == 0
+
+
+def test_debug_port_flag_serves_health(tmp_path):
+    """--debug-port wires the pprof-:6060 analog into a crawl run."""
+    import threading
+    import urllib.request
+
+    from crawler_amd.utils import debugserver
+
+    started = {}
+    real = debugserver.maybe_start
+
+    def spy(port, metrics=None):
+        # substitute an ephemeral port so the test never collides
+        srv = (debugserver.DebugServer(0, metrics=metrics).start()
+               if port else None)
+        started["srv"] = srv
+        return srv
+
+    debugserver.maybe_start = spy
+    try:
+        from crawler_amd.cli import main
+
+        rc = main([
+            "--mode", "standalone", "--sampling", "channel",
+            "--urls", "c0000000001", "--storage-root", str(tmp_path),
+            "--crawl-id", "dbg1", "--synthetic-universe", "50",
+            "--synthetic-posts", "8", "--min-users", "1", "--skip-media",
+            "--debug-port", "6060", "--disable-rate-limits",
+        ])
+        assert rc == 0
+        srv = started["srv"]
+        assert srv is not None
+        body = urllib.request.urlopen(
+            f"http://127.0.0.1:{srv.port}/healthz").read()
+        assert body == b"ok\n"
+        srv.stop()
+    finally:
+        debugserver.maybe_start = real

@@ -328,3 +328,19 @@ def test_depth_advance_waits_for_processing_pages(tmp_path):
     assert orch.current_depth in (0, 1)  # advances iff a next layer exists
     if orch.current_depth == 0:
         assert orch.done
+
+
+def test_work_result_count_only_roundtrip():
+    """discovered_count survives the JSON hop (orchestrated path sends
+    counts; names travel via the RCCL all-gather instead)."""
+    r = M.WorkResult(work_item_id="w1", worker_id="r0",
+                     posts_stored=5, discovered_count=123456)
+    r2 = M.WorkResult.from_json(r.to_json())
+    assert r2.discovered_count == 123456
+    assert r2.discovered == []
+    # old-style payloads (no field) still parse
+    import json as _json
+    d = _json.loads(r.to_json())
+    del d["discovered_count"]
+    r3 = M.WorkResult.from_json(_json.dumps(d))
+    assert r3.discovered_count == 0
