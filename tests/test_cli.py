@@ -192,3 +192,40 @@ def test_debug_port_flag_serves_health(tmp_path):
         srv.stop()
     finally:
         debugserver.maybe_start = real
+
+
+def test_env_overrides_full_mapping():
+    """CRAWLER_* env overrides (viper env parity, main.go:231-261)."""
+    from crawler_amd.config import CrawlerConfig, config_from_env
+
+    env = {
+        "CRAWLER_CONCURRENCY": "9",
+        "CRAWLER_STORAGE_ROOT": "/tmp/envroot",
+        "CRAWLER_MAX_DEPTH": "4",
+        "CRAWLER_MAX_POSTS": "123",
+        "CRAWLER_MAX_PAGES": "77",
+        "CRAWLER_MAX_COMMENTS": "5",
+        "CRAWLER_PLATFORM": "youtube",
+        "CRAWLER_SAMPLING": "random",
+        "CRAWLER_CRAWL_ID": "envcrawl",
+        "CRAWLER_CRAWL_LABEL": "lbl",
+        "CRAWLER_MIN_USERS": "42",
+        "CRAWLER_SKIP_MEDIA": "true",
+        "CRAWLER_SEED_SIZE": "11",
+        "CRAWLER_WALKBACK_RATE": "33",
+    }
+    cfg = config_from_env(CrawlerConfig(crawl_id="x",
+                                        storage_root="/tmp/a"), env=env)
+    assert cfg.concurrency == 9
+    assert cfg.storage_root == "/tmp/envroot"
+    assert cfg.max_depth == 4 and cfg.max_posts == 123
+    assert cfg.max_pages == 77 and cfg.max_comments == 5
+    assert cfg.platform == "youtube" and cfg.sampling_method == "random"
+    assert cfg.crawl_id == "envcrawl" and cfg.crawl_label == "lbl"
+    assert cfg.min_users == 42 and cfg.skip_media_download is True
+    assert cfg.seed_size == 11 and cfg.walkback_rate == 33
+    # falsy boolean spellings
+    cfg2 = config_from_env(CrawlerConfig(crawl_id="x",
+                                         storage_root="/tmp/a"),
+                           env={"CRAWLER_SKIP_MEDIA": "0"})
+    assert cfg2.skip_media_download is False
