@@ -158,14 +158,18 @@ class GpuCrawlEngine:
             if nz.numel():
                 rows = nz[:, 0]
                 cols = nz[:, 1]
-                names = res.link_name[rows, cols].cpu().numpy()
-                lens = res.link_len[rows, cols].cpu().numpy()
-                w = names.shape[1]
-                # link_name rows carry garbage past len: zero-pad so the
-                # rows sort/compare as fixed-width byte strings
-                pad_mask = np.arange(w, dtype=np.uint8)[None, :] < \
-                    lens[:, None]
-                padded = np.where(pad_mask, names, 0)
+                names_g = res.link_name[rows, cols]
+                lens_g = res.link_len[rows, cols]
+                w = names_g.shape[1]
+                # link_name rows carry garbage past len: zero-pad ON THE
+                # GPU so the D2H ships clean fixed-width byte strings
+                col_idx = torch.arange(w, device=names_g.device,
+                                       dtype=lens_g.dtype)
+                padded_g = torch.where(col_idx[None, :] < lens_g[:, None],
+                                       names_g,
+                                       torch.zeros_like(names_g))
+                padded = padded_g.cpu().numpy()
+                lens = lens_g.cpu().numpy()
                 if as_arrays:
                     discovered_arrays.append(padded)
                 else:

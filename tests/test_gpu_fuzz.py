@@ -105,7 +105,15 @@ def rand_comments(rng, n_msgs):
     return out
 
 
-@pytest.mark.parametrize("seed", [11, 22, 33])
+import os
+
+_SEEDS = [11, 22, 33]
+if os.environ.get("CRAWL_FUZZ_SEEDS"):
+    # extended campaigns: CRAWL_FUZZ_SEEDS="1,2,3,..." (evidence runs)
+    _SEEDS = [int(x) for x in os.environ["CRAWL_FUZZ_SEEDS"].split(",")]
+
+
+@pytest.mark.parametrize("seed", _SEEDS)
 def test_fuzz_batches_byte_identical(seed):
     from crawler_amd.ops import gpu
 
