@@ -1,0 +1,242 @@
+// FanoutCore — the pybind-free heart of the native spill sink.
+//
+// Split out of fanout_sink.cc so the SAME concurrency logic can be
+// stress-tested standalone under ThreadSanitizer
+// (tools/tsan_sink_stress.cc; SURVEY §5.2 — the reference relies on
+// convention, we add a sanitizer gate). The pybind wrapper
+// (fanout_sink.cc) only converts buffers and releases the GIL around
+// the blocking calls.
+//
+// Concurrency contract (locks documented per structure, reference
+// convention):
+//   mu_     guards q_, pending_, ticket_pending_, ticket_errors_, stop_
+//   fd_mu_  guards fds_, lru_ (workers open files in parallel)
+//   errors_/bytes_ are atomics; ticket_seq_ is an atomic counter.
+// Eviction runs only when no write is in flight (drain / sync batch).
+#pragma once
+
+#include <fcntl.h>
+#include <sys/stat.h>
+#include <sys/types.h>
+#include <unistd.h>
+
+#include <atomic>
+#include <condition_variable>
+#include <cstring>
+#include <mutex>
+#include <queue>
+#include <stdexcept>
+#include <string>
+#include <thread>
+#include <unordered_map>
+#include <vector>
+
+namespace crawl_native {
+
+inline void make_parent_dirs(const std::string& path) {
+  std::string dir = path.substr(0, path.find_last_of('/'));
+  if (dir.empty()) return;
+  std::string cur;
+  size_t pos = 0;
+  while (pos != std::string::npos) {
+    pos = dir.find('/', pos + 1);
+    cur = dir.substr(0, pos);
+    if (!cur.empty()) ::mkdir(cur.c_str(), 0755);  // EEXIST is fine
+  }
+}
+
+struct Task {
+  std::string path;  // resolved to an fd by the WORKER (parallel opens)
+  const char* data;
+  size_t len;
+  uint64_t ticket;  // 0 = untracked
+};
+
+class FanoutCore {
+ public:
+  explicit FanoutCore(int n_threads = 4, int max_open = 256)
+      : max_open_(max_open), stop_(false), pending_(0), errors_(0),
+        bytes_(0) {
+    if (n_threads < 1) n_threads = 1;
+    for (int i = 0; i < n_threads; ++i)
+      workers_.emplace_back([this] { this->worker(); });
+  }
+
+  ~FanoutCore() { close(); }
+
+  // Enqueue one batch; ticket==0 -> untracked. Caller guarantees the
+  // data pointers stay valid until the matching wait/drain.
+  void enqueue(std::vector<Task>&& tasks, uint64_t ticket) {
+    std::unique_lock<std::mutex> lk(mu_);
+    if (ticket != 0) {
+      ticket_pending_[ticket] = tasks.size();
+      ticket_errors_[ticket] = 0;
+    }
+    for (auto& t : tasks) q_.push(std::move(t));
+    pending_ += tasks.size();
+    lk.unlock();
+    cv_.notify_all();
+  }
+
+  uint64_t next_ticket() { return ++ticket_seq_; }
+
+  // Returns the error count for the ticket (0 = clean).
+  size_t wait_ticket(uint64_t ticket) {
+    std::unique_lock<std::mutex> lk(mu_);
+    done_cv_.wait(lk, [this, ticket] {
+      auto it = ticket_pending_.find(ticket);
+      return it == ticket_pending_.end() || it->second == 0;
+    });
+    size_t errs = ticket_errors_[ticket];
+    ticket_pending_.erase(ticket);
+    ticket_errors_.erase(ticket);
+    return errs;
+  }
+
+  // Wait until EVERYTHING queued has been written. Returns untracked
+  // error count since the last drain and evicts fds beyond the cap.
+  size_t drain() {
+    size_t errs;
+    {
+      std::unique_lock<std::mutex> lk(mu_);
+      done_cv_.wait(lk, [this] { return pending_ == 0; });
+      errs = errors_.exchange(0);
+    }
+    evict_to_cap();
+    return errs;
+  }
+
+  void flush() {
+    std::unique_lock<std::mutex> lk(mu_);
+    done_cv_.wait(lk, [this] { return pending_ == 0; });
+    std::unique_lock<std::mutex> flk(fd_mu_);
+    for (auto& kv : fds_) ::fsync(kv.second);
+  }
+
+  void close() {
+    {
+      std::unique_lock<std::mutex> lk(mu_);
+      if (stop_) return;
+      done_cv_.wait(lk, [this] { return pending_ == 0; });
+      stop_ = true;
+    }
+    cv_.notify_all();
+    for (auto& t : workers_) t.join();
+    workers_.clear();
+    std::unique_lock<std::mutex> flk(fd_mu_);
+    for (auto& kv : fds_) ::close(kv.second);
+    fds_.clear();
+    lru_.clear();
+  }
+
+  size_t bytes_written() const { return bytes_.load(); }
+  size_t open_files() {
+    std::unique_lock<std::mutex> lk(fd_mu_);
+    return fds_.size();
+  }
+
+ private:
+  // Worker-side open with its own lock: O_CREAT opens + mkdirs are
+  // metadata-heavy; resolving in the pool parallelizes them. -1 = fail.
+  int fd_for(const std::string& path) {
+    {
+      std::unique_lock<std::mutex> lk(fd_mu_);
+      auto it = fds_.find(path);
+      if (it != fds_.end()) return it->second;
+    }
+    int fd = ::open(path.c_str(), O_WRONLY | O_CREAT | O_APPEND, 0644);
+    if (fd < 0 && errno == ENOENT) {
+      make_parent_dirs(path);
+      fd = ::open(path.c_str(), O_WRONLY | O_CREAT | O_APPEND, 0644);
+    }
+    if (fd < 0) return -1;
+    std::unique_lock<std::mutex> lk(fd_mu_);
+    auto it = fds_.find(path);
+    if (it != fds_.end()) {  // raced: another worker opened it first
+      ::close(fd);
+      return it->second;
+    }
+    fds_[path] = fd;
+    lru_.push_back(path);
+    return fd;
+  }
+
+  // Runs ONLY when no write is in flight (a same-batch eviction could
+  // close an fd a queued write still holds).
+  void evict_to_cap() {
+    std::unique_lock<std::mutex> lk(fd_mu_);
+    while (fds_.size() > static_cast<size_t>(max_open_) &&
+           !lru_.empty()) {
+      std::string old = lru_.front();          // oldest first; O_APPEND
+      lru_.erase(lru_.begin());                // makes reopen safe
+      auto oit = fds_.find(old);
+      if (oit != fds_.end()) {
+        ::close(oit->second);
+        fds_.erase(oit);
+      }
+    }
+  }
+
+  void worker() {
+    for (;;) {
+      Task t;
+      {
+        std::unique_lock<std::mutex> lk(mu_);
+        cv_.wait(lk, [this] { return stop_ || !q_.empty(); });
+        if (q_.empty()) {
+          if (stop_) return;
+          continue;
+        }
+        t = std::move(q_.front());
+        q_.pop();
+      }
+      int fd = fd_for(t.path);
+      const char* p = t.data;
+      size_t left = t.len;
+      bool ok = (fd >= 0);
+      while (ok && left > 0) {
+        ssize_t w = ::write(fd, p, left);
+        if (w < 0) {
+          if (errno == EINTR) continue;
+          ok = false;
+          break;
+        }
+        p += w;
+        left -= static_cast<size_t>(w);
+      }
+      if (ok)
+        bytes_.fetch_add(t.len);
+      else if (t.ticket == 0)
+        errors_.fetch_add(1);  // ticketed failures live in ticket_errors_
+      {
+        std::unique_lock<std::mutex> lk(mu_);
+        if (t.ticket != 0) {
+          auto it = ticket_pending_.find(t.ticket);
+          if (it != ticket_pending_.end() && it->second > 0) --it->second;
+          if (!ok) ++ticket_errors_[t.ticket];
+        }
+        if (--pending_ == 0 ||
+            (t.ticket != 0 && ticket_pending_[t.ticket] == 0))
+          done_cv_.notify_all();
+      }
+    }
+  }
+
+  int max_open_;
+  bool stop_;
+  size_t pending_;
+  std::atomic<uint64_t> ticket_seq_{0};
+  std::unordered_map<uint64_t, size_t> ticket_pending_;   // under mu_
+  std::unordered_map<uint64_t, size_t> ticket_errors_;    // under mu_
+  std::atomic<size_t> errors_;
+  std::atomic<size_t> bytes_;
+  std::mutex fd_mu_;  // guards fds_/lru_ (workers open in parallel)
+  std::unordered_map<std::string, int> fds_;
+  std::vector<std::string> lru_;
+  std::queue<Task> q_;
+  std::mutex mu_;
+  std::condition_variable cv_, done_cv_;
+  std::vector<std::thread> workers_;
+};
+
+}  // namespace crawl_native

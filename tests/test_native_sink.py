@@ -201,3 +201,33 @@ def test_state_ticket_api_roundtrip(tmp_path):
     for ch, (lo, hi) in zip(channels, bounds):
         data = (tmp_path / "n6" / ch / "posts" / "posts.jsonl").read_bytes()
         assert data == bytes(buf[lo:hi]) * 2
+
+
+def test_tsan_stress_harness(tmp_path):
+    """Compile FanoutCore's stress harness under ThreadSanitizer and run
+    it (the sanitizer gate the reference lacks). Any data
+    race fails the run (halt_on_error)."""
+    import shutil
+    import subprocess
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    src = os.path.join(root, "tools", "tsan_sink_stress.cc")
+    exe = str(tmp_path / "tsan_sink")
+    if shutil.which("g++") is None:
+        pytest.skip("no g++")
+    build = subprocess.run(
+        ["g++", "-fsanitize=thread", "-O1", "-g", "-std=c++17",
+         "-pthread", src, "-o", exe],
+        capture_output=True, text=True, timeout=120,
+    )
+    if build.returncode != 0 and "tsan" in build.stderr.lower():
+        pytest.skip("libtsan unavailable")
+    assert build.returncode == 0, build.stderr
+    run = subprocess.run(
+        [exe, str(tmp_path / "out")],
+        env={**os.environ, "TSAN_OPTIONS": "halt_on_error=1"},
+        capture_output=True, text=True, timeout=120,
+    )
+    assert run.returncode == 0, run.stdout + run.stderr
+    assert "tsan-stress OK" in run.stdout
+    assert "ThreadSanitizer" not in run.stderr
