@@ -135,7 +135,10 @@ def main():
             log("pinned allocation failed; using pageable host memory")
             return torch.empty(nbytes, dtype=torch.uint8)
 
-    pinned = [alloc_host(int(chunk_posts * 2600)) for _ in range(2)]
+    # worst-case line sizes differ by platform: telegram ~2.1KB/post,
+    # youtube ~3.0KB/video (richer channel block per line)
+    per_post = 2600 if my_platform == "telegram" else 3600
+    pinned = [alloc_host(int(chunk_posts * per_post)) for _ in range(2)]
     compute_stream = torch.cuda.Stream()
     # two copy streams: MI355X has multiple SDMA engines; splitting the
     # 2.5GB D2H in half across streams uses two of them
