@@ -154,25 +154,30 @@ class GpuCrawlEngine:
             # D2H + ONE ascii decode of the packed name block, then
             # string slices — a per-name bytes().decode() loop costs
             # seconds at ~1M discoveries per crawl.
-            nz = new_mask.nonzero()
-            if nz.numel():
-                rows = nz[:, 0]
-                cols = nz[:, 1]
-                names_g = res.link_name[rows, cols]
-                lens_g = res.link_len[rows, cols]
-                w = names_g.shape[1]
-                # link_name rows carry garbage past len: zero-pad ON THE
-                # GPU so the D2H ships clean fixed-width byte strings
-                col_idx = torch.arange(w, device=names_g.device,
-                                       dtype=lens_g.dtype)
-                padded_g = torch.where(col_idx[None, :] < lens_g[:, None],
-                                       names_g,
-                                       torch.zeros_like(names_g))
-                padded = padded_g.cpu().numpy()
-                lens = lens_g.cpu().numpy()
-                if as_arrays:
-                    discovered_arrays.append(padded)
-                else:
+            if as_arrays:
+                # device-side densification (claim_compact_kernel):
+                # ships exactly M zero-padded rows, no host gather
+                names_g, _hashes_g = self.seen.compact_claimed(res,
+                                                               new_mask)
+                if names_g.shape[0]:
+                    discovered_arrays.append(names_g.cpu().numpy())
+            else:
+                nz = new_mask.nonzero()
+                if nz.numel():
+                    rows = nz[:, 0]
+                    cols = nz[:, 1]
+                    names_g = res.link_name[rows, cols]
+                    lens_g = res.link_len[rows, cols]
+                    w = names_g.shape[1]
+                    # zero-pad ON THE GPU so the D2H ships clean
+                    # fixed-width byte strings
+                    col_idx = torch.arange(w, device=names_g.device,
+                                           dtype=lens_g.dtype)
+                    padded_g = torch.where(
+                        col_idx[None, :] < lens_g[:, None], names_g,
+                        torch.zeros_like(names_g))
+                    padded = padded_g.cpu().numpy()
+                    lens = lens_g.cpu().numpy()
                     blob = padded.tobytes().decode("ascii", "replace")
                     discovered.extend(
                         blob[i * w:i * w + ln]

@@ -59,6 +59,33 @@ bloom_update_kernel(const unsigned long long* hashes, const int* cnt, int n,
   }
 }
 
+// Compact the claim winners into a dense block: out_names[pos] gets the
+// ZERO-PADDED 32-byte name, out_hashes[pos] its hash. Row order is the
+// atomic claim order (nondeterministic) — consumers sort/unique host-side
+// (engine/gpu_runner.py admission). Replaces a host-side
+// nonzero+gather+pad round-trip over ~1M candidate slots per chunk.
+__global__ void __launch_bounds__(256)
+claim_compact_kernel(const unsigned char* new_mask,
+                     const unsigned char* names,
+                     const unsigned char* lens,
+                     const unsigned long long* hashes, int n, int maxl,
+                     unsigned char* out_names,
+                     unsigned long long* out_hashes,
+                     unsigned int* cursor, unsigned int cap) {
+  int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  int stride = gridDim.x * blockDim.x;
+  for (int idx = tid; idx < n * maxl; idx += stride) {
+    if (!new_mask[idx]) continue;
+    unsigned int pos = atomicAdd(cursor, 1u);
+    if (pos >= cap) continue;  // caller sizes cap; truncation guarded
+    int ln = lens[idx];
+    const unsigned char* src = names + (long long)idx * 32;
+    unsigned char* dst = out_names + (long long)pos * 32;
+    for (int j = 0; j < 32; ++j) dst[j] = (j < ln) ? src[j] : 0;
+    out_hashes[pos] = hashes[idx];
+  }
+}
+
 // Bulk-insert pre-hashed values (e.g. a merged remote seen-set after an
 // all-gather) without producing claim output.
 __global__ void __launch_bounds__(256)
@@ -101,6 +128,23 @@ int crawl_bloom_update(const void* hashes, const void* cnt, int n, int maxl,
                      (hipStream_t)stream,
                      (const unsigned long long*)hashes, (const int*)cnt, n,
                      maxl, (unsigned int*)bloom, bloom_bits - 1);
+  return (int)hipGetLastError();
+}
+
+int crawl_claim_compact(const void* new_mask, const void* names,
+                        const void* lens, const void* hashes, int n,
+                        int maxl, void* out_names, void* out_hashes,
+                        void* cursor, unsigned int cap, int grid,
+                        void* stream) {
+  hipLaunchKernelGGL(crawl::claim_compact_kernel, dim3(grid), dim3(256), 0,
+                     (hipStream_t)stream,
+                     (const unsigned char*)new_mask,
+                     (const unsigned char*)names,
+                     (const unsigned char*)lens,
+                     (const unsigned long long*)hashes, n, maxl,
+                     (unsigned char*)out_names,
+                     (unsigned long long*)out_hashes,
+                     (unsigned int*)cursor, cap);
   return (int)hipGetLastError();
 }
 

@@ -639,6 +639,36 @@ class SeenSet:
     def new_count(self) -> int:
         return int(self._n_new.item())
 
+    def compact_claimed(self, res: EncodeResult, new_mask: torch.Tensor):
+        """Densify the claim winners ON DEVICE: returns
+        (names uint8[M, 32] zero-padded, hashes int64[M]) — row order is
+        the atomic claim order (consumers sort host-side). Replaces a
+        host nonzero+gather+pad over N*MAX_LINKS candidate slots."""
+        n = res.link_cnt.shape[0]
+        cap = n * MAX_LINKS
+        out_names = torch.empty((cap, 32), dtype=torch.uint8,
+                                device=self.device)
+        out_hashes = torch.empty(cap, dtype=torch.int64,
+                                 device=self.device)
+        cursor = torch.zeros(1, dtype=torch.int32, device=self.device)
+        grid = min((cap + 255) // 256, 4096)
+        rc = self.lib.crawl_claim_compact(
+            ctypes.c_void_p(new_mask.data_ptr()),
+            ctypes.c_void_p(res.link_name.data_ptr()),
+            ctypes.c_void_p(res.link_len.data_ptr()),
+            ctypes.c_void_p(res.link_hash.data_ptr()),
+            n, MAX_LINKS,
+            ctypes.c_void_p(out_names.data_ptr()),
+            ctypes.c_void_p(out_hashes.data_ptr()),
+            ctypes.c_void_p(cursor.data_ptr()),
+            ctypes.c_uint(cap), grid, self._stream(),
+        )
+        if rc != 0:
+            raise RuntimeError(
+                f"crawl_claim_compact failed: hip error {rc}")
+        m = int(cursor.item())  # syncs the stream
+        return out_names[:m], out_hashes[:m]
+
     def insert_hashes(self, hashes: torch.Tensor) -> None:
         """Bulk-insert merged remote hashes (post all-gather)."""
         n = hashes.numel()

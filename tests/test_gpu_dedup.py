@@ -96,3 +96,33 @@ def test_fnv_hash_matches_python_oracle():
             assert int(hashes[i, k]) == gpu.fnv1a64(name.encode()), name
             checked += 1
     assert checked > 0
+
+
+def test_compact_claimed_matches_mask_gather():
+    """claim_compact_kernel densification == host nonzero+gather over the
+    same new_mask (set equality; compact order is atomic)."""
+    from crawler_amd.ops import gpu as gpu_mod
+
+    feed = SyntheticFeed(FeedConfig(seed=5, universe=1000))
+    batch = feed.build_batch(np.arange(300, 340), posts_per_channel=100)
+    res = gpu_mod.parse_encode(batch.to("cuda:0"), now=NOW)
+    seen = gpu_mod.SeenSet(torch.device("cuda:0"))
+    new_mask = seen.claim(res)
+    torch.cuda.synchronize()
+    names_c, hashes_c = seen.compact_claimed(res, new_mask)
+    # host reference gather
+    nz = new_mask.nonzero()
+    rows, cols = nz[:, 0], nz[:, 1]
+    names_h = res.link_name[rows, cols].cpu().numpy()
+    lens_h = res.link_len[rows, cols].cpu().numpy()
+    ref = {bytes(names_h[i][: lens_h[i]]) for i in range(len(lens_h))}
+    got_rows = names_c.cpu().numpy()
+    got = {bytes(r).rstrip(b"\0") for r in got_rows}
+    assert got == ref
+    assert names_c.shape[0] == len(lens_h)
+    ref_h = set(res.link_hash[rows, cols].cpu().tolist())
+    assert set(hashes_c.cpu().tolist()) == ref_h
+    # second claim of the same batch claims nothing
+    mask2 = seen.claim(res)
+    n2, _ = seen.compact_claimed(res, mask2)
+    assert n2.shape[0] == 0
