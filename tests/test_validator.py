@@ -349,3 +349,44 @@ def test_walkback_exhausted_leaves_batch_for_recovery(tmp_path):
     sm.add_discovered_channel("backstop")
     assert v.pump_walkback()
     assert rw.pending_batches[bid].status == "completed"
+
+
+# ---------- synthetic title-rule matrix (channelvalidator.go:132-153) ----
+
+def _html(title, noindex=False):
+    meta = b'<meta name="robots" content="noindex, nofollow">' \
+        if noindex else b""
+    return (b"<html><head><title>" + title + b"</title>" + meta +
+            b"</head><body>x</body></html>")
+
+
+@pytest.mark.parametrize("title,noindex,status,reason", [
+    (b"Telegram: View @somechan", False, "valid", ""),
+    (b"View @somechan", False, "valid", ""),
+    (b"Telegram: Contact @someone", True, "invalid",
+     "username_not_occupied"),
+    (b"Contact @someone", True, "invalid", "username_not_occupied"),
+    (b"Telegram: Contact @someone", False, "not_channel",
+     "not_supergroup"),
+    (b"Telegram Messenger", False, "invalid", "not_found"),
+    (b"Some Random Page", False, "invalid", "unrecognized"),
+    (b"", False, "invalid", "unrecognized"),
+])
+def test_title_rule_matrix(title, noindex, status, reason):
+    r = parse_channel_html(_html(title, noindex))
+    assert r.status == status
+    assert r.reason == reason
+
+
+def test_body_cap_64kb():
+    """Only the first 64KB are parsed (channelvalidator.go:103 cap);
+    a title appearing after the cap never classifies."""
+    filler = b"<!-- " + b"x" * (64 * 1024) + b" -->"
+    body = filler + _html(b"Telegram: View @late")
+    r = parse_channel_html(body)
+    assert r.status == "invalid" and r.reason == "unrecognized"
+
+
+def test_missing_title_tag():
+    r = parse_channel_html(b"<html><body>no title here</body></html>")
+    assert r.status == "invalid" and r.reason == "unrecognized"
