@@ -123,22 +123,6 @@ def main():
     now = dt.datetime(2026, 1, 1, tzinfo=dt.timezone.utc)
     seen = gpu.SeenSet(device)
 
-    # Untimed setup priming: a freshly-idle MI355X starts at low DVFS
-    # clocks and takes seconds of sustained load to ramp; short runs
-    # (small --steps/--warmup) would otherwise under-read by ~2x. This
-    # only loads the compute pipeline (no claims, no D2H) and runs
-    # BEFORE the W warmup steps; the timed region is unchanged.
-    t_pr = time.time()
-    prim = 0
-    while time.time() - t_pr < 3.0 and prim < 60:
-        if my_platform == "telegram":
-            gpu.parse_encode(chunks[0], now=now)
-        else:
-            gpu.yt_parse_encode(chunks[0], now=now)
-        torch.cuda.synchronize()
-        prim += 1
-    log(f"primed {prim} iterations ({time.time() - t_pr:.1f}s)")
-
     # Double-buffered pinned host ring for the JSONL output (the storage
     # boundary): D2H of step k overlaps the kernels of step k+1 on a
     # separate copy stream. Worst case ~2.6KB/post.
@@ -267,6 +251,23 @@ def main():
         p50_channel_ms = round(statistics.median(lat), 3)
         log(f"p50 channel latency: {p50_channel_ms} ms "
             f"({args.posts} posts/channel)")
+
+    # Untimed priming IMMEDIATELY before the timed region: a
+    # freshly-idle MI355X starts at low DVFS clocks, and clocks decay
+    # within the idle gaps of setup/p50 measurement — short runs (small
+    # --steps) would otherwise under-read by ~2x while the first timed
+    # steps re-ramp. This only loads the compute pipeline (no claims,
+    # no D2H side effects); the timed region is unchanged.
+    t_pr = time.time()
+    prim = 0
+    while time.time() - t_pr < 3.0 and prim < 60:
+        if my_platform == "telegram":
+            gpu.parse_encode(chunks[0], now=now)
+        else:
+            gpu.yt_parse_encode(chunks[0], now=now)
+        torch.cuda.synchronize()
+        prim += 1
+    log(f"primed {prim} iterations ({time.time() - t_pr:.1f}s)")
 
     # ---- timed region ----
     if world > 1:
