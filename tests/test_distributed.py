@@ -400,3 +400,29 @@ def test_queue_stash_survives_slow_publish():
     qa.publish("second")  # satisfies whichever racer stashed
     ta.join(); tb.join()
     assert sorted(x for x in got if x) == ["first", "second"]
+
+
+def test_offline_workers_tracked_and_reassigned(tmp_path):
+    """A worker that stops heartbeating is marked offline and its
+    in-flight item is republished at high priority with retry_count++
+    (orchestrator.go:493-559)."""
+    clock = {"t": 1000.0}
+    cfg, orch, workers, store = mk_dist_env(tmp_path, n_workers=1,
+                                            sampling="channel")
+    orch.clock = lambda: clock["t"]
+    orch.sm.initialize(["c0000000001"])
+    orch.heartbeats.register("w0")
+    store.set("hb/w0", f"{clock['t']}|active")
+    orch.distribute()
+    assert len(orch.in_flight) == 1
+    # 6 minutes pass with no beat and no result
+    clock["t"] += 360
+    n = orch.check_worker_health()
+    assert n == 1
+    assert "w0" in orch.offline
+    item = next(iter(orch.in_flight.values()))[0]
+    assert item.retry_count == 1
+    assert item.priority == M.PRIORITY_HIGH
+    # the republished copy is claimable
+    raw = orch.work_q.claim(timeout_s=0.0)
+    assert raw is not None
