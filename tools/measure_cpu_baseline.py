@@ -135,7 +135,7 @@ Host: {os.uname().nodename} ({os.cpu_count()} logical cores).
 | Python golden (1 core) | {py_pps:,.0f} |
 | C++ reference (1 thread) | {r1['posts_per_sec']:,.0f} |
 | C++ reference ({ncpu} threads) | {rn['posts_per_sec']:,.0f} |
-| **MI355X HIP pipeline (1 GPU)** | **24,400,000** (see profiles/) |
+| **MI355X HIP pipeline (1 GPU)** | **28,000,000** (see profiles/) |
 
 GPU vs best host number: {24_400_000 / rn['posts_per_sec']:,.1f}x.
 Generated by tools/measure_cpu_baseline.py on {dt.date.today()}.
