@@ -22,7 +22,7 @@ per step) so a channel is crawled by exactly one rank per crawl.
 from __future__ import annotations
 
 import datetime as _dt
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 import numpy as np
 import torch
