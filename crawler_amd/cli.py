@@ -337,9 +337,47 @@ def main(argv: Optional[List[str]] = None) -> int:
         if cfg.seed_size and not urls:
             urls = [feed.username_of(i) for i in range(cfg.seed_size)]
         runner.seed(urls)
-        stats = runner.run(
-            max_seconds=cfg.max_crawl_duration_s or None
-        )
+        vthread = stop_evt = None
+        if cfg.tandem_crawl:
+            # The reference runs validator PODS against the shared
+            # Postgres (dapr/standalone.go:276-314); single-node analog:
+            # an in-process validator thread over the same store
+            # (identical claim semantics, test_tandem_e2e pattern).
+            import threading
+
+            from .engine.validator import TandemValidator
+            from .feed.client import TokenBucket
+            from .feed.tme import MockTMe
+
+            rl = None
+            if not cfg.disable_rate_limits:
+                rl = TokenBucket(cfg.validator_request_rate * 60, 100)
+            v = TandemValidator(
+                cfg, sm, rw,
+                fetcher=MockTMe(universe=args.synthetic_universe),
+                rate_limiter=rl,
+            )
+            v.recover()
+            stop_evt = threading.Event()
+
+            def vloop():
+                while not stop_evt.is_set():
+                    n = v.pump_edges()
+                    b = v.pump_walkback()
+                    if not n and not b:
+                        stop_evt.wait(0.01)
+
+            vthread = threading.Thread(target=vloop, daemon=True,
+                                       name="tandem-validator")
+            vthread.start()
+        try:
+            stats = runner.run(
+                max_seconds=cfg.max_crawl_duration_s or None
+            )
+        finally:
+            if vthread is not None:
+                stop_evt.set()
+                vthread.join(timeout=10)
     else:
         runner = StandaloneRunner(cfg, sm, pool)
         stats = runner.run(urls)

@@ -229,3 +229,20 @@ def test_env_overrides_full_mapping():
                                          storage_root="/tmp/a"),
                            env={"CRAWLER_SKIP_MEDIA": "0"})
     assert cfg2.skip_media_download is False
+
+
+def test_e2e_tandem_crawl_cli(tmp_path):
+    """--tandem-crawl through the CLI runs the crawler with an
+    in-process validator thread (the single-node validator-pod analog):
+    the crawl progresses past batch barriers and completes."""
+    rc = main([
+        "--mode", "standalone", "--sampling", "random-walk",
+        "--tandem-crawl", "--seed-size", "3",
+        "--storage-root", str(tmp_path), "--crawl-id", "tnd1",
+        "--synthetic-universe", "300", "--synthetic-posts", "12",
+        "--disable-rate-limits", "--min-users", "1", "--max-pages", "8",
+        "--max-crawl-duration", "30s", "--validator-timeout", "20s",
+    ])
+    assert rc == 0
+    jsonls = list((tmp_path / "tnd1").rglob("posts.jsonl"))
+    assert len(jsonls) >= 3  # seeds + validator-fed hops crawled
