@@ -193,6 +193,19 @@ class Orchestrator:
                 time.sleep(idle_sleep)
             rounds += 1
         self.broadcast_stop(n_workers)
+        # Grace period: hold the store (this process hosts the TCPStore
+        # master) until every worker acked with WORKER_STOPPING or the
+        # timeout passes — otherwise workers crash on a dead store
+        # mid-poll instead of consuming their poison pill.
+        deadline = self.clock() + 15.0
+        stopping = 0
+        while stopping < n_workers and self.clock() < deadline:
+            raw = self.status_q.claim(timeout_s=0.2)
+            if raw is None:
+                continue
+            msg = M.StatusMessage.from_json(raw)
+            if msg.message_type == M.MSG_WORKER_STOPPING:
+                stopping += 1
         return dict(self.stats)
 
 

@@ -43,7 +43,10 @@ class Worker:
 
         def loop():
             while not self._stop.is_set():
-                self.heartbeats.beat(self.worker_id, M.WORKER_ACTIVE)
+                try:
+                    self.heartbeats.beat(self.worker_id, M.WORKER_ACTIVE)
+                except Exception:
+                    return  # store gone: main loop is exiting too
                 self._stop.wait(interval_s)
 
         self._hb_thread = threading.Thread(target=loop, daemon=True)
@@ -104,7 +107,14 @@ class Worker:
 
     def run_once(self, timeout_s: float = 0.5) -> bool:
         """Claim and process one item. Returns False on poison pill/empty."""
-        raw = self.work_q.claim(timeout_s=timeout_s)
+        try:
+            raw = self.work_q.claim(timeout_s=timeout_s)
+        except Exception:
+            # The store master (orchestrator) is gone: the crawl is over
+            # — exit cleanly instead of crashing mid-poll (the reference
+            # worker likewise dies with its sidecar; we prefer a clean
+            # stop so state files close properly).
+            return False
         if raw is None:
             return True  # idle, keep polling
         item = M.WorkItem.from_json(raw)
@@ -127,10 +137,13 @@ class Worker:
                     break
         finally:
             self._stop.set()
-            self.status_q.publish(M.StatusMessage(
-                message_type=M.MSG_WORKER_STOPPING,
-                worker_id=self.worker_id, processed=self.processed,
-            ).to_json())
+            try:
+                self.status_q.publish(M.StatusMessage(
+                    message_type=M.MSG_WORKER_STOPPING,
+                    worker_id=self.worker_id, processed=self.processed,
+                ).to_json())
+            except Exception:
+                pass  # store may already be gone
             self.sm.close()
 
 

@@ -426,3 +426,48 @@ def test_offline_workers_tracked_and_reassigned(tmp_path):
     # the republished copy is claimable
     raw = orch.work_q.claim(timeout_s=0.0)
     assert raw is not None
+
+
+def test_cli_orchestrator_worker_two_processes(tmp_path):
+    """Real cross-process run of --mode orchestrator + --mode worker over
+    a TCPStore on 127.0.0.1 (the reference's pod topology, one host)."""
+    import os
+    import subprocess
+    import sys
+
+    port = 29877
+    env = {**os.environ, "MASTER_ADDR": "127.0.0.1",
+           "MASTER_PORT": str(port), "CRAWLER_NUM_WORKERS": "1"}
+    common = [
+        "--storage-root", str(tmp_path), "--crawl-id", "dist1",
+        "--synthetic-universe", "100", "--synthetic-posts", "10",
+        "--disable-rate-limits", "--min-users", "1",
+        "--sampling", "channel", "--urls", "c0000000001,c0000000002",
+    ]
+    orch = subprocess.Popen(
+        [sys.executable, "-m", "crawler_amd.cli", "--mode",
+         "orchestrator"] + common,
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        text=True,
+    )
+    worker = subprocess.Popen(
+        [sys.executable, "-m", "crawler_amd.cli", "--mode", "worker",
+         "--worker-id", "wA"] + common,
+        env=env, stdout=subprocess.PIPE, stderr=subprocess.STDOUT,
+        text=True,
+    )
+    try:
+        out_o, _ = orch.communicate(timeout=90)
+        out_w, _ = worker.communicate(timeout=90)
+    finally:
+        for p in (orch, worker):
+            if p.poll() is None:
+                p.kill()
+    assert orch.returncode == 0, out_o
+    assert worker.returncode == 0, out_w
+    assert "orchestrator complete" in out_o
+    # the worker's state manager stored the posts
+    jsonls = list(tmp_path.rglob("posts.jsonl"))
+    assert jsonls, out_w
+    total = sum(p.read_bytes().count(b"\n") for p in jsonls)
+    assert total == 2 * 10
