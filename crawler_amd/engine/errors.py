@@ -23,6 +23,11 @@ class WalkbackExhausted(Exception):
     the buffer; dapr/standalone.go:902)."""
 
 
+class ConnectionDropped(Exception):
+    """Transport-level failure: the pooled session must be destroyed and
+    recreated (HandleConnectionError, connection_pool.go:346-413)."""
+
+
 class PoolExhausted(Exception):
     """Connection pool has no available connections."""
 
@@ -57,3 +62,11 @@ def is_tdlib_400(err_msg: str) -> bool:
         or "400 USERNAME_INVALID" in err_msg
         or "no messages found in the chat" in err_msg
     )
+
+
+def is_connection_error(err_msg: str) -> bool:
+    """Transport-failure classification (the substrings the reference's
+    HandleConnectionError call sites key on)."""
+    m = err_msg.lower()
+    return ("connection" in m or "timeout" in m or "socket" in m
+            or "broken pipe" in m)

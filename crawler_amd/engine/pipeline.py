@@ -117,21 +117,35 @@ def run_for_channel_with_pool(pool, page: Page, sm, cfg, rw=None,
     """Checkout -> run -> release|retire (runner.go:506-561)."""
     client = pool.get_connection()
     retire = False
+    recreate = False
     try:
         return run_for_channel(client, page, sm, cfg, rw=rw, seen=seen,
                                mode_hooks=mode_hooks, rng=rng, now=now)
     except E.FloodWaitRetire:
         retire = True
         raise
+    except E.ConnectionDropped:
+        recreate = True
+        raise
+    except TelegramAPIError as err:
+        # transport errors escaping from mid-channel calls (comments
+        # fetch etc.) also destroy+recreate the session
+        if E.is_connection_error(str(err)):
+            recreate = True
+        raise
     finally:
         if retire:
             pool.retire_connection(client)
+        elif recreate:
+            pool.handle_connection_error(client)
         else:
             pool.release_connection(client)
 
 
 def _classify_api_error(err: Exception):
     msg = str(err)
+    if E.is_connection_error(msg):
+        raise E.ConnectionDropped(msg)
     secs, is_flood = E.parse_flood_wait_secs(msg)
     if is_flood:
         if secs >= E.FLOOD_WAIT_RETIRE_THRESHOLD_SECS:

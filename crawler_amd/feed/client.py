@@ -57,6 +57,7 @@ class FaultConfig:
     flood_wait_secs: int = 30             # injected retry-after
     long_flood_permille: int = 0          # odds of a >=300s FLOOD_WAIT
     invalid_channel_permille: int = 0     # odds a username 400s
+    conn_reset_permille: int = 0          # odds of a transport failure
 
 
 class SyntheticTelegramClient:
@@ -112,6 +113,8 @@ class SyntheticTelegramClient:
             raise TelegramAPIError(
                 f"[429] FLOOD_WAIT_{f.flood_wait_secs}"
             )
+        if f.conn_reset_permille and (h >> 20) % 1000 < f.conn_reset_permille:
+            raise TelegramAPIError("connection reset by peer")
 
     # -- TDLib facade (crawler/crawler.go:109-126 subset the engine uses) --
 

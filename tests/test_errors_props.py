@@ -119,3 +119,62 @@ def test_link_extraction_agrees_with_regex(s):
     msg = SynthMessage(content_type="messageText",
                        text=FormattedText(text=s))
     assert extract_channel_links(msg) == expected
+
+
+def test_connection_error_recreates_session(tmp_path):
+    """Transport failure destroys + recreates the pooled session
+    (HandleConnectionError, connection_pool.go:346-413): the pool stays
+    at full strength with a FRESH conn id, nothing is retired."""
+    from crawler_amd.config import CrawlerConfig
+    from crawler_amd.engine import LocalStateManager, Page
+    from crawler_amd.engine import errors as E
+    from crawler_amd.engine.pipeline import run_for_channel_with_pool
+    from crawler_amd.feed import FeedConfig, SyntheticFeed
+    from crawler_amd.feed.client import ConnectionPool, FaultConfig
+
+    cfg = CrawlerConfig(crawl_id="cr1", storage_root=str(tmp_path),
+                        min_users=1, disable_rate_limits=True)
+    feed = SyntheticFeed(FeedConfig(seed=3, universe=100,
+                                    posts_per_channel=10))
+    pool = ConnectionPool(
+        feed, 2, cfg.rate_limit, posts_per_channel=10,
+        disable_rate_limits=True,
+        faults=FaultConfig(conn_reset_permille=1000),  # always reset
+    )
+    before = pool.stats()
+    ids_before = set(pool.available.keys())
+    sm = LocalStateManager(cfg)
+    with pytest.raises(E.ConnectionDropped):
+        run_for_channel_with_pool(pool, Page(id="p", url="c0000000001"),
+                                  sm, cfg)
+    after = pool.stats()
+    assert after["available"] == before["available"]  # recreated, not lost
+    assert after.get("retired", 0) == before.get("retired", 0)
+    assert set(pool.available.keys()) != ids_before   # fresh session dir
+    sm.close()
+
+
+def test_connection_error_contained_by_runner(tmp_path):
+    """A transport failure marks the page error and the crawl proceeds
+    with the recreated session."""
+    from crawler_amd.config import CrawlerConfig
+    from crawler_amd.engine import LocalStateManager
+    from crawler_amd.engine.runner import StandaloneRunner
+    from crawler_amd.feed import FeedConfig, SyntheticFeed
+    from crawler_amd.feed.client import ConnectionPool, FaultConfig
+
+    cfg = CrawlerConfig(crawl_id="cr2", storage_root=str(tmp_path),
+                        min_users=1, sampling_method="channel",
+                        disable_rate_limits=True)
+    feed = SyntheticFeed(FeedConfig(seed=3, universe=100,
+                                    posts_per_channel=10))
+    # ~30% of API calls drop the connection
+    pool = ConnectionPool(
+        feed, 2, cfg.rate_limit, posts_per_channel=10,
+        disable_rate_limits=True,
+        faults=FaultConfig(conn_reset_permille=300),
+    )
+    runner = StandaloneRunner(cfg, LocalStateManager(cfg), pool)
+    stats = runner.run(["c%010d" % i for i in range(1, 9)])
+    assert stats["pages"] + stats["errors"] == 8
+    assert pool.stats()["available"] + pool.stats()["in_use"] == 2
