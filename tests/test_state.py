@@ -275,3 +275,17 @@ def test_concurrent_edge_claims_stress():
     [t.join() for t in ts]
     assert len(claimed) == 400
     assert len(set(claimed)) == 400
+
+
+def test_store_file_moves_into_media_layout(tmp_path):
+    """StoreFile (state interface parity): move a fetched file into
+    crawlID/media/channel/ (storageproviders.go paths)."""
+    sm = mk_sm(tmp_path)
+    src = tmp_path / "dl.tmp"
+    src.write_bytes(b"blobdata")
+    dst, name = sm.store_file("chanZ", str(src), "photo1.jpg")
+    assert name == "photo1.jpg"
+    assert not src.exists()
+    assert open(dst, "rb").read() == b"blobdata"
+    assert "/media/chanZ/" in dst
+    sm.close()
