@@ -55,3 +55,17 @@ def test_urls_required_unless_dapr_job():
     ok(platform="telegram", sampling_method="channel", mode="dapr-job")
     # random never needs URLs
     ok(platform="youtube", sampling_method="random", mode="")
+
+
+def test_generate_crawl_id_format():
+    """GenerateCrawlID 'YYYYMMDDHHMMSS' (common/utils.go:103-111)."""
+    import datetime as dt
+
+    from crawler_amd.config import generate_crawl_id
+
+    t = dt.datetime(2026, 3, 4, 5, 6, 7, tzinfo=dt.timezone.utc)
+    assert generate_crawl_id(t) == "20260304050607"
+    # default: now-based, 14 digits, parseable back
+    cid = generate_crawl_id()
+    assert len(cid) == 14 and cid.isdigit()
+    dt.datetime.strptime(cid, "%Y%m%d%H%M%S")
