@@ -753,10 +753,31 @@ class LocalStateManager(BaseStateManager):
             "id": media_id, "firstSeen": _now().isoformat(),
         }
 
+    MEDIA_CACHE_TTL_DAYS = 30  # daprstate.go:1252-1678 shard expiry
+
+    def _expire_media_cache(self) -> int:
+        """Drop cache entries older than the TTL (the reference's
+        sharded-cache cleanup that runs on Close)."""
+        cutoff = _now() - _dt.timedelta(days=self.MEDIA_CACHE_TTL_DAYS)
+        dropped = 0
+        for k in list(self.media_cache):
+            v = self.media_cache[k]
+            try:
+                seen = _dt.datetime.fromisoformat(v["firstSeen"])
+            except (KeyError, TypeError, ValueError):
+                continue  # legacy entry without a timestamp: keep
+            if seen.tzinfo is None:
+                seen = seen.replace(tzinfo=UTC)
+            if seen < cutoff:
+                del self.media_cache[k]
+                dropped += 1
+        return dropped
+
     def close(self) -> None:
         for f in self._post_files.values():
             f.close()
         self._post_files.clear()
+        self._expire_media_cache()
         if self._native_sink is not None:
             self._native_sink.close()
             self._native_sink = None

@@ -289,3 +289,24 @@ def test_store_file_moves_into_media_layout(tmp_path):
     assert open(dst, "rb").read() == b"blobdata"
     assert "/media/chanZ/" in dst
     sm.close()
+
+
+def test_media_cache_ttl_expiry_on_close(tmp_path):
+    """30-day media-cache expiry at Close (daprstate.go:1252-1678)."""
+    import datetime as dt
+
+    sm = mk_sm(tmp_path)
+    sm.mark_media_as_processed("fresh1")
+    old = (dt.datetime.now(dt.timezone.utc)
+           - dt.timedelta(days=31)).isoformat()
+    sm.media_cache["stale1"] = {"id": "stale1", "firstSeen": old}
+    sm.media_cache["legacy"] = {"id": "legacy"}  # no timestamp: kept
+    sm.close()
+    import json
+    cache = json.load(open(sm._media_cache_path()))
+    sm2 = mk_sm(tmp_path)
+    sm2.media_cache = cache
+    assert sm2.has_processed_media("fresh1")
+    assert not sm2.has_processed_media("stale1")
+    assert sm2.has_processed_media("legacy")
+    sm2.close()
