@@ -138,3 +138,43 @@ def test_media_flags_skip_media(small_batch):
         assert obj["thumb_url"] == ""  # skip_media drops thumbs
         if "messageDocument" in obj["post_type"]:
             assert obj["media_url"].startswith("AgAD")
+
+
+def test_max_comments_cap_through_pipeline(tmp_path):
+    """--max-comments caps the per-message comments fetch
+    (GetMessageComments pagination cap, telegramutils.go:311-747)."""
+    import json
+
+    from crawler_amd.config import CrawlerConfig
+    from crawler_amd.engine import LocalStateManager, Page
+    from crawler_amd.engine.pipeline import run_for_channel_with_pool
+    from crawler_amd.feed.client import ConnectionPool
+
+    feed = SyntheticFeed(FeedConfig(seed=17, universe=60,
+                                    posts_per_channel=30,
+                                    comment_rate=1.0,
+                                    max_comments_per_post=9))
+    results = {}
+    for cap in (3, -1):
+        cfg = CrawlerConfig(crawl_id=f"mc{cap}",
+                            storage_root=str(tmp_path), min_users=1,
+                            disable_rate_limits=True, max_comments=cap)
+        pool = ConnectionPool(feed, 1, cfg.rate_limit,
+                              posts_per_channel=30,
+                              disable_rate_limits=True)
+        sm = LocalStateManager(cfg)
+        res = run_for_channel_with_pool(
+            pool, Page(id="p", url="c0000000002"), sm, cfg)
+        assert res.posts_stored == 30
+        sm.close()
+        path = (tmp_path / f"mc{cap}" / "c0000000002" / "posts" /
+                "posts.jsonl")
+        lens = []
+        for line in path.read_bytes().splitlines():
+            obj = json.loads(line)
+            if obj["comments"] is not None:
+                lens.append(len(obj["comments"]))
+        assert lens, "feed must produce commented posts"
+        results[cap] = max(lens)
+    assert results[3] == 3          # capped exactly at --max-comments
+    assert 4 <= results[-1] <= 9    # uncapped exceeds the cap
