@@ -12,7 +12,9 @@
 //   mu_     guards q_, pending_, ticket_pending_, ticket_errors_, stop_
 //   fd_mu_  guards fds_, lru_ (workers open files in parallel)
 //   errors_/bytes_ are atomics; ticket_seq_ is an atomic counter.
-// Eviction runs only when no write is in flight (drain / sync batch).
+// Eviction is refcount-guarded: fds a worker holds mid-write are
+// pinned (FdEntry.refs) and skipped, so drain()/eviction is safe even
+// with concurrent producers (a TSAN-found hole; see fd_acquire).
 #pragma once
 
 #include <fcntl.h>
@@ -110,7 +112,7 @@ class FanoutCore {
     std::unique_lock<std::mutex> lk(mu_);
     done_cv_.wait(lk, [this] { return pending_ == 0; });
     std::unique_lock<std::mutex> flk(fd_mu_);
-    for (auto& kv : fds_) ::fsync(kv.second);
+    for (auto& kv : fds_) ::fsync(kv.second.fd);
   }
 
   void close() {
@@ -124,7 +126,7 @@ class FanoutCore {
     for (auto& t : workers_) t.join();
     workers_.clear();
     std::unique_lock<std::mutex> flk(fd_mu_);
-    for (auto& kv : fds_) ::close(kv.second);
+    for (auto& kv : fds_) ::close(kv.second.fd);
     fds_.clear();
     lru_.clear();
   }
@@ -137,12 +139,18 @@ class FanoutCore {
 
  private:
   // Worker-side open with its own lock: O_CREAT opens + mkdirs are
-  // metadata-heavy; resolving in the pool parallelizes them. -1 = fail.
-  int fd_for(const std::string& path) {
+  // metadata-heavy; resolving in the pool parallelizes them. The entry
+  // is REFCOUNTED: acquire pins it against eviction until fd_release
+  // (TSAN found the hole: a concurrent producer's drain() could evict
+  // an fd another worker was mid-write(2) on). -1 = open failure.
+  int fd_acquire(const std::string& path) {
     {
       std::unique_lock<std::mutex> lk(fd_mu_);
       auto it = fds_.find(path);
-      if (it != fds_.end()) return it->second;
+      if (it != fds_.end()) {
+        ++it->second.refs;
+        return it->second.fd;
+      }
     }
     int fd = ::open(path.c_str(), O_WRONLY | O_CREAT | O_APPEND, 0644);
     if (fd < 0 && errno == ENOENT) {
@@ -154,26 +162,40 @@ class FanoutCore {
     auto it = fds_.find(path);
     if (it != fds_.end()) {  // raced: another worker opened it first
       ::close(fd);
-      return it->second;
+      ++it->second.refs;
+      return it->second.fd;
     }
-    fds_[path] = fd;
+    fds_[path] = FdEntry{fd, 1};
     lru_.push_back(path);
     return fd;
   }
 
-  // Runs ONLY when no write is in flight (a same-batch eviction could
-  // close an fd a queued write still holds).
+  void fd_release(const std::string& path) {
+    std::unique_lock<std::mutex> lk(fd_mu_);
+    auto it = fds_.find(path);
+    if (it != fds_.end() && it->second.refs > 0) --it->second.refs;
+  }
+
+  // Evict down to the cap, skipping fds a worker currently holds
+  // (refs > 0). Safe to call concurrently with writes.
   void evict_to_cap() {
     std::unique_lock<std::mutex> lk(fd_mu_);
+    size_t i = 0;
     while (fds_.size() > static_cast<size_t>(max_open_) &&
-           !lru_.empty()) {
-      std::string old = lru_.front();          // oldest first; O_APPEND
-      lru_.erase(lru_.begin());                // makes reopen safe
-      auto oit = fds_.find(old);
-      if (oit != fds_.end()) {
-        ::close(oit->second);
-        fds_.erase(oit);
+           i < lru_.size()) {
+      const std::string& old = lru_[i];        // oldest first; O_APPEND
+      auto oit = fds_.find(old);               // makes reopen safe
+      if (oit == fds_.end()) {
+        lru_.erase(lru_.begin() + i);
+        continue;
       }
+      if (oit->second.refs > 0) {              // pinned: skip
+        ++i;
+        continue;
+      }
+      ::close(oit->second.fd);
+      fds_.erase(oit);
+      lru_.erase(lru_.begin() + i);
     }
   }
 
@@ -190,7 +212,7 @@ class FanoutCore {
         t = std::move(q_.front());
         q_.pop();
       }
-      int fd = fd_for(t.path);
+      int fd = fd_acquire(t.path);
       const char* p = t.data;
       size_t left = t.len;
       bool ok = (fd >= 0);
@@ -204,6 +226,7 @@ class FanoutCore {
         p += w;
         left -= static_cast<size_t>(w);
       }
+      if (fd >= 0) fd_release(t.path);
       if (ok)
         bytes_.fetch_add(t.len);
       else if (t.ticket == 0)
@@ -230,8 +253,12 @@ class FanoutCore {
   std::unordered_map<uint64_t, size_t> ticket_errors_;    // under mu_
   std::atomic<size_t> errors_;
   std::atomic<size_t> bytes_;
+  struct FdEntry {
+    int fd;
+    int refs;  // workers mid-write pin the entry against eviction
+  };
   std::mutex fd_mu_;  // guards fds_/lru_ (workers open in parallel)
-  std::unordered_map<std::string, int> fds_;
+  std::unordered_map<std::string, FdEntry> fds_;
   std::vector<std::string> lru_;
   std::queue<Task> q_;
   std::mutex mu_;
