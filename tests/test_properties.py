@@ -134,3 +134,57 @@ def test_post_jsonl_always_parses(desc, title, views):
     assert obj["description"] == desc
     assert obj["channel_data"]["channel_name"] == title
     assert obj["view_count"] == views
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.lists(
+    st.tuples(
+        hazard_text,                                   # body text
+        st.sampled_from(["messageText", "messagePhoto", "messagePoll",
+                         "messageSticker", "messageDocument"]),
+        st.integers(0, 2**31 - 1),                     # views
+        st.integers(0, 10_000),                        # forwards
+        st.dictionaries(st.sampled_from(["👍", "🔥", "❤"]),
+                        st.integers(1, 1000), max_size=3),
+    ),
+    min_size=1, max_size=12,
+))
+def test_pack_unpack_roundtrip(rows):
+    """SoA batch pack -> unpack returns the same message fields
+    (the packer is what every GPU test trusts for inputs)."""
+    from crawler_amd.ops import batch as B
+    from crawler_amd.ops import golden as G
+
+    msgs = []
+    for k, (text, ct, views, fwd, reacts) in enumerate(rows):
+        kw = {}
+        if ct == "messageText":
+            kw["text"] = G.FormattedText(text=text)
+        elif ct == "messagePhoto":
+            kw["caption"] = G.FormattedText(text=text)
+        elif ct == "messagePoll":
+            kw["poll_question"] = text
+        elif ct == "messageDocument":
+            kw["document_name"] = text
+        msgs.append(G.SynthMessage(
+            chat_id=-1001000000042, msg_id=(k + 1) << 20,
+            date=1_700_000_000 + k, content_type=ct, views=views,
+            forwards=fwd, reactions=dict(reacts),
+            poster_handle=f"user{k:04d}", **kw,
+        ))
+    ch = [B.ChannelRow(chat_id=-1001000000042, username="propchan",
+                       title="P", member_count=1, post_count=len(msgs),
+                       total_views=0)]
+    batch = B.pack(msgs, ch, [0] * len(msgs))
+    for i, orig in enumerate(msgs):
+        back = B.unpack_message(batch, i)
+        assert back.content_type == orig.content_type
+        assert back.views == orig.views and back.forwards == orig.forwards
+        assert back.reactions == orig.reactions
+        assert back.msg_id == orig.msg_id and back.date == orig.date
+        if orig.text:
+            assert back.text.text == orig.text.text
+        if orig.caption:
+            assert back.caption.text == orig.caption.text
+        assert back.poll_question == orig.poll_question
+        assert back.document_name == orig.document_name
