@@ -291,7 +291,15 @@ def main():
         elapsed = float(e.item())
 
     posts_done = args.steps * chunk_posts
-    value = world * posts_done / elapsed
+    if world > 1:
+        # whole-job aggregate: SUM the per-rank work (ranks differ in
+        # --platform mixed, where telegram and youtube step sizes differ)
+        pd = torch.tensor([float(posts_done)], device=device)
+        torch.distributed.all_reduce(pd)  # default op = SUM
+        total_done = float(pd.item())
+    else:
+        total_done = float(posts_done)
+    value = total_done / elapsed
 
     if sink_f is not None:
         sink_f.close()
@@ -315,7 +323,7 @@ def main():
                           "(BASELINE config #2, weak-scaled)"
                           if args.platform == "telegram" else
                           f"{args.platform} crawl (BASELINE config #4 dispatch)"),
-                "global_batch": world * chunk_posts,
+                "global_batch": int(total_done / args.steps),
                 "seq_len": 0,
                 "platform": args.platform,
                 "parallelism": f"dp{world} (channel-sharded, RCCL discovery "
