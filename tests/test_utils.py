@@ -96,3 +96,30 @@ def test_debug_server_maybe_start_disabled():
 
     assert maybe_start(0) is None
     assert maybe_start(None) is None
+
+
+def test_latency_tracker_window_trims():
+    from crawler_amd.utils.metrics import LatencyTracker
+
+    lt = LatencyTracker(window=8)
+    for i in range(20):
+        lt.observe(float(i))
+    assert len(lt.samples) == 8
+    assert lt.samples == [float(i) for i in range(12, 20)]
+    # p50 over the surviving window only
+    assert lt.percentile(50) in (15.0, 16.0)
+
+
+def test_float_encoding_go_rules():
+    """Go encoding/json: NaN/Inf are unsupported values; integral floats
+    print bare (model/post.py _enc_float)."""
+    from crawler_amd.models.post import _enc_float
+
+    assert _enc_float(2.0) == "2"
+    assert _enc_float(0.5) == "0.5"
+    assert _enc_float(-3.0) == "-3"
+    import pytest
+
+    for bad in (float("nan"), float("inf"), float("-inf")):
+        with pytest.raises(ValueError):
+            _enc_float(bad)
