@@ -471,3 +471,17 @@ def test_cli_orchestrator_worker_two_processes(tmp_path):
     assert jsonls, out_w
     total = sum(p.read_bytes().count(b"\n") for p in jsonls)
     assert total == 2 * 10
+
+
+def test_queue_drain_returns_fifo_and_empties():
+    from crawler_amd.parallel.queue import InMemoryStore, StoreQueue
+
+    store = InMemoryStore()
+    q = StoreQueue(store, "t-dr")
+    for i in range(7):
+        q.publish(f"m{i}")
+    assert q.size() == 7
+    got = q.drain()
+    assert got == [f"m{i}" for i in range(7)]
+    assert q.size() == 0
+    assert q.drain() == []
