@@ -182,3 +182,25 @@ def test_client_pool_rotation():
         pool.get()
     assert pool.retired >= 2
     assert 3 <= len(made) <= 6  # ~200/50 rotations
+
+
+def test_build_corpus_deterministic_bytes():
+    """Same seed -> byte-identical encoded corpus (the bench relies on
+    deterministic regeneration across processes/boxes)."""
+    import datetime as dt
+
+    from crawler_amd.youtube.batch import build_corpus, encode_yt_batch
+    from crawler_amd.youtube.synth import SyntheticYouTubeIndex
+
+    NOW = dt.datetime(2026, 1, 1, tzinfo=dt.timezone.utc)
+    a = build_corpus(SyntheticYouTubeIndex(seed=99), 50,
+                     crawl_label="det")
+    b = build_corpus(SyntheticYouTubeIndex(seed=99), 50,
+                     crawl_label="det")
+    la, _ = encode_yt_batch(a, now=NOW)
+    lb, _ = encode_yt_batch(b, now=NOW)
+    assert b"".join(la) == b"".join(lb)
+    c = build_corpus(SyntheticYouTubeIndex(seed=100), 50,
+                     crawl_label="det")
+    lc, _ = encode_yt_batch(c, now=NOW)
+    assert b"".join(la) != b"".join(lc)
