@@ -197,10 +197,10 @@ def test_build_corpus_deterministic_bytes():
                      crawl_label="det")
     b = build_corpus(SyntheticYouTubeIndex(seed=99), 50,
                      crawl_label="det")
-    la, _ = encode_yt_batch(a, now=NOW)
-    lb, _ = encode_yt_batch(b, now=NOW)
+    la = encode_yt_batch(a, now=NOW)
+    lb = encode_yt_batch(b, now=NOW)
     assert b"".join(la) == b"".join(lb)
     c = build_corpus(SyntheticYouTubeIndex(seed=100), 50,
                      crawl_label="det")
-    lc, _ = encode_yt_batch(c, now=NOW)
+    lc = encode_yt_batch(c, now=NOW)
     assert b"".join(la) != b"".join(lc)
