@@ -211,3 +211,23 @@ def test_build_link_private_channel_empty():
     link, public_id = build_telegram_link_and_message_id("", m)
     assert link == ""
     assert public_id == 5
+
+
+def test_extract_nonoverlap_cursor_go_findall():
+    """Go FindAllSubmatch non-overlap: after a match the scan resumes at
+    the END of the match — the embedded 't.me/xyzzy' starting inside the
+    first match is skipped, later candidates still match."""
+    from crawler_amd.ops import golden as G
+    import re as _re
+
+    ft = G.FormattedText(text="t.me/abcdet.me/xyzzy and t.me/tttttme")
+    names = G.extract_channel_links(
+        G.SynthMessage(content_type="messageText", text=ft))
+    # exact oracle pin: Python finditer has the same non-overlap cursor
+    pat = _re.compile(r"(https?://)?t\.me/([a-zA-Z][a-zA-Z0-9_]{4,31})")
+    expect = []
+    for m in pat.finditer(ft.text):
+        nm = m.group(2).lower()
+        if nm not in G.RESERVED_PATHS and nm not in expect:
+            expect.append(nm)
+    assert names == expect == ["abcdet", "tttttme"]
