@@ -340,3 +340,33 @@ def test_outlinks_cached_chat_id_fast_path(tmp_path):
     got = RW.validate_outlinks(links("cachedc"), "src", sm, rw, client,
                                cfg)
     assert got == {"cachedc": True} and client.calls == []
+
+
+def test_tandem_tail_skips_self_and_invalid(tmp_path):
+    """Tandem pending-edge stream applies the same pre-filters as
+    standard mode: self-links, regex-invalid names and cached-invalid
+    channels never become pending edges (runner.go:1252-1306)."""
+    cfg, sm, rw = mk(tmp_path)
+    rw.mark_invalid_channel("badcha")
+    owner = mk_page("selfchan", seq="sq-t")
+    links = [DiscoveredLink(name=n, source_type="message_text")
+             for n in ("selfchan", "ab", "badcha", "goodch1", "goodch2")]
+    bid = RW.tandem_tail(owner, links, sm, rw, cfg, random.Random(1))
+    assert bid is not None
+    edges = rw.edges_of_batch(bid)
+    assert {e.destination_channel for e in edges} == \
+        {"goodch1", "goodch2"}
+    assert rw.pending_batches[bid].status == "closed"
+
+
+def test_tandem_tail_forced_walkback_when_nothing_survives(tmp_path):
+    cfg, sm, rw = mk(tmp_path)
+    seed_discovered(sm, ["backstop"])
+    owner = mk_page("selfchan", seq="sq-u")
+    links = [DiscoveredLink(name="selfchan", source_type="message_text")]
+    bid = RW.tandem_tail(owner, links, sm, rw, cfg, random.Random(2))
+    assert bid is None                      # no batch opened
+    pages = rw.get_pages(5)
+    assert len(pages) == 1 and pages[0].url == "backstop"
+    wb = [e for e in edges_for(rw, "sq-u") if e.walkback]
+    assert len(wb) == 1
