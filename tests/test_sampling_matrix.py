@@ -69,3 +69,24 @@ def test_generate_crawl_id_format():
     cid = generate_crawl_id()
     assert len(cid) == 14 and cid.isdigit()
     dt.datetime.strptime(cid, "%Y%m%d%H%M%S")
+
+
+def test_parse_time_ago_units():
+    """parseTimeAgo units (main.go parseTimeAgo): Nd / Nw / Nm / Ny and
+    hour forms resolve to a past timestamp."""
+    import datetime as dt
+
+    from crawler_amd.config import parse_time_ago
+
+    now = dt.datetime(2026, 6, 15, 12, 0, 0, tzinfo=dt.timezone.utc)
+    assert parse_time_ago("30d", now=now) == now - dt.timedelta(days=30)
+    assert parse_time_ago("2w", now=now) == now - dt.timedelta(weeks=2)
+    assert parse_time_ago("6h", now=now) == now - dt.timedelta(hours=6)
+    m = parse_time_ago("3m", now=now)
+    assert (now - m).days in (89, 90, 91, 92)        # calendar months
+    y = parse_time_ago("1y", now=now)
+    assert (now - y).days in (365, 366)
+    import pytest as _pt
+
+    with _pt.raises(ValueError):
+        parse_time_ago("5x", now=now)
