@@ -133,3 +133,25 @@ def test_jsonl_line_roundtrip():
     assert obj["description"] == "hello <world> & 'stuff'\nnewline"
     assert "\\u003cworld\\u003e" in line  # raw bytes show Go-style escapes
     assert obj["outlinks"] == ["abcde", "fghij"]
+
+
+def test_reactions_sorted_map_keys():
+    """Go encoding/json sorts map keys; reactions must serialize in
+    byte-sorted emoji order regardless of insertion order."""
+    import json as _json
+
+    p = Post(
+        post_link="x", channel_id="1", post_uid="u", url="x",
+        published_at=dt.datetime(2024, 1, 1, tzinfo=UTC),
+        created_at=dt.datetime(2024, 1, 1, tzinfo=UTC),
+        reactions={"🔥": 2, "❤": 5, "👍": 1},
+    )
+    line = p.to_jsonl()
+    obj = _json.loads(line)
+    assert obj["reactions"] == {"🔥": 2, "❤": 5, "👍": 1}
+    # raw byte order: keys appear sorted by their UTF-8 bytes
+    raw = line[line.index('"reactions"'):]
+    pos = {e: raw.index(_json.dumps(e, ensure_ascii=False))
+           for e in ("❤", "👍", "🔥")}
+    order = sorted(pos, key=pos.get)
+    assert order == sorted(order, key=lambda s: s.encode())
