@@ -126,13 +126,19 @@ def main():
     # Double-buffered pinned host ring for the JSONL output (the storage
     # boundary): D2H of step k overlaps the kernels of step k+1 on a
     # separate copy stream. Worst case ~2.6KB/post.
+    host_alloc_state = {"pinned": True}
+
     def alloc_host(nbytes):
         # 8 ranks x 2 buffers of pinned memory can brush against host
         # lockable-memory limits; fall back to pageable rather than fail
+        # — but RECORD it (result JSON carries pinned per rank: a
+        # pageable fallback silently changes what is measured)
         try:
             return torch.empty(nbytes, dtype=torch.uint8, pin_memory=True)
         except RuntimeError:
-            log("pinned allocation failed; using pageable host memory")
+            host_alloc_state["pinned"] = False
+            print(f"[bench rank{rank}] pinned allocation failed; "
+                  "using pageable host memory", file=sys.stderr, flush=True)
             return torch.empty(nbytes, dtype=torch.uint8)
 
     # worst-case line sizes differ by platform: telegram ~2.1KB/post,
@@ -166,31 +172,21 @@ def main():
                 out_t = res.out
                 new_mask = seen.claim(res)
                 if world > 1:
+                    # count-sized exchange + bloom OR-union — no cap
+                    # (round 1 silently truncated at 64k hashes/step)
                     flat = new_mask.flatten().bool()
                     new_hashes = res.link_hash.flatten()[flat]
-                    cap = 65536
-                    buf = torch.zeros(cap, dtype=torch.int64, device=device)
-                    k = min(new_hashes.numel(), cap)
-                    buf[:k] = new_hashes[:k]
-                    gathered = [torch.empty_like(buf) for _ in range(world)]
-                    torch.distributed.all_gather(gathered, buf)
-                    for r, g in enumerate(gathered):
-                        if r != rank:
-                            seen.insert_hashes(g[g != 0])
+                    seen.merge_remote(new_hashes, torch.distributed,
+                                      world)
                 nd = seen.new_count()  # syncs within the compute stream
             else:
                 out_t, _off, _len = gpu.yt_parse_encode(chunk, now=now)
                 if world > 1:
                     # mixed dispatch still exchanges (empty) discovery
                     # buffers so collectives stay symmetric across ranks
-                    buf = torch.zeros(65536, dtype=torch.int64,
-                                      device=device)
-                    gathered = [torch.empty_like(buf)
-                                for _ in range(world)]
-                    torch.distributed.all_gather(gathered, buf)
-                    for r, g in enumerate(gathered):
-                        if r != rank:
-                            seen.insert_hashes(g[g != 0])
+                    empty = torch.zeros(0, dtype=torch.int64,
+                                        device=device)
+                    seen.merge_remote(empty, torch.distributed, world)
                 nd = 0
         nbytes = out_t.numel()
         assert nbytes <= pinned[slot].numel(), (
@@ -290,6 +286,26 @@ def main():
         torch.distributed.all_reduce(e, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(e.item())
 
+    # per-rank hygiene record: pinned-or-pageable + achieved D2H GB/s
+    # (JSONL bytes moved device->host inside the timed region / elapsed)
+    timed_bytes = args.steps * (total_out_bytes
+                                // max(1, args.warmup + args.steps))
+    my_d2h_gbs = timed_bytes / elapsed / 1e9
+    if world > 1:
+        info = torch.tensor(
+            [1.0 if host_alloc_state["pinned"] else 0.0, my_d2h_gbs],
+            device=device)
+        g_info = [torch.empty_like(info) for _ in range(world)]
+        torch.distributed.all_gather(g_info, info)
+        per_rank_host = [
+            {"pinned": bool(g[0].item() > 0.5),
+             "d2h_gbs": round(float(g[1].item()), 2)}
+            for g in g_info
+        ]
+    else:
+        per_rank_host = [{"pinned": host_alloc_state["pinned"],
+                          "d2h_gbs": round(my_d2h_gbs, 2)}]
+
     posts_done = args.steps * chunk_posts
     if world > 1:
         # whole-job aggregate: SUM the per-rank work (ranks differ in
@@ -337,6 +353,8 @@ def main():
                 "p50_channel_latency_ms": p50_channel_ms,
                 "new_discoveries": new_discoveries,
                 "sink": args.sink,
+                "pinned": all(r["pinned"] for r in per_rank_host),
+                "per_rank_host": per_rank_host,
             },
         }
         print(json.dumps(result), flush=True)

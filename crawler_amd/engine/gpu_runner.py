@@ -28,6 +28,7 @@ import numpy as np
 import torch
 
 from ..feed.synth import SyntheticFeed
+from ..parallel import collectives as C
 from .state import LocalStateManager, Page
 
 
@@ -52,6 +53,7 @@ class GpuCrawlEngine:
         self.timings = collections.defaultdict(float)  # phase seconds
         self.stats = {"pages": 0, "posts": 0, "jsonl_bytes": 0,
                       "discovered": 0, "deadends": 0}
+        self.last_deadends = set()
 
     # ---- helpers ----
 
@@ -79,6 +81,10 @@ class GpuCrawlEngine:
         valid = [(u, self._cid_of(u)) for u in usernames]
         bad = [u for u, c in valid if c is None]
         ok = [(u, c) for u, c in valid if c is not None]
+        # deadend channels of THIS call (invalid username or zero posts
+        # after filters) — run() uses this to set page status so the
+        # add_layer deadend-replacement budget activates (base.go:284)
+        self.last_deadends = set(bad)
         for u in bad:
             self.stats["deadends"] += 1
         for ci, i in enumerate(range(0, len(ok), self.chunk_channels)):
@@ -139,6 +145,12 @@ class GpuCrawlEngine:
                     # when a crash forces the layer to re-process
                     self.sm.truncate_posts(uname)
                     items.append((uname, lo, hi))
+                else:
+                    # valid channel but every post filtered out -> the
+                    # CPU pipeline's deadend status (runner.py status
+                    # machine; ref isChannelActiveWithinPeriod)
+                    self.last_deadends.add(uname)
+                    self.stats["deadends"] += 1
             # one fan-out call for the whole chunk: the native sink
             # (crawler_amd/native) appends all channels in parallel with
             # the GIL released; the ticket is awaited when this ring slot
@@ -216,36 +228,49 @@ class GpuCrawlEngine:
             sm.initialize(seed_urls)
         depth = 0
         while True:
-            layer = [p for p in sm.get_layer_by_depth(depth)
-                     if p.status == "unfetched"]
-            if not layer:
+            all_layer = sm.get_layer_by_depth(depth)
+            if not all_layer:
                 break
-            names = [p.url for p in layer]
-            if comm is not None:
-                rank = dist.get_rank()
-                world = dist.get_world_size()
-                mine = names[rank::world]
-            else:
-                mine = names
+            # Resume fix: a depth whose pages are all
+            # fetched (crash happened after its save_state) must NOT end
+            # the crawl — skip processing and advance to the next depth,
+            # like the CPU runner's all_layer check.
+            layer = [p for p in all_layer if p.status == "unfetched"]
             import time as _time
-            discovered, _ = self.process_channels(mine, as_arrays=True)
-            t0 = _time.perf_counter()
-            for p in layer:
-                p.status = "fetched"
-                sm.update_page(p)
-            if comm is not None:
-                # exchange discoveries (names, fixed-width padded):
-                # materialize strings only for the cross-rank hop
-                w = discovered.shape[1] if len(discovered) else 32
-                blob = discovered.tobytes().decode("ascii", "replace")
-                as_strs = [blob[i * w:i * w + 32].rstrip("\x00")
-                           for i in range(len(discovered))]
-                names_x = self._allgather_names(as_strs)
-                discovered = np.zeros((len(names_x), 32), dtype=np.uint8)
-                for i, nm in enumerate(names_x):
-                    b = nm.encode()[:32]
-                    discovered[i, :len(b)] = np.frombuffer(
-                        b, dtype=np.uint8)
+            if layer:
+                names = [p.url for p in layer]
+                if comm is not None:
+                    rank = dist.get_rank()
+                    world = dist.get_world_size()
+                    mine = names[rank::world]
+                else:
+                    mine = names
+                discovered, _ = self.process_channels(mine,
+                                                      as_arrays=True)
+                dead = set(self.last_deadends)
+                t0 = _time.perf_counter()
+                if comm is not None:
+                    # exchange discoveries as packed uint8 rows on the
+                    # backend's device (RCCL: cuda; gloo: cpu) —
+                    # count-sized, no cap
+                    rows = torch.from_numpy(
+                        np.ascontiguousarray(discovered))
+                    out_rows = C.allgather_rows(
+                        rows, dist, world, device=self.device)
+                    discovered = out_rows.numpy()
+                    # deadend statuses must agree across ranks: the
+                    # next layer's deadend-replacement budget is
+                    # computed independently by every rank
+                    dead = set(C.rows_to_names(C.allgather_rows(
+                        C.names_to_rows(sorted(dead)), dist, world,
+                        device=self.device)))
+                for p in layer:
+                    p.status = ("deadend" if p.url in dead
+                                else "fetched")
+                    sm.update_page(p)
+            else:
+                discovered = np.zeros((0, 32), dtype=np.uint8)
+                t0 = _time.perf_counter()
             if (self.cfg.sampling_method == "snowball"
                     and (self.cfg.max_depth < 0
                          or depth < self.cfg.max_depth)
@@ -298,31 +323,3 @@ class GpuCrawlEngine:
         sm.save_state()
         sm.close()
         return dict(self.stats)
-
-    @staticmethod
-    def _allgather_names(names: List[str], width: int = 32) -> List[str]:
-        import torch.distributed as dist
-
-        world = dist.get_world_size()
-        n = len(names)
-        counts = torch.tensor([n], dtype=torch.int64)
-        all_counts = [torch.zeros(1, dtype=torch.int64)
-                      for _ in range(world)]
-        dist.all_gather(all_counts, counts)
-        max_n = max(int(c.item()) for c in all_counts)
-        buf = torch.zeros(max_n, width, dtype=torch.uint8)
-        for i, name in enumerate(names):
-            b = name.encode()[:width]
-            buf[i, :len(b)] = torch.frombuffer(bytearray(b),
-                                               dtype=torch.uint8)
-        gathered = [torch.empty_like(buf) for _ in range(world)]
-        dist.all_gather(gathered, buf)
-        out = []
-        for r, g in enumerate(gathered):
-            cnt = int(all_counts[r].item())
-            arr = g.numpy()
-            for i in range(cnt):
-                row = arr[i]
-                ln = int((row != 0).sum())
-                out.append(bytes(row[:ln]).decode())
-        return out

@@ -437,6 +437,15 @@ class LocalStateManager(BaseStateManager):
 
     def save_state(self) -> None:
         os.makedirs(self._crawl_dir(), exist_ok=True)
+        # Durability contract (reference writes each post synchronously
+        # BEFORE status updates, daprstate.go:1106-1147): a checkpoint
+        # must not mark pages 'fetched' while their JSONL posts are still
+        # sitting in buffered handles or in the native sink's queue — a
+        # crash after the checkpoint would lose posts resume never
+        # re-emits. Flush everything first.
+        for f in self._post_files.values():
+            f.flush()
+        self.drain_post_writes()
         with self._lock:
             state = {
                 "layers": [

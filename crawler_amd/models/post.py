@@ -90,12 +90,34 @@ def _enc_time(t: Optional[_dt.datetime]) -> str:
 
 
 def _enc_float(v: float) -> str:
-    # Go: shortest round-trip representation; integral floats print bare.
+    # Go encoding/json floatEncoder (encode.go): shortest round-trip
+    # digits; 'f' format for 1e-6 <= |v| < 1e21, 'e' otherwise with the
+    # e-0X -> e-X exponent cleanup. Integral floats under 1e21 print
+    # bare. Python's repr produces the same shortest digit string but
+    # its own format choice (switches to exponent at |v| < 1e-4), so
+    # re-format per Go's rules.
     if v != v or v in (float("inf"), float("-inf")):
         raise ValueError("json: unsupported value: " + repr(v))
-    if float(v).is_integer() and abs(v) < 1e21:
+    v = float(v)
+    if v.is_integer() and abs(v) < 1e21:
         return str(int(v))
-    return repr(float(v))
+    s = repr(v)
+    if "e" not in s:
+        return s  # python chose 'f' here; Go agrees in this range
+    mant, _, es = s.partition("e")
+    exp = int(es)
+    if 1e-6 <= abs(v) < 1e21:
+        # python chose 'e' (only happens for |v| < 1e-4 among
+        # non-integral floats) but Go uses 'f': expand manually.
+        sign = "-" if mant.startswith("-") else ""
+        core = mant.lstrip("-")
+        digits = core.replace(".", "")
+        point = len(core.partition(".")[0]) + exp  # always <= 0 here
+        return sign + "0." + "0" * (-point) + digits
+    if -9 <= exp < 0:
+        # Go's exponent cleanup: e-07 -> e-7 (2-digit negatives only)
+        return f"{mant}e-{-exp}"
+    return s
 
 
 def _enc_opt_int(v: Optional[int]) -> str:

@@ -669,6 +669,31 @@ class SeenSet:
         m = int(cursor.item())  # syncs the stream
         return out_names[:m], out_hashes[:m]
 
+    def merge_remote(self, res_or_hashes, dist, world: int,
+                     group=None) -> int:
+        """Cross-rank seen-set union: exchange this
+        rank's newly-claimed hashes with every peer (count-sized
+        all-gather over RCCL — NO cap, round 1's silent 64k truncation
+        is gone) and insert the remote ones into the local table. Also
+        OR-unions the bloom pre-filter. Returns the number of remote
+        hashes inserted."""
+        from ..parallel import collectives as C
+
+        local = (res_or_hashes if isinstance(res_or_hashes, torch.Tensor)
+                 else res_or_hashes.link_hash.flatten())
+        per_rank = C.allgather_hashes(local, dist, world,
+                                      device=self.device, group=group)
+        rank = C._safe_rank(dist, group)
+        inserted = 0
+        for r, h in enumerate(per_rank):
+            if r == rank:
+                continue
+            h = h[h != 0].to(self.device)
+            self.insert_hashes(h)
+            inserted += int(h.numel())
+        C.bloom_union(self.bloom, dist, world, group=group)
+        return inserted
+
     def insert_hashes(self, hashes: torch.Tensor) -> None:
         """Bulk-insert merged remote hashes (post all-gather)."""
         n = hashes.numel()

@@ -1,0 +1,167 @@
+"""Backend-aware collective exchange helpers (the RCCL/xGMI data plane).
+
+Every cross-rank exchange in the framework goes through this module so
+tensor placement is decided in ONE place: NCCL (= RCCL on ROCm) requires
+device tensors, gloo requires CPU tensors. Round 1 shipped CPU tensors
+into NCCL collectives (crash at world>1); these helpers make that
+impossible — `collective_device` picks the device the backend demands
+and `check_collective_device` asserts before every call.
+
+Exchange shapes (SURVEY §2.6 comm table, ref daprstate.go:550-657 exact
+dedup semantics):
+- `allgather_rows`   — variable-count fixed-width byte rows (channel
+  names), count-sized: counts are all-gathered first, payloads padded to
+  the max count; NO silent cap.
+- `allgather_hashes` — variable-count int64 hash lists (seen-set claim
+  exchange), same count-sized protocol.
+- `bloom_union`      — bitwise-OR union of per-rank bloom segments.
+  NCCL/RCCL has no BOR reduction, so the union is an all-gather +
+  on-device `bitwise_or_` (8 MB bloom x 8 ranks = 64 MB over xGMI,
+  sub-millisecond); gloo uses the native BOR all-reduce.
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+
+
+def collective_device(dist, device: Optional[torch.device] = None,
+                      group=None) -> torch.device:
+    """The device tensors must live on for this process group's backend.
+
+    `device` is the caller's compute device (e.g. the engine's cuda:N);
+    it is used when the backend wants device tensors. Fake dists used in
+    unit tests (no get_backend) default to CPU.
+    """
+    try:
+        backend = str(dist.get_backend(group))
+    except (AttributeError, RuntimeError, ValueError):
+        return torch.device("cpu")
+    if backend.startswith("nccl"):
+        if device is not None and device.type == "cuda":
+            return device
+        return torch.device("cuda", torch.cuda.current_device())
+    return torch.device("cpu")
+
+
+def check_collective_device(t: torch.Tensor, dist, group=None) -> None:
+    """Fail fast (with a useful message) instead of crashing inside
+    NCCL: device tensors for nccl, CPU tensors for gloo."""
+    try:
+        backend = str(dist.get_backend(group))
+    except (AttributeError, RuntimeError, ValueError):
+        return
+    if backend.startswith("nccl") and t.device.type != "cuda":
+        raise RuntimeError(
+            f"collective tensor on {t.device} but backend is {backend}: "
+            "NCCL/RCCL requires device tensors")
+    if backend.startswith("gloo") and t.device.type != "cpu":
+        raise RuntimeError(
+            f"collective tensor on {t.device} but backend is {backend}: "
+            "gloo requires CPU tensors")
+
+
+def _allgather_counts(n: int, dist, world: int,
+                      device: torch.device, group=None) -> List[int]:
+    counts = torch.tensor([n], dtype=torch.int64, device=device)
+    check_collective_device(counts, dist, group)
+    out = [torch.zeros(1, dtype=torch.int64, device=device)
+           for _ in range(world)]
+    dist.all_gather(out, counts)
+    return [int(c.item()) for c in out]
+
+
+def allgather_rows(rows: torch.Tensor, dist, world: int,
+                   device: Optional[torch.device] = None,
+                   group=None) -> torch.Tensor:
+    """All-gather variable-count uint8[N, W] rows; returns the
+    concatenated uint8[sum(N_r), W] on `rows`'s original device.
+    Count-sized — no cap, no drop."""
+    dev = collective_device(dist, device, group)
+    width = rows.shape[1] if rows.ndim == 2 else 32
+    counts = _allgather_counts(rows.shape[0], dist, world, dev, group)
+    max_n = max(counts)
+    if max_n == 0:
+        return rows[:0]
+    buf = torch.zeros(max_n, width, dtype=torch.uint8, device=dev)
+    if rows.shape[0]:
+        buf[:rows.shape[0]] = rows.to(dev)
+    check_collective_device(buf, dist, group)
+    gathered = [torch.empty_like(buf) for _ in range(world)]
+    dist.all_gather(gathered, buf)
+    parts = [gathered[r][:counts[r]] for r in range(world) if counts[r]]
+    out = torch.cat(parts) if parts else buf[:0]
+    return out.to(rows.device)
+
+
+def allgather_hashes(local: torch.Tensor, dist, world: int,
+                     device: Optional[torch.device] = None,
+                     group=None) -> List[torch.Tensor]:
+    """All-gather variable-count int64 hash vectors; returns one tensor
+    per rank (caller usually skips its own), each on the collective
+    device. Count-sized — replaces round 1's silent 64k cap."""
+    dev = collective_device(dist, device, group)
+    counts = _allgather_counts(local.numel(), dist, world, dev, group)
+    max_n = max(counts)
+    if max_n == 0:
+        return [torch.zeros(0, dtype=torch.int64, device=dev)
+                for _ in range(world)]
+    buf = torch.zeros(max_n, dtype=torch.int64, device=dev)
+    if local.numel():
+        buf[:local.numel()] = local.to(dev)
+    check_collective_device(buf, dist, group)
+    gathered = [torch.empty_like(buf) for _ in range(world)]
+    dist.all_gather(gathered, buf)
+    return [gathered[r][:counts[r]] for r in range(world)]
+
+
+def bloom_union(bloom: torch.Tensor, dist, world: int,
+                group=None) -> None:
+    """In-place cross-rank OR-union of a bloom filter's int32 words
+    ('merge via bitwise-OR all-reduce on bloom segments').
+    Works on gloo (native BOR) and NCCL/RCCL (gather + device OR)."""
+    try:
+        backend = str(dist.get_backend(group))
+    except (AttributeError, RuntimeError, ValueError):
+        backend = "gloo"
+    check_collective_device(bloom, dist, group)
+    if backend.startswith("nccl"):
+        gathered = [torch.empty_like(bloom) for _ in range(world)]
+        dist.all_gather(gathered, bloom)
+        for r, g in enumerate(gathered):
+            if r != _safe_rank(dist, group):
+                bloom.bitwise_or_(g)
+    else:
+        dist.all_reduce(bloom, op=dist.ReduceOp.BOR)
+
+
+def _safe_rank(dist, group=None) -> int:
+    try:
+        return int(dist.get_rank(group))
+    except (AttributeError, RuntimeError, ValueError):
+        return 0
+
+
+def names_to_rows(names: List[str], width: int = 32,
+                  device=None) -> torch.Tensor:
+    """Pack python strings as zero-padded uint8[N, width] rows."""
+    out = torch.zeros(len(names), width, dtype=torch.uint8)
+    for i, name in enumerate(names):
+        b = name.encode()[:width]
+        if b:
+            out[i, :len(b)] = torch.frombuffer(bytearray(b),
+                                               dtype=torch.uint8)
+    return out.to(device) if device is not None else out
+
+
+def rows_to_names(rows: torch.Tensor) -> List[str]:
+    """Decode zero-padded uint8[N, W] rows to strings with ONE bulk
+    decode (a per-row bytes().decode() loop costs seconds at ~1M)."""
+    if rows.shape[0] == 0:
+        return []
+    arr = rows.cpu().numpy()
+    w = arr.shape[1]
+    lens = (arr != 0).sum(axis=1)
+    blob = arr.tobytes().decode("ascii", "replace")
+    return [blob[i * w:i * w + int(l)] for i, l in enumerate(lens)]

@@ -29,6 +29,7 @@ import uuid
 from typing import Callable, List, Tuple
 
 from ..engine.state import LocalStateManager, Page
+from . import collectives as C
 from . import messages as M
 from .queue import Heartbeats, StoreQueue
 
@@ -37,10 +38,11 @@ class OrchestratedCrawl:
     def __init__(self, cfg, sm: LocalStateManager, store, rank: int,
                  world: int,
                  process_fn: Callable[[List[str]], Tuple[List[str], int]],
-                 chunk_channels: int = 64, dist=None):
+                 chunk_channels: int = 64, dist=None, device=None):
         """process_fn(names) -> (discovered_names, posts_stored).
         `dist` is torch.distributed (injected so CPU tests can pass gloo
-        and unit tests can fake it)."""
+        and unit tests can fake it); `device` is the compute device used
+        for collective buffers when the backend is NCCL/RCCL."""
         self.cfg = cfg
         self.sm = sm
         self.rank = rank
@@ -51,44 +53,22 @@ class OrchestratedCrawl:
             import torch.distributed as dist  # noqa: PLC0415
 
         self.dist = dist
+        self.device = device
         self.work_q = StoreQueue(store, M.TOPIC_WORK_QUEUE)
         self.result_q = StoreQueue(store, M.TOPIC_RESULTS)
         self.heartbeats = Heartbeats(store)
         self.stats = {"chunks": 0, "pages": 0, "posts": 0,
                       "discovered": 0, "layers": 0}
 
-    # ---- name exchange (fixed-width padded all-gather) ----
+    # ---- name exchange (count-sized fixed-width all-gather, placed on
+    # the backend's device: RCCL wants device tensors, gloo wants CPU) ----
 
     def _allgather_names(self, names: List[str], width: int = 32
                          ) -> List[str]:
-        import torch
-
-        dist = self.dist
-        counts = torch.tensor([len(names)], dtype=torch.int64)
-        all_counts = [torch.zeros(1, dtype=torch.int64)
-                      for _ in range(self.world)]
-        dist.all_gather(all_counts, counts)
-        max_n = max(int(c.item()) for c in all_counts) or 1
-        buf = torch.zeros(max_n, width, dtype=torch.uint8)
-        for i, name in enumerate(names):
-            b = name.encode()[:width]
-            buf[i, :len(b)] = torch.frombuffer(bytearray(b),
-                                               dtype=torch.uint8)
-        gathered = [torch.empty_like(buf) for _ in range(self.world)]
-        dist.all_gather(gathered, buf)
-        out: List[str] = []
-        for r, g in enumerate(gathered):
-            cnt = int(all_counts[r].item())
-            arr = g.numpy()[:cnt]
-            if not cnt:
-                continue
-            lens = (arr != 0).sum(axis=1)
-            # one decode of the packed block, then slices (a per-row
-            # bytes().decode() loop costs seconds at ~1M names)
-            blob = arr.tobytes().decode("ascii", "replace")
-            out.extend(blob[i * width:i * width + int(l)]
-                       for i, l in enumerate(lens))
-        return out
+        rows = C.names_to_rows(names, width)
+        out_rows = C.allgather_rows(rows, self.dist, self.world,
+                                    device=self.device)
+        return C.rows_to_names(out_rows)
 
     # ---- one layer ----
 

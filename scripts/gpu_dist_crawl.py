@@ -65,6 +65,7 @@ def main():
         process_fn=lambda names: eng.process_channels(names),
         chunk_channels=args.chunk_channels,
         dist=dist if world > 1 else _SoloDist(),
+        device=torch.device("cuda", local_rank),
     )
     seeds = [feed.username_of(i) for i in range(args.seeds)]
     t0 = time.perf_counter()

@@ -155,3 +155,39 @@ def test_reactions_sorted_map_keys():
            for e in ("❤", "👍", "🔥")}
     order = sorted(pos, key=pos.get)
     assert order == sorted(order, key=lambda s: s.encode())
+
+
+def test_go_float_encoder_parity():
+    """Go encoding/json floatEncoder rules: 'f' format for
+    1e-6 <= |v| < 1e21, 'e' otherwise with e-0X -> e-X cleanup.
+    Expected strings computed with Go strconv semantics."""
+    from crawler_amd.models.post import _enc_float
+
+    cases = [
+        (0.0, "0"),
+        (1.5, "1.5"),
+        (-2.25, "-2.25"),
+        (0.001, "0.001"),
+        (0.0001, "0.0001"),
+        (1e-05, "0.00001"),         # python repr says 1e-05
+        (1.5e-05, "0.000015"),
+        (-3.25e-05, "-0.0000325"),
+        (1e-06, "0.000001"),        # boundary: still 'f' in Go
+        (9.9e-07, "9.9e-7"),        # below 1e-6: 'e' + cleanup
+        (1e-07, "1e-7"),
+        (-1e-07, "-1e-7"),
+        (1.25e-08, "1.25e-8"),
+        (1e-10, "1e-10"),           # 2-digit exponent: no cleanup
+        (5e-324, "5e-324"),         # denormal min
+        (1e21, "1e+21"),            # >= 1e21: 'e' form
+        (2.5e22, "2.5e+22"),
+        (1e20, "100000000000000000000"),  # integral < 1e21: bare
+        (123456789.5, "123456789.5"),
+    ]
+    for v, want in cases:
+        assert _enc_float(v) == want, (v, _enc_float(v), want)
+    import pytest
+
+    for bad in (float("nan"), float("inf"), float("-inf")):
+        with pytest.raises(ValueError):
+            _enc_float(bad)
