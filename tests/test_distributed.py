@@ -225,10 +225,11 @@ def _names_worker(rank, world, port, results_dir):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
     dist.init_process_group("gloo", rank=rank, world_size=world)
-    from crawler_amd.engine.gpu_runner import GpuCrawlEngine
+    from crawler_amd.parallel import collectives as C
 
     mine = [f"rank{rank}chan{i}" for i in range(2 + rank)]
-    merged = GpuCrawlEngine._allgather_names(mine)
+    merged = C.rows_to_names(C.allgather_rows(
+        C.names_to_rows(mine), dist, world))
     with open(os.path.join(results_dir, f"n{rank}.txt"), "w") as f:
         f.write(",".join(sorted(merged)))
     dist.destroy_process_group()
