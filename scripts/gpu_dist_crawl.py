@@ -38,16 +38,22 @@ def main():
     ap.add_argument("--universe", type=int, default=1_000_000)
     ap.add_argument("--storage", default="/tmp/gpu-dist-crawl")
     ap.add_argument("--chunk-channels", type=int, default=64)
+    ap.add_argument("--fake-engine", action="store_true",
+                    help="CPU dry run: gloo backend + a stub hot path "
+                         "(exercises the exact launcher wiring at "
+                         "world>1 without a GPU)")
+    ap.add_argument("--store-port", type=int, default=29761)
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-    torch.cuda.set_device(local_rank)
+    if not args.fake_engine:
+        torch.cuda.set_device(local_rank)
     if world > 1:
-        dist.init_process_group("nccl")
+        dist.init_process_group("gloo" if args.fake_engine else "nccl")
     host = os.environ.get("MASTER_ADDR", "127.0.0.1")
-    store = TCPStore(host, 29761, is_master=(rank == 0),
+    store = TCPStore(host, args.store_port, is_master=(rank == 0),
                      wait_for_workers=False)
 
     cfg = CrawlerConfig(
@@ -58,14 +64,20 @@ def main():
     feed = SyntheticFeed(FeedConfig(seed=2026, universe=args.universe,
                                     posts_per_channel=args.posts))
     sm = LocalStateManager(cfg)
-    eng = GpuCrawlEngine(cfg, sm, feed, device=f"cuda:{local_rank}",
-                         posts_per_channel=args.posts, chunk_channels=256)
+    if args.fake_engine:
+        eng = _FakeEngine(feed, args.posts)
+        device = None
+    else:
+        eng = GpuCrawlEngine(cfg, sm, feed, device=f"cuda:{local_rank}",
+                             posts_per_channel=args.posts,
+                             chunk_channels=256)
+        device = torch.device("cuda", local_rank)
     crawl = OrchestratedCrawl(
         cfg, sm, store, rank, world,
         process_fn=lambda names: eng.process_channels(names),
         chunk_channels=args.chunk_channels,
         dist=dist if world > 1 else _SoloDist(),
-        device=torch.device("cuda", local_rank),
+        device=device,
     )
     seeds = [feed.username_of(i) for i in range(args.seeds)]
     t0 = time.perf_counter()
@@ -93,6 +105,27 @@ class _SoloDist:
     @staticmethod
     def all_gather(out_list, t):
         out_list[0].copy_(t)
+
+
+class _FakeEngine:
+    """--fake-engine stub hot path: deterministic discovery fan-out so
+    a CPU dry run exercises the full launcher wiring (TCPStore queue,
+    chunk claiming, gloo collectives, checkpoints) at world>1."""
+
+    def __init__(self, feed, posts):
+        self.feed = feed
+        self.posts = posts
+        self.stats = {"fake_pages": 0}
+
+    def process_channels(self, names):
+        discovered = []
+        for n in names:
+            cid = int(n[1:]) if n[1:].isdigit() else 0
+            for k in range(2):
+                discovered.append(self.feed.username_of(
+                    (cid * 31 + 7 * (k + 1)) % self.feed.cfg.universe))
+        self.stats["fake_pages"] += len(names)
+        return discovered, len(names) * self.posts
 
 
 if __name__ == "__main__":

@@ -486,3 +486,34 @@ def test_queue_drain_returns_fifo_and_empties():
     assert got == [f"m{i}" for i in range(7)]
     assert q.size() == 0
     assert q.drain() == []
+
+
+def test_gpu_dist_crawl_dry_run_world2(tmp_path):
+    """World>1-shaped dry run of scripts/gpu_dist_crawl.py: the exact
+    launcher wiring (torchrun -> TCPStore queue -> OrchestratedCrawl ->
+    collectives) with --fake-engine on gloo."""
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29752", os.path.join(repo, "scripts",
+         "gpu_dist_crawl.py"), "--fake-engine", "--seeds", "12",
+         "--posts", "50", "--max-depth", "1", "--max-pages", "100",
+         "--storage", str(tmp_path), "--store-port", "29773"],
+        capture_output=True, text=True, timeout=120, cwd=repo,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    import json
+
+    line = [l for l in out.stdout.splitlines()
+            if l.startswith("{")][-1]
+    stats = json.loads(line)
+    assert stats["world"] == 2
+    assert stats["layers"] == 2
+    prog_path = os.path.join(str(tmp_path), "r0", "dist-crawl",
+                             "progress.json")
+    with open(prog_path) as f:
+        assert json.load(f)["status"] == "completed"
