@@ -257,9 +257,6 @@ def main(argv: Optional[List[str]] = None) -> int:
     dbg = maybe_start(args.debug_port)
 
     from .feed import FeedConfig, SyntheticFeed
-    from .feed.client import ConnectionPool
-    from .engine import LocalStateManager, RandomWalkStore
-    from .engine.runner import RandomWalkRunner, StandaloneRunner
 
     feed = SyntheticFeed(FeedConfig(
         seed=args.synthetic_seed, universe=args.synthetic_universe,
@@ -275,115 +272,20 @@ def main(argv: Optional[List[str]] = None) -> int:
 
         return worker_main(cfg, feed)
 
-    if cfg.platform == "youtube":
-        from .youtube.runner import run_youtube
+    # Platform dispatch is ONE registry lookup (crawler/crawler.go
+    # Crawler interface + DefaultCrawlerFactory:79-106): adding a
+    # platform means registering a module under crawler_amd/platforms/,
+    # never touching this file.
+    from .registry import CrawlContext, get_crawler, register_all_crawlers
 
-        stats = run_youtube(cfg, urls)
-        print(f"youtube crawl complete: {stats}", file=sys.stderr)
-        return 0
-
-    db_urls = [u for u in args.tdlib_database_urls.split(",") if u]
-    pool_size = len(db_urls) if db_urls else args.pool_size
-    pool = ConnectionPool(
-        feed, pool_size, cfg.rate_limit,
-        posts_per_channel=args.synthetic_posts,
-        disable_rate_limits=cfg.disable_rate_limits,
-    )
-    sm = LocalStateManager(cfg)
-
-    chunker = None
-    if cfg.combine_files:
-        # CombineFiles mode (dapr/standalone.go:253-271): posts go through
-        # the temp->watch combiner; combined files land under the crawl's
-        # combined/ dir (the upload-binding mock)
-        import shutil
-
-        from .engine.chunker import Chunker
-
-        combined_dir = os.path.join(cfg.storage_root, cfg.crawl_id,
-                                    "combined")
-        os.makedirs(combined_dir, exist_ok=True)
-        chunker = Chunker(
-            cfg.combine_temp_dir, cfg.combine_watch_dir,
-            cfg.combine_write_dir,
-            upload=lambda p: shutil.copy(p, combined_dir),
-            trigger_bytes=cfg.combine_trigger_size * 1024 * 1024,
-            hard_cap_bytes=cfg.combine_hard_cap * 1024 * 1024,
-        )
-        chunker.verify_cleanup()  # crash recovery at startup
-        chunker.start()
-        sm.attach_chunker(chunker)
-
-    if cfg.validate_only:
-        from .engine.validator import run_validation_loop
-
-        rw = RandomWalkStore()
-        run_validation_loop(cfg, sm, rw)
-        return 0
-
-    if args.gpu and cfg.sampling_method in ("channel", "snowball"):
-        # MI355X execution mode: whole layers through the HIP kernels
-        from .engine.gpu_runner import GpuCrawlEngine
-
-        engine = GpuCrawlEngine(cfg, sm, feed,
-                                posts_per_channel=args.synthetic_posts)
-        stats = engine.run(urls)
-        print(f"gpu crawl complete: {stats}", file=sys.stderr)
-        return 0
-
-    if cfg.sampling_method == "random-walk":
-        rw = RandomWalkStore()
-        runner = RandomWalkRunner(cfg, sm, rw, pool)
-        if cfg.seed_size and not urls:
-            urls = [feed.username_of(i) for i in range(cfg.seed_size)]
-        runner.seed(urls)
-        vthread = stop_evt = None
-        if cfg.tandem_crawl:
-            # The reference runs validator PODS against the shared
-            # Postgres (dapr/standalone.go:276-314); single-node analog:
-            # an in-process validator thread over the same store
-            # (identical claim semantics, test_tandem_e2e pattern).
-            import threading
-
-            from .engine.validator import TandemValidator
-            from .feed.client import TokenBucket
-            from .feed.tme import MockTMe
-
-            rl = None
-            if not cfg.disable_rate_limits:
-                rl = TokenBucket(cfg.validator_request_rate * 60, 100)
-            v = TandemValidator(
-                cfg, sm, rw,
-                fetcher=MockTMe(universe=args.synthetic_universe),
-                rate_limiter=rl,
-            )
-            v.recover()
-            stop_evt = threading.Event()
-
-            def vloop():
-                while not stop_evt.is_set():
-                    n = v.pump_edges()
-                    b = v.pump_walkback()
-                    if not n and not b:
-                        stop_evt.wait(0.01)
-
-            vthread = threading.Thread(target=vloop, daemon=True,
-                                       name="tandem-validator")
-            vthread.start()
-        try:
-            stats = runner.run(
-                max_seconds=cfg.max_crawl_duration_s or None
-            )
-        finally:
-            if vthread is not None:
-                stop_evt.set()
-                vthread.join(timeout=10)
-    else:
-        runner = StandaloneRunner(cfg, sm, pool)
-        stats = runner.run(urls)
-    if chunker is not None:
-        chunker.stop()
-    print(f"crawl complete: {stats}", file=sys.stderr)
+    register_all_crawlers()
+    crawler = get_crawler(cfg.platform)
+    ctx = CrawlContext(cfg=cfg, args=args, urls=urls, feed=feed)
+    crawler.initialize(ctx)
+    try:
+        crawler.run(ctx)
+    finally:
+        crawler.close()
     return 0
 
 
