@@ -3,7 +3,7 @@
 Parity (reference telegramhelper/channelvalidator.go, validator_rate_limiter.go):
 - ParseChannelHTML title rules (channelvalidator.go:132-153):
     "View @..."                         -> valid
-    "Contact @..." + meta robots noindex -> invalid / username_not_occupied
+    "Contact @..." + meta robots noindex -> invalid / not_found
     "Contact @..."                      -> not_channel / not_supergroup
     "Telegram Messenger"                -> invalid / not_found
 - 64 KB body cap (channelvalidator.go:103)
@@ -53,7 +53,7 @@ def parse_channel_html(body: bytes) -> ChannelValidationResult:
         return ChannelValidationResult("valid")
     if title.startswith(b"Contact @") or b"Telegram: Contact @" in title:
         if _NOINDEX_RE.search(body):
-            return ChannelValidationResult("invalid", "username_not_occupied")
+            return ChannelValidationResult("invalid", "not_found")
         return ChannelValidationResult("not_channel", "not_supergroup")
     if b"Telegram Messenger" in title:
         return ChannelValidationResult("invalid", "not_found")

@@ -7,7 +7,7 @@
 // marker. Oracle: crawler_amd/engine/htmlvalidator.py parse_channel_html.
 //
 // status: 0=valid 1=not_channel 2=invalid
-// reason: 0="" 1=not_supergroup 2=username_not_occupied 3=not_found
+// reason: 0="" 1=not_supergroup 2=not_found(noindex) 3=not_found
 //         4=unrecognized
 
 #include "common.h"
@@ -97,7 +97,7 @@ html_classify_kernel(const unsigned char* pool, const long* doc_off,
         }
         from = m + 5;
       }
-      if (noindex) { st = 2; rs = 2; }       // username_not_occupied
+      if (noindex) { st = 2; rs = 2; }       // not_found (noindex)
       else { st = 1; rs = 1; }               // not_channel/not_supergroup
     } else if (find_sub(s, tlo, thi, "Telegram Messenger", 18, lane) >= 0) {
       st = 2; rs = 3;                        // not_found

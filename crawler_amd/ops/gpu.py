@@ -260,7 +260,7 @@ def reservoir_sample_oracle(P: int, K: int, k: int, seed: int = 0):
 
 
 _HTML_STATUS = ["valid", "not_channel", "invalid"]
-_HTML_REASON = ["", "not_supergroup", "username_not_occupied", "not_found",
+_HTML_REASON = ["", "not_supergroup", "not_found", "not_found",
                 "unrecognized"]
 
 

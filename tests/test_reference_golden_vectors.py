@@ -1,0 +1,198 @@
+"""Golden vectors transcribed LITERALLY from the reference's own tests.
+
+Round-1 verdict item 7: all byte-exactness in this repo was
+GPU <-> Python-oracle <-> C++-reference, every leg builder-authored.
+These expected values are copied from the reference test files (cited
+per block), so the oracle is pinned to the reference's published
+behavior, not to itself.
+
+Sources:
+- /root/reference/telegramhelper/channel_links_test.go:65-330
+- /root/reference/telegramhelper/username_filter_test.go:5-68
+- /root/reference/telegramhelper/channelvalidator_test.go:15-85
+- /root/reference/telegramhelper/tdutils.go:1005-1031 (msg-id >> 20)
+"""
+import pytest
+
+from crawler_amd.engine.htmlvalidator import parse_channel_html
+from crawler_amd.ops.golden import (Entity, FormattedText, SynthMessage,
+                                    build_telegram_link_and_message_id,
+                                    extract_channel_links,
+                                    filter_username)
+
+
+def _msg(content_type, text=None, entities=None):
+    ft = FormattedText(text=text or "", entities=entities or [])
+    if content_type == "messageText":
+        return SynthMessage(content_type=content_type, text=ft)
+    return SynthMessage(content_type=content_type, caption=ft)
+
+
+# --- channel_links_test.go: plain-text regex scan ---
+
+LINK_VECTORS = [
+    # (test name, content_type, text, entities, expected sorted names)
+    ("PlainTextTmeLink", "messageText",
+     "Check out https://t.me/channelname for news", [], ["channelname"]),
+    ("PlainTextTmeLinkNoScheme", "messageText",
+     "Visit t.me/somechan today", [], ["somechan"]),
+    ("PlainTextMultipleLinks", "messageText",
+     "t.me/chanone and t.me/chantwo", [], ["chanone", "chantwo"]),
+    ("TextEntityTypeTextUrl", "messageText", "click here",
+     [Entity("text_url", 0, 10, url="https://t.me/linkedchan")],
+     ["linkedchan"]),
+    ("TextEntityTypeTextUrl_NonTme", "messageText", "link",
+     [Entity("text_url", 0, 4, url="https://example.com/page")], []),
+    # UTF-16 offset regressions (channel_links_test.go:115-175)
+    ("Mention_ASCII", "messageText", "Hello @testchan!",
+     [Entity("mention", 6, 9)], ["testchan"]),
+    ("Mention_CyrillicPrefix", "messageText", "Привет @testchan",
+     [Entity("mention", 7, 9)], ["testchan"]),
+    ("Mention_EmojiPrefix", "messageText", "😀 @testchan",
+     [Entity("mention", 3, 9)], ["testchan"]),
+    ("Mention_ArabicPrefix", "messageText", "مرحبا @testchan",
+     [Entity("mention", 6, 9)], ["testchan"]),
+    ("TextEntityTypeUrl_TmeLink", "messageText",
+     "See https://t.me/urlchan for details",
+     [Entity("url", 4, 20)], ["urlchan"]),
+    # media captions (channel_links_test.go:193-230)
+    ("PhotoCaption", "messagePhoto", "t.me/photochan", [], ["photochan"]),
+    ("VideoCaption", "messageVideo", "t.me/videochan", [], ["videochan"]),
+    ("DocumentCaption", "messageDocument", "t.me/docchan", [],
+     ["docchan"]),
+    ("AnimationCaption", "messageAnimation", "t.me/animchan", [],
+     ["animchan"]),
+    ("AudioCaption", "messageAudio", "t.me/audiochan", [], ["audiochan"]),
+    ("VoiceNoteCaption", "messageVoiceNote", "t.me/voicechan", [],
+     ["voicechan"]),
+    # reserved paths (channel_links_test.go:233-255)
+    ("ReservedPath_Joinchat", "messageText",
+     "https://t.me/joinchat/abc123", [], []),
+    ("ReservedPath_Share", "messageText",
+     "https://t.me/share/url?url=x", [], []),
+    ("ReservedPath_Proxy", "messageText",
+     "https://t.me/proxy?server=x", [], []),
+    # case normalization (channel_links_test.go:281-285)
+    ("CaseNormalization", "messageText", "t.me/MixedCase", [],
+     ["mixedcase"]),
+    # unsupported content type returns empty (channel_links_test.go:289)
+    ("UnknownContentType", "messageSticker", "t.me/stickchan", [], []),
+    # short names filtered by the {4,32} regex (channel_links_test.go:300)
+    ("TooShortName", "messageText", "t.me/abc", [], []),
+    # mention at end of string (channel_links_test.go:310)
+    ("Mention_AtEndOfString", "messageText", "@endchan",
+     [Entity("mention", 0, 8)], ["endchan"]),
+]
+
+
+@pytest.mark.parametrize(
+    "name,ctype,text,entities,want",
+    LINK_VECTORS, ids=[v[0] for v in LINK_VECTORS])
+def test_channel_links_reference_vectors(name, ctype, text, entities,
+                                         want):
+    got = sorted(extract_channel_links(_msg(ctype, text, entities)))
+    assert got == sorted(want)
+
+
+def test_channel_links_deduplication_reference_vector():
+    """channel_links_test.go:259-277: same channel as plain URL and as
+    a TextUrl entity yields one entry."""
+    msg = _msg("messageText", "Check t.me/samechan",
+               [Entity("text_url", 6, 13, url="https://t.me/samechan")])
+    got = extract_channel_links(msg)
+    assert got == ["samechan"]
+
+
+# --- username_filter_test.go:5-68 literal matrix ---
+
+USERNAME_VECTORS = [
+    ("valid simple", "testchannel", True, ""),
+    ("valid with underscore", "test_channel", True, ""),
+    ("valid with numbers", "channel123", True, ""),
+    ("valid min length", "abcde", True, ""),
+    ("valid 32 chars", "abcdefghijklmnopqrstuvwxyz123456", True, ""),
+    ("too short 4 chars", "abcd", False, "too_short"),
+    ("too short 1 char", "a", False, "too_short"),
+    ("too short empty", "", False, "too_short"),
+    ("too long 33 chars", "abcdefghijklmnopqrstuvwxyz1234567", False,
+     "too_long"),
+    ("starts with number", "1channel", False, "invalid_start_char"),
+    ("starts with underscore", "_channel", False, "invalid_start_char"),
+    ("starts with non-ASCII letter", "échannel", False,
+     "invalid_start_char"),
+    ("ends with underscore", "channel_", False, "ends_with_underscore"),
+    ("contains space", "test channel", False, "invalid_char"),
+    ("contains dash", "test-channel", False, "invalid_char"),
+    ("contains dot", "test.channel", False, "invalid_char"),
+    ("contains unicode", "téstchannel", False, "invalid_char"),
+    ("ends with _bot", "some_bot", False, "bot_suffix"),
+    ("ends with Bot", "SomeBot", False, "bot_suffix"),
+    ("ends with BOT", "SomeBOT", False, "bot_suffix"),
+    ("ends with _Bot", "Test_Bot", False, "bot_suffix"),
+    ("looks like path", "usr/local", False, "invalid_char"),
+    ("contains tilde", "home~user", False, "invalid_char"),
+    ("contains dot path", "file.name", False, "invalid_char"),
+]
+
+
+@pytest.mark.parametrize(
+    "name,username,want_valid,want_reason",
+    USERNAME_VECTORS, ids=[v[0].replace(" ", "_")
+                           for v in USERNAME_VECTORS])
+def test_filter_username_reference_vectors(name, username, want_valid,
+                                           want_reason):
+    valid, reason = filter_username(username)
+    assert valid == want_valid, (username, reason)
+    if not want_valid:
+        assert reason == want_reason, (username, reason, want_reason)
+
+
+# --- channelvalidator_test.go:15-85 fixture classification ---
+
+HTML_VECTORS = [
+    ("valid-channel.html", "valid", ""),
+    ("not-a-supergroup.html", "not_channel", "not_supergroup"),
+    ("invalid-channel.html", "invalid", "not_found"),
+    ("username-not-occupied.html", "invalid", "not_found"),
+]
+
+
+@pytest.mark.parametrize("fixture,status,reason", HTML_VECTORS,
+                         ids=[v[0] for v in HTML_VECTORS])
+def test_parse_channel_html_reference_vectors(fixture, status, reason):
+    import os
+
+    path = os.path.join(os.path.dirname(__file__), "..", "fixtures",
+                        "telegram-html", fixture)
+    with open(path, "rb") as f:
+        body = f.read()
+    res = parse_channel_html(body)
+    assert res.status == status
+    assert res.reason == reason
+
+
+def test_parse_channel_html_unrecognised_title_errors():
+    """channelvalidator_test.go:71-77."""
+    with pytest.raises(Exception):
+        r = parse_channel_html(
+            b"<html><head><title>Something Unexpected</title></head>"
+            b"</html>")
+        if r is not None:  # impls may return error-status instead
+            assert r.status not in ("valid", "not_channel", "invalid")
+
+
+# --- tdutils.go:1005-1031 BuildTelegramLinkAndMessageID semantics ---
+
+def test_build_link_msg_id_shift20():
+    """public id = internal id / 1048576 (i.e. >> 20)."""
+    msg = SynthMessage(msg_id=5 << 20)
+    link, pid = build_telegram_link_and_message_id("chan", msg)
+    assert pid == 5
+    assert link == "https://t.me/chan/5"
+    # media album appends ?single (tdutils.go:1022-1025)
+    msg2 = SynthMessage(msg_id=7 << 20, media_album_id=99)
+    link2, pid2 = build_telegram_link_and_message_id("chan", msg2)
+    assert link2 == "https://t.me/chan/7?single"
+    # private channel (no username) -> empty link
+    link3, pid3 = build_telegram_link_and_message_id("", msg)
+    assert link3 == "" and pid3 == 5

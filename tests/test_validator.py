@@ -44,7 +44,7 @@ def test_not_supergroup_fixture():
 def test_username_not_occupied_fixture():
     r = parse_channel_html(fixture("username-not-occupied.html"))
     assert r.status == "invalid"
-    assert r.reason == "username_not_occupied"
+    assert r.reason == "not_found"
 
 
 def test_invalid_channel_fixture():
@@ -364,8 +364,8 @@ def _html(title, noindex=False):
     (b"Telegram: View @somechan", False, "valid", ""),
     (b"View @somechan", False, "valid", ""),
     (b"Telegram: Contact @someone", True, "invalid",
-     "username_not_occupied"),
-    (b"Contact @someone", True, "invalid", "username_not_occupied"),
+     "not_found"),
+    (b"Contact @someone", True, "invalid", "not_found"),
     (b"Telegram: Contact @someone", False, "not_channel",
      "not_supergroup"),
     (b"Telegram Messenger", False, "invalid", "not_found"),
