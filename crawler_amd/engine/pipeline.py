@@ -235,8 +235,11 @@ def run_for_channel(client, page: Page, sm, cfg, rw=None, seen=None,
         try:
             comments = None
             if m.reply_count > 0:
+                # paginated thread walk; reply_count plays the
+                # reference's commentcount role (tdutils.go:434)
                 comments = client.get_message_comments(
-                    m.chat_id, m.msg_id, cfg.max_comments
+                    m.chat_id, m.msg_id, cfg.max_comments,
+                    comment_count=m.reply_count,
                 )
             post = G.parse_message(
                 m,

@@ -189,14 +189,61 @@ class SyntheticTelegramClient:
                 return m
         raise TelegramAPIError(f"[404] message not found: {message_id}")
 
-    def get_message_comments(self, chat_id: int, message_id: int,
-                             max_comments: int = -1) -> list:
+    def get_message_thread_history(self, chat_id: int, message_id: int,
+                                   from_message_id: int = 0,
+                                   limit: int = 100) -> list:
+        """One page of a comment thread, newest-first
+        (TDLib GetMessageThreadHistory as used by
+        telegramutils.go:594-600): returns [(thread_msg_id, Comment)].
+        Thread ids descend from len(comments) to 1; from_message_id=0
+        starts at the newest."""
         self._call_count += 1
         m = self.get_message(chat_id, message_id)
         coms = getattr(m, "_comments", [])
-        if max_comments is not None and max_comments >= 0:
-            coms = coms[:max_comments]
-        return coms
+        n = len(coms)
+        # slot s holds thread id n-s (slot 0 = newest)
+        start = 0 if not from_message_id else n - from_message_id + 1
+        limit = max(0, min(limit, 100))
+        page = []
+        for s in range(start, min(n, start + limit)):
+            page.append((n - s, coms[s]))
+        return page
+
+    def get_message_comments(self, chat_id: int, message_id: int,
+                             max_comments: int = -1,
+                             comment_count: int = 0) -> list:
+        """Paginated thread walk (GetMessageComments,
+        telegramutils.go:311-747): 100/batch from the newest, batch
+        size shrunk to the remaining need, stop on empty page / no
+        progress / enough comments. maxcomments==0 means none;
+        comment_count (when known, < maxcomments) caps the walk like
+        the reference's commentcount argument."""
+        if max_comments == 0:
+            return []
+        comments: list = []
+        from_message_id = 0
+        while True:
+            batch = 100
+            if max_comments and max_comments > 0:
+                remaining = max_comments - len(comments)
+                if remaining <= 0:
+                    break
+                if 0 < comment_count < max_comments:
+                    remaining = comment_count - len(comments)
+                    if remaining <= 0:
+                        break
+                if remaining < batch:
+                    batch = remaining
+            page = self.get_message_thread_history(
+                chat_id, message_id, from_message_id, batch)
+            if not page:
+                break
+            comments.extend(c for _tid, c in page)
+            prev = from_message_id
+            from_message_id = page[-1][0]
+            if from_message_id == 0 or from_message_id == prev:
+                break
+        return comments
 
     def close(self):
         self.closed = True
@@ -284,8 +331,15 @@ class RateLimitedClient:
             self.message_bucket.acquire()
         return out
 
-    def get_message_comments(self, chat_id, message_id, max_comments=-1):
-        return self.c.get_message_comments(chat_id, message_id, max_comments)
+    def get_message_thread_history(self, chat_id, message_id,
+                                   from_message_id=0, limit=100):
+        return self.c.get_message_thread_history(
+            chat_id, message_id, from_message_id, limit)
+
+    def get_message_comments(self, chat_id, message_id, max_comments=-1,
+                             comment_count=0):
+        return self.c.get_message_comments(chat_id, message_id,
+                                           max_comments, comment_count)
 
     def close(self):
         self.c.close()

@@ -178,3 +178,70 @@ def test_max_comments_cap_through_pipeline(tmp_path):
         results[cap] = max(lens)
     assert results[3] == 3          # capped exactly at --max-comments
     assert 4 <= results[-1] <= 9    # uncapped exceeds the cap
+
+
+# ---- comment-thread pagination (telegramutils.go:311-747; VERDICT r01
+# item 6: the 100/batch walk, not a pre-baked list) ----
+
+def _comment_heavy_client():
+    from crawler_amd.config import TelegramRateLimitConfig
+    from crawler_amd.feed.client import SyntheticTelegramClient
+    from crawler_amd.feed.synth import FeedConfig as FC
+    from crawler_amd.feed.synth import SyntheticFeed as SF
+
+    feed = SF(FC(seed=7, universe=1000, comment_rate=1.0,
+                 max_comments_per_post=250))
+    return SyntheticTelegramClient(feed, "conn0", posts_per_channel=6)
+
+
+def _first_msg_with_comments(client, min_comments=101):
+    msgs = client.get_chat_history(-1001000000000 - 5)
+    for m in msgs:
+        coms = getattr(m, "_comments", [])
+        if len(coms) >= min_comments:
+            return m, coms
+    raise AssertionError("no comment-heavy message in the synth feed")
+
+
+def test_thread_history_paginates_100_per_batch():
+    client = _comment_heavy_client()
+    m, coms = _first_msg_with_comments(client)
+    page1 = client.get_message_thread_history(m.chat_id, m.msg_id,
+                                              0, 100)
+    assert len(page1) == 100
+    # newest-first: descending thread ids starting at len(coms)
+    assert page1[0][0] == len(coms)
+    assert [tid for tid, _ in page1] == list(
+        range(len(coms), len(coms) - 100, -1))
+    page2 = client.get_message_thread_history(
+        m.chat_id, m.msg_id, page1[-1][0], 100)
+    assert page2[0][0] == page1[-1][0] - 1
+    # limit is clamped to 100 like GetChatHistory
+    big = client.get_message_thread_history(m.chat_id, m.msg_id, 0, 500)
+    assert len(big) == 100
+
+
+def test_get_message_comments_walks_pages():
+    client = _comment_heavy_client()
+    m, coms = _first_msg_with_comments(client)
+    calls0 = client._call_count
+    got = client.get_message_comments(m.chat_id, m.msg_id, 1000)
+    # multiple thread-history calls happened (paginated, not pre-baked)
+    assert client._call_count - calls0 >= 2 * ((len(coms) // 100) + 1)
+    assert len(got) == len(coms)
+    assert [c.to_json() for c in got] == [c.to_json() for c in coms]
+
+
+def test_get_message_comments_respects_max_and_count():
+    client = _comment_heavy_client()
+    m, coms = _first_msg_with_comments(client)
+    assert client.get_message_comments(m.chat_id, m.msg_id, 0) == []
+    got = client.get_message_comments(m.chat_id, m.msg_id, 150)
+    assert len(got) == 150
+    # commentcount < maxcomments caps the walk (telegramutils.go:459-466)
+    got2 = client.get_message_comments(m.chat_id, m.msg_id, 150,
+                                       comment_count=120)
+    assert len(got2) == 120
+    # unlimited (-1) returns everything
+    got3 = client.get_message_comments(m.chat_id, m.msg_id, -1)
+    assert len(got3) == len(coms)
