@@ -180,34 +180,47 @@ class GpuRandomWalk:
                 nm = nb.decode()
                 ok, _ = filter_username(nm)
                 cache[nb] = (nm, ok, self._cid_of(nm))
+        # Outlink validation (SearchPublicChat equivalent + filter +
+        # invalid cache; runner.go:1310-1383), hoisted out of the walker
+        # loop: one invalid-set snapshot per hop (is_invalid_channel
+        # costs a lock + _now() per call at ~160k calls/hop), validated
+        # names admitted to the discovered/seed tables in bulk.
+        import datetime as _dtm
+        hop_now = _dtm.datetime.now(_dtm.timezone.utc)
+        ttl_cut = hop_now - _dtm.timedelta(days=30)
+        inv = {u for u, t in self.rw.invalid_channels.items()
+               if t > ttl_cut}
+        per_walker = []
+        admitted = []
         for k, p in enumerate(live):
-            # outlink validation order preserved from the scalar path
-            # (SearchPublicChat equivalent + filter + invalid cache;
-            # runner.go:1310-1383); name order is sorted — walk_tail
-            # itself sorts before choosing, so decisions are unchanged
-            new_channels = {}
+            nc = []
             for nb in un[bounds[k]:bounds[k + 1]]:
                 nm, ok, cid = cache[nb]
                 if nm == p.url:
                     continue
-                if not ok or self.rw.is_invalid_channel(nm):
+                if not ok or nm in inv:
                     continue
                 if cid is None:
                     self.rw.mark_invalid_channel(nm)
+                    inv.add(nm)
                     continue
-                self.sm.add_discovered_channel(nm)
-                self.rw.upsert_seed_channel(nm)
-                new_channels[nm] = True
+                nc.append(nm)
+            per_walker.append(nc)
+            admitted.extend(nc)
+        self.sm.add_discovered_channels_bulk(admitted)
+        self.rw.upsert_seed_channels_bulk(admitted)
+        for k, p in enumerate(live):
             try:
-                randomwalk.walk_tail(p, new_channels, self.sm, self.rw,
-                                     self.cfg, self.rng)
+                randomwalk.walk_tail_fast(p, per_walker[k], self.sm,
+                                          self.rw, self.cfg, self.rng,
+                                          hop_now)
             except randomwalk.E.WalkbackExhausted:
                 self.stats["walkback_exhausted"] += 1
                 continue  # page left in buffer
             self.rw.mark_channel_crawled(p.url, 0)
             self.rw.delete_pages([p.id])
             self.stats["pages"] += 1
-        self.stats["edges"] = len(self.rw.edge_records)
+        self.stats["edges"] = self.rw.edge_count()
         return len(live)
 
     def run(self, max_pages: int, now: Optional[_dt.datetime] = None) -> dict:

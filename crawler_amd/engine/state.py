@@ -217,6 +217,16 @@ class DiscoveredChannels:
             self._keys.append(name)
             return True
 
+    def add_bulk(self, names) -> int:
+        """Bulk insert (the GPU walk path adds ~160k names per hop —
+        one lock + set ops instead of 160k add() calls)."""
+        with self._lock:
+            missing = [n for n in names if n not in self._items]
+            for n in missing:
+                self._items[n] = True
+            self._keys.extend(missing)
+            return len(missing)
+
     def contains(self, name: str) -> bool:
         with self._lock:
             return name in self._items
@@ -347,6 +357,9 @@ class BaseStateManager:
 
     def add_discovered_channel(self, name: str) -> None:
         self.discovered.add(name)
+
+    def add_discovered_channels_bulk(self, names) -> int:
+        return self.discovered.add_bulk(names)
 
     def get_random_discovered_channel(self, rng=None) -> Optional[str]:
         return self.discovered.random(rng)
@@ -812,7 +825,12 @@ class RandomWalkStore:
     def __init__(self):
         self._lock = threading.RLock()
         self.page_buffer: Dict[str, Page] = {}       # page_id -> Page
-        self.edge_records: List[EdgeRecord] = []
+        self._edge_list: List[EdgeRecord] = []
+        # skipped-edge BLOCKS (source, sequence_id, time, names): the
+        # GPU walk appends ~160k skipped edges per hop — one tuple per
+        # walker here, expanded to EdgeRecords lazily on first read
+        # ('append to device edge log; host spill')
+        self._edge_blocks: List[tuple] = []
         self.seed_channels: Dict[str, dict] = {}     # username -> row
         self.invalid_channels: Dict[str, _dt.datetime] = {}
         self.pending_batches: Dict[str, PendingEdgeBatch] = {}
@@ -845,12 +863,53 @@ class RandomWalkStore:
 
     # ---- edge_records ----
 
+    @property
+    def edge_records(self) -> List[EdgeRecord]:
+        with self._lock:
+            self._materialize_edges()
+            return self._edge_list
+
+    @edge_records.setter
+    def edge_records(self, value: List[EdgeRecord]) -> None:
+        with self._lock:
+            self._edge_blocks = []
+            self._edge_list = value
+
+    def _materialize_edges(self) -> None:
+        if not self._edge_blocks:
+            return
+        for (src, seq, t, names) in self._edge_blocks:
+            self._edge_list.extend(
+                EdgeRecord(n, t, src, False, True, seq, "")
+                for n in names)
+        self._edge_blocks = []
+
+    def save_skipped_edges_block(self, source: str, sequence_id: str,
+                                 names, now) -> None:
+        """O(1) append of one walker's skipped-edge set; expanded to
+        EdgeRecord rows on first edge_records read."""
+        with self._lock:
+            if names:
+                self._edge_blocks.append((source, sequence_id, now,
+                                          names))
+
+    def edge_count(self) -> int:
+        """len(edge_records) without forcing materialization."""
+        with self._lock:
+            return (len(self._edge_list)
+                    + sum(len(b[3]) for b in self._edge_blocks))
+
     def save_edge_records(self, edges: List[EdgeRecord]) -> None:
         with self._lock:
             for e in edges:
                 if e.discovery_time is None:
                     e.discovery_time = _now()
-                self.edge_records.append(e)
+                self._edge_list.append(e)
+
+    def save_edge_records_fast(self, edges: List[EdgeRecord]) -> None:
+        """Bulk append; caller guarantees discovery_time is set."""
+        with self._lock:
+            self._edge_list.extend(edges)
 
     def get_random_skipped_edge(self, exclude: set, rng=None,
                                 sequence_id: Optional[str] = None,
@@ -910,6 +969,17 @@ class RandomWalkStore:
             })
             if chat_id:
                 row["chat_id"] = chat_id
+
+    def upsert_seed_channels_bulk(self, usernames) -> None:
+        """Bulk upsert without chat ids (the ~500k-row seed_channels
+        table of sql/random-walk-schema.sql fills at hop granularity)."""
+        with self._lock:
+            sc = self.seed_channels
+            for u in usernames:
+                if u not in sc:
+                    sc[u] = {"username": u, "chat_id": 0,
+                             "last_crawled_at": None,
+                             "invalidated_at": None}
 
     def mark_channel_crawled(self, username: str, chat_id: int) -> None:
         with self._lock:
