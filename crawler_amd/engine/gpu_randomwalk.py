@@ -56,6 +56,8 @@ class GpuRandomWalk:
         # are pure functions of the name; the invalid-channel cache is
         # NOT cached here because 400 handling mutates it between hops)
         self._name_cache = {}
+        import collections
+        self.timings = collections.defaultdict(float)  # phase seconds
 
     def seed(self, urls: List[str]):
         import uuid
@@ -96,13 +98,21 @@ class GpuRandomWalk:
         if not live:
             return 0
 
+        import time as _time
+        t0 = _time.perf_counter()
         cids = np.array([self._cid_of(p.url) for p in live],
                         dtype=np.int64)
         batch = self.feed.build_batch_device(
             cids, self.device, posts_per_channel=self.ppc
         )
+        torch.cuda.synchronize()
+        self.timings["feedgen"] += _time.perf_counter() - t0
+        t0 = _time.perf_counter()
         res = self.gpu.parse_encode(batch, now=now)
         self.seen.claim(res)
+        torch.cuda.synchronize()
+        self.timings["kernels"] += _time.perf_counter() - t0
+        t0 = _time.perf_counter()
 
         # Zero-pad link names ON DEVICE so the host sees clean
         # fixed-width byte strings (one where(), no per-name slicing)
@@ -133,6 +143,8 @@ class GpuRandomWalk:
         padded = padded_g.cpu().numpy()
         cnts = res.link_cnt.cpu().numpy()
         torch.cuda.synchronize()
+        self.timings["d2h"] += _time.perf_counter() - t0
+        t0 = _time.perf_counter()
 
         # spill JSONL per channel (K x P layout) in ONE native-sink call
         P = self.ppc
@@ -158,6 +170,8 @@ class GpuRandomWalk:
         self._spill_tickets[slot] = self.sm.store_post_lines_batch(
             items, memoryview(out_host.numpy()), ticket=True
         )
+        self.timings["spill-issue"] += _time.perf_counter() - t0
+        t0 = _time.perf_counter()
 
         # Per-walker unique link names, vectorized: a (walker, name)
         # structured np.unique replaces the per-post python scan (the
@@ -170,6 +184,8 @@ class GpuRandomWalk:
         pairs["n"] = np.ascontiguousarray(padded[mi, si]).view(
             f"S{w}").ravel()
         uniq = np.unique(pairs)  # sorted by walker, then name
+        self.timings["np-uniq"] += _time.perf_counter() - t0
+        t0 = _time.perf_counter()
         uw, un = uniq["w"], uniq["n"]
         bounds = np.searchsorted(uw, np.arange(K + 1))
         # validate each distinct name once (filter + universe are pure;
@@ -220,6 +236,7 @@ class GpuRandomWalk:
             self.rw.mark_channel_crawled(p.url, 0)
             self.rw.delete_pages([p.id])
             self.stats["pages"] += 1
+        self.timings["walk-tail"] += _time.perf_counter() - t0
         self.stats["edges"] = self.rw.edge_count()
         return len(live)
 

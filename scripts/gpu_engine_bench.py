@@ -103,6 +103,7 @@ def run_randomwalk(args):
         "posts": stats["posts"],
         "edges": stats["edges"],
         "walkback_exhausted": stats["walkback_exhausted"],
+        "phase_s": {k: round(v, 3) for k, v in eng.timings.items()},
     }))
 
 
