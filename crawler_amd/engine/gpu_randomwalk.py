@@ -24,8 +24,7 @@ import numpy as np
 import torch
 
 from ..feed.synth import SyntheticFeed
-from ..ops.golden import filter_username
-from . import randomwalk
+from . import randomwalk, vecvalidate
 from .state import LocalStateManager, Page, RandomWalkStore
 
 
@@ -52,10 +51,6 @@ class GpuRandomWalk:
         self._spill_tickets = [None, None]  # native-sink tickets
         self._inflight_paths = {}           # slot -> channel set
         self._hop_idx = 0
-        # per-name validation caches (filter_username + universe check
-        # are pure functions of the name; the invalid-channel cache is
-        # NOT cached here because 400 handling mutates it between hops)
-        self._name_cache = {}
         import collections
         self.timings = collections.defaultdict(float)  # phase seconds
 
@@ -187,44 +182,44 @@ class GpuRandomWalk:
         self.timings["np-uniq"] += _time.perf_counter() - t0
         t0 = _time.perf_counter()
         uw, un = uniq["w"], uniq["n"]
-        bounds = np.searchsorted(uw, np.arange(K + 1))
-        # validate each distinct name once (filter + universe are pure;
-        # cached across hops)
-        cache = self._name_cache
-        for nb in np.unique(un):
-            if nb not in cache:
-                nm = nb.decode()
-                ok, _ = filter_username(nm)
-                cache[nb] = (nm, ok, self._cid_of(nm))
         # Outlink validation (SearchPublicChat equivalent + filter +
-        # invalid cache; runner.go:1310-1383), hoisted out of the walker
-        # loop: one invalid-set snapshot per hop (is_invalid_channel
-        # costs a lock + _now() per call at ~160k calls/hop), validated
-        # names admitted to the discovered/seed tables in bulk.
+        # invalid cache; runner.go:1310-1383), fully vectorized
+        # (engine/vecvalidate.py — oracle-equal by test): per-name
+        # FilterUsername + universe id, invalid-cache via np.isin on
+        # the UNIQUE names, self-links excluded per pair.
         import datetime as _dtm
         hop_now = _dtm.datetime.now(_dtm.timezone.utc)
         ttl_cut = hop_now - _dtm.timedelta(days=30)
-        inv = {u for u, t in self.rw.invalid_channels.items()
-               if t > ttl_cut}
-        per_walker = []
-        admitted = []
-        for k, p in enumerate(live):
-            nc = []
-            for nb in un[bounds[k]:bounds[k + 1]]:
-                nm, ok, cid = cache[nb]
-                if nm == p.url:
-                    continue
-                if not ok or nm in inv:
-                    continue
-                if cid is None:
-                    self.rw.mark_invalid_channel(nm)
-                    inv.add(nm)
-                    continue
-                nc.append(nm)
-            per_walker.append(nc)
-            admitted.extend(nc)
-        self.sm.add_discovered_channels_bulk(admitted)
-        self.rw.upsert_seed_channels_bulk(admitted)
+        unames, inv_idx = np.unique(un, return_inverse=True)
+        ok_v, cid_ok_v, _cids_v = vecvalidate.validate_names(
+            unames, self.feed.cfg.universe)
+        inv_set = {u for u, t in self.rw.invalid_channels.items()
+                   if t > ttl_cut}
+        if inv_set:
+            inv_arr = np.array(sorted(inv_set),
+                               dtype=unames.dtype)
+            not_invalid = ~np.isin(unames, inv_arr)
+        else:
+            not_invalid = np.ones(len(unames), dtype=bool)
+        # filter-ok but not a live channel -> invalid-cache insert
+        # (runner.go:1310-1383's 400 branch)
+        to_mark = unames[ok_v & not_invalid & ~cid_ok_v]
+        for nm in vecvalidate.decode_names(to_mark):
+            self.rw.mark_invalid_channel(nm)
+        admit_name = ok_v & cid_ok_v & not_invalid
+        # decoded python strings only for admitted names
+        name_py = np.empty(len(unames), dtype=object)
+        adm_idx = np.nonzero(admit_name)[0]
+        name_py[adm_idx] = vecvalidate.decode_names(unames[adm_idx])
+        owner_b = np.array([p.url for p in live], dtype=unames.dtype)
+        admit_pair = admit_name[inv_idx] & (un != owner_b[uw])
+        uw_s = uw[admit_pair]
+        names_s = name_py[inv_idx[admit_pair]]
+        bounds = np.searchsorted(uw_s, np.arange(K + 1))
+        per_walker = [list(names_s[bounds[k]:bounds[k + 1]])
+                      for k in range(K)]
+        self.sm.add_discovered_channels_bulk(names_s.tolist())
+        self.rw.upsert_seed_channels_bulk(names_s.tolist())
         for k, p in enumerate(live):
             try:
                 randomwalk.walk_tail_fast(p, per_walker[k], self.sm,
