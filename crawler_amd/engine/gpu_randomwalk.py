@@ -51,6 +51,12 @@ class GpuRandomWalk:
         self._spill_tickets = [None, None]  # native-sink tickets
         self._inflight_paths = {}           # slot -> channel set
         self._hop_idx = 0
+        # cross-hop validation cache keyed by link hash (sorted arrays;
+        # validation is a pure function of the name, so hits skip both
+        # validate_names and the decode)
+        self._vc_hashes = np.zeros(0, dtype=np.int64)
+        self._vc_names = np.zeros(0, dtype=object)  # str if admitted
+        self._vc_adm = np.zeros(0, dtype=bool)
         import collections
         self.timings = collections.defaultdict(float)  # phase seconds
 
@@ -71,6 +77,11 @@ class GpuRandomWalk:
             if cid < self.feed.cfg.universe:
                 return cid
         return None
+
+    def _inv_bytes_items(self, ttl_cut):
+        for u, t in self.rw.invalid_channels.items():
+            if t > ttl_cut:
+                yield u.encode(), t
 
     def _hop(self, pages: List[Page],
              now: Optional[_dt.datetime] = None) -> int:
@@ -109,15 +120,37 @@ class GpuRandomWalk:
         self.timings["kernels"] += _time.perf_counter() - t0
         t0 = _time.perf_counter()
 
-        # Zero-pad link names ON DEVICE so the host sees clean
-        # fixed-width byte strings (one where(), no per-name slicing)
+        # Per-walker link dedup ON DEVICE: sort
+        # (walker, hash) pairs, keep the first of each run, gather ONE
+        # zero-padded name row per unique pair — the host never sees
+        # the ~1M raw link slots, only ~150k unique (walker, name)
+        # rows (5 MB instead of 65 MB D2H, no host np.unique).
         lens_g = res.link_len
         names_g = res.link_name
         w = names_g.shape[2]
-        col = torch.arange(w, device=names_g.device,
-                           dtype=lens_g.dtype)
-        padded_g = torch.where(col[None, None, :] < lens_g[..., None],
-                               names_g, torch.zeros_like(names_g))
+        P = self.ppc
+        dev = names_g.device
+        L = names_g.shape[1]
+        cnt_g = res.link_cnt
+        mask_g = (torch.arange(L, device=dev)[None, :]
+                  < cnt_g[:, None])
+        nz = mask_g.nonzero()
+        mi_g, si_g = nz[:, 0], nz[:, 1]
+        wk_g = torch.div(mi_g, P, rounding_mode="floor")
+        h_g = res.link_hash[mi_g, si_g]
+        ord1 = torch.argsort(h_g)
+        ord2 = torch.argsort(wk_g[ord1], stable=True)
+        perm = ord1[ord2]
+        ws_g, hs_g = wk_g[perm], h_g[perm]
+        keep = torch.ones(ws_g.numel(), dtype=torch.bool, device=dev)
+        if ws_g.numel() > 1:
+            keep[1:] = (ws_g[1:] != ws_g[:-1]) | (hs_g[1:] != hs_g[:-1])
+        sel = perm[keep]
+        col = torch.arange(w, device=dev, dtype=lens_g.dtype)
+        sel_lens = lens_g[mi_g[sel], si_g[sel]]
+        sel_names = names_g[mi_g[sel], si_g[sel]]
+        u_rows_g = torch.where(col[None, :] < sel_lens[:, None],
+                               sel_names, torch.zeros_like(sel_names))
 
         # D2H through a reusable pinned ring (2-deep: this hop's disk
         # writes overlap the next hop's kernels via the native sink)
@@ -135,14 +168,14 @@ class GpuRandomWalk:
         out_host.copy_(res.out, non_blocking=True)
         line_off = res.line_off.cpu().numpy()
         line_len = res.line_len.cpu().numpy()
-        padded = padded_g.cpu().numpy()
-        cnts = res.link_cnt.cpu().numpy()
+        u_w = ws_g[keep].to(torch.int32).cpu().numpy()
+        u_h = hs_g[keep].cpu().numpy()
+        u_rows = u_rows_g.cpu().numpy()
         torch.cuda.synchronize()
         self.timings["d2h"] += _time.perf_counter() - t0
         t0 = _time.perf_counter()
 
         # spill JSONL per channel (K x P layout) in ONE native-sink call
-        P = self.ppc
         K = len(live)
         off2 = line_off.reshape(K, P)
         len2 = line_len.reshape(K, P)
@@ -168,58 +201,71 @@ class GpuRandomWalk:
         self.timings["spill-issue"] += _time.perf_counter() - t0
         t0 = _time.perf_counter()
 
-        # Per-walker unique link names, vectorized: a (walker, name)
-        # structured np.unique replaces the per-post python scan (the
-        # round-1 hot spot — NEXT_STEPS #6 / VERDICT item 5)
-        N, L = cnts.shape[0], padded.shape[1]
-        cnt_mask = np.arange(L)[None, :] < cnts[:, None]
-        mi, si = np.nonzero(cnt_mask)
-        pairs = np.empty(len(mi), dtype=[("w", np.int32), ("n", f"S{w}")])
-        pairs["w"] = mi // P
-        pairs["n"] = np.ascontiguousarray(padded[mi, si]).view(
-            f"S{w}").ravel()
-        uniq = np.unique(pairs)  # sorted by walker, then name
-        self.timings["np-uniq"] += _time.perf_counter() - t0
-        t0 = _time.perf_counter()
-        uw, un = uniq["w"], uniq["n"]
         # Outlink validation (SearchPublicChat equivalent + filter +
-        # invalid cache; runner.go:1310-1383), fully vectorized
-        # (engine/vecvalidate.py — oracle-equal by test): per-name
-        # FilterUsername + universe id, invalid-cache via np.isin on
-        # the UNIQUE names, self-links excluded per pair.
+        # invalid cache; runner.go:1310-1383) on the device-deduped
+        # (walker, name) rows. A cross-hop cache keyed by link hash
+        # skips validate+decode for names seen in earlier hops (the
+        # seen-set/bloom already key on this same fnv1a64).
         import datetime as _dtm
         hop_now = _dtm.datetime.now(_dtm.timezone.utc)
         ttl_cut = hop_now - _dtm.timedelta(days=30)
-        unames, inv_idx = np.unique(un, return_inverse=True)
-        ok_v, cid_ok_v, _cids_v = vecvalidate.validate_names(
-            unames, self.feed.cfg.universe)
-        inv_set = {u for u, t in self.rw.invalid_channels.items()
-                   if t > ttl_cut}
-        if inv_set:
-            inv_arr = np.array(sorted(inv_set),
-                               dtype=unames.dtype)
-            not_invalid = ~np.isin(unames, inv_arr)
+        U = len(u_h)
+        un = np.ascontiguousarray(u_rows).view(f"S{w}").ravel()
+        vh = self._vc_hashes
+        if len(vh):
+            pos = np.searchsorted(vh, u_h)
+            pos_c = np.clip(pos, 0, len(vh) - 1)
+            hit = vh[pos_c] == u_h
         else:
-            not_invalid = np.ones(len(unames), dtype=bool)
-        # filter-ok but not a live channel -> invalid-cache insert
-        # (runner.go:1310-1383's 400 branch)
-        to_mark = unames[ok_v & not_invalid & ~cid_ok_v]
-        for nm in vecvalidate.decode_names(to_mark):
-            self.rw.mark_invalid_channel(nm)
-        admit_name = ok_v & cid_ok_v & not_invalid
-        # decoded python strings only for admitted names
-        name_py = np.empty(len(unames), dtype=object)
-        adm_idx = np.nonzero(admit_name)[0]
-        name_py[adm_idx] = vecvalidate.decode_names(unames[adm_idx])
-        owner_b = np.array([p.url for p in live], dtype=unames.dtype)
-        admit_pair = admit_name[inv_idx] & (un != owner_b[uw])
-        uw_s = uw[admit_pair]
-        names_s = name_py[inv_idx[admit_pair]]
+            pos_c = np.zeros(U, dtype=np.int64)
+            hit = np.zeros(U, dtype=bool)
+        res_names = np.empty(U, dtype=object)
+        res_adm = np.zeros(U, dtype=bool)
+        if hit.any():
+            res_names[hit] = self._vc_names[pos_c[hit]]
+            res_adm[hit] = self._vc_adm[pos_c[hit]]
+        miss = ~hit
+        if miss.any():
+            mh, m_first = np.unique(u_h[miss], return_index=True)
+            m_rows = un[miss][m_first]
+            ok_v, cid_ok_v, _cids = vecvalidate.validate_names(
+                m_rows, self.feed.cfg.universe)
+            adm_v = ok_v & cid_ok_v
+            m_names = np.empty(len(mh), dtype=object)
+            adm_i = np.nonzero(adm_v)[0]
+            m_names[adm_i] = vecvalidate.decode_names(m_rows[adm_i])
+            # filter-ok but not a live channel -> invalid-cache insert
+            # (runner.go:1310-1383's 400 branch); marked once, cached
+            for nm in vecvalidate.decode_names(m_rows[ok_v & ~cid_ok_v]):
+                self.rw.mark_invalid_channel(nm)
+            mpos = np.searchsorted(mh, u_h[miss])
+            res_names[miss] = m_names[mpos]
+            res_adm[miss] = adm_v[mpos]
+            # fold into the sorted cross-hop cache
+            all_h = np.concatenate([vh, mh])
+            order = np.argsort(all_h, kind="mergesort")
+            self._vc_hashes = all_h[order]
+            self._vc_names = np.concatenate(
+                [self._vc_names, m_names])[order]
+            self._vc_adm = np.concatenate(
+                [self._vc_adm, adm_v])[order]
+        # dynamic invalid-channel cache (mutated by 400 handling)
+        inv_set = {b for b, t in self._inv_bytes_items(ttl_cut)}
+        not_inv = (~np.isin(un, np.array(sorted(inv_set),
+                                         dtype=un.dtype))
+                   if inv_set else np.ones(U, dtype=bool))
+        owner_b = np.array([p.url for p in live], dtype=un.dtype)
+        final = res_adm & not_inv & (un != owner_b[u_w])
+        uw_s = u_w[final]
+        names_s = res_names[final]
         bounds = np.searchsorted(uw_s, np.arange(K + 1))
-        per_walker = [list(names_s[bounds[k]:bounds[k + 1]])
+        # per-walker lists sorted by NAME (walk_tail sorts before
+        # choosing; device order within a walker is hash order)
+        per_walker = [sorted(names_s[bounds[k]:bounds[k + 1]])
                       for k in range(K)]
-        self.sm.add_discovered_channels_bulk(names_s.tolist())
-        self.rw.upsert_seed_channels_bulk(names_s.tolist())
+        uniq_admitted = set(names_s.tolist())
+        self.sm.add_discovered_channels_bulk(uniq_admitted)
+        self.rw.upsert_seed_channels_bulk(uniq_admitted)
         for k, p in enumerate(live):
             try:
                 randomwalk.walk_tail_fast(p, per_walker[k], self.sm,
