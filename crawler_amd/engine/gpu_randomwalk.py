@@ -51,12 +51,13 @@ class GpuRandomWalk:
         self._spill_tickets = [None, None]  # native-sink tickets
         self._inflight_paths = {}           # slot -> channel set
         self._hop_idx = 0
-        # cross-hop validation cache keyed by link hash (sorted arrays;
-        # validation is a pure function of the name, so hits skip both
-        # validate_names and the decode)
-        self._vc_hashes = np.zeros(0, dtype=np.int64)
-        self._vc_names = np.zeros(0, dtype=object)  # str if admitted
-        self._vc_adm = np.zeros(0, dtype=bool)
+        # cross-hop validation cache keyed by link hash — two sorted
+        # levels (main + pending) so each hop merges only the small
+        # level; validation is a pure function of the name, so hits
+        # skip validate_names AND the decode
+        self._vc = [[np.zeros(0, dtype=np.int64),
+                     np.zeros(0, dtype=object),
+                     np.zeros(0, dtype=bool)] for _ in range(2)]
         import collections
         self.timings = collections.defaultdict(float)  # phase seconds
 
@@ -82,6 +83,41 @@ class GpuRandomWalk:
         for u, t in self.rw.invalid_channels.items():
             if t > ttl_cut:
                 yield u.encode(), t
+
+    def _vc_lookup(self, hashes, out_names, out_adm):
+        """Fill out_names/out_adm from both cache levels; returns the
+        still-missing mask."""
+        miss = np.ones(len(hashes), dtype=bool)
+        for h_arr, n_arr, a_arr in self._vc:
+            if not len(h_arr) or not miss.any():
+                continue
+            pos = np.searchsorted(h_arr, hashes)
+            pos_c = np.clip(pos, 0, len(h_arr) - 1)
+            hit = (h_arr[pos_c] == hashes) & miss
+            if hit.any():
+                out_names[hit] = n_arr[pos_c[hit]]
+                out_adm[hit] = a_arr[pos_c[hit]]
+                miss &= ~hit
+        return miss
+
+    def _vc_insert(self, new_h, new_n, new_a):
+        """Merge new entries into the pending level; spill pending into
+        main when it grows past a quarter of main (amortized O(n))."""
+        main, pend = self._vc
+        all_h = np.concatenate([pend[0], new_h])
+        order = np.argsort(all_h, kind="mergesort")
+        pend[0] = all_h[order]
+        pend[1] = np.concatenate([pend[1], new_n])[order]
+        pend[2] = np.concatenate([pend[2], new_a])[order]
+        if len(pend[0]) > max(len(main[0]) // 4, 1 << 18):
+            all_h = np.concatenate([main[0], pend[0]])
+            order = np.argsort(all_h, kind="mergesort")
+            main[0] = all_h[order]
+            main[1] = np.concatenate([main[1], pend[1]])[order]
+            main[2] = np.concatenate([main[2], pend[2]])[order]
+            pend[0] = np.zeros(0, dtype=np.int64)
+            pend[1] = np.zeros(0, dtype=object)
+            pend[2] = np.zeros(0, dtype=bool)
 
     def _hop(self, pages: List[Page],
              now: Optional[_dt.datetime] = None) -> int:
@@ -211,20 +247,10 @@ class GpuRandomWalk:
         ttl_cut = hop_now - _dtm.timedelta(days=30)
         U = len(u_h)
         un = np.ascontiguousarray(u_rows).view(f"S{w}").ravel()
-        vh = self._vc_hashes
-        if len(vh):
-            pos = np.searchsorted(vh, u_h)
-            pos_c = np.clip(pos, 0, len(vh) - 1)
-            hit = vh[pos_c] == u_h
-        else:
-            pos_c = np.zeros(U, dtype=np.int64)
-            hit = np.zeros(U, dtype=bool)
         res_names = np.empty(U, dtype=object)
         res_adm = np.zeros(U, dtype=bool)
-        if hit.any():
-            res_names[hit] = self._vc_names[pos_c[hit]]
-            res_adm[hit] = self._vc_adm[pos_c[hit]]
-        miss = ~hit
+        miss = self._vc_lookup(u_h, res_names, res_adm)
+        inv_set = {b for b, t in self._inv_bytes_items(ttl_cut)}
         if miss.any():
             mh, m_first = np.unique(u_h[miss], return_index=True)
             m_rows = un[miss][m_first]
@@ -233,7 +259,8 @@ class GpuRandomWalk:
             adm_v = ok_v & cid_ok_v
             m_names = np.empty(len(mh), dtype=object)
             adm_i = np.nonzero(adm_v)[0]
-            m_names[adm_i] = vecvalidate.decode_names(m_rows[adm_i])
+            new_admitted = vecvalidate.decode_names(m_rows[adm_i])
+            m_names[adm_i] = new_admitted
             # filter-ok but not a live channel -> invalid-cache insert
             # (runner.go:1310-1383's 400 branch); marked once, cached
             for nm in vecvalidate.decode_names(m_rows[ok_v & ~cid_ok_v]):
@@ -241,16 +268,18 @@ class GpuRandomWalk:
             mpos = np.searchsorted(mh, u_h[miss])
             res_names[miss] = m_names[mpos]
             res_adm[miss] = adm_v[mpos]
-            # fold into the sorted cross-hop cache
-            all_h = np.concatenate([vh, mh])
-            order = np.argsort(all_h, kind="mergesort")
-            self._vc_hashes = all_h[order]
-            self._vc_names = np.concatenate(
-                [self._vc_names, m_names])[order]
-            self._vc_adm = np.concatenate(
-                [self._vc_adm, adm_v])[order]
+            self._vc_insert(mh, m_names, adm_v)
+            # discovered/seed admission only needs FIRST-SEEN names:
+            # every admitted name from an earlier hop was admitted the
+            # hop it was first validated (400-marked names are never
+            # statically admissible, so the inv filter is a no-op here
+            # beyond correctness hygiene)
+            if inv_set:
+                new_admitted = [nm for nm in new_admitted
+                                if nm.encode() not in inv_set]
+            self.sm.add_discovered_channels_bulk(new_admitted)
+            self.rw.upsert_seed_channels_bulk(new_admitted)
         # dynamic invalid-channel cache (mutated by 400 handling)
-        inv_set = {b for b, t in self._inv_bytes_items(ttl_cut)}
         not_inv = (~np.isin(un, np.array(sorted(inv_set),
                                          dtype=un.dtype))
                    if inv_set else np.ones(U, dtype=bool))
@@ -263,9 +292,6 @@ class GpuRandomWalk:
         # choosing; device order within a walker is hash order)
         per_walker = [sorted(names_s[bounds[k]:bounds[k + 1]])
                       for k in range(K)]
-        uniq_admitted = set(names_s.tolist())
-        self.sm.add_discovered_channels_bulk(uniq_admitted)
-        self.rw.upsert_seed_channels_bulk(uniq_admitted)
         for k, p in enumerate(live):
             try:
                 randomwalk.walk_tail_fast(p, per_walker[k], self.sm,
