@@ -180,9 +180,21 @@ struct JsonEmit {
   uint8_t* out;
   int cur;
 
+  // Striped copy, dword-granular: the emitters are memory-op bound on
+  // the literal/template bytes (~1.3 KB of raw copies per 2 KB line) —
+  // unaligned 4-byte loads/stores quarter the op count vs byte stripes.
   DEV void raw(const uint8_t* s, int n) {
-    if (W)
-      for (int j = lane_id(); j < n; j += WAVE) out[cur + j] = s[j];
+    if (W) {
+      int lane = lane_id();
+      int nw = n >> 2;
+      for (int k = lane; k < nw; k += WAVE) {
+        unsigned w;
+        __builtin_memcpy(&w, s + 4 * k, 4);
+        __builtin_memcpy(out + cur + 4 * k, &w, 4);
+      }
+      for (int j = (nw << 2) + lane; j < n; j += WAVE)
+        out[cur + j] = s[j];
+    }
     cur += n;
   }
 
@@ -211,16 +223,17 @@ struct JsonEmit {
     return false;
   }
 
-  // Escaped JSON string content (no quotes). The measure pass (!W) scans
-  // clean 256-byte blocks 4 bytes/lane with a SWAR special-byte probe
-  // (the emitters are VALU-bound, not bandwidth-bound — see
-  // profiles/r01_pmc_counters.csv); dirty or boundary-risk blocks and the
-  // write pass fall back to the exact 64-byte stripe path below.
+  // Escaped JSON string content (no quotes). BOTH passes scan clean
+  // 256-byte blocks 4 bytes/lane with a SWAR special-byte probe; a
+  // clean block is copied verbatim with dword stores on the write pass
+  // (same ballot -> measure and write can never disagree on length).
+  // Dirty or boundary-risk blocks fall back to the exact 64-byte
+  // stripe path below.
   DEV void esc(const uint8_t* s, int n) {
     int lane = lane_id();
     int start = 0;
     while (start < n) {
-      if (!W && start + 4 * WAVE <= n) {
+      if (start + 4 * WAVE <= n) {
         // a U+2028/29 leader just before the block would swallow bytes
         // INSIDE it; route those rare blocks to the exact path
         bool risk = (start >= 1 && s[start - 1] == 0xE2) ||
@@ -230,6 +243,8 @@ struct JsonEmit {
           unsigned w;
           __builtin_memcpy(&w, s + start + 4 * lane, 4);
           if (__ballot(swar_special(w)) == 0) {
+            if (W)
+              __builtin_memcpy(out + cur + 4 * lane, &w, 4);
             cur += 4 * WAVE;   // every byte emits verbatim
             start += 4 * WAVE;
             continue;
@@ -268,23 +283,26 @@ struct JsonEmit {
 
   // Wave-parallel decimal emission: lane j computes digit j-from-the-right
   // ((v / 10^j) % 10) so a 10-digit number costs ~1 division per LANE in
-  // parallel instead of a serial divide chain on lane 0 (the emitters are
-  // VALU-instruction-bound; profiles/r01_pmc_counters.csv).
+  // parallel instead of a serial divide chain on lane 0. 10^lane comes
+  // from a constant table (the per-lane multiply loop cost ~lane VALU
+  // ops per call).
   DEV void u64(unsigned long long v) {
+    static const unsigned long long POW10[20] = {
+        1ULL, 10ULL, 100ULL, 1000ULL, 10000ULL, 100000ULL, 1000000ULL,
+        10000000ULL, 100000000ULL, 1000000000ULL, 10000000000ULL,
+        100000000000ULL, 1000000000000ULL, 10000000000000ULL,
+        100000000000000ULL, 1000000000000000ULL, 10000000000000000ULL,
+        100000000000000000ULL, 1000000000000000000ULL,
+        10000000000000000000ULL};
     int n = u64_dec_len(v);
     if (W) {
       int lane = lane_id();
       if (lane < n) {
         unsigned d;
-        if (v <= 0xFFFFFFFFULL) {
-          unsigned p = 1;
-          for (int k = 0; k < lane; ++k) p *= 10u;   // lane<10: cheap
-          d = ((unsigned)v / p) % 10u;
-        } else {
-          unsigned long long p = 1;
-          for (int k = 0; k < lane; ++k) p *= 10ULL;
-          d = (unsigned)((v / p) % 10ULL);
-        }
+        if (v <= 0xFFFFFFFFULL)
+          d = ((unsigned)v / (unsigned)POW10[lane]) % 10u;
+        else
+          d = (unsigned)((v / POW10[lane]) % 10ULL);
         out[cur + n - 1 - lane] = (uint8_t)('0' + d);
       }
     }
