@@ -42,20 +42,46 @@ DEV int u16_units_of_byte(uint8_t b) {
 // Output bytes for input byte s[p] (with lookahead for U+2028/U+2029).
 // 1 = verbatim; 2 = two-char escape; 6 = \u00xx / \u202x; 0 = swallowed
 // (continuation bytes of an escaped U+2028/29 sequence).
+// Per-byte Go-JSON escape class: 1 verbatim, 2 two-char escape, 6
+// \u00xx, 0xFF = positional (0xE2/0x80/0xA8/0xA9 around U+2028/29).
+// Generated from the rules below; keep in sync with escape_write_at.
+__device__ const uint8_t ESC_CLS[256] = {
+    0x06, 0x06, 0x06, 0x06, 0x06, 0x06, 0x06, 0x06, 0x06, 0x02, 0x02, 0x06, 0x06, 0x02, 0x06, 0x06,
+    0x06, 0x06, 0x06, 0x06, 0x06, 0x06, 0x06, 0x06, 0x06, 0x06, 0x06, 0x06, 0x06, 0x06, 0x06, 0x06,
+    0x01, 0x01, 0x02, 0x01, 0x01, 0x01, 0x06, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01,
+    0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x06, 0x01, 0x06, 0x01,
+    0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01,
+    0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x02, 0x01, 0x01, 0x01,
+    0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01,
+    0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01,
+    0xFF, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01,
+    0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01,
+    0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0xFF, 0xFF, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01,
+    0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01,
+    0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01,
+    0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01,
+    0x01, 0x01, 0xFF, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01,
+    0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01, 0x01,
+};
+
 DEV int escape_len_at(const uint8_t* s, int n, int p) {
   uint8_t c = s[p];
-  if (c == '"' || c == '\\' || c == '\n' || c == '\r' || c == '\t') return 2;
-  if (c < 0x20) return 6;
-  if (c == '<' || c == '>' || c == '&') return 6;
-  if (c == 0xE2 && p + 2 < n && s[p + 1] == 0x80 &&
-      (s[p + 2] == 0xA8 || s[p + 2] == 0xA9))
-    return 6;
-  if (c == 0x80 && p >= 1 && s[p - 1] == 0xE2 && p + 1 < n &&
-      (s[p + 1] == 0xA8 || s[p + 1] == 0xA9))
-    return 0;
-  if ((c == 0xA8 || c == 0xA9) && p >= 2 && s[p - 2] == 0xE2 &&
-      s[p - 1] == 0x80)
-    return 0;
+  uint8_t v = ESC_CLS[c];
+  if (v != 0xFF) return v;
+  if (c == 0xE2) {
+    if (p + 2 < n && s[p + 1] == 0x80 &&
+        (s[p + 2] == 0xA8 || s[p + 2] == 0xA9))
+      return 6;
+    return 1;
+  }
+  if (c == 0x80) {
+    if (p >= 1 && s[p - 1] == 0xE2 && p + 1 < n &&
+        (s[p + 1] == 0xA8 || s[p + 1] == 0xA9))
+      return 0;
+    return 1;
+  }
+  // c == 0xA8 || c == 0xA9
+  if (p >= 2 && s[p - 2] == 0xE2 && s[p - 1] == 0x80) return 0;
   return 1;
 }
 
