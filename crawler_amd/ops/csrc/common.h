@@ -183,20 +183,29 @@ DEV void rfc3339_write(uint8_t* out, int cur, long long secs) {
   long long d = doy - (153 * mp + 2) / 5 + 1;
   long long m = mp + (mp < 10 ? 3 : -9);
   y += (m <= 2);
-  char buf[20];
-  int yy = (int)y;
-  buf[0] = '0' + (yy / 1000) % 10; buf[1] = '0' + (yy / 100) % 10;
-  buf[2] = '0' + (yy / 10) % 10;   buf[3] = '0' + yy % 10;
-  buf[4] = '-';
-  buf[5] = '0' + ((int)m) / 10; buf[6] = '0' + ((int)m) % 10;
-  buf[7] = '-';
-  buf[8] = '0' + ((int)d) / 10; buf[9] = '0' + ((int)d) % 10;
-  buf[10] = 'T';
-  buf[11] = '0' + hh / 10; buf[12] = '0' + hh % 10; buf[13] = ':';
-  buf[14] = '0' + mm / 10; buf[15] = '0' + mm % 10; buf[16] = ':';
-  buf[17] = '0' + ss / 10; buf[18] = '0' + ss % 10; buf[19] = 'Z';
+  // Pack the 20 chars into 5 dwords in REGISTERS (a char buf[20] read
+  // with a lane-indexed subscript spills to scratch — the write kernel
+  // was latency-bound on private-segment traffic, not VALU).
+  int yy = (int)y, mo = (int)m, dd = (int)d;
+  auto pk = [](int a, int b, int c, int e) {
+    return (unsigned)a | ((unsigned)b << 8) | ((unsigned)c << 16) |
+           ((unsigned)e << 24);
+  };
+  unsigned w0 = pk('0' + (yy / 1000) % 10, '0' + (yy / 100) % 10,
+                   '0' + (yy / 10) % 10, '0' + yy % 10);
+  unsigned w1 = pk('-', '0' + mo / 10, '0' + mo % 10, '-');
+  unsigned w2 = pk('0' + dd / 10, '0' + dd % 10, 'T', '0' + hh / 10);
+  unsigned w3 = pk('0' + hh % 10, ':', '0' + mm / 10, '0' + mm % 10);
+  unsigned w4 = pk(':', '0' + ss / 10, '0' + ss % 10, 'Z');
   int lane = lane_id();
-  if (lane < 20) out[cur + lane] = (uint8_t)buf[lane];
+  if (lane < 5) {
+    unsigned v = w0;
+    if (lane == 1) v = w1;
+    else if (lane == 2) v = w2;
+    else if (lane == 3) v = w3;
+    else if (lane == 4) v = w4;
+    __builtin_memcpy(out + cur + 4 * lane, &v, 4);
+  }
 }
 
 #define LIT(e, s) (e).raw((const unsigned char*)(s), (int)sizeof(s) - 1)
@@ -298,9 +307,10 @@ struct JsonEmit {
       if (W) {
         int off = wave_prefix_excl(el);
         if (p < n && el > 0) {
-          uint8_t tmp[6];
-          int w = escape_write_at(s, n, p, tmp);
-          for (int j = 0; j < w; ++j) out[cur + off + j] = tmp[j];
+          // write the expansion straight to its global slot — a local
+          // tmp[6] staging array spills to scratch (private-segment
+          // latency dominated the write kernel)
+          escape_write_at(s, n, p, out + cur + off);
         }
       }
       cur += wave_sum(el);
