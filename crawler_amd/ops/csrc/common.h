@@ -208,39 +208,16 @@ DEV void rfc3339_write(uint8_t* out, int cur, long long secs) {
   }
 }
 
-// Literal emission as IMMEDIATE stores: the write kernel is
-// latency-bound on its ~1.3 KB/line of constant-template bytes — a
-// load-from-rodata + store chain per fragment serializes on s_waitcnt.
-// The unrolled constant-index reads of a string literal fold to
-// immediates at -O3, so each fragment becomes mov+store pairs with no
-// loads at all (single-lane; stores never block issue).
-#define LIT(e, s) (e).template lit<(int)sizeof(s) - 1>(s)
+#define LIT(e, s) (e).raw((const unsigned char*)(s), (int)sizeof(s) - 1)
 
 template <bool W>
 struct JsonEmit {
   uint8_t* out;
   int cur;
 
-  template <int N, int M>
-  DEV void lit(const char (&str)[M]) {
-    if (W && lane_id() == 0) {
-      int k = 0;
-#pragma unroll
-      for (; k + 4 <= N; k += 4) {
-        unsigned v = (unsigned)(unsigned char)str[k] |
-                     ((unsigned)(unsigned char)str[k + 1] << 8) |
-                     ((unsigned)(unsigned char)str[k + 2] << 16) |
-                     ((unsigned)(unsigned char)str[k + 3] << 24);
-        __builtin_memcpy(out + cur + k, &v, 4);
-      }
-#pragma unroll
-      for (; k < N; ++k) out[cur + k] = (unsigned char)str[k];
-    }
-    cur += N;
-  }
-
-  // Striped copy, dword-granular: unaligned 4-byte loads/stores quarter
-  // the memory-op count vs byte stripes (variable-content fields).
+  // Striped copy, dword-granular: the emitters are memory-op bound on
+  // the literal/template bytes (~1.3 KB of raw copies per 2 KB line) —
+  // unaligned 4-byte loads/stores quarter the op count vs byte stripes.
   DEV void raw(const uint8_t* s, int n) {
     if (W) {
       int lane = lane_id();
