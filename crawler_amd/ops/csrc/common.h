@@ -12,6 +12,11 @@
 
 #define WAVE 64
 #define DEV __device__ __forceinline__
+// Out-of-line: emit_line expands ~85 helper calls; fully inlined the
+// write kernel is ~77k instructions (~300 KB) of straight-line code
+// against a 32 KB L1I -> instruction-fetch bound. Shared noinline
+// bodies keep the hot loop I-cache resident.
+#define DEVNI __device__ __noinline__
 
 namespace crawl {
 
@@ -166,7 +171,7 @@ DEV int i64_dec_len(int64_t v) {
   return (v < 0) ? 1 + u64_dec_len((uint64_t)(-v)) : u64_dec_len((uint64_t)v);
 }
 
-DEV void rfc3339_write(uint8_t* out, int cur, long long secs) {
+DEVNI void rfc3339_write(uint8_t* out, int cur, long long secs) {
   long long days = secs / 86400;
   long long rem = secs % 86400;
   if (rem < 0) { rem += 86400; days -= 1; }
@@ -218,7 +223,7 @@ struct JsonEmit {
   // Striped copy, dword-granular: the emitters are memory-op bound on
   // the literal/template bytes (~1.3 KB of raw copies per 2 KB line) —
   // unaligned 4-byte loads/stores quarter the op count vs byte stripes.
-  DEV void raw(const uint8_t* s, int n) {
+  DEVNI void raw(const uint8_t* s, int n) {
     if (W) {
       int lane = lane_id();
       int nw = n >> 2;
@@ -264,7 +269,7 @@ struct JsonEmit {
   // (same ballot -> measure and write can never disagree on length).
   // Dirty or boundary-risk blocks fall back to the exact 64-byte
   // stripe path below.
-  DEV void esc(const uint8_t* s, int n) {
+  DEVNI void esc(const uint8_t* s, int n) {
     int lane = lane_id();
     int start = 0;
     while (start < n) {
@@ -322,7 +327,7 @@ struct JsonEmit {
   // parallel instead of a serial divide chain on lane 0. 10^lane comes
   // from a constant table (the per-lane multiply loop cost ~lane VALU
   // ops per call).
-  DEV void u64(unsigned long long v) {
+  DEVNI void u64(unsigned long long v) {
     static const unsigned long long POW10[20] = {
         1ULL, 10ULL, 100ULL, 1000ULL, 10000ULL, 100000ULL, 1000000ULL,
         10000000ULL, 100000000ULL, 1000000000ULL, 10000000000ULL,
