@@ -12,11 +12,19 @@
 
 #define WAVE 64
 #define DEV __device__ __forceinline__
-// Out-of-line: emit_line expands ~85 helper calls; fully inlined the
-// write kernel is ~77k instructions (~300 KB) of straight-line code
-// against a 32 KB L1I -> instruction-fetch bound. Shared noinline
-// bodies keep the hot loop I-cache resident.
-#define DEVNI __device__ __noinline__
+
+// Non-temporal (L2-streaming) stores for the JSONL output: the line
+// bytes are written once and never read on device, and 2.5 GB/batch of
+// regular stores evicts the literal/template and text-pool lines the
+// emitters re-read every line — the write kernel stalls ~one
+// HBM-latency per emitter call without this.
+typedef unsigned int __attribute__((aligned(1))) u32_unal;
+DEV void nt_store_u32(unsigned char* p, unsigned v) {
+  __builtin_nontemporal_store(v, (u32_unal*)p);
+}
+DEV void nt_store_u8(unsigned char* p, unsigned char v) {
+  __builtin_nontemporal_store(v, p);
+}
 
 namespace crawl {
 
@@ -171,7 +179,7 @@ DEV int i64_dec_len(int64_t v) {
   return (v < 0) ? 1 + u64_dec_len((uint64_t)(-v)) : u64_dec_len((uint64_t)v);
 }
 
-DEVNI void rfc3339_write(uint8_t* out, int cur, long long secs) {
+DEV void rfc3339_write(uint8_t* out, int cur, long long secs) {
   long long days = secs / 86400;
   long long rem = secs % 86400;
   if (rem < 0) { rem += 86400; days -= 1; }
@@ -209,7 +217,7 @@ DEVNI void rfc3339_write(uint8_t* out, int cur, long long secs) {
     else if (lane == 2) v = w2;
     else if (lane == 3) v = w3;
     else if (lane == 4) v = w4;
-    __builtin_memcpy(out + cur + 4 * lane, &v, 4);
+    nt_store_u32(out + cur + 4 * lane, v);
   }
 }
 
@@ -223,17 +231,17 @@ struct JsonEmit {
   // Striped copy, dword-granular: the emitters are memory-op bound on
   // the literal/template bytes (~1.3 KB of raw copies per 2 KB line) —
   // unaligned 4-byte loads/stores quarter the op count vs byte stripes.
-  DEVNI void raw(const uint8_t* s, int n) {
+  DEV void raw(const uint8_t* s, int n) {
     if (W) {
       int lane = lane_id();
       int nw = n >> 2;
       for (int k = lane; k < nw; k += WAVE) {
         unsigned w;
         __builtin_memcpy(&w, s + 4 * k, 4);
-        __builtin_memcpy(out + cur + 4 * k, &w, 4);
+        nt_store_u32(out + cur + 4 * k, w);
       }
       for (int j = (nw << 2) + lane; j < n; j += WAVE)
-        out[cur + j] = s[j];
+        nt_store_u8(out + cur + j, s[j]);
     }
     cur += n;
   }
@@ -269,7 +277,7 @@ struct JsonEmit {
   // (same ballot -> measure and write can never disagree on length).
   // Dirty or boundary-risk blocks fall back to the exact 64-byte
   // stripe path below.
-  DEVNI void esc(const uint8_t* s, int n) {
+  DEV void esc(const uint8_t* s, int n) {
     int lane = lane_id();
     int start = 0;
     while (start < n) {
@@ -284,7 +292,7 @@ struct JsonEmit {
           __builtin_memcpy(&w, s + start + 4 * lane, 4);
           if (__ballot(swar_special(w)) == 0) {
             if (W)
-              __builtin_memcpy(out + cur + 4 * lane, &w, 4);
+              nt_store_u32(out + cur + 4 * lane, w);
             cur += 4 * WAVE;   // every byte emits verbatim
             start += 4 * WAVE;
             continue;
@@ -305,7 +313,7 @@ struct JsonEmit {
       int el = (p < n) ? escape_len_at(s, n, p) : 0;
       unsigned long long dirty = __ballot(p < n && el != 1);
       if (dirty == 0) {
-        if (W && p < n) out[cur + p - start] = s[p];
+        if (W && p < n) nt_store_u8(out + cur + p - start, s[p]);
         cur += span;
         return;
       }
@@ -327,7 +335,7 @@ struct JsonEmit {
   // parallel instead of a serial divide chain on lane 0. 10^lane comes
   // from a constant table (the per-lane multiply loop cost ~lane VALU
   // ops per call).
-  DEVNI void u64(unsigned long long v) {
+  DEV void u64(unsigned long long v) {
     static const unsigned long long POW10[20] = {
         1ULL, 10ULL, 100ULL, 1000ULL, 10000ULL, 100000ULL, 1000000ULL,
         10000000ULL, 100000000ULL, 1000000000ULL, 10000000000ULL,
@@ -344,7 +352,7 @@ struct JsonEmit {
           d = ((unsigned)v / (unsigned)POW10[lane]) % 10u;
         else
           d = (unsigned)((v / POW10[lane]) % 10ULL);
-        out[cur + n - 1 - lane] = (uint8_t)('0' + d);
+        nt_store_u8(out + cur + n - 1 - lane, (uint8_t)('0' + d));
       }
     }
     cur += n;
