@@ -42,12 +42,20 @@ def synth_blob_size(media_id: str, seed: int = 0) -> int:
     return 10 * 1024 + (h % (2 * 1024 * 1024))  # thumbnail-ish
 
 
+_IDX_CACHE = {}  # device -> cached arange (grown on demand)
+
+
 def device_blob(media_id: str, size: int, device,
                 seed: int = 0) -> torch.Tensor:
     """Deterministic device-resident blob (uint8[size])."""
     base = (_stable_hash(f"{media_id}:{seed}:blob") & 0x7FFFFFFF) or 1
-    idx = torch.arange(size, device=device, dtype=torch.int64)
-    return ((idx * 2654435761 + base) >> 7).to(torch.uint8)
+    key = str(device)
+    idx = _IDX_CACHE.get(key)
+    if idx is None or idx.numel() < size:
+        idx = torch.arange(max(size, 8 << 20), device=device,
+                           dtype=torch.int64)
+        _IDX_CACHE[key] = idx
+    return ((idx[:size] * 2654435761 + base) >> 7).to(torch.uint8)
 
 
 class MediaEngine:
@@ -137,14 +145,18 @@ class MediaEngine:
         blocking get() with all slots in flight would deadlock)."""
         import collections
 
+        import numpy as np
+
         blob = device_blob(media_id, size, self.device, self.seed)
         pending = collections.deque()  # (buf, n, ev, dst_off)
         data = bytearray(size)
+        view = np.frombuffer(data, dtype=np.uint8)
 
         def drain_one():
             buf, n, ev, dst = pending.popleft()
             ev.synchronize()
-            data[dst:dst + n] = bytes(buf[:n].numpy())
+            # single copy pinned->bytearray (bytes() cost a second one)
+            view[dst:dst + n] = buf[:n].numpy()
             self._ring_free.put(buf)
 
         for off in range(0, size, self._slot_bytes):
@@ -159,7 +171,9 @@ class MediaEngine:
             pending.append((buf, n, ev, off))
         while pending:
             drain_one()
-        self._q.put((path, bytes(data), None))
+        # hand the bytearray itself to the writer (no bytes() copy; we
+        # never touch it again)
+        self._q.put((path, data, None))
 
     def flush(self):
         import time
