@@ -56,6 +56,12 @@ def main():
                     help="videos per GPU for the youtube platform")
     ap.add_argument("--chunk-videos", type=int, default=100_000)
     ap.add_argument("--max-comments", type=int, default=1000)
+    ap.add_argument("--comment-rate", type=float, default=0.02,
+                    help="telegram feed: fraction of posts with comment "
+                         "threads (comment-heavy variant: 1.0)")
+    ap.add_argument("--max-comments-per-post", type=int, default=3,
+                    help="telegram feed: thread size cap (comment-heavy "
+                         "variant: 250 exercises >100-comment posts)")
     ap.add_argument("--cpu", action="store_true",
                     help="debug: run the Python golden path on CPU (tiny)")
     args = ap.parse_args()
@@ -90,8 +96,10 @@ def main():
         n_chunks = max(1, args.channels // args.chunk_channels)
         chunk_ch = args.chunk_channels
         posts = args.posts
-        feed = SyntheticFeed(FeedConfig(seed=1234 + rank,
-                                        universe=1_000_000))
+        feed = SyntheticFeed(FeedConfig(
+            seed=1234 + rank, universe=1_000_000,
+            comment_rate=args.comment_rate,
+            max_comments_per_post=args.max_comments_per_post))
         for c in range(n_chunks):
             ids = (np.arange(c * chunk_ch, (c + 1) * chunk_ch)
                    + rank * args.channels)
@@ -142,8 +150,13 @@ def main():
             return torch.empty(nbytes, dtype=torch.uint8)
 
     # worst-case line sizes differ by platform: telegram ~2.1KB/post,
-    # youtube ~3.0KB/video (richer channel block per line)
+    # youtube ~3.0KB/video (richer channel block per line). Comment
+    # threads add ~230B per comment (avg thread = half the cap) — the
+    # ring assert below catches any underestimate loudly.
     per_post = 2600 if my_platform == "telegram" else 3600
+    if my_platform == "telegram" and args.comment_rate > 0:
+        avg_thread = args.max_comments_per_post / 2 + 1
+        per_post += int(args.comment_rate * avg_thread * 230 * 1.4)
     pinned = [alloc_host(int(chunk_posts * per_post)) for _ in range(2)]
     compute_stream = torch.cuda.Stream()
     # two copy streams: MI355X has multiple SDMA engines; splitting the
@@ -266,9 +279,32 @@ def main():
     log(f"primed {prim} iterations ({time.time() - t_pr:.1f}s)")
 
     # ---- timed region ----
+    # In-loop GPU-busy sampling (sysfs gpu_busy_percent, ~50 ms cadence)
+    # so short runs carry their own utilization record — a single
+    # post-hoc SMI sample can miss a 5 s run entirely.
+    import glob as _glob
+    import threading as _threading
+    busy_samples = []
+    stop_busy = _threading.Event()
+    busy_paths = sorted(_glob.glob(
+        "/sys/class/drm/card*/device/gpu_busy_percent"))
+
+    def _busy_sampler():
+        while not stop_busy.is_set():
+            for p in busy_paths:
+                try:
+                    with open(p) as f:
+                        busy_samples.append(int(f.read().strip()))
+                except (OSError, ValueError):
+                    pass
+            stop_busy.wait(0.05)
+
+    sampler = _threading.Thread(target=_busy_sampler, daemon=True)
+
     if world > 1:
         torch.distributed.barrier()
     torch.cuda.synchronize()
+    sampler.start()
     step_times = []
     t0 = time.perf_counter()
     for si in range(args.steps):
@@ -276,6 +312,7 @@ def main():
         step(args.warmup + si)
         step_times.append((time.perf_counter() - ts) * 1000)
     torch.cuda.synchronize()
+    stop_busy.set()
     if world > 1:
         torch.distributed.barrier()
     t1 = time.perf_counter()
@@ -355,6 +392,14 @@ def main():
                 "sink": args.sink,
                 "pinned": all(r["pinned"] for r in per_rank_host),
                 "per_rank_host": per_rank_host,
+                "gpu_busy_sampled": (
+                    {"mean": round(sum(busy_samples) / len(busy_samples),
+                                   1),
+                     "max": max(busy_samples),
+                     "n": len(busy_samples)}
+                    if busy_samples else None),
+                "comment_rate": args.comment_rate,
+                "max_comments_per_post": args.max_comments_per_post,
             },
         }
         print(json.dumps(result), flush=True)
