@@ -248,7 +248,7 @@ def main():
         one = feed.build_batch_device(
             np.array([999_990 + rank]), device, posts_per_channel=args.posts
         )
-        lat_pin = alloc_host(args.posts * 3000)
+        lat_pin = alloc_host(args.posts * (per_post + 400))
         for _ in range(11):
             torch.cuda.synchronize()
             t = time.perf_counter()
