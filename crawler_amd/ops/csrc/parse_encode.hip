@@ -25,74 +25,74 @@ namespace crawl {
 
 struct BatchView {
   // per-message
-  const long* chat_id;
-  const long* msg_id;
-  const long* text_off;
-  const int* date;
-  const int* content_type;
-  const int* views;
-  const int* forwards;
-  const int* media_album_id;
-  const int* channel_idx;
-  const int* flags;
-  const int* text_len;
-  const int* aux_off;
-  const int* aux_len;
-  const int* ent_off;
-  const int* ent_cnt;
-  const int* react_off;
-  const int* react_cnt;
-  const int* com_off;
-  const int* com_cnt;
-  const int* poster_off;
-  const int* poster_len;
+  const long* __restrict__ chat_id;
+  const long* __restrict__ msg_id;
+  const long* __restrict__ text_off;
+  const int* __restrict__ date;
+  const int* __restrict__ content_type;
+  const int* __restrict__ views;
+  const int* __restrict__ forwards;
+  const int* __restrict__ media_album_id;
+  const int* __restrict__ channel_idx;
+  const int* __restrict__ flags;
+  const int* __restrict__ text_len;
+  const int* __restrict__ aux_off;
+  const int* __restrict__ aux_len;
+  const int* __restrict__ ent_off;
+  const int* __restrict__ ent_cnt;
+  const int* __restrict__ react_off;
+  const int* __restrict__ react_cnt;
+  const int* __restrict__ com_off;
+  const int* __restrict__ com_cnt;
+  const int* __restrict__ poster_off;
+  const int* __restrict__ poster_len;
   // pools / tables
-  const unsigned char* pool;
-  const int* entities;  // [E,5] etype,off16,len16,url_off,url_len
-  const int* react_emoji;
-  const int* react_count;
-  const int* com_text_off;
-  const int* com_text_len;
-  const int* com_handle_off;
-  const int* com_handle_len;
-  const int* com_views;
-  const int* com_replies;
-  const int* com_react_off;
-  const int* com_react_cnt;
+  const unsigned char* __restrict__ pool;
+  const int* __restrict__ entities;  // [E,5] etype,off16,len16,url_off,url_len
+  const int* __restrict__ react_emoji;
+  const int* __restrict__ react_count;
+  const int* __restrict__ com_text_off;
+  const int* __restrict__ com_text_len;
+  const int* __restrict__ com_handle_off;
+  const int* __restrict__ com_handle_len;
+  const int* __restrict__ com_views;
+  const int* __restrict__ com_replies;
+  const int* __restrict__ com_react_off;
+  const int* __restrict__ com_react_cnt;
   // channel table
-  const long* ch_chat_id;
-  const int* ch_member;
-  const int* ch_postcount;
-  const int* ch_totalviews;
-  const int* ch_user_off;
-  const int* ch_user_len;
-  const int* ch_title_off;
-  const int* ch_title_len;
+  const long* __restrict__ ch_chat_id;
+  const int* __restrict__ ch_member;
+  const int* __restrict__ ch_postcount;
+  const int* __restrict__ ch_totalviews;
+  const int* __restrict__ ch_user_off;
+  const int* __restrict__ ch_user_len;
+  const int* __restrict__ ch_title_off;
+  const int* __restrict__ ch_title_len;
   // emoji vocabulary
-  const unsigned char* emoji_pool;
-  const int* emoji_off;
-  const int* emoji_len;
+  const unsigned char* __restrict__ emoji_pool;
+  const int* __restrict__ emoji_off;
+  const int* __restrict__ emoji_len;
   // timestamps (preformatted Go time strings)
-  const unsigned char* created_str;
+  const unsigned char* __restrict__ created_str;
   int created_len;
-  const unsigned char* capture_str;
+  const unsigned char* __restrict__ capture_str;
   int capture_len;
   // config
   int n;
   int skip_media;
   long min_post_date;  // unix secs; <= filters nothing when INT64_MIN
   // content-type name table
-  const unsigned char* ctname_pool;
-  const int* ctname_off;
-  const int* ctname_len;
+  const unsigned char* __restrict__ ctname_pool;
+  const int* __restrict__ ctname_off;
+  const int* __restrict__ ctname_len;
 };
 
 struct LinkOut {
-  unsigned char* name;     // [N, MAX_LINKS, 32]
-  unsigned char* name_len; // [N, MAX_LINKS]
-  unsigned char* src;      // [N, MAX_LINKS] 0=mention 1=text_url 2=url 3=plain
-  int* cnt;                // [N]
-  unsigned long long* hash;// [N, MAX_LINKS] fnv1a64 of name
+  unsigned char* __restrict__ name;     // [N, MAX_LINKS, 32]
+  unsigned char* __restrict__ name_len; // [N, MAX_LINKS]
+  unsigned char* __restrict__ src;      // [N, MAX_LINKS] 0=mention 1=text_url 2=url 3=plain
+  int* __restrict__ cnt;                // [N]
+  unsigned long long* __restrict__ hash;// [N, MAX_LINKS] fnv1a64 of name
 };
 
 // ---------- UTF-16 offset resolution (oracle: golden.utf16_offset_to_bytes)
@@ -457,7 +457,7 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
 // ---------- kernels ----------
 
 __global__ void __launch_bounds__(256, 6)
-measure_extract_kernel(BatchView B, LinkOut LO, int* line_len) {
+measure_extract_kernel(BatchView B, LinkOut LO, int* __restrict__ line_len) {
   const int lane = lane_id();
   const int wave = wave_id();
   const int waves_per_grid = gridDim.x * 4;
@@ -481,8 +481,8 @@ measure_extract_kernel(BatchView B, LinkOut LO, int* line_len) {
 }
 
 __global__ void __launch_bounds__(256, 4)
-write_kernel(BatchView B, LinkOut LO, const long* line_off,
-             const int* line_len, unsigned char* out) {
+write_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
+             const int* __restrict__ line_len, unsigned char* __restrict__ out) {
   const int lane = lane_id();
   const int wave = wave_id();
   const int waves_per_grid = gridDim.x * 4;
@@ -507,8 +507,8 @@ DEV void copy_line(const unsigned char* src, unsigned char* dst, int n,
                    int lane);
 
 __global__ void __launch_bounds__(256)
-write_lds_kernel(BatchView B, LinkOut LO, const long* line_off,
-                 const int* line_len, unsigned char* out) {
+write_lds_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
+                 const int* __restrict__ line_len, unsigned char* __restrict__ out) {
   __shared__ unsigned char lbuf[4][LDS_LINE_BYTES];
   const int lane = lane_id();
   const int wave = wave_id();
@@ -535,8 +535,8 @@ write_lds_kernel(BatchView B, LinkOut LO, const long* line_off,
 // measuring pass; compaction is pure memcpy-rate).
 
 __global__ void __launch_bounds__(256)
-write_scratch_kernel(BatchView B, LinkOut LO, unsigned char* scratch,
-                     long stride, int* line_len, int* overflow) {
+write_scratch_kernel(BatchView B, LinkOut LO, unsigned char* __restrict__ scratch,
+                     long stride, int* __restrict__ line_len, int* __restrict__ overflow) {
   const int lane = lane_id();
   const int wave = wave_id();
   const int waves_per_grid = gridDim.x * 4;
