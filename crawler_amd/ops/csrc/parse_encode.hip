@@ -290,6 +290,10 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
   Emit<W> e{};
   e.out = out;
   e.cur = 0;
+  // Load EVERY per-line scalar BEFORE the first store: vmcnt is a
+  // single FIFO over loads AND stores on CDNA, so a load issued after
+  // stores waits for those stores to retire (the write kernel spent
+  // ~85 full-pipeline waits per line on lazily-loaded fields).
   const int c = B.channel_idx[i];
   const unsigned char* user = B.pool + B.ch_user_off[c];
   const int user_n = B.ch_user_len[c];
@@ -299,6 +303,26 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
   const int ct = B.content_type[i];
   const int ncom = B.com_cnt[i];
   const bool has_user = user_n > 0;
+  const long chat_id_v = B.chat_id[i];
+  const int media_album = B.media_album_id[i];
+  const int date_v = B.date[i];
+  const int views_v = B.views[i];
+  const int forwards_v = B.forwards[i];
+  const int ch_member_v = B.ch_member[c];
+  const int ch_postcount_v = B.ch_postcount[c];
+  const int ch_totalviews_v = B.ch_totalviews[c];
+  const int flags_v = B.flags[i];
+  const int com0_v = B.com_off[i];
+  const int react0_v = B.react_off[i];
+  const int reactc_v = B.react_cnt[i];
+  const int poster_off_v = B.poster_off[i];
+  const int poster_len_v = B.poster_len[i];
+  const int text_off_v = B.text_off[i];
+  const int text_len_v = B.text_len[i];
+  const int aux_off_v = B.aux_off[i];
+  const int aux_len_v = B.aux_len[i];
+  const int ctname_off_v = B.ctname_off[ct];
+  const int ctname_len_v = B.ctname_len[ct];
 
   // post_link / url (tdutils.go:1005-1031; empty for private channels)
   auto post_link = [&]() {
@@ -307,14 +331,14 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
     e.esc(user, user_n);
     LIT(e, "/");
     e.i64(pub_id);
-    if (B.media_album_id[i] != 0) LIT(e, "?single");
+    if (media_album != 0) LIT(e, "?single");
     LIT(e, "\"");
   };
 
   LIT(e, "{\"post_link\":");
   post_link();
   LIT(e, ",\"channel_id\":\"");
-  e.i64(B.chat_id[i]);
+  e.i64(chat_id_v);
   LIT(e, "\",\"post_uid\":\"");
   e.i64(pub_id);
   LIT(e, "-");
@@ -322,15 +346,15 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
   LIT(e, "\",\"url\":");
   post_link();
   LIT(e, ",\"published_at\":\"");
-  e.rfc3339(B.date[i]);
+  e.rfc3339(date_v);
   LIT(e, "\",\"created_at\":\"");
   e.raw(B.created_str, B.created_len);
   LIT(e, "\",\"language_code\":\"\",\"engagement\":");
-  e.i64(B.views[i]);
+  e.i64(views_v);
   LIT(e, ",\"view_count\":");
-  e.i64(B.views[i]);
+  e.i64(views_v);
   LIT(e, ",\"like_count\":0,\"share_count\":");
-  e.i64(B.forwards[i]);
+  e.i64(forwards_v);
   LIT(e, ",\"comment_count\":");
   e.i64(ncom);
   LIT(e, ",\"crawl_label\":\"\",\"list_ids\":null,\"channel_name\":");
@@ -341,16 +365,16 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
         "\"all_labels\":null,\"label_ids\":null,\"is_ad\":false,"
         "\"transcript_text\":\"\",\"image_text\":\"\",\"video_length\":null,"
         "\"is_verified\":null,\"channel_data\":{\"channel_id\":\"");
-  e.i64(B.chat_id[i]);
+  e.i64(chat_id_v);
   LIT(e, "\",\"channel_name\":");
   e.qesc(title, title_n);
   LIT(e, ",\"channel_description\":\"\",\"channel_profile_image\":\"\","
         "\"channel_engagement_data\":{\"follower_count\":");
-  e.i64(B.ch_member[c]);
+  e.i64(ch_member_v);
   LIT(e, ",\"following_count\":0,\"like_count\":0,\"post_count\":");
-  e.i64(B.ch_postcount[c]);
+  e.i64(ch_postcount_v);
   LIT(e, ",\"views_count\":");
-  e.i64(B.ch_totalviews[c]);
+  e.i64(ch_totalviews_v);
   LIT(e, ",\"comment_count\":0,\"share_count\":0},"
         "\"channel_url_external\":\"https://t.me/c/");
   e.esc(user, user_n);
@@ -368,11 +392,11 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
     const unsigned char* d;
     int dn;
     if (ct == 0 || ct == 1 || ct == 2 || ct == 4 || ct == 14) {
-      d = B.pool + B.text_off[i];
-      dn = B.text_len[i];
+      d = B.pool + text_off_v;
+      dn = text_len_v;
     } else if (ct == 3 || ct == 9 || ct == 10 || ct == 11) {
-      d = B.pool + B.aux_off[i];
-      dn = B.aux_len[i];
+      d = B.pool + aux_off_v;
+      dn = aux_len_v;
     } else {
       d = nullptr;
       dn = 0;
@@ -380,29 +404,29 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
     e.qesc(d, dn);
   }
   LIT(e, ",\"repost_channel_data\":null,\"post_type\":[\"");
-  e.raw(B.ctname_pool + B.ctname_off[ct], B.ctname_len[ct]);
+  e.raw(B.ctname_pool + ctname_off_v, ctname_len_v);
   LIT(e, "\"],\"inner_link\":{},\"post_title\":null,\"media_data\":"
         "{\"document_name\":\"\"},\"is_reply\":null,\"ad_fields\":null,"
         "\"likes_count\":0,\"shares_count\":");
-  e.i64(B.forwards[i]);
+  e.i64(forwards_v);
   LIT(e, ",\"comments_count\":");
   e.i64(ncom);
   LIT(e, ",\"views_count\":");
-  e.i64(B.views[i]);
+  e.i64(views_v);
   LIT(e, ",\"searchable_text\":\"\",\"all_text\":\"\","
         "\"contrast_agent_project_ids\":null,\"agent_ids\":null,"
         "\"segment_ids\":null,\"thumb_url\":\"");
   // media (fetchAndUploadMedia skip rules, tdutils.go:233-239): GPU path
   // always runs skip_media (media-on is staged host-side).
   LIT(e, "\",\"media_url\":\"");
-  if ((ct == 3 || ct == 8) && (B.flags[i] & 2)) {
+  if ((ct == 3 || ct == 8) && (flags_v & 2)) {
     LIT(e, "AgAD");
     e.i64(pub_id);
     LIT(e, "v");
   }
   LIT(e, "\",\"comments\":[");
   {
-    const int c0 = B.com_off[i];
+    const int c0 = com0_v;
     for (int k = 0; k < ncom; ++k) {
       if (k) LIT(e, ",");
       const int cc = c0 + k;
@@ -429,7 +453,7 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
   }
   LIT(e, "],\"reactions\":{");
   {
-    const int r0 = B.react_off[i], rc = B.react_cnt[i];
+    const int r0 = react0_v, rc = reactc_v;
     for (int r = 0; r < rc; ++r) {
       if (r) LIT(e, ",");
       int em = B.react_emoji[r0 + r];
@@ -449,7 +473,7 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
   LIT(e, "],\"capture_time\":\"");
   e.raw(B.capture_str, B.capture_len);
   LIT(e, "\",\"handle\":");
-  e.qesc(B.pool + B.poster_off[i], B.poster_len[i]);
+  e.qesc(B.pool + poster_off_v, poster_len_v);
   LIT(e, "}\n");
   return e.cur;
 }
