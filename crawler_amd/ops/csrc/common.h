@@ -227,6 +227,27 @@ template <bool W>
 struct JsonEmit {
   uint8_t* out;
   int cur;
+  const uint8_t* lds;  // LDS literal pool base (tg_lits.h), may be null
+
+  // Pooled constant fragment served from LDS: ds_read waits on
+  // lgkmcnt, NOT the loads+stores vmcnt FIFO — so a literal after line
+  // stores doesn't stall on those stores retiring (the write kernel's
+  // dominant stall; see profiles/r02_valu_diet.md).
+  DEV void lds_lit(int off, int n) {
+    if (W) {
+      const uint8_t* s = lds + off;
+      int lane = lane_id();
+      int nw = n >> 2;
+      for (int k = lane; k < nw; k += WAVE) {
+        unsigned w;
+        __builtin_memcpy(&w, s + 4 * k, 4);
+        nt_store_u32(out + cur + 4 * k, w);
+      }
+      for (int j = (nw << 2) + lane; j < n; j += WAVE)
+        nt_store_u8(out + cur + j, s[j]);
+    }
+    cur += n;
+  }
 
   // Striped copy, dword-granular: the emitters are memory-op bound on
   // the literal/template bytes (~1.3 KB of raw copies per 2 KB line) —

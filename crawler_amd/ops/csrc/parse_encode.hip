@@ -18,6 +18,7 @@
 // crawler_amd/ops/golden_batch.py (byte-identical output required).
 
 #include "common.h"
+#include "tg_lits.h"
 
 namespace crawl {
 
@@ -285,11 +286,12 @@ struct Emit : JsonEmit<W> {
 
 template <bool W>
 DEV int emit_line(const BatchView& B, int i, unsigned char* out,
-                  LinkList& L) {
+                  LinkList& L, const unsigned char* lds_lits) {
   const int lane = lane_id();
   Emit<W> e{};
   e.out = out;
   e.cur = 0;
+  e.lds = lds_lits;
   // Load EVERY per-line scalar BEFORE the first store: vmcnt is a
   // single FIFO over loads AND stores on CDNA, so a load issued after
   // stores waits for those stores to retire (the write kernel spent
@@ -327,66 +329,53 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
   // post_link / url (tdutils.go:1005-1031; empty for private channels)
   auto post_link = [&]() {
     if (!has_user) { LIT(e, "\"\""); return; }
-    LIT(e, "\"https://t.me/");
+    TGLIT(e, TGL_0);
     e.esc(user, user_n);
     LIT(e, "/");
     e.i64(pub_id);
-    if (media_album != 0) LIT(e, "?single");
+    if (media_album != 0) TGLIT(e, TGL_1);
     LIT(e, "\"");
   };
 
-  LIT(e, "{\"post_link\":");
+  TGLIT(e, TGL_2);
   post_link();
-  LIT(e, ",\"channel_id\":\"");
+  TGLIT(e, TGL_3);
   e.i64(chat_id_v);
-  LIT(e, "\",\"post_uid\":\"");
+  TGLIT(e, TGL_4);
   e.i64(pub_id);
   LIT(e, "-");
   e.esc(user, user_n);
-  LIT(e, "\",\"url\":");
+  TGLIT(e, TGL_5);
   post_link();
-  LIT(e, ",\"published_at\":\"");
+  TGLIT(e, TGL_6);
   e.rfc3339(date_v);
-  LIT(e, "\",\"created_at\":\"");
+  TGLIT(e, TGL_7);
   e.raw(B.created_str, B.created_len);
-  LIT(e, "\",\"language_code\":\"\",\"engagement\":");
+  TGLIT(e, TGL_8);
   e.i64(views_v);
-  LIT(e, ",\"view_count\":");
+  TGLIT(e, TGL_9);
   e.i64(views_v);
-  LIT(e, ",\"like_count\":0,\"share_count\":");
+  TGLIT(e, TGL_10);
   e.i64(forwards_v);
-  LIT(e, ",\"comment_count\":");
+  TGLIT(e, TGL_11);
   e.i64(ncom);
-  LIT(e, ",\"crawl_label\":\"\",\"list_ids\":null,\"channel_name\":");
+  TGLIT(e, TGL_12);
   e.qesc(title, title_n);
-  LIT(e, ",\"search_terms\":null,\"search_term_ids\":null,\"project_ids\":null,"
-        "\"exercise_ids\":null,\"label_data\":null,\"labels_metadata\":null,"
-        "\"project_labeled_post_ids\":null,\"labeler_ids\":null,"
-        "\"all_labels\":null,\"label_ids\":null,\"is_ad\":false,"
-        "\"transcript_text\":\"\",\"image_text\":\"\",\"video_length\":null,"
-        "\"is_verified\":null,\"channel_data\":{\"channel_id\":\"");
+  TGLIT(e, TGL_13);
   e.i64(chat_id_v);
-  LIT(e, "\",\"channel_name\":");
+  TGLIT(e, TGL_14);
   e.qesc(title, title_n);
-  LIT(e, ",\"channel_description\":\"\",\"channel_profile_image\":\"\","
-        "\"channel_engagement_data\":{\"follower_count\":");
+  TGLIT(e, TGL_15);
   e.i64(ch_member_v);
-  LIT(e, ",\"following_count\":0,\"like_count\":0,\"post_count\":");
+  TGLIT(e, TGL_16);
   e.i64(ch_postcount_v);
-  LIT(e, ",\"views_count\":");
+  TGLIT(e, TGL_17);
   e.i64(ch_totalviews_v);
-  LIT(e, ",\"comment_count\":0,\"share_count\":0},"
-        "\"channel_url_external\":\"https://t.me/c/");
+  TGLIT(e, TGL_18);
   e.esc(user, user_n);
-  LIT(e, "\",\"channel_url\":\"https://t.me/c/");
+  TGLIT(e, TGL_19);
   e.esc(user, user_n);
-  LIT(e, "\",\"country_code\":\"\",\"published_at\":\"0001-01-01T00:00:00Z\"},"
-        "\"platform_name\":\"Telegram\",\"shared_id\":null,"
-        "\"quoted_id\":null,\"replied_id\":null,\"ai_label\":null,"
-        "\"root_post_id\":null,\"engagement_steps_count\":0,\"ocr_data\":null,"
-        "\"performance_scores\":{\"likes\":null,\"shares\":null,"
-        "\"comments\":null,\"views\":0},\"has_embed_media\":null,"
-        "\"description\":");
+  TGLIT(e, TGL_20);
   // description per content switch (tdutils.go:443-587)
   {
     const unsigned char* d;
@@ -403,36 +392,32 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
     }
     e.qesc(d, dn);
   }
-  LIT(e, ",\"repost_channel_data\":null,\"post_type\":[\"");
+  TGLIT(e, TGL_21);
   e.raw(B.ctname_pool + ctname_off_v, ctname_len_v);
-  LIT(e, "\"],\"inner_link\":{},\"post_title\":null,\"media_data\":"
-        "{\"document_name\":\"\"},\"is_reply\":null,\"ad_fields\":null,"
-        "\"likes_count\":0,\"shares_count\":");
+  TGLIT(e, TGL_22);
   e.i64(forwards_v);
-  LIT(e, ",\"comments_count\":");
+  TGLIT(e, TGL_23);
   e.i64(ncom);
-  LIT(e, ",\"views_count\":");
+  TGLIT(e, TGL_17);
   e.i64(views_v);
-  LIT(e, ",\"searchable_text\":\"\",\"all_text\":\"\","
-        "\"contrast_agent_project_ids\":null,\"agent_ids\":null,"
-        "\"segment_ids\":null,\"thumb_url\":\"");
+  TGLIT(e, TGL_24);
   // media (fetchAndUploadMedia skip rules, tdutils.go:233-239): GPU path
   // always runs skip_media (media-on is staged host-side).
-  LIT(e, "\",\"media_url\":\"");
+  TGLIT(e, TGL_25);
   if ((ct == 3 || ct == 8) && (flags_v & 2)) {
     LIT(e, "AgAD");
     e.i64(pub_id);
     LIT(e, "v");
   }
-  LIT(e, "\",\"comments\":[");
+  TGLIT(e, TGL_26);
   {
     const int c0 = com0_v;
     for (int k = 0; k < ncom; ++k) {
       if (k) LIT(e, ",");
       const int cc = c0 + k;
-      LIT(e, "{\"text\":");
+      TGLIT(e, TGL_27);
       e.qesc(B.pool + B.com_text_off[cc], B.com_text_len[cc]);
-      LIT(e, ",\"reactions\":{");
+      TGLIT(e, TGL_28);
       const int r0 = B.com_react_off[cc], rc = B.com_react_cnt[cc];
       for (int r = 0; r < rc; ++r) {
         if (r) LIT(e, ",");
@@ -442,16 +427,16 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
         LIT(e, "\":");
         e.i64(B.react_count[r0 + r]);
       }
-      LIT(e, "},\"view_count\":");
+      TGLIT(e, TGL_29);
       e.i64(B.com_views[cc]);
-      LIT(e, ",\"reply_count\":");
+      TGLIT(e, TGL_30);
       e.i64(B.com_replies[cc]);
-      LIT(e, ",\"handle\":");
+      TGLIT(e, TGL_31);
       e.qesc(B.pool + B.com_handle_off[cc], B.com_handle_len[cc]);
       LIT(e, "}");
     }
   }
-  LIT(e, "],\"reactions\":{");
+  TGLIT(e, TGL_32);
   {
     const int r0 = react0_v, rc = reactc_v;
     for (int r = 0; r < rc; ++r) {
@@ -463,16 +448,16 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
       e.i64(B.react_count[r0 + r]);
     }
   }
-  LIT(e, "},\"outlinks\":[");
+  TGLIT(e, TGL_33);
   for (int k = 0; k < L.cnt; ++k) {
     if (k) LIT(e, ",");
     LIT(e, "\"");
     e.raw(L.names + k * 32, L.lens[k]);
     LIT(e, "\"");
   }
-  LIT(e, "],\"capture_time\":\"");
+  TGLIT(e, TGL_34);
   e.raw(B.capture_str, B.capture_len);
-  LIT(e, "\",\"handle\":");
+  TGLIT(e, TGL_35);
   e.qesc(B.pool + poster_off_v, poster_len_v);
   LIT(e, "}\n");
   return e.cur;
@@ -482,6 +467,10 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
 
 __global__ void __launch_bounds__(256, 6)
 measure_extract_kernel(BatchView B, LinkOut LO, int* __restrict__ line_len) {
+  __shared__ unsigned char s_lits[TG_POOL_BYTES];
+  for (int t = threadIdx.x; t < TG_POOL_BYTES; t += blockDim.x)
+    s_lits[t] = (unsigned char)tg_lit_pool.v[t];
+  __syncthreads();
   const int lane = lane_id();
   const int wave = wave_id();
   const int waves_per_grid = gridDim.x * 4;
@@ -496,7 +485,7 @@ measure_extract_kernel(BatchView B, LinkOut LO, int* __restrict__ line_len) {
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, 0};
     extract_links(B, i, L, lane);
-    int len = emit_line<false>(B, i, nullptr, L);
+    int len = emit_line<false>(B, i, nullptr, L, s_lits);
     if (lane == 0) {
       line_len[i] = len;
       LO.cnt[i] = L.cnt;
@@ -507,6 +496,10 @@ measure_extract_kernel(BatchView B, LinkOut LO, int* __restrict__ line_len) {
 __global__ void __launch_bounds__(256, 4)
 write_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
              const int* __restrict__ line_len, unsigned char* __restrict__ out) {
+  __shared__ unsigned char s_lits[TG_POOL_BYTES];
+  for (int t = threadIdx.x; t < TG_POOL_BYTES; t += blockDim.x)
+    s_lits[t] = (unsigned char)tg_lit_pool.v[t];
+  __syncthreads();
   const int lane = lane_id();
   const int wave = wave_id();
   const int waves_per_grid = gridDim.x * 4;
@@ -516,7 +509,7 @@ write_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
                LO.name_len + (size_t)i * MAX_LINKS,
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, LO.cnt[i]};
-    emit_line<true>(B, i, out + line_off[i], L);
+    emit_line<true>(B, i, out + line_off[i], L, s_lits);
   }
 }
 
@@ -534,6 +527,10 @@ __global__ void __launch_bounds__(256)
 write_lds_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
                  const int* __restrict__ line_len, unsigned char* __restrict__ out) {
   __shared__ unsigned char lbuf[4][LDS_LINE_BYTES];
+  __shared__ unsigned char s_lits[TG_POOL_BYTES];
+  for (int t = threadIdx.x; t < TG_POOL_BYTES; t += blockDim.x)
+    s_lits[t] = (unsigned char)tg_lit_pool.v[t];
+  __syncthreads();
   const int lane = lane_id();
   const int wave = wave_id();
   const int waves_per_grid = gridDim.x * 4;
@@ -545,10 +542,10 @@ write_lds_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, LO.cnt[i]};
     if (len > LDS_LINE_BYTES) {
-      emit_line<true>(B, i, out + line_off[i], L);
+      emit_line<true>(B, i, out + line_off[i], L, s_lits);
       continue;
     }
-    emit_line<true>(B, i, &lbuf[wave][0], L);
+    emit_line<true>(B, i, &lbuf[wave][0], L, s_lits);
     copy_line(&lbuf[wave][0], out + line_off[i], len, lane);
   }
 }
@@ -561,6 +558,10 @@ write_lds_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
 __global__ void __launch_bounds__(256)
 write_scratch_kernel(BatchView B, LinkOut LO, unsigned char* __restrict__ scratch,
                      long stride, int* __restrict__ line_len, int* __restrict__ overflow) {
+  __shared__ unsigned char s_lits[TG_POOL_BYTES];
+  for (int t = threadIdx.x; t < TG_POOL_BYTES; t += blockDim.x)
+    s_lits[t] = (unsigned char)tg_lit_pool.v[t];
+  __syncthreads();
   const int lane = lane_id();
   const int wave = wave_id();
   const int waves_per_grid = gridDim.x * 4;
@@ -574,7 +575,7 @@ write_scratch_kernel(BatchView B, LinkOut LO, unsigned char* __restrict__ scratc
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, 0};
     extract_links(B, i, L, lane);
-    int len = emit_line<true>(B, i, scratch + (size_t)i * stride, L);
+    int len = emit_line<true>(B, i, scratch + (size_t)i * stride, L, s_lits);
     if (lane == 0) {
       line_len[i] = len;
       LO.cnt[i] = L.cnt;
