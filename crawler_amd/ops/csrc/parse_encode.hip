@@ -23,6 +23,9 @@
 namespace crawl {
 
 #define MAX_LINKS 8
+// per-wave LDS staging budget for user+title+poster+description
+// (oversize fields fall back to their global pointers)
+#define TG_STAGE_BYTES 3968
 
 struct BatchView {
   // per-message
@@ -286,7 +289,8 @@ struct Emit : JsonEmit<W> {
 
 template <bool W>
 DEV int emit_line(const BatchView& B, int i, unsigned char* out,
-                  LinkList& L, const unsigned char* lds_lits) {
+                  LinkList& L, const unsigned char* lds_lits,
+                  unsigned char* lds_stage) {
   const int lane = lane_id();
   Emit<W> e{};
   e.out = out;
@@ -325,6 +329,40 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
   const int aux_len_v = B.aux_len[i];
   const int ctname_off_v = B.ctname_off[ct];
   const int ctname_len_v = B.ctname_len[ct];
+  // description source per content switch (tdutils.go:443-587)
+  const unsigned char* desc_p;
+  int desc_n;
+  if (ct == 0 || ct == 1 || ct == 2 || ct == 4 || ct == 14) {
+    desc_p = B.pool + text_off_v;
+    desc_n = text_len_v;
+  } else if (ct == 3 || ct == 9 || ct == 10 || ct == 11) {
+    desc_p = B.pool + aux_off_v;
+    desc_n = aux_len_v;
+  } else {
+    desc_p = nullptr;
+    desc_n = 0;
+  }
+  const unsigned char* poster_p = B.pool + poster_off_v;
+  // Stage the escape-scanned variable fields in LDS at line START: the
+  // staging loads batch into ONE vmcnt wait before any store exists,
+  // and every later esc() read is a ds_read (lgkmcnt) that never waits
+  // on the line's store stream (the vmcnt FIFO orders loads AFTER
+  // stores; see profiles/r02_valu_diet.md).
+  int stage_o = 0;
+  auto staged = [&](const unsigned char* ptr,
+                    int n) -> const unsigned char* {
+    if (!W || lds_stage == nullptr || n <= 0 ||
+        stage_o + n > TG_STAGE_BYTES)
+      return ptr;
+    unsigned char* dst = lds_stage + stage_o;
+    stage_o += n;
+    for (int j = lane; j < n; j += WAVE) dst[j] = ptr[j];
+    return dst;
+  };
+  user = staged(user, user_n);
+  title = staged(title, title_n);
+  poster_p = staged(poster_p, poster_len_v);
+  desc_p = staged(desc_p, desc_n);
 
   // post_link / url (tdutils.go:1005-1031; empty for private channels)
   auto post_link = [&]() {
@@ -376,22 +414,7 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
   TGLIT(e, TGL_19);
   e.esc(user, user_n);
   TGLIT(e, TGL_20);
-  // description per content switch (tdutils.go:443-587)
-  {
-    const unsigned char* d;
-    int dn;
-    if (ct == 0 || ct == 1 || ct == 2 || ct == 4 || ct == 14) {
-      d = B.pool + text_off_v;
-      dn = text_len_v;
-    } else if (ct == 3 || ct == 9 || ct == 10 || ct == 11) {
-      d = B.pool + aux_off_v;
-      dn = aux_len_v;
-    } else {
-      d = nullptr;
-      dn = 0;
-    }
-    e.qesc(d, dn);
-  }
+  e.qesc(desc_p, desc_n);
   TGLIT(e, TGL_21);
   e.raw(B.ctname_pool + ctname_off_v, ctname_len_v);
   TGLIT(e, TGL_22);
@@ -458,7 +481,7 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
   TGLIT(e, TGL_34);
   e.raw(B.capture_str, B.capture_len);
   TGLIT(e, TGL_35);
-  e.qesc(B.pool + poster_off_v, poster_len_v);
+  e.qesc(poster_p, poster_len_v);
   LIT(e, "}\n");
   return e.cur;
 }
@@ -485,7 +508,7 @@ measure_extract_kernel(BatchView B, LinkOut LO, int* __restrict__ line_len) {
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, 0};
     extract_links(B, i, L, lane);
-    int len = emit_line<false>(B, i, nullptr, L, s_lits);
+    int len = emit_line<false>(B, i, nullptr, L, s_lits, nullptr);
     if (lane == 0) {
       line_len[i] = len;
       LO.cnt[i] = L.cnt;
@@ -497,6 +520,7 @@ __global__ void __launch_bounds__(256, 4)
 write_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
              const int* __restrict__ line_len, unsigned char* __restrict__ out) {
   __shared__ unsigned char s_lits[TG_POOL_BYTES];
+  __shared__ unsigned char s_stage[4][TG_STAGE_BYTES];
   for (int t = threadIdx.x; t < TG_POOL_BYTES; t += blockDim.x)
     s_lits[t] = (unsigned char)tg_lit_pool.v[t];
   __syncthreads();
@@ -509,7 +533,8 @@ write_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
                LO.name_len + (size_t)i * MAX_LINKS,
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, LO.cnt[i]};
-    emit_line<true>(B, i, out + line_off[i], L, s_lits);
+    emit_line<true>(B, i, out + line_off[i], L, s_lits,
+                    &s_stage[wave][0]);
   }
 }
 
@@ -542,10 +567,10 @@ write_lds_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, LO.cnt[i]};
     if (len > LDS_LINE_BYTES) {
-      emit_line<true>(B, i, out + line_off[i], L, s_lits);
+      emit_line<true>(B, i, out + line_off[i], L, s_lits, nullptr);
       continue;
     }
-    emit_line<true>(B, i, &lbuf[wave][0], L, s_lits);
+    emit_line<true>(B, i, &lbuf[wave][0], L, s_lits, nullptr);
     copy_line(&lbuf[wave][0], out + line_off[i], len, lane);
   }
 }
@@ -559,6 +584,7 @@ __global__ void __launch_bounds__(256)
 write_scratch_kernel(BatchView B, LinkOut LO, unsigned char* __restrict__ scratch,
                      long stride, int* __restrict__ line_len, int* __restrict__ overflow) {
   __shared__ unsigned char s_lits[TG_POOL_BYTES];
+  __shared__ unsigned char s_stage[4][TG_STAGE_BYTES];
   for (int t = threadIdx.x; t < TG_POOL_BYTES; t += blockDim.x)
     s_lits[t] = (unsigned char)tg_lit_pool.v[t];
   __syncthreads();
@@ -575,7 +601,8 @@ write_scratch_kernel(BatchView B, LinkOut LO, unsigned char* __restrict__ scratc
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, 0};
     extract_links(B, i, L, lane);
-    int len = emit_line<true>(B, i, scratch + (size_t)i * stride, L, s_lits);
+    int len = emit_line<true>(B, i, scratch + (size_t)i * stride, L, s_lits,
+                              &s_stage[wave][0]);
     if (lane == 0) {
       line_len[i] = len;
       LO.cnt[i] = L.cnt;
