@@ -19,11 +19,22 @@
 // emitters re-read every line — the write kernel stalls ~one
 // HBM-latency per emitter call without this.
 typedef unsigned int __attribute__((aligned(1))) u32_unal;
+#ifndef CRAWL_NT_STORES
+#define CRAWL_NT_STORES 0
+#endif
 DEV void nt_store_u32(unsigned char* p, unsigned v) {
+#if CRAWL_NT_STORES
   __builtin_nontemporal_store(v, (u32_unal*)p);
+#else
+  __builtin_memcpy(p, &v, 4);
+#endif
 }
 DEV void nt_store_u8(unsigned char* p, unsigned char v) {
+#if CRAWL_NT_STORES
   __builtin_nontemporal_store(v, p);
+#else
+  *p = v;
+#endif
 }
 
 namespace crawl {
