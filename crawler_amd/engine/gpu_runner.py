@@ -258,6 +258,22 @@ class GpuCrawlEngine:
                     out_rows = C.allgather_rows(
                         rows, dist, world, device=self.device)
                     discovered = out_rows.numpy()
+                    # seen-set union: fold every rank's
+                    # claims into the local table + OR-union the bloom
+                    # so later layers dedup identically on all ranks
+                    if discovered.shape[0]:
+                        from . import vecvalidate as _vv
+                        hs = torch.from_numpy(
+                            _vv.fnv1a64_rows(discovered))
+                        self.seen.insert_hashes(hs.to(self.device))
+                    bl = self.seen.bloom
+                    cdev = C.collective_device(dist, self.device)
+                    if bl.device == cdev:
+                        C.bloom_union(bl, dist, world)
+                    else:
+                        host_bl = bl.to(cdev)
+                        C.bloom_union(host_bl, dist, world)
+                        bl.copy_(host_bl)
                     # deadend statuses must agree across ranks: the
                     # next layer's deadend-replacement budget is
                     # computed independently by every rank

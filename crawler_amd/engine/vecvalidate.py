@@ -68,6 +68,26 @@ def validate_names(unames: np.ndarray, universe: int):
     return ok, cid_ok, val
 
 
+def fnv1a64_rows(rows: np.ndarray) -> np.ndarray:
+    """Vectorized FNV-1a 64 over NUL-padded uint8[N, W] name rows —
+    matches the device hash (csrc/common.h fnv1a64) and the scalar
+    oracle (ops/golden.fnv1a64): hash of the unpadded bytes."""
+    if rows.ndim != 2:
+        rows = np.ascontiguousarray(rows).view(np.uint8).reshape(
+            len(rows), rows.dtype.itemsize)
+    n, w = rows.shape
+    # NOTE: the repo-wide basis (csrc/common.h, ops/golden.py) is
+    # 1469598103934665603 — not the standard FNV offset basis
+    h = np.full(n, 1469598103934665603, dtype=np.uint64)
+    prime = np.uint64(1099511628211)
+    lens = (rows != 0).sum(axis=1)
+    for j in range(w):
+        live = j < lens
+        hj = (h ^ rows[:, j].astype(np.uint64)) * prime
+        h = np.where(live, hj, h)
+    return h.view(np.int64)
+
+
 def decode_names(unames: np.ndarray) -> list:
     """One bulk ascii decode of NUL-padded S<w> rows -> python strings."""
     n = len(unames)

@@ -70,3 +70,16 @@ def test_decode_names_roundtrip():
     arr = np.array(names, dtype="S32")
     assert decode_names(arr) == names
     assert decode_names(arr[:0]) == []
+
+
+def test_fnv1a64_rows_matches_scalar_oracle():
+    from crawler_amd.engine.vecvalidate import fnv1a64_rows
+    from crawler_amd.ops.gpu import fnv1a64
+
+    names = ["chan1", "c0000000042", "x", "", "a" * 32]
+    arr = np.array(names, dtype="S32")
+    got = fnv1a64_rows(arr)
+    for i, nm in enumerate(names):
+        want = fnv1a64(nm.encode())
+        # device/oracle hash is uint64; rows variant returns int64 view
+        assert int(got.view(np.uint64)[i]) == want, nm
