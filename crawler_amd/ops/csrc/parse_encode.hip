@@ -287,7 +287,7 @@ struct Emit : JsonEmit<W> {
   }
 };
 
-template <bool W>
+template <bool W, bool STAGED>
 DEV int emit_line(const BatchView& B, int i, unsigned char* out,
                   LinkList& L, const unsigned char* lds_lits,
                   unsigned char* lds_stage) {
@@ -348,21 +348,26 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
   // and every later esc() read is a ds_read (lgkmcnt) that never waits
   // on the line's store stream (the vmcnt FIFO orders loads AFTER
   // stores; see profiles/r02_valu_diet.md).
-  int stage_o = 0;
-  auto staged = [&](const unsigned char* ptr,
-                    int n) -> const unsigned char* {
-    if (!W || lds_stage == nullptr || n <= 0 ||
-        stage_o + n > TG_STAGE_BYTES)
-      return ptr;
-    unsigned char* dst = lds_stage + stage_o;
-    stage_o += n;
-    for (int j = lane; j < n; j += WAVE) dst[j] = ptr[j];
-    return dst;
-  };
-  user = staged(user, user_n);
-  title = staged(title, title_n);
-  poster_p = staged(poster_p, poster_len_v);
-  desc_p = staged(desc_p, desc_n);
+  // STAGED instantiations copy unconditionally (the HOST verified the
+  // batch's max field lengths fit TG_STAGE_BYTES), so every staged
+  // pointer is provably addrspace(3): the escape reads become ds_read
+  // (lgkmcnt) instead of flat_load, which waits BOTH counters — the
+  // mixed-provenance select of an earlier revision generated exactly
+  // those flat loads and re-serialized the emitter.
+  if (STAGED) {
+    int stage_o = 0;
+    auto staged = [&](const unsigned char* ptr,
+                      int n) -> const unsigned char* {
+      unsigned char* dst = lds_stage + stage_o;
+      stage_o += n;
+      for (int j = lane; j < n; j += WAVE) dst[j] = ptr[j];
+      return dst;
+    };
+    user = staged(user, user_n);
+    title = staged(title, title_n);
+    poster_p = staged(poster_p, poster_len_v);
+    desc_p = staged(desc_p, desc_n);
+  }
 
   // post_link / url (tdutils.go:1005-1031; empty for private channels)
   auto post_link = [&]() {
@@ -508,7 +513,7 @@ measure_extract_kernel(BatchView B, LinkOut LO, int* __restrict__ line_len) {
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, 0};
     extract_links(B, i, L, lane);
-    int len = emit_line<false>(B, i, nullptr, L, s_lits, nullptr);
+    int len = emit_line<false, false>(B, i, nullptr, L, s_lits, nullptr);
     if (lane == 0) {
       line_len[i] = len;
       LO.cnt[i] = L.cnt;
@@ -519,6 +524,30 @@ measure_extract_kernel(BatchView B, LinkOut LO, int* __restrict__ line_len) {
 __global__ void __launch_bounds__(256, 4)
 write_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
              const int* __restrict__ line_len, unsigned char* __restrict__ out) {
+  __shared__ unsigned char s_lits[TG_POOL_BYTES];
+  for (int t = threadIdx.x; t < TG_POOL_BYTES; t += blockDim.x)
+    s_lits[t] = (unsigned char)tg_lit_pool.v[t];
+  __syncthreads();
+  const int lane = lane_id();
+  const int wave = wave_id();
+  const int waves_per_grid = gridDim.x * 4;
+  for (int i = blockIdx.x * 4 + wave; i < B.n; i += waves_per_grid) {
+    if (line_len[i] == 0) continue;
+    LinkList L{LO.name + (size_t)i * MAX_LINKS * 32,
+               LO.name_len + (size_t)i * MAX_LINKS,
+               LO.src + (size_t)i * MAX_LINKS,
+               LO.hash + (size_t)i * MAX_LINKS, LO.cnt[i]};
+    emit_line<true, false>(B, i, out + line_off[i], L, s_lits, nullptr);
+  }
+}
+
+// Staged write: the host launches this when the batch's max field
+// lengths fit TG_STAGE_BYTES (flagship corpus always does); escape
+// reads come from LDS so no in-line load waits on the store FIFO.
+__global__ void __launch_bounds__(256, 4)
+write_staged_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
+                    const int* __restrict__ line_len,
+                    unsigned char* __restrict__ out) {
   __shared__ unsigned char s_lits[TG_POOL_BYTES];
   __shared__ unsigned char s_stage[4][TG_STAGE_BYTES];
   for (int t = threadIdx.x; t < TG_POOL_BYTES; t += blockDim.x)
@@ -533,8 +562,8 @@ write_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
                LO.name_len + (size_t)i * MAX_LINKS,
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, LO.cnt[i]};
-    emit_line<true>(B, i, out + line_off[i], L, s_lits,
-                    &s_stage[wave][0]);
+    emit_line<true, true>(B, i, out + line_off[i], L, s_lits,
+                          &s_stage[wave][0]);
   }
 }
 
@@ -543,7 +572,11 @@ write_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
 // exact global offset with the funnel-shift dword copy. Lines larger than
 // the LDS budget (rare) fall back to direct global emission.
 
-#define LDS_LINE_BYTES (12 * 1024)
+// 3 KB covers the typical ~2 KB line at FULL occupancy (4 waves x 3KB
+// + literal pool ~= 14 KB/block, 8 blocks/CU); bigger lines (comment
+// threads) take the direct-global fallback. The round-1 12 KB buffer
+// capped occupancy at 3 blocks/CU and lost 2x.
+#define LDS_LINE_BYTES 3072
 
 DEV void copy_line(const unsigned char* src, unsigned char* dst, int n,
                    int lane);
@@ -567,10 +600,10 @@ write_lds_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, LO.cnt[i]};
     if (len > LDS_LINE_BYTES) {
-      emit_line<true>(B, i, out + line_off[i], L, s_lits, nullptr);
+      emit_line<true, false>(B, i, out + line_off[i], L, s_lits, nullptr);
       continue;
     }
-    emit_line<true>(B, i, &lbuf[wave][0], L, s_lits, nullptr);
+    emit_line<true, false>(B, i, &lbuf[wave][0], L, s_lits, nullptr);
     copy_line(&lbuf[wave][0], out + line_off[i], len, lane);
   }
 }
@@ -584,7 +617,6 @@ __global__ void __launch_bounds__(256)
 write_scratch_kernel(BatchView B, LinkOut LO, unsigned char* __restrict__ scratch,
                      long stride, int* __restrict__ line_len, int* __restrict__ overflow) {
   __shared__ unsigned char s_lits[TG_POOL_BYTES];
-  __shared__ unsigned char s_stage[4][TG_STAGE_BYTES];
   for (int t = threadIdx.x; t < TG_POOL_BYTES; t += blockDim.x)
     s_lits[t] = (unsigned char)tg_lit_pool.v[t];
   __syncthreads();
@@ -601,8 +633,8 @@ write_scratch_kernel(BatchView B, LinkOut LO, unsigned char* __restrict__ scratc
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, 0};
     extract_links(B, i, L, lane);
-    int len = emit_line<true>(B, i, scratch + (size_t)i * stride, L, s_lits,
-                              &s_stage[wave][0]);
+    int len = emit_line<true, false>(B, i, scratch + (size_t)i * stride, L,
+                                     s_lits, nullptr);
     if (lane == 0) {
       line_len[i] = len;
       LO.cnt[i] = L.cnt;
@@ -760,6 +792,22 @@ int crawl_write(void** batch_ptrs, const long* scalars, void** link_ptrs,
                      (const int*)line_len, (unsigned char*)out);
   return (int)hipGetLastError();
 }
+
+int crawl_write_staged(void** batch_ptrs, const long* scalars,
+                       void** link_ptrs, const void* line_off,
+                       const void* line_len, void* out, int grid,
+                       void* stream) {
+  crawl::BatchView B = crawl::make_view(batch_ptrs, scalars);
+  crawl::LinkOut LO = crawl::make_links(link_ptrs);
+  hipLaunchKernelGGL(crawl::write_staged_kernel, dim3(grid), dim3(256), 0,
+                     (hipStream_t)stream, B, LO, (const long*)line_off,
+                     (const int*)line_len, (unsigned char*)out);
+  return (int)hipGetLastError();
+}
+
+// max per-line staged bytes the staged writer supports (host checks the
+// batch's max field lengths against this before choosing it)
+int crawl_stage_budget() { return TG_STAGE_BYTES; }
 
 int crawl_write_scratch(void** batch_ptrs, const long* scalars,
                         void** link_ptrs, void* scratch, long stride,

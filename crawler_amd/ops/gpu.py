@@ -63,6 +63,9 @@ def load_lib():
         ctypes.POINTER(ctypes.c_void_p), ctypes.c_void_p, ctypes.c_void_p,
         ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
     ]
+    lib.crawl_write_staged.restype = ctypes.c_int
+    lib.crawl_write_staged.argtypes = lib.crawl_write.argtypes
+    lib.crawl_stage_budget.restype = ctypes.c_int
     _bind_dedup(lib)
     _bind_feedgen(lib)
     _bind_v2(lib)
@@ -444,6 +447,30 @@ def _stride_bound(batch: B.MessageBatch) -> int:
     return int(bound.max().item())
 
 
+def _stage_fits(batch, lib) -> bool:
+    """True when every line's staged fields (user+title+poster+desc)
+    fit the staged writer's per-wave LDS budget; cached per batch."""
+    if os.environ.get("CRAWL_NO_STAGED") == "1":
+        return False
+    fit = getattr(batch, "_stage_fit", None)
+    if fit is None:
+        m = batch.meta
+        mx = 0
+        if batch.n:
+            mx = max(int(m["text_len"].max().item()),
+                     int(m["aux_len"].max().item()),
+                     0) + int(m["poster_len"].max().item())
+        if batch.n_channels:
+            mx += (int(batch.ch_user_len.max().item())
+                   + int(batch.ch_title_len.max().item()))
+        fit = mx <= int(lib.crawl_stage_budget())
+        try:
+            batch._stage_fit = fit
+        except AttributeError:
+            pass
+    return fit
+
+
 def parse_encode(
     batch: B.MessageBatch,
     now: Optional[_dt.datetime] = None,
@@ -558,9 +585,15 @@ def parse_encode(
         torch.cumsum(line_len.to(torch.int64)[:-1], 0, out=line_off[1:])
         total = int(line_off[-1].item() + line_len[-1].item()) if n else 0
         out = torch.empty(total, dtype=torch.uint8, device=dev)
-        writer = (lib.crawl_write_lds if os.environ.get("CRAWL_LDS_WRITE")
-                  else lib.crawl_write)  # LDS staging measured 2x slower
-                  # (48KB static shared tanks occupancy; see profiles/)
+        if os.environ.get("CRAWL_LDS_WRITE"):
+            writer = lib.crawl_write_lds  # experimental LDS line buffer
+        elif _stage_fits(batch, lib):
+            # staged writer: escape-scanned fields go through LDS so
+            # in-line loads never wait the store FIFO (profiles/
+            # r02_valu_diet.md); host verified the batch fits
+            writer = lib.crawl_write_staged
+        else:
+            writer = lib.crawl_write
         rc = writer(
             batch_ptrs, scalars, link_ptrs,
             ctypes.c_void_p(line_off.data_ptr()),
