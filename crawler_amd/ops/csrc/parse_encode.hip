@@ -287,10 +287,25 @@ struct Emit : JsonEmit<W> {
   }
 };
 
+// Block-LDS copies of the per-batch constant tables (staged kernel):
+// emoji vocabulary, created/capture timestamp strings, content-type
+// names. All tiny; staging them turns their per-line loads (issued
+// AFTER line stores -> full store-FIFO drains) into lgkmcnt ds_reads.
+struct StagedTabs {
+  const unsigned char* emoji_pool;
+  const int* emoji_off;
+  const int* emoji_len;
+  const unsigned char* created;
+  const unsigned char* capture;
+  const unsigned char* ctname_pool;
+  const int* ctname_off;
+  const int* ctname_len;
+};
+
 template <bool W, bool STAGED>
 DEV int emit_line(const BatchView& B, int i, unsigned char* out,
                   LinkList& L, const unsigned char* lds_lits,
-                  unsigned char* lds_stage) {
+                  unsigned char* lds_stage, const StagedTabs* T) {
   const int lane = lane_id();
   Emit<W> e{};
   e.out = out;
@@ -354,6 +369,10 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
   // (lgkmcnt) instead of flat_load, which waits BOTH counters — the
   // mixed-provenance select of an earlier revision generated exactly
   // those flat loads and re-serialized the emitter.
+  const int* re_src = B.react_emoji + react0_v;
+  const int* rcnt_src = B.react_count + react0_v;
+  const unsigned char* lnames_src = L.names;
+  const unsigned char* llens_src = L.lens;
   if (STAGED) {
     int stage_o = 0;
     auto staged = [&](const unsigned char* ptr,
@@ -367,7 +386,27 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
     title = staged(title, title_n);
     poster_p = staged(poster_p, poster_len_v);
     desc_p = staged(desc_p, desc_n);
+    lnames_src = staged(L.names, L.cnt * 32);
+    llens_src = staged(L.lens, L.cnt);
+    // reaction id/count arrays (int32), 4B-aligned carve
+    stage_o = (stage_o + 3) & ~3;
+    int* s_re = (int*)(lds_stage + stage_o);
+    stage_o += reactc_v * 4;
+    int* s_rc = (int*)(lds_stage + stage_o);
+    stage_o += reactc_v * 4;
+    for (int j = lane; j < reactc_v; j += WAVE) {
+      s_re[j] = B.react_emoji[react0_v + j];
+      s_rc[j] = B.react_count[react0_v + j];
+    }
+    re_src = s_re;
+    rcnt_src = s_rc;
   }
+  const unsigned char* emoji_pool_s = STAGED ? T->emoji_pool : B.emoji_pool;
+  const int* emoji_off_s = STAGED ? T->emoji_off : B.emoji_off;
+  const int* emoji_len_s = STAGED ? T->emoji_len : B.emoji_len;
+  const unsigned char* created_s = STAGED ? T->created : B.created_str;
+  const unsigned char* capture_s = STAGED ? T->capture : B.capture_str;
+  const unsigned char* ctpool_s = STAGED ? T->ctname_pool : B.ctname_pool;
 
   // post_link / url (tdutils.go:1005-1031; empty for private channels)
   auto post_link = [&]() {
@@ -393,7 +432,7 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
   TGLIT(e, TGL_6);
   e.rfc3339(date_v);
   TGLIT(e, TGL_7);
-  e.raw(B.created_str, B.created_len);
+  e.raw(created_s, B.created_len);
   TGLIT(e, TGL_8);
   e.i64(views_v);
   TGLIT(e, TGL_9);
@@ -421,7 +460,8 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
   TGLIT(e, TGL_20);
   e.qesc(desc_p, desc_n);
   TGLIT(e, TGL_21);
-  e.raw(B.ctname_pool + ctname_off_v, ctname_len_v);
+  e.raw(ctpool_s + (STAGED ? T->ctname_off[ct] : ctname_off_v),
+        STAGED ? T->ctname_len[ct] : ctname_len_v);
   TGLIT(e, TGL_22);
   e.i64(forwards_v);
   TGLIT(e, TGL_23);
@@ -451,7 +491,7 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
         if (r) LIT(e, ",");
         int em = B.react_emoji[r0 + r];
         LIT(e, "\"");
-        e.raw(B.emoji_pool + B.emoji_off[em], B.emoji_len[em]);
+        e.raw(emoji_pool_s + emoji_off_s[em], emoji_len_s[em]);
         LIT(e, "\":");
         e.i64(B.react_count[r0 + r]);
       }
@@ -466,25 +506,25 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
   }
   TGLIT(e, TGL_32);
   {
-    const int r0 = react0_v, rc = reactc_v;
+    const int rc = reactc_v;
     for (int r = 0; r < rc; ++r) {
       if (r) LIT(e, ",");
-      int em = B.react_emoji[r0 + r];
+      int em = re_src[r];
       LIT(e, "\"");
-      e.raw(B.emoji_pool + B.emoji_off[em], B.emoji_len[em]);
+      e.raw(emoji_pool_s + emoji_off_s[em], emoji_len_s[em]);
       LIT(e, "\":");
-      e.i64(B.react_count[r0 + r]);
+      e.i64(rcnt_src[r]);
     }
   }
   TGLIT(e, TGL_33);
   for (int k = 0; k < L.cnt; ++k) {
     if (k) LIT(e, ",");
     LIT(e, "\"");
-    e.raw(L.names + k * 32, L.lens[k]);
+    e.raw(lnames_src + k * 32, llens_src[k]);
     LIT(e, "\"");
   }
   TGLIT(e, TGL_34);
-  e.raw(B.capture_str, B.capture_len);
+  e.raw(capture_s, B.capture_len);
   TGLIT(e, TGL_35);
   e.qesc(poster_p, poster_len_v);
   LIT(e, "}\n");
@@ -513,7 +553,8 @@ measure_extract_kernel(BatchView B, LinkOut LO, int* __restrict__ line_len) {
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, 0};
     extract_links(B, i, L, lane);
-    int len = emit_line<false, false>(B, i, nullptr, L, s_lits, nullptr);
+    int len = emit_line<false, false>(B, i, nullptr, L, s_lits,
+                                     nullptr, nullptr);
     if (lane == 0) {
       line_len[i] = len;
       LO.cnt[i] = L.cnt;
@@ -537,22 +578,57 @@ write_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
                LO.name_len + (size_t)i * MAX_LINKS,
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, LO.cnt[i]};
-    emit_line<true, false>(B, i, out + line_off[i], L, s_lits, nullptr);
+    emit_line<true, false>(B, i, out + line_off[i], L, s_lits, nullptr,
+                           nullptr);
   }
 }
 
 // Staged write: the host launches this when the batch's max field
 // lengths fit TG_STAGE_BYTES (flagship corpus always does); escape
 // reads come from LDS so no in-line load waits on the store FIFO.
+#define TG_EMOJI_POOL_CAP 512
+#define TG_EMOJI_CAP 64
+#define TG_TS_CAP 64
+#define TG_CTNAME_POOL_CAP 512
+#define TG_CTNAME_CAP 32
+
 __global__ void __launch_bounds__(256, 4)
 write_staged_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
                     const int* __restrict__ line_len,
-                    unsigned char* __restrict__ out) {
+                    unsigned char* __restrict__ out,
+                    int n_emoji, int emoji_pool_bytes,
+                    int n_ctnames, int ctname_pool_bytes) {
   __shared__ unsigned char s_lits[TG_POOL_BYTES];
   __shared__ unsigned char s_stage[4][TG_STAGE_BYTES];
+  __shared__ unsigned char s_emoji_pool[TG_EMOJI_POOL_CAP];
+  __shared__ int s_emoji_off[TG_EMOJI_CAP];
+  __shared__ int s_emoji_len[TG_EMOJI_CAP];
+  __shared__ unsigned char s_created[TG_TS_CAP];
+  __shared__ unsigned char s_capture[TG_TS_CAP];
+  __shared__ unsigned char s_ctpool[TG_CTNAME_POOL_CAP];
+  __shared__ int s_ctoff[TG_CTNAME_CAP];
+  __shared__ int s_ctlen[TG_CTNAME_CAP];
   for (int t = threadIdx.x; t < TG_POOL_BYTES; t += blockDim.x)
     s_lits[t] = (unsigned char)tg_lit_pool.v[t];
+  for (int t = threadIdx.x; t < emoji_pool_bytes; t += blockDim.x)
+    s_emoji_pool[t] = B.emoji_pool[t];
+  for (int t = threadIdx.x; t < n_emoji; t += blockDim.x) {
+    s_emoji_off[t] = B.emoji_off[t];
+    s_emoji_len[t] = B.emoji_len[t];
+  }
+  for (int t = threadIdx.x; t < B.created_len; t += blockDim.x)
+    s_created[t] = B.created_str[t];
+  for (int t = threadIdx.x; t < B.capture_len; t += blockDim.x)
+    s_capture[t] = B.capture_str[t];
+  for (int t = threadIdx.x; t < ctname_pool_bytes; t += blockDim.x)
+    s_ctpool[t] = B.ctname_pool[t];
+  for (int t = threadIdx.x; t < n_ctnames; t += blockDim.x) {
+    s_ctoff[t] = B.ctname_off[t];
+    s_ctlen[t] = B.ctname_len[t];
+  }
   __syncthreads();
+  StagedTabs T{s_emoji_pool, s_emoji_off, s_emoji_len, s_created,
+               s_capture, s_ctpool, s_ctoff, s_ctlen};
   const int lane = lane_id();
   const int wave = wave_id();
   const int waves_per_grid = gridDim.x * 4;
@@ -563,7 +639,7 @@ write_staged_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, LO.cnt[i]};
     emit_line<true, true>(B, i, out + line_off[i], L, s_lits,
-                          &s_stage[wave][0]);
+                          &s_stage[wave][0], &T);
   }
 }
 
@@ -600,10 +676,12 @@ write_lds_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, LO.cnt[i]};
     if (len > LDS_LINE_BYTES) {
-      emit_line<true, false>(B, i, out + line_off[i], L, s_lits, nullptr);
+      emit_line<true, false>(B, i, out + line_off[i], L, s_lits,
+                             nullptr, nullptr);
       continue;
     }
-    emit_line<true, false>(B, i, &lbuf[wave][0], L, s_lits, nullptr);
+    emit_line<true, false>(B, i, &lbuf[wave][0], L, s_lits, nullptr,
+                           nullptr);
     copy_line(&lbuf[wave][0], out + line_off[i], len, lane);
   }
 }
@@ -634,7 +712,7 @@ write_scratch_kernel(BatchView B, LinkOut LO, unsigned char* __restrict__ scratc
                LO.hash + (size_t)i * MAX_LINKS, 0};
     extract_links(B, i, L, lane);
     int len = emit_line<true, false>(B, i, scratch + (size_t)i * stride, L,
-                                     s_lits, nullptr);
+                                     s_lits, nullptr, nullptr);
     if (lane == 0) {
       line_len[i] = len;
       LO.cnt[i] = L.cnt;
@@ -795,13 +873,17 @@ int crawl_write(void** batch_ptrs, const long* scalars, void** link_ptrs,
 
 int crawl_write_staged(void** batch_ptrs, const long* scalars,
                        void** link_ptrs, const void* line_off,
-                       const void* line_len, void* out, int grid,
-                       void* stream) {
+                       const void* line_len, void* out,
+                       int n_emoji, int emoji_pool_bytes,
+                       int n_ctnames, int ctname_pool_bytes,
+                       int grid, void* stream) {
   crawl::BatchView B = crawl::make_view(batch_ptrs, scalars);
   crawl::LinkOut LO = crawl::make_links(link_ptrs);
   hipLaunchKernelGGL(crawl::write_staged_kernel, dim3(grid), dim3(256), 0,
                      (hipStream_t)stream, B, LO, (const long*)line_off,
-                     (const int*)line_len, (unsigned char*)out);
+                     (const int*)line_len, (unsigned char*)out,
+                     n_emoji, emoji_pool_bytes, n_ctnames,
+                     ctname_pool_bytes);
   return (int)hipGetLastError();
 }
 

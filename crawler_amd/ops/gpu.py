@@ -64,7 +64,12 @@ def load_lib():
         ctypes.c_void_p, ctypes.c_int, ctypes.c_void_p,
     ]
     lib.crawl_write_staged.restype = ctypes.c_int
-    lib.crawl_write_staged.argtypes = lib.crawl_write.argtypes
+    lib.crawl_write_staged.argtypes = [
+        ctypes.POINTER(ctypes.c_void_p), ctypes.POINTER(ctypes.c_long),
+        ctypes.POINTER(ctypes.c_void_p), ctypes.c_void_p, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+        ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
+    ]
     lib.crawl_stage_budget.restype = ctypes.c_int
     _bind_dedup(lib)
     _bind_feedgen(lib)
@@ -447,19 +452,34 @@ def _stride_bound(batch: B.MessageBatch) -> int:
     return int(bound.max().item())
 
 
-def _stage_fits(batch, lib) -> bool:
-    """True when every line's staged fields (user+title+poster+desc)
-    fit the staged writer's per-wave LDS budget; cached per batch."""
+# per-line staged extras beyond the text fields: outlink names
+# (8 x 32 + 8) + reaction id/count ints (2 x 64 x 4) + alignment slack
+_STAGE_FIXED = 812
+# staged kernel's block-table caps (parse_encode.hip TG_*_CAP)
+_EMOJI_POOL_CAP, _EMOJI_CAP, _TS_CAP = 512, 64, 64
+_CTNAME_POOL_CAP, _CTNAME_CAP = 512, 32
+
+
+def _stage_fits(batch, lib, tables, created, capture) -> bool:
+    """True when every line's staged fields (user+title+poster+desc +
+    links/reactions) fit the staged writer's per-wave LDS budget AND
+    the per-batch tables fit its block caps; cached per batch."""
     if os.environ.get("CRAWL_NO_STAGED") == "1":
+        return False
+    if (tables["emoji_pool"].numel() > _EMOJI_POOL_CAP
+            or tables["emoji_off"].numel() > _EMOJI_CAP
+            or tables["ctname_pool"].numel() > _CTNAME_POOL_CAP
+            or tables["ctname_off"].numel() > _CTNAME_CAP
+            or len(created) > _TS_CAP or len(capture) > _TS_CAP):
         return False
     fit = getattr(batch, "_stage_fit", None)
     if fit is None:
         m = batch.meta
-        mx = 0
+        mx = _STAGE_FIXED
         if batch.n:
-            mx = max(int(m["text_len"].max().item()),
-                     int(m["aux_len"].max().item()),
-                     0) + int(m["poster_len"].max().item())
+            mx += (max(int(m["text_len"].max().item()),
+                       int(m["aux_len"].max().item()))
+                   + int(m["poster_len"].max().item()))
         if batch.n_channels:
             mx += (int(batch.ch_user_len.max().item())
                    + int(batch.ch_title_len.max().item()))
@@ -585,21 +605,32 @@ def parse_encode(
         torch.cumsum(line_len.to(torch.int64)[:-1], 0, out=line_off[1:])
         total = int(line_off[-1].item() + line_len[-1].item()) if n else 0
         out = torch.empty(total, dtype=torch.uint8, device=dev)
-        if os.environ.get("CRAWL_LDS_WRITE"):
-            writer = lib.crawl_write_lds  # experimental LDS line buffer
-        elif _stage_fits(batch, lib):
-            # staged writer: escape-scanned fields go through LDS so
-            # in-line loads never wait the store FIFO (profiles/
-            # r02_valu_diet.md); host verified the batch fits
-            writer = lib.crawl_write_staged
+        if (not os.environ.get("CRAWL_LDS_WRITE")
+                and _stage_fits(batch, lib, tables, created, capture)):
+            # staged writer: escape-scanned fields + reactions +
+            # outlinks + per-batch tables go through LDS so in-line
+            # loads never wait the store FIFO (profiles/
+            # r02_valu_diet.md); host verified everything fits
+            rc = lib.crawl_write_staged(
+                batch_ptrs, scalars, link_ptrs,
+                ctypes.c_void_p(line_off.data_ptr()),
+                ctypes.c_void_p(line_len.data_ptr()),
+                ctypes.c_void_p(out.data_ptr()),
+                tables["emoji_off"].numel(),
+                tables["emoji_pool"].numel(),
+                tables["ctname_off"].numel(),
+                tables["ctname_pool"].numel(), grid, stream_ptr,
+            )
         else:
-            writer = lib.crawl_write
-        rc = writer(
-            batch_ptrs, scalars, link_ptrs,
-            ctypes.c_void_p(line_off.data_ptr()),
-            ctypes.c_void_p(line_len.data_ptr()),
-            ctypes.c_void_p(out.data_ptr()), grid, stream_ptr,
-        )
+            writer = (lib.crawl_write_lds
+                      if os.environ.get("CRAWL_LDS_WRITE")
+                      else lib.crawl_write)
+            rc = writer(
+                batch_ptrs, scalars, link_ptrs,
+                ctypes.c_void_p(line_off.data_ptr()),
+                ctypes.c_void_p(line_len.data_ptr()),
+                ctypes.c_void_p(out.data_ptr()), grid, stream_ptr,
+            )
         if rc != 0:
             raise RuntimeError(f"crawl_write launch failed: hip error {rc}")
 
