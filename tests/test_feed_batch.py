@@ -245,3 +245,29 @@ def test_get_message_comments_respects_max_and_count():
     # unlimited (-1) returns everything
     got3 = client.get_message_comments(m.chat_id, m.msg_id, -1)
     assert len(got3) == len(coms)
+
+
+def test_thousand_comment_threads_paginate():
+    """BASELINE config #4's --max-comments 1000 shape: threads larger
+    than 1000 comments walk 10+ pages and cap exactly at 1000."""
+    from crawler_amd.feed.client import SyntheticTelegramClient
+    from crawler_amd.feed.synth import FeedConfig as FC
+    from crawler_amd.feed.synth import SyntheticFeed as SF
+
+    feed = SF(FC(seed=9, universe=500, comment_rate=1.0,
+                 max_comments_per_post=1200))
+    client = SyntheticTelegramClient(feed, "conn0", posts_per_channel=8)
+    m, coms = _first_msg_with_comments_from(client, 1001)
+    got = client.get_message_comments(m.chat_id, m.msg_id, 1000)
+    assert len(got) == 1000
+    assert [c.to_json() for c in got] == [c.to_json()
+                                          for c in coms[:1000]]
+
+
+def _first_msg_with_comments_from(client, min_comments):
+    for cid in range(40):
+        for m in client.get_chat_history(-1001000000000 - cid):
+            coms = getattr(m, "_comments", [])
+            if len(coms) >= min_comments:
+                return m, coms
+    raise AssertionError("no sufficiently comment-heavy message")
