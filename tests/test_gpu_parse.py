@@ -247,3 +247,59 @@ def test_comment_heavy_device_feedgen_matches_host(gpu_mod):
     res = gpu_mod.parse_encode(dev, now=NOW)
     torch.cuda.synchronize()
     assert bytes(res.out.cpu().numpy()) == b"".join(gl_host)
+
+
+def _writer_pick_corpus():
+    """A batch with one >8KB text (over the staged writer's LDS budget)
+    plus normal posts — forces the plain-writer fallback."""
+    big = ("Huge " + "x" * 8000 + ' with "escapes" <&> and t.me/bigchan1 '
+           + "я" * 120)
+    msgs = []
+    for k, t in enumerate([big, "small t.me/smallchan1", "plain"]):
+        msgs.append(G.SynthMessage(
+            chat_id=-100777, msg_id=(k + 1) << 20,
+            date=1_700_000_100 + k, content_type="messageText",
+            text=G.FormattedText(text=t), views=k, forwards=k,
+            reactions={"🔥": k + 2}, poster_handle=f"user{k:04d}",
+        ))
+    ch = [B.ChannelRow(chat_id=-100777, username="bigtextchan",
+                       title="Big", member_count=9, post_count=len(msgs),
+                       total_views=50)]
+    return B.pack(msgs, ch, [0] * len(msgs))
+
+
+def test_oversize_text_uses_plain_writer_and_matches(gpu_mod):
+    """The host gate must route oversize batches to the plain writer
+    (staged LDS budget exceeded) with identical bytes."""
+    from crawler_amd.ops.gpu import _stage_fits, require_lib
+
+    batch = _writer_pick_corpus()
+    dev_batch = batch.to("cuda:0")
+    lib = require_lib()
+    # verify the gate actually rejects this batch
+    from crawler_amd.ops.gpu import const_tables
+    tables = const_tables(dev_batch.device)
+    assert not _stage_fits(dev_batch, lib, tables, b"x" * 30, b"y" * 35)
+    golden_lines, _ = encode_batch(batch, now=NOW)
+    res = gpu_mod.parse_encode(dev_batch, now=NOW)
+    torch.cuda.synchronize()
+    assert bytes(res.out.cpu().numpy()) == b"".join(golden_lines)
+
+
+def test_plain_writer_forced_matches_staged(gpu_mod, feed,
+                                            monkeypatch):
+    """CRAWL_NO_STAGED=1 (plain writer) and the default staged writer
+    must produce identical bytes on a staged-eligible corpus."""
+    import os
+
+    batch = feed.build_batch(np.arange(4), posts_per_channel=64)
+    dev1 = batch.to("cuda:0")
+    res_staged = gpu_mod.parse_encode(dev1, now=NOW)
+    torch.cuda.synchronize()
+    monkeypatch.setenv("CRAWL_NO_STAGED", "1")
+    dev2 = batch.to("cuda:0")
+    res_plain = gpu_mod.parse_encode(dev2, now=NOW)
+    torch.cuda.synchronize()
+    monkeypatch.delenv("CRAWL_NO_STAGED")
+    assert bytes(res_staged.out.cpu().numpy()) == bytes(
+        res_plain.out.cpu().numpy())
