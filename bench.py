@@ -165,7 +165,9 @@ def main():
     n_copy = 2 if os.environ.get("CRAWL_ONE_SDMA", "") != "1" else 1
     inflight = [None, None]  # keep res tensors alive while copying
     sink_f = None
-    if args.sink == "file" and rank == 0:
+    if args.sink == "file":
+        # rank-sharded sink files: every rank writes its own shard so
+        # an 8-rank file-sink run measures true aggregate disk pressure
         os.makedirs(args.sink_dir, exist_ok=True)
         sink_f = open(os.path.join(args.sink_dir, f"rank{rank}.jsonl"), "wb")
 
