@@ -278,32 +278,8 @@ DEV void extract_links(const BatchView& B, int i, LinkList& L, int lane) {
 // JsonEmit<W> (common.h) provides raw/esc/u64/i64/rfc3339 and the LIT
 // macro; Emit adds the quoted-escape shorthand used below.
 
-// MODE: 0 = plain; 1 = RECORD (measure writes each variable
-// fragment's end cursor to frag_row); 2 = USE (writer re-syncs its
-// cursor from the recorded table staged in LDS). The SAME emit body
-// instantiates all modes, so recorder and user can never drift. In
-// USE mode every mark() breaks the data->address dependency chain:
-// downstream store addresses come from the LDS table instead of the
-// accumulated (load-dependent) cursor, so the CDNA vmcnt FIFO stops
-// serializing fragments (profiles/r02_valu_diet.md).
-#define FRAG_SLOTS 32
-
-template <bool W, int MODE>
+template <bool W>
 struct Emit : JsonEmit<W> {
-  int* frag_row;        // MODE 1: global [FRAG_SLOTS] row for line i
-  const int* frag_lds;  // MODE 2: LDS copy of that row
-  int mj = 0;
-
-  DEV void mark() {
-    if (MODE == 1) {
-      if (lane_id() == 0) frag_row[mj] = this->cur;
-      ++mj;
-    } else if (MODE == 2) {
-      this->cur = frag_lds[mj];
-      ++mj;
-    }
-  }
-
   DEV void qesc(const unsigned char* s, int n) {
     LIT(*this, "\"");
     this->esc(s, n);
@@ -326,18 +302,15 @@ struct StagedTabs {
   const int* ctname_len;
 };
 
-template <bool W, bool STAGED, int MODE>
+template <bool W, bool STAGED>
 DEV int emit_line(const BatchView& B, int i, unsigned char* out,
                   LinkList& L, const unsigned char* lds_lits,
-                  unsigned char* lds_stage, const StagedTabs* T,
-                  int* frag_row) {
+                  unsigned char* lds_stage, const StagedTabs* T) {
   const int lane = lane_id();
-  Emit<W, MODE> e{};
+  Emit<W> e{};
   e.out = out;
   e.cur = 0;
   e.lds = lds_lits;
-  e.frag_row = frag_row;
-  e.frag_lds = nullptr;
   // Load EVERY per-line scalar BEFORE the first store: vmcnt is a
   // single FIFO over loads AND stores on CDNA, so a load issued after
   // stores waits for those stores to retire (the write kernel spent
@@ -402,15 +375,6 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
   const unsigned char* llens_src = L.lens;
   if (STAGED) {
     int stage_o = 0;
-    if (MODE == 2) {
-      // the recorded cursor table, staged first (reads at mark() are
-      // then lgkmcnt ds_reads: no store-FIFO waits)
-      int* ft = (int*)lds_stage;
-      for (int j = lane; j < FRAG_SLOTS; j += WAVE)
-        ft[j] = frag_row[j];
-      e.frag_lds = ft;
-      stage_o = FRAG_SLOTS * 4;
-    }
     auto staged = [&](const unsigned char* ptr,
                       int n) -> const unsigned char* {
       unsigned char* dst = lds_stage + stage_o;
@@ -457,76 +421,53 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
 
   TGLIT(e, TGL_2);
   post_link();
-  e.mark();
   TGLIT(e, TGL_3);
   e.i64(chat_id_v);
-  e.mark();
   TGLIT(e, TGL_4);
   e.i64(pub_id);
   LIT(e, "-");
   e.esc(user, user_n);
-  e.mark();
   TGLIT(e, TGL_5);
   post_link();
-  e.mark();
   TGLIT(e, TGL_6);
   e.rfc3339(date_v);
-  e.mark();
   TGLIT(e, TGL_7);
   e.raw(created_s, B.created_len);
-  e.mark();
   TGLIT(e, TGL_8);
   e.i64(views_v);
-  e.mark();
   TGLIT(e, TGL_9);
   e.i64(views_v);
-  e.mark();
   TGLIT(e, TGL_10);
   e.i64(forwards_v);
-  e.mark();
   TGLIT(e, TGL_11);
   e.i64(ncom);
-  e.mark();
   TGLIT(e, TGL_12);
   e.qesc(title, title_n);
-  e.mark();
   TGLIT(e, TGL_13);
   e.i64(chat_id_v);
-  e.mark();
   TGLIT(e, TGL_14);
   e.qesc(title, title_n);
-  e.mark();
   TGLIT(e, TGL_15);
   e.i64(ch_member_v);
-  e.mark();
   TGLIT(e, TGL_16);
   e.i64(ch_postcount_v);
-  e.mark();
   TGLIT(e, TGL_17);
   e.i64(ch_totalviews_v);
-  e.mark();
   TGLIT(e, TGL_18);
   e.esc(user, user_n);
-  e.mark();
   TGLIT(e, TGL_19);
   e.esc(user, user_n);
-  e.mark();
   TGLIT(e, TGL_20);
   e.qesc(desc_p, desc_n);
-  e.mark();
   TGLIT(e, TGL_21);
   e.raw(ctpool_s + (STAGED ? T->ctname_off[ct] : ctname_off_v),
         STAGED ? T->ctname_len[ct] : ctname_len_v);
-  e.mark();
   TGLIT(e, TGL_22);
   e.i64(forwards_v);
-  e.mark();
   TGLIT(e, TGL_23);
   e.i64(ncom);
-  e.mark();
   TGLIT(e, TGL_17);
   e.i64(views_v);
-  e.mark();
   TGLIT(e, TGL_24);
   // media (fetchAndUploadMedia skip rules, tdutils.go:233-239): GPU path
   // always runs skip_media (media-on is staged host-side).
@@ -536,7 +477,6 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
     e.i64(pub_id);
     LIT(e, "v");
   }
-  e.mark();
   TGLIT(e, TGL_26);
   {
     const int c0 = com0_v;
@@ -594,8 +534,7 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
 // ---------- kernels ----------
 
 __global__ void __launch_bounds__(256, 6)
-measure_extract_kernel(BatchView B, LinkOut LO, int* __restrict__ line_len,
-                       int* __restrict__ frag_tab) {
+measure_extract_kernel(BatchView B, LinkOut LO, int* __restrict__ line_len) {
   __shared__ unsigned char s_lits[TG_POOL_BYTES];
   for (int t = threadIdx.x; t < TG_POOL_BYTES; t += blockDim.x)
     s_lits[t] = (unsigned char)tg_lit_pool.v[t];
@@ -614,9 +553,8 @@ measure_extract_kernel(BatchView B, LinkOut LO, int* __restrict__ line_len,
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, 0};
     extract_links(B, i, L, lane);
-    int len = emit_line<false, false, 1>(B, i, nullptr, L, s_lits, nullptr,
-                                         nullptr,
-                                         frag_tab + (size_t)i * FRAG_SLOTS);
+    int len = emit_line<false, false>(B, i, nullptr, L, s_lits,
+                                     nullptr, nullptr);
     if (lane == 0) {
       line_len[i] = len;
       LO.cnt[i] = L.cnt;
@@ -640,8 +578,8 @@ write_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
                LO.name_len + (size_t)i * MAX_LINKS,
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, LO.cnt[i]};
-    emit_line<true, false, 0>(B, i, out + line_off[i], L, s_lits, nullptr,
-                              nullptr, nullptr);
+    emit_line<true, false>(B, i, out + line_off[i], L, s_lits, nullptr,
+                           nullptr);
   }
 }
 
@@ -658,7 +596,6 @@ __global__ void __launch_bounds__(256, 4)
 write_staged_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
                     const int* __restrict__ line_len,
                     unsigned char* __restrict__ out,
-                    int* __restrict__ frag_tab,
                     int n_emoji, int emoji_pool_bytes,
                     int n_ctnames, int ctname_pool_bytes) {
   __shared__ unsigned char s_lits[TG_POOL_BYTES];
@@ -701,9 +638,8 @@ write_staged_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
                LO.name_len + (size_t)i * MAX_LINKS,
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, LO.cnt[i]};
-    emit_line<true, true, 2>(B, i, out + line_off[i], L, s_lits,
-                             &s_stage[wave][0], &T,
-                             frag_tab + (size_t)i * FRAG_SLOTS);
+    emit_line<true, true>(B, i, out + line_off[i], L, s_lits,
+                          &s_stage[wave][0], &T);
   }
 }
 
@@ -740,12 +676,12 @@ write_lds_kernel(BatchView B, LinkOut LO, const long* __restrict__ line_off,
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, LO.cnt[i]};
     if (len > LDS_LINE_BYTES) {
-      emit_line<true, false, 0>(B, i, out + line_off[i], L, s_lits,
-                                nullptr, nullptr, nullptr);
+      emit_line<true, false>(B, i, out + line_off[i], L, s_lits,
+                             nullptr, nullptr);
       continue;
     }
-    emit_line<true, false, 0>(B, i, &lbuf[wave][0], L, s_lits, nullptr,
-                              nullptr, nullptr);
+    emit_line<true, false>(B, i, &lbuf[wave][0], L, s_lits, nullptr,
+                           nullptr);
     copy_line(&lbuf[wave][0], out + line_off[i], len, lane);
   }
 }
@@ -775,8 +711,8 @@ write_scratch_kernel(BatchView B, LinkOut LO, unsigned char* __restrict__ scratc
                LO.src + (size_t)i * MAX_LINKS,
                LO.hash + (size_t)i * MAX_LINKS, 0};
     extract_links(B, i, L, lane);
-    int len = emit_line<true, false, 0>(B, i, scratch + (size_t)i * stride, L,
-                                        s_lits, nullptr, nullptr, nullptr);
+    int len = emit_line<true, false>(B, i, scratch + (size_t)i * stride, L,
+                                     s_lits, nullptr, nullptr);
     if (lane == 0) {
       line_len[i] = len;
       LO.cnt[i] = L.cnt;
@@ -915,17 +851,14 @@ extern "C" {
 int crawl_batch_ptr_count() { return 49; }
 
 int crawl_measure_extract(void** batch_ptrs, const long* scalars,
-                          void** link_ptrs, void* line_len, void* frag_tab,
-                          int grid, void* stream) {
+                          void** link_ptrs, void* line_len, int grid,
+                          void* stream) {
   crawl::BatchView B = crawl::make_view(batch_ptrs, scalars);
   crawl::LinkOut LO = crawl::make_links(link_ptrs);
   hipLaunchKernelGGL(crawl::measure_extract_kernel, dim3(grid), dim3(256), 0,
-                     (hipStream_t)stream, B, LO, (int*)line_len,
-                     (int*)frag_tab);
+                     (hipStream_t)stream, B, LO, (int*)line_len);
   return (int)hipGetLastError();
 }
-
-int crawl_frag_slots() { return FRAG_SLOTS; }
 
 int crawl_write(void** batch_ptrs, const long* scalars, void** link_ptrs,
                 const void* line_off, const void* line_len, void* out,
@@ -940,7 +873,7 @@ int crawl_write(void** batch_ptrs, const long* scalars, void** link_ptrs,
 
 int crawl_write_staged(void** batch_ptrs, const long* scalars,
                        void** link_ptrs, const void* line_off,
-                       const void* line_len, void* out, void* frag_tab,
+                       const void* line_len, void* out,
                        int n_emoji, int emoji_pool_bytes,
                        int n_ctnames, int ctname_pool_bytes,
                        int grid, void* stream) {
@@ -949,7 +882,7 @@ int crawl_write_staged(void** batch_ptrs, const long* scalars,
   hipLaunchKernelGGL(crawl::write_staged_kernel, dim3(grid), dim3(256), 0,
                      (hipStream_t)stream, B, LO, (const long*)line_off,
                      (const int*)line_len, (unsigned char*)out,
-                     (int*)frag_tab, n_emoji, emoji_pool_bytes, n_ctnames,
+                     n_emoji, emoji_pool_bytes, n_ctnames,
                      ctname_pool_bytes);
   return (int)hipGetLastError();
 }

@@ -54,10 +54,9 @@ def load_lib():
     lib.crawl_measure_extract.restype = ctypes.c_int
     lib.crawl_measure_extract.argtypes = [
         ctypes.POINTER(ctypes.c_void_p), ctypes.POINTER(ctypes.c_long),
-        ctypes.POINTER(ctypes.c_void_p), ctypes.c_void_p, ctypes.c_void_p,
-        ctypes.c_int, ctypes.c_void_p,
+        ctypes.POINTER(ctypes.c_void_p), ctypes.c_void_p, ctypes.c_int,
+        ctypes.c_void_p,
     ]
-    lib.crawl_frag_slots.restype = ctypes.c_int
     lib.crawl_write.restype = ctypes.c_int
     lib.crawl_write.argtypes = [
         ctypes.POINTER(ctypes.c_void_p), ctypes.POINTER(ctypes.c_long),
@@ -68,8 +67,8 @@ def load_lib():
     lib.crawl_write_staged.argtypes = [
         ctypes.POINTER(ctypes.c_void_p), ctypes.POINTER(ctypes.c_long),
         ctypes.POINTER(ctypes.c_void_p), ctypes.c_void_p, ctypes.c_void_p,
-        ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int, ctypes.c_int,
-        ctypes.c_int, ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
+        ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_int,
+        ctypes.c_int, ctypes.c_int, ctypes.c_void_p,
     ]
     lib.crawl_stage_budget.restype = ctypes.c_int
     _bind_dedup(lib)
@@ -594,12 +593,9 @@ def parse_encode(
         if rc != 0:
             raise RuntimeError(f"crawl_compact failed: hip error {rc}")
     else:
-        frag_tab = torch.zeros(n * lib.crawl_frag_slots(),
-                               dtype=torch.int32, device=dev)
         rc = lib.crawl_measure_extract(
             batch_ptrs, scalars, link_ptrs,
-            ctypes.c_void_p(line_len.data_ptr()),
-            ctypes.c_void_p(frag_tab.data_ptr()), grid, stream_ptr,
+            ctypes.c_void_p(line_len.data_ptr()), grid, stream_ptr,
         )
         if rc != 0:
             raise RuntimeError(
@@ -620,7 +616,6 @@ def parse_encode(
                 ctypes.c_void_p(line_off.data_ptr()),
                 ctypes.c_void_p(line_len.data_ptr()),
                 ctypes.c_void_p(out.data_ptr()),
-                ctypes.c_void_p(frag_tab.data_ptr()),
                 tables["emoji_off"].numel(),
                 tables["emoji_pool"].numel(),
                 tables["ctname_off"].numel(),
