@@ -132,6 +132,7 @@ class FanoutCore {
   }
 
   size_t bytes_written() const { return bytes_.load(); }
+  int last_errno() const { return last_errno_.load(); }
   size_t open_files() {
     std::unique_lock<std::mutex> lk(fd_mu_);
     return fds_.size();
@@ -157,7 +158,10 @@ class FanoutCore {
       make_parent_dirs(path);
       fd = ::open(path.c_str(), O_WRONLY | O_CREAT | O_APPEND, 0644);
     }
-    if (fd < 0) return -1;
+    if (fd < 0) {
+      last_errno_.store(errno);
+      return -1;
+    }
     std::unique_lock<std::mutex> lk(fd_mu_);
     auto it = fds_.find(path);
     if (it != fds_.end()) {  // raced: another worker opened it first
@@ -220,6 +224,7 @@ class FanoutCore {
         ssize_t w = ::write(fd, p, left);
         if (w < 0) {
           if (errno == EINTR) continue;
+          last_errno_.store(errno);
           ok = false;
           break;
         }
@@ -253,6 +258,7 @@ class FanoutCore {
   std::unordered_map<uint64_t, size_t> ticket_errors_;    // under mu_
   std::atomic<size_t> errors_;
   std::atomic<size_t> bytes_;
+  std::atomic<int> last_errno_{0};
   struct FdEntry {
     int fd;
     int refs;  // workers mid-write pin the entry against eviction

@@ -42,7 +42,8 @@ class FanoutSink {
     }
     if (errs != 0)
       throw std::runtime_error(
-          "fanout sink: write(2) failures in batch: " +
+          "fanout sink: write(2) failures in batch (errno=" +
+          std::to_string(core_.last_errno()) + "): " +
           std::to_string(errs));
   }
 
@@ -70,7 +71,9 @@ class FanoutSink {
       errs = core_.wait_ticket(ticket);
     }
     if (errs != 0)
-      throw std::runtime_error("fanout sink: write(2) failures: " +
+      throw std::runtime_error(
+          "fanout sink: write(2) failures (errno=" +
+          std::to_string(core_.last_errno()) + "): " +
                                std::to_string(errs));
   }
 
@@ -81,7 +84,9 @@ class FanoutSink {
       errs = core_.drain();
     }
     if (errs != 0)
-      throw std::runtime_error("fanout sink: write(2) failures: " +
+      throw std::runtime_error(
+          "fanout sink: write(2) failures (errno=" +
+          std::to_string(core_.last_errno()) + "): " +
                                std::to_string(errs));
   }
 
