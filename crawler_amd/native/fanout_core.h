@@ -84,11 +84,19 @@ class FanoutCore {
 
   // Returns the error count for the ticket (0 = clean).
   size_t wait_ticket(uint64_t ticket) {
+    {
+      std::unique_lock<std::mutex> lk(mu_);
+      done_cv_.wait(lk, [this, ticket] {
+        auto it = ticket_pending_.find(ticket);
+        return it == ticket_pending_.end() || it->second == 0;
+      });
+    }
+    // Evict surplus fds HERE too: a long ticketed pipeline (the GPU
+    // random-walk touches ~1k new channel files per hop and only
+    // drains at crawl end) otherwise accumulates open fds until the
+    // process hits EMFILE. Pinned fds (mid-write) are skipped.
+    evict_to_cap();
     std::unique_lock<std::mutex> lk(mu_);
-    done_cv_.wait(lk, [this, ticket] {
-      auto it = ticket_pending_.find(ticket);
-      return it == ticket_pending_.end() || it->second == 0;
-    });
     size_t errs = ticket_errors_[ticket];
     ticket_pending_.erase(ticket);
     ticket_errors_.erase(ticket);

@@ -231,3 +231,27 @@ def test_tsan_stress_harness(tmp_path):
     assert run.returncode == 0, run.stdout + run.stderr
     assert "tsan-stress OK" in run.stdout
     assert "ThreadSanitizer" not in run.stderr
+
+
+def test_ticketed_pipeline_evicts_fds(tmp_path):
+    """A long ticketed pipeline must not accumulate open fds past the
+    cap (EMFILE regression: the GPU random-walk opens ~1k new channel
+    files per hop and only drains at crawl end)."""
+    from crawler_amd.native import load
+
+    sink = load().FanoutSink(4, 32)  # cap at 32 open files
+    data = b"x" * 64
+    last = None
+    for hop in range(8):
+        paths = [str(tmp_path / f"h{hop}" / f"c{k}" / "posts.jsonl")
+                 for k in range(64)]
+        t = sink.write_batch_ticket(paths, data,
+                                    [0] * len(paths),
+                                    [len(data)] * len(paths))
+        if last is not None:
+            sink.wait_ticket(last)
+            assert sink.open_files <= 32 + 64  # cap + one in-flight batch
+        last = t
+    sink.wait_ticket(last)
+    sink.drain()
+    assert sink.open_files <= 32
