@@ -232,7 +232,18 @@ DEV void rfc3339_write(uint8_t* out, int cur, long long secs) {
   }
 }
 
-#define LIT(e, s) (e).raw((const unsigned char*)(s), (int)sizeof(s) - 1)
+// Tiny literals (<= 4 bytes: separators, quotes) become ONE immediate
+// store instead of raw()'s striped-loop setup (~6 instrs + a global
+// load per byte); longer literals keep the wave-cooperative copy. The
+// all-immediate variant was measured SLOWER for large literals
+// (single-lane serialization) — this is scoped to the tiny ones.
+#define LIT(e, s)                                              \
+  do {                                                         \
+    if ((int)sizeof(s) - 1 <= 4)                               \
+      (e).template lit_small<(int)sizeof(s) - 1>(s);           \
+    else                                                       \
+      (e).raw((const unsigned char*)(s), (int)sizeof(s) - 1);  \
+  } while (0)
 
 template <bool W>
 struct JsonEmit {
@@ -258,6 +269,31 @@ struct JsonEmit {
         nt_store_u8(out + cur + j, s[j]);
     }
     cur += n;
+  }
+
+  template <int N, int M>
+  DEV void lit_small(const char (&str)[M]) {
+    if (W && lane_id() == 0) {
+      if (N == 1) {
+        out[cur] = (unsigned char)str[0];
+      } else if (N == 2) {
+        unsigned short v = (unsigned short)((unsigned char)str[0] |
+                                            ((unsigned char)str[1] << 8));
+        __builtin_memcpy(out + cur, &v, 2);
+      } else {
+        unsigned v = (unsigned)(unsigned char)str[0] |
+                     ((unsigned)(unsigned char)str[1] << 8) |
+                     ((N > 2 ? (unsigned)(unsigned char)str[2] : 0u) << 16) |
+                     ((N > 3 ? (unsigned)(unsigned char)str[3] : 0u) << 24);
+        if (N == 4)
+          __builtin_memcpy(out + cur, &v, 4);
+        else {
+          __builtin_memcpy(out + cur, &v, 2);
+          out[cur + 2] = (unsigned char)str[2];
+        }
+      }
+    }
+    cur += N;
   }
 
   // Striped copy, dword-granular: the emitters are memory-op bound on
