@@ -35,7 +35,8 @@ from .state import LocalStateManager, Page
 class GpuCrawlEngine:
     def __init__(self, cfg, sm: LocalStateManager, feed: SyntheticFeed,
                  device="cuda:0", posts_per_channel: Optional[int] = None,
-                 chunk_channels: int = 256, use_device_gen: bool = True):
+                 chunk_channels: int = 256, use_device_gen: bool = True,
+                 fixed_now: Optional[_dt.datetime] = None):
         from ..ops import gpu as gpu_mod
 
         self.gpu = gpu_mod
@@ -47,6 +48,9 @@ class GpuCrawlEngine:
         self.ppc = posts_per_channel or feed.cfg.posts_per_channel
         self.chunk_channels = chunk_channels
         self.seen = gpu_mod.SeenSet(self.device)
+        # pin the capture timestamp for byte-reproducible crawls
+        # (tools/verify_jsonl.py re-derives expected bytes from it)
+        self.fixed_now = fixed_now
         self._pin_ring = [None, None]      # reusable pinned host slots
         self._spill_tickets = [None, None]  # sink ticket per ring slot
         import collections
@@ -72,7 +76,7 @@ class GpuCrawlEngine:
         zero-padded uint8[M, 32] array instead of python strings — the
         BFS loop sorts/dedups those at numpy speed (857k string objects
         per layer cost ~0.5s to build and sort)."""
-        now = now or _dt.datetime.now(_dt.timezone.utc)
+        now = now or self.fixed_now or _dt.datetime.now(_dt.timezone.utc)
         discovered: List[str] = []
         discovered_arrays = []
         posts_total = 0

@@ -34,6 +34,9 @@ def main():
                     default="snowball")
     ap.add_argument("--blobs", type=int, default=2000,
                     help="media mode: number of blobs to fetch+upload")
+    ap.add_argument("--now", default=None,
+                    help="pin the capture timestamp (ISO, UTC) so "
+                         "tools/verify_jsonl.py can re-derive the bytes")
     ap.add_argument("--walkers", type=int, default=512,
                     help="randomwalk: concurrent walker chains per hop")
     ap.add_argument("--walkback-rate", type=int, default=15)
@@ -53,8 +56,13 @@ def main():
     feed = SyntheticFeed(FeedConfig(seed=2026, universe=args.universe,
                                     posts_per_channel=args.posts))
     sm = LocalStateManager(cfg)
+    fixed_now = None
+    if args.now:
+        import datetime as dt
+        fixed_now = dt.datetime.fromisoformat(args.now).replace(
+            tzinfo=dt.timezone.utc)
     eng = GpuCrawlEngine(cfg, sm, feed, posts_per_channel=args.posts,
-                         chunk_channels=512)
+                         chunk_channels=512, fixed_now=fixed_now)
     seeds = [feed.username_of(i) for i in range(args.seeds)]
     t0 = time.perf_counter()
     stats = eng.run(seeds)
