@@ -205,12 +205,15 @@ struct LinkList {
       if (eq) return;
     }
     if (cnt >= MAX_LINKS) return;
-    // All lanes store identical values (name is wave-uniform): keeps the
-    // next add()'s dedup reads coherent without a cross-lane fence.
-    for (int j = 0; j < len; ++j) names[cnt * 32 + j] = name[j];
-    lens[cnt] = (unsigned char)len;
-    srcs[cnt] = (unsigned char)src;
-    hashes[cnt] = fnv1a64(name, len);
+    // lane 0 stores; same-address stores from all 64 lanes serialize
+    // (64-way conflict in LDS, redundant traffic in global). Reads
+    // broadcast, and every lane keeps `cnt` in step.
+    if (lane == 0) {
+      for (int j = 0; j < len; ++j) names[cnt * 32 + j] = name[j];
+      lens[cnt] = (unsigned char)len;
+      srcs[cnt] = (unsigned char)src;
+      hashes[cnt] = fnv1a64(name, len);
+    }
     ++cnt;
   }
 };
@@ -536,6 +539,13 @@ DEV int emit_line(const BatchView& B, int i, unsigned char* out,
 __global__ void __launch_bounds__(256, 6)
 measure_extract_kernel(BatchView B, LinkOut LO, int* __restrict__ line_len) {
   __shared__ unsigned char s_lits[TG_POOL_BYTES];
+  // LDS-resident LinkList per wave: the extractor's dedup scans and
+  // name stores stay in LDS (single-lane writes, broadcast reads) and
+  // spill to the global LinkOut arrays ONCE per message.
+  __shared__ unsigned char s_ln[4][MAX_LINKS * 32];
+  __shared__ unsigned char s_ll[4][MAX_LINKS];
+  __shared__ unsigned char s_ls[4][MAX_LINKS];
+  __shared__ unsigned long long s_lh[4][MAX_LINKS];
   for (int t = threadIdx.x; t < TG_POOL_BYTES; t += blockDim.x)
     s_lits[t] = (unsigned char)tg_lit_pool.v[t];
   __syncthreads();
@@ -548,13 +558,20 @@ measure_extract_kernel(BatchView B, LinkOut LO, int* __restrict__ line_len) {
       if (lane == 0) { line_len[i] = 0; LO.cnt[i] = 0; }
       continue;
     }
-    LinkList L{LO.name + (size_t)i * MAX_LINKS * 32,
-               LO.name_len + (size_t)i * MAX_LINKS,
-               LO.src + (size_t)i * MAX_LINKS,
-               LO.hash + (size_t)i * MAX_LINKS, 0};
+    LinkList L{&s_ln[wave][0], &s_ll[wave][0], &s_ls[wave][0],
+               &s_lh[wave][0], 0};
     extract_links(B, i, L, lane);
     int len = emit_line<false, false>(B, i, nullptr, L, s_lits,
                                      nullptr, nullptr);
+    // spill the wave's links to the global LinkOut (lane-striped)
+    const int nb = L.cnt * 32;
+    unsigned char* gname = LO.name + (size_t)i * MAX_LINKS * 32;
+    for (int j = lane; j < nb; j += WAVE) gname[j] = s_ln[wave][j];
+    if (lane < L.cnt) {
+      LO.name_len[(size_t)i * MAX_LINKS + lane] = s_ll[wave][lane];
+      LO.src[(size_t)i * MAX_LINKS + lane] = s_ls[wave][lane];
+      LO.hash[(size_t)i * MAX_LINKS + lane] = s_lh[wave][lane];
+    }
     if (lane == 0) {
       line_len[i] = len;
       LO.cnt[i] = L.cnt;
