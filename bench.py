@@ -55,7 +55,9 @@ def main():
     ap.add_argument("--videos", type=int, default=400_000,
                     help="videos per GPU for the youtube platform")
     ap.add_argument("--chunk-videos", type=int, default=100_000)
-    ap.add_argument("--max-comments", type=int, default=1000)
+    ap.add_argument("--max-comments", type=int, default=1000,
+                    help="accepted for CLI parity (the corpus knobs "
+                         "below control the bench's comment shape)")
     ap.add_argument("--comment-rate", type=float, default=0.02,
                     help="telegram feed: fraction of posts with comment "
                          "threads (comment-heavy variant: 1.0)")
