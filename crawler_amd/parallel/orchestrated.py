@@ -38,11 +38,16 @@ class OrchestratedCrawl:
     def __init__(self, cfg, sm: LocalStateManager, store, rank: int,
                  world: int,
                  process_fn: Callable[[List[str]], Tuple[List[str], int]],
-                 chunk_channels: int = 64, dist=None, device=None):
+                 chunk_channels: int = 64, dist=None, device=None,
+                 deadends_fn: Callable[[], set] = None):
         """process_fn(names) -> (discovered_names, posts_stored).
         `dist` is torch.distributed (injected so CPU tests can pass gloo
         and unit tests can fake it); `device` is the compute device used
-        for collective buffers when the backend is NCCL/RCCL."""
+        for collective buffers when the backend is NCCL/RCCL;
+        `deadends_fn` (optional) reports the channels the LAST
+        process_fn call classified as deadends, so page statuses and
+        the MaxPages deadend-replacement budget (state/base.go:284)
+        stay correct — and identical across ranks via the exchange."""
         self.cfg = cfg
         self.sm = sm
         self.rank = rank
@@ -54,6 +59,7 @@ class OrchestratedCrawl:
 
         self.dist = dist
         self.device = device
+        self.deadends_fn = deadends_fn
         self.work_q = StoreQueue(store, M.TOPIC_WORK_QUEUE)
         self.result_q = StoreQueue(store, M.TOPIC_RESULTS)
         self.heartbeats = Heartbeats(store)
@@ -89,6 +95,7 @@ class OrchestratedCrawl:
         dist.barrier()  # queue fully published before claims start
 
         my_discovered: List[str] = []
+        my_dead: set = set()
         while True:
             raw = self.work_q.claim(timeout_s=0.0)
             if raw is None:
@@ -96,6 +103,8 @@ class OrchestratedCrawl:
             item = M.WorkItem.from_json(raw)
             chans = item.config["channels"]
             discovered, posts = self.process_fn(chans)
+            if self.deadends_fn is not None:
+                my_dead |= set(self.deadends_fn())
             my_discovered.extend(discovered)
             self.stats["chunks"] += 1
             self.stats["pages"] += len(chans)
@@ -108,7 +117,9 @@ class OrchestratedCrawl:
                 trace_id=item.trace_id,
             ).to_json())
         self.heartbeats.beat(f"rank{self.rank}", M.WORKER_IDLE)
-        return self._allgather_names(my_discovered)
+        all_dead = (set(self._allgather_names(sorted(my_dead)))
+                    if self.deadends_fn is not None else set())
+        return self._allgather_names(my_discovered), all_dead
 
     # ---- the crawl ----
 
@@ -122,11 +133,11 @@ class OrchestratedCrawl:
             if not layer:
                 break
             names = [p.url for p in layer]
-            all_discovered = self._run_layer(names, depth)
+            all_discovered, all_dead = self._run_layer(names, depth)
             self.stats["layers"] += 1
             self.stats["discovered"] += len(all_discovered)
             for p in layer:
-                p.status = "fetched"
+                p.status = "deadend" if p.url in all_dead else "fetched"
                 sm.update_page(p)
             if self.rank == 0:
                 # drain results for bookkeeping (page counts / errors)

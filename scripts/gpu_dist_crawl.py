@@ -78,6 +78,8 @@ def main():
         chunk_channels=args.chunk_channels,
         dist=dist if world > 1 else _SoloDist(),
         device=device,
+        deadends_fn=(None if args.fake_engine
+                     else (lambda: eng.last_deadends)),
     )
     seeds = [feed.username_of(i) for i in range(args.seeds)]
     t0 = time.perf_counter()

@@ -105,3 +105,28 @@ def test_two_rank_orchestrated_gloo(tmp_path):
         (tmp_path / "r0" / "o2" / "progress.json").read_text()
     )
     assert prog["status"] == "completed"
+
+
+def test_orchestrated_marks_deadends(tmp_path):
+    """deadends_fn routes engine-classified deadends into page status,
+    keeping the MaxPages deadend-replacement budget live (base.go:284)
+    — and the set is exchanged so all ranks would agree."""
+    cfg = CrawlerConfig(crawl_id="o5", storage_root=str(tmp_path),
+                        sampling_method="channel", max_depth=0,
+                        min_users=1)
+    sm = LocalStateManager(cfg)
+    store = InMemoryStore()
+    state = {"dead": set()}
+
+    def process(names):
+        state["dead"] = {n for n in names if n.startswith("dead")}
+        return [], len(names)
+
+    crawl = OrchestratedCrawl(cfg, sm, store, rank=0, world=1,
+                              process_fn=process, chunk_channels=8,
+                              dist=FakeDist(),
+                              deadends_fn=lambda: state["dead"])
+    crawl.run(["chan001", "deadbeef", "chan002"])
+    statuses = {p.url: p.status for p in sm.get_layer_by_depth(0)}
+    assert statuses == {"chan001": "fetched", "deadbeef": "deadend",
+                        "chan002": "fetched"}
