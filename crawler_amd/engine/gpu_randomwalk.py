@@ -173,20 +173,34 @@ class GpuRandomWalk:
         nz = mask_g.nonzero()
         mi_g, si_g = nz[:, 0], nz[:, 1]
         wk_g = torch.div(mi_g, P, rounding_mode="floor")
-        h_g = res.link_hash[mi_g, si_g]
-        ord1 = torch.argsort(h_g)
-        ord2 = torch.argsort(wk_g[ord1], stable=True)
-        perm = ord1[ord2]
-        ws_g, hs_g = wk_g[perm], h_g[perm]
+        # zero-padded rows for ALL pairs, then a byte-LEXICOGRAPHIC sort
+        # within each walker (names pack into 4 big-endian int64 keys:
+        # lexicographic byte order == big-endian integer order). Exact
+        # name dedup AND pre-sorted per-walker output: the host's
+        # per-walker sorted() over ~300-name lists disappears, and the
+        # cross-hop cache's hash keying no longer decides dedup.
+        col = torch.arange(w, device=dev, dtype=lens_g.dtype)
+        all_lens = lens_g[mi_g, si_g]
+        all_rows = torch.where(col[None, :] < all_lens[:, None],
+                               names_g[mi_g, si_g],
+                               torch.zeros_like(names_g[mi_g, si_g]))
+        be = (256 ** torch.arange(7, -1, -1, device=dev,
+                                  dtype=torch.int64))
+        keys = (all_rows.view(-1, 4, 8).to(torch.int64) * be).sum(-1)
+        perm = torch.argsort(keys[:, 3])
+        for kcol in (2, 1, 0):
+            perm = perm[torch.argsort(keys[perm, kcol], stable=True)]
+        perm = perm[torch.argsort(wk_g[perm], stable=True)]
+        ws_g = wk_g[perm]
+        ks_g = keys[perm]
         keep = torch.ones(ws_g.numel(), dtype=torch.bool, device=dev)
         if ws_g.numel() > 1:
-            keep[1:] = (ws_g[1:] != ws_g[:-1]) | (hs_g[1:] != hs_g[:-1])
+            keep[1:] = ((ws_g[1:] != ws_g[:-1])
+                        | (ks_g[1:] != ks_g[:-1]).any(dim=1))
         sel = perm[keep]
-        col = torch.arange(w, device=dev, dtype=lens_g.dtype)
-        sel_lens = lens_g[mi_g[sel], si_g[sel]]
-        sel_names = names_g[mi_g[sel], si_g[sel]]
-        u_rows_g = torch.where(col[None, :] < sel_lens[:, None],
-                               sel_names, torch.zeros_like(sel_names))
+        h_g = res.link_hash[mi_g, si_g]
+        hs_sel_g = h_g[sel]
+        u_rows_g = all_rows[sel]
 
         # D2H through a reusable pinned ring (2-deep: this hop's disk
         # writes overlap the next hop's kernels via the native sink)
@@ -205,7 +219,7 @@ class GpuRandomWalk:
         line_off = res.line_off.cpu().numpy()
         line_len = res.line_len.cpu().numpy()
         u_w = ws_g[keep].to(torch.int32).cpu().numpy()
-        u_h = hs_g[keep].cpu().numpy()
+        u_h = hs_sel_g.cpu().numpy()
         u_rows = u_rows_g.cpu().numpy()
         torch.cuda.synchronize()
         self.timings["d2h"] += _time.perf_counter() - t0
@@ -288,9 +302,10 @@ class GpuRandomWalk:
         uw_s = u_w[final]
         names_s = res_names[final]
         bounds = np.searchsorted(uw_s, np.arange(K + 1))
-        # per-walker lists sorted by NAME (walk_tail sorts before
-        # choosing; device order within a walker is hash order)
-        per_walker = [sorted(names_s[bounds[k]:bounds[k + 1]])
+        # per-walker lists arrive NAME-SORTED from the device sort
+        # (byte-lexicographic == python str sort for ASCII names), so
+        # no host re-sort (walk_tail_fast consumes sorted lists)
+        per_walker = [list(names_s[bounds[k]:bounds[k + 1]])
                       for k in range(K)]
         for k, p in enumerate(live):
             try:
