@@ -80,7 +80,11 @@ def main():
     device = torch.device("cuda", local_rank)
 
     if world > 1:
-        torch.distributed.init_process_group("nccl")
+        # CRAWL_BENCH_BACKEND=gloo lets a single-GPU box dry-run the
+        # world>1 path (two ranks sharing cuda:0; collectives bounce
+        # through CPU) before a real RCCL multi-GPU run
+        torch.distributed.init_process_group(
+            os.environ.get("CRAWL_BENCH_BACKEND", "nccl"))
 
     from crawler_amd.feed import FeedConfig, SyntheticFeed
     from crawler_amd.ops import gpu

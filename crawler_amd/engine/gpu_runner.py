@@ -270,14 +270,7 @@ class GpuCrawlEngine:
                         hs = torch.from_numpy(
                             _vv.fnv1a64_rows(discovered))
                         self.seen.insert_hashes(hs.to(self.device))
-                    bl = self.seen.bloom
-                    cdev = C.collective_device(dist, self.device)
-                    if bl.device == cdev:
-                        C.bloom_union(bl, dist, world)
-                    else:
-                        host_bl = bl.to(cdev)
-                        C.bloom_union(host_bl, dist, world)
-                        bl.copy_(host_bl)
+                    C.bloom_union(self.seen.bloom, dist, world)
                     # deadend statuses must agree across ranks: the
                     # next layer's deadend-replacement budget is
                     # computed independently by every rank

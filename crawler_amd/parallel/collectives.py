@@ -120,20 +120,27 @@ def bloom_union(bloom: torch.Tensor, dist, world: int,
                 group=None) -> None:
     """In-place cross-rank OR-union of a bloom filter's int32 words
     ('merge via bitwise-OR all-reduce on bloom segments').
-    Works on gloo (native BOR) and NCCL/RCCL (gather + device OR)."""
+    Works on gloo (native BOR) and NCCL/RCCL (gather + device OR).
+    Handles placement itself: a CUDA bloom under a gloo group (or vice
+    versa) is bounced through the backend's device."""
     try:
         backend = str(dist.get_backend(group))
     except (AttributeError, RuntimeError, ValueError):
         backend = "gloo"
-    check_collective_device(bloom, dist, group)
+    cdev = collective_device(dist, bloom.device if
+                             bloom.device.type == "cuda" else None, group)
+    buf = bloom if bloom.device == cdev else bloom.to(cdev)
+    check_collective_device(buf, dist, group)
     if backend.startswith("nccl"):
-        gathered = [torch.empty_like(bloom) for _ in range(world)]
-        dist.all_gather(gathered, bloom)
+        gathered = [torch.empty_like(buf) for _ in range(world)]
+        dist.all_gather(gathered, buf)
         for r, g in enumerate(gathered):
             if r != _safe_rank(dist, group):
-                bloom.bitwise_or_(g)
+                buf.bitwise_or_(g)
     else:
-        dist.all_reduce(bloom, op=dist.ReduceOp.BOR)
+        dist.all_reduce(buf, op=dist.ReduceOp.BOR)
+    if buf is not bloom:
+        bloom.copy_(buf)
 
 
 def _safe_rank(dist, group=None) -> int:
