@@ -76,8 +76,13 @@ def main():
         return run_cpu_debug(args)
 
     assert torch.cuda.is_available(), "bench requires a ROCm GPU"
-    torch.cuda.set_device(local_rank)
-    device = torch.device("cuda", local_rank)
+    n_dev = torch.cuda.device_count()
+    dev_idx = local_rank % n_dev
+    if dev_idx != local_rank:
+        log(f"rank {rank}: wrapping local_rank {local_rank} onto "
+            f"cuda:{dev_idx} ({n_dev} visible devices — dry-run mode)")
+    torch.cuda.set_device(dev_idx)
+    device = torch.device("cuda", dev_idx)
 
     if world > 1:
         # CRAWL_BENCH_BACKEND=gloo lets a single-GPU box dry-run the
