@@ -49,9 +49,15 @@ def main():
     world = int(os.environ.get("WORLD_SIZE", "1"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     if not args.fake_engine:
+        # wrap onto visible devices so a 1-GPU box can dry-run world>1
+        # (two ranks share cuda:0; use CRAWL_DIST_BACKEND=gloo then)
+        local_rank = local_rank % torch.cuda.device_count()
         torch.cuda.set_device(local_rank)
     if world > 1:
-        dist.init_process_group("gloo" if args.fake_engine else "nccl")
+        backend = os.environ.get(
+            "CRAWL_DIST_BACKEND",
+            "gloo" if args.fake_engine else "nccl")
+        dist.init_process_group(backend)
     host = os.environ.get("MASTER_ADDR", "127.0.0.1")
     store = TCPStore(host, args.store_port, is_master=(rank == 0),
                      wait_for_workers=False)
