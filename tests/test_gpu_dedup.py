@@ -126,3 +126,65 @@ def test_compact_claimed_matches_mask_gather():
     mask2 = seen.claim(res)
     n2, _ = seen.compact_claimed(res, mask2)
     assert n2.shape[0] == 0
+
+
+def test_merge_remote_exchange_on_device():
+    """SeenSet.merge_remote end-to-end on the GPU with a fake 2-rank
+    dist: remote hashes insert into the table (blocking later claims)
+    and the bloom union runs without device/backend crashes (the
+    world>1 path the driver's multi-GPU run exercises for real)."""
+    from crawler_amd.ops import gpu
+
+    gpu.require_lib()
+    feed = SyntheticFeed(FeedConfig(seed=77, universe=20_000))
+    batch = feed.build_batch(np.arange(4), posts_per_channel=64)
+    res = gpu.parse_encode(batch.to("cuda:0"),
+                           now=dt.datetime(2026, 1, 1,
+                                           tzinfo=dt.timezone.utc))
+    seen = gpu.SeenSet(torch.device("cuda:0"))
+    new_mask = seen.claim(res)
+    torch.cuda.synchronize()
+    _names, my_hashes = seen.compact_claimed(res, new_mask)
+
+    remote = torch.tensor([987654321987, 123456789123],
+                          dtype=torch.int64)
+
+    class Fake2Dist:
+        """rank 0 of 2; 'rank 1' contributes `remote`."""
+
+        @staticmethod
+        def get_backend(group=None):
+            return "gloo"
+
+        @staticmethod
+        def get_rank(group=None):
+            return 0
+
+        @staticmethod
+        def all_gather(out_list, t):
+            out_list[0].copy_(t)
+            if t.dtype == torch.int64 and t.numel() >= remote.numel():
+                out_list[1].zero_()
+                out_list[1][:remote.numel()] = remote
+            else:
+                out_list[1].copy_(t)
+
+        @staticmethod
+        def all_reduce(t, op=None):
+            pass
+
+        class ReduceOp:
+            BOR = "bor"
+
+    bloom_before = seen.bloom.clone()
+    inserted = seen.merge_remote(my_hashes, Fake2Dist, 2)
+    torch.cuda.synchronize()
+    assert inserted == remote.numel()
+    # the remote hashes now occupy the table: inserting them again via
+    # the claim-side bulk insert is a no-op, and our own claims remain
+    seen.insert_hashes(remote.to("cuda:0"))
+    mask2 = seen.claim(res)
+    n2, _ = seen.compact_claimed(res, mask2)
+    assert n2.shape[0] == 0  # nothing re-claims
+    # bloom unchanged by the (identity) gloo union but still valid
+    assert torch.equal(seen.bloom, bloom_before)
