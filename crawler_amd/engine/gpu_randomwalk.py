@@ -264,6 +264,8 @@ class GpuRandomWalk:
         res_names = np.empty(U, dtype=object)
         res_adm = np.zeros(U, dtype=bool)
         miss = self._vc_lookup(u_h, res_names, res_adm)
+        self.timings["t-cache"] += _time.perf_counter() - t0
+        t0 = _time.perf_counter()
         inv_set = {b for b, t in self._inv_bytes_items(ttl_cut)}
         if miss.any():
             mh, m_first = np.unique(u_h[miss], return_index=True)
@@ -293,6 +295,8 @@ class GpuRandomWalk:
                                 if nm.encode() not in inv_set]
             self.sm.add_discovered_channels_bulk(new_admitted)
             self.rw.upsert_seed_channels_bulk(new_admitted)
+        self.timings["t-validate"] += _time.perf_counter() - t0
+        t0 = _time.perf_counter()
         # dynamic invalid-channel cache (mutated by 400 handling)
         not_inv = (~np.isin(un, np.array(sorted(inv_set),
                                          dtype=un.dtype))
@@ -307,6 +311,8 @@ class GpuRandomWalk:
         # no host re-sort (walk_tail_fast consumes sorted lists)
         per_walker = [list(names_s[bounds[k]:bounds[k + 1]])
                       for k in range(K)]
+        self.timings["t-group"] += _time.perf_counter() - t0
+        t0 = _time.perf_counter()
         for k, p in enumerate(live):
             try:
                 randomwalk.walk_tail_fast(p, per_walker[k], self.sm,
