@@ -51,11 +51,13 @@ class GpuRandomWalk:
         self._spill_tickets = [None, None]  # native-sink tickets
         self._inflight_paths = {}           # slot -> channel set
         self._hop_idx = 0
-        # cross-hop validation cache: hash -> decoded name (admitted)
-        # or None (statically rejected). A plain dict beats the sorted
-        # numpy levels (searchsorted + object gathers cost ~170 ms/hop
-        # at 160k probes; dict gets ~40 ms).
-        self._vc = {}
+        # cross-hop validation cache keyed by link hash — two sorted
+        # levels (main + pending) so each hop merges only the small
+        # level; validation is a pure function of the name, so hits
+        # skip validate_names AND the decode
+        self._vc = [[np.zeros(0, dtype=np.int64),
+                     np.zeros(0, dtype=object),
+                     np.zeros(0, dtype=bool)] for _ in range(2)]
         import collections
         self.timings = collections.defaultdict(float)  # phase seconds
 
@@ -82,29 +84,40 @@ class GpuRandomWalk:
             if t > ttl_cut:
                 yield u.encode(), t
 
-    _VC_MISS = object()  # sentinel distinct from None (= rejected)
-
     def _vc_lookup(self, hashes, out_names, out_adm):
-        """Fill out_names/out_adm from the dict cache; returns the
+        """Fill out_names/out_adm from both cache levels; returns the
         still-missing mask."""
-        vc = self._vc
-        sent = self._VC_MISS
-        vals = [vc.get(h, sent) for h in hashes.tolist()]
-        miss = np.fromiter((v is sent for v in vals), dtype=bool,
-                           count=len(vals))
-        adm = np.fromiter(
-            (v is not sent and v is not None for v in vals),
-            dtype=bool, count=len(vals))
-        out_adm[:] = adm
-        if adm.any():
-            idx = np.nonzero(adm)[0]
-            out_names[idx] = [vals[i] for i in idx]
+        miss = np.ones(len(hashes), dtype=bool)
+        for h_arr, n_arr, a_arr in self._vc:
+            if not len(h_arr) or not miss.any():
+                continue
+            pos = np.searchsorted(h_arr, hashes)
+            pos_c = np.clip(pos, 0, len(h_arr) - 1)
+            hit = (h_arr[pos_c] == hashes) & miss
+            if hit.any():
+                out_names[hit] = n_arr[pos_c[hit]]
+                out_adm[hit] = a_arr[pos_c[hit]]
+                miss &= ~hit
         return miss
 
     def _vc_insert(self, new_h, new_n, new_a):
-        vc = self._vc
-        for h, nm, ok in zip(new_h.tolist(), new_n, new_a.tolist()):
-            vc[h] = nm if ok else None
+        """Merge new entries into the pending level; spill pending into
+        main when it grows past a quarter of main (amortized O(n))."""
+        main, pend = self._vc
+        all_h = np.concatenate([pend[0], new_h])
+        order = np.argsort(all_h, kind="mergesort")
+        pend[0] = all_h[order]
+        pend[1] = np.concatenate([pend[1], new_n])[order]
+        pend[2] = np.concatenate([pend[2], new_a])[order]
+        if len(pend[0]) > max(len(main[0]) // 4, 1 << 18):
+            all_h = np.concatenate([main[0], pend[0]])
+            order = np.argsort(all_h, kind="mergesort")
+            main[0] = all_h[order]
+            main[1] = np.concatenate([main[1], pend[1]])[order]
+            main[2] = np.concatenate([main[2], pend[2]])[order]
+            pend[0] = np.zeros(0, dtype=np.int64)
+            pend[1] = np.zeros(0, dtype=object)
+            pend[2] = np.zeros(0, dtype=bool)
 
     def _hop(self, pages: List[Page],
              now: Optional[_dt.datetime] = None) -> int:
