@@ -183,3 +183,31 @@ def test_crash_resume_does_not_duplicate_posts(tmp_path):
     batch = feed.build_batch(np.array([21]), posts_per_channel=64)
     lines, _ = encode_batch(batch, now=NOW)
     assert data == b"".join(lines)
+
+
+def test_bench_world2_dry_run_on_one_gpu(tmp_path):
+    """The bench's world>1 path (discovery exchange, bloom union,
+    per-rank host info) executes end-to-end with two ranks sharing
+    cuda:0 over gloo — the shape the driver's multi-GPU SCALE run
+    exercises over RCCL."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, CRAWL_BENCH_BACKEND="gloo")
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29781", os.path.join(repo, "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1",
+         "--channels", "50", "--posts", "200", "--chunk-channels", "25"],
+        capture_output=True, text=True, timeout=300, cwd=repo, env=env,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    res = json.loads(line)
+    assert res["n_gpus"] == 2
+    assert res["value"] > 0
+    assert len(res["config"]["per_rank_host"]) == 2
