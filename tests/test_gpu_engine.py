@@ -211,3 +211,34 @@ def test_bench_world2_dry_run_on_one_gpu(tmp_path):
     assert res["n_gpus"] == 2
     assert res["value"] > 0
     assert len(res["config"]["per_rank_host"]) == 2
+
+
+def test_dist_crawl_world2_real_engine_on_one_gpu(tmp_path):
+    """OrchestratedCrawl + the REAL GPU engine at world 2 on one GPU
+    (gloo collectives): dynamic chunk claiming, discovery + deadend
+    exchange, rank-sharded JSONL (BASELINE config #3 shape)."""
+    import json
+    import os
+    import subprocess
+    import sys
+
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ, CRAWL_DIST_BACKEND="gloo")
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29783", os.path.join(repo, "scripts",
+         "gpu_dist_crawl.py"), "--seeds", "40", "--posts", "200",
+         "--max-depth", "1", "--max-pages", "200",
+         "--storage", str(tmp_path), "--store-port", "29784"],
+        capture_output=True, text=True, timeout=300, cwd=repo, env=env,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.splitlines() if l.startswith("{")][-1]
+    res = json.loads(line)
+    assert res["world"] == 2
+    assert res["layers"] >= 2
+    assert res["posts"] > 0
+    with open(os.path.join(str(tmp_path), "r0", "dist-crawl",
+                           "progress.json")) as f:
+        assert json.load(f)["status"] == "completed"
