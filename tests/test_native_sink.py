@@ -255,3 +255,25 @@ def test_ticketed_pipeline_evicts_fds(tmp_path):
     sink.wait_ticket(last)
     sink.drain()
     assert sink.open_files <= 32
+
+
+def test_write_failures_surface_with_errno(tmp_path):
+    """Unwritable targets raise with the captured errno instead of
+    silently dropping data (the EMFILE incident made errno part of the
+    contract)."""
+    import os
+
+    import pytest
+
+    from crawler_amd.native import load
+
+    sink = load().FanoutSink(2, 8)
+    # a FILE in the parent position defeats even root (ENOTDIR)
+    blocker = tmp_path / "blocker"
+    blocker.write_text("file, not a directory")
+    data = b"x" * 16
+    with pytest.raises(RuntimeError, match=r"errno=\d+"):
+        sink.write_batch([str(blocker / "sub" / "posts.jsonl")],
+                         data, [0], [len(data)])
+    sink.drain()
+    assert os.path.exists(blocker)
