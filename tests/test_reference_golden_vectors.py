@@ -196,3 +196,62 @@ def test_build_link_msg_id_shift20():
     # private channel (no username) -> empty link
     link3, pid3 = build_telegram_link_and_message_id("", msg)
     assert link3 == "" and pid3 == 5
+
+
+# --- main_test.go:9-205 TestValidateSamplingMethod literal matrix ---
+
+SAMPLING_VECTORS = [
+    # (platform, method, kwargs, expect_error_substring or None)
+    ("telegram", "channel", dict(url_list=["https://t.me/test"]), None),
+    ("telegram", "snowball",
+     dict(url_list=["https://t.me/test1", "https://t.me/test2"]), None),
+    ("telegram", "random", {}, "not supported for platform 'telegram'"),
+    ("youtube", "channel",
+     dict(url_list=["https://youtube.com/c/test"]), None),
+    ("youtube", "random", {}, None),
+    ("youtube", "snowball",
+     dict(url_list=["https://youtube.com/c/seed"]), None),
+    ("youtube", "channel", {}, "channel sampling requires URLs"),
+    ("telegram", "snowball", {}, "snowball sampling requires URLs"),
+    ("youtube", "channel", dict(url_file="/path/to/urls.txt"), None),
+    ("unsupported", "channel",
+     dict(url_list=["https://example.com"]), "unsupported platform"),
+    ("youtube", "invalid",
+     dict(url_list=["https://youtube.com/c/test"]),
+     "not supported for platform 'youtube'"),
+    ("telegram", "random-walk",
+     dict(seed_size=100, crawl_id="my-crawl"), None),
+    ("telegram", "random-walk",
+     dict(url_list=["chan1", "chan2"], crawl_id="my-crawl"), None),
+    ("telegram", "random-walk",
+     dict(url_file_url="https://example.com/seeds.txt",
+          crawl_id="my-crawl"), None),
+    ("telegram", "random-walk", dict(crawl_id="my-crawl"),
+     "must provide either seed urls or seed size"),
+    ("telegram", "random-walk",
+     dict(url_list=["chan1"], seed_size=100, crawl_id="my-crawl"),
+     "must provide either seed urls or seed size"),
+    ("telegram", "random-walk",
+     dict(seed_size=100,
+          crawl_id="this-crawl-id-is-way-too-long-and-exceeds-32-chars"),
+     "crawl IDs cannot exceed 32 characters"),
+    # dapr-job mode relaxations (TestValidateSamplingMethodDaprJobMode)
+    ("telegram", "channel", dict(mode="dapr-job"), None),
+    ("youtube", "snowball", dict(mode="dapr-job"), None),
+    ("telegram", "channel", dict(mode="standalone"),
+     "channel sampling requires URLs"),
+    ("telegram", "channel", {}, "channel sampling requires URLs"),
+]
+
+
+@pytest.mark.parametrize("platform,method,kwargs,err", SAMPLING_VECTORS)
+def test_sampling_validation_reference_vectors(platform, method, kwargs,
+                                               err):
+    from crawler_amd.config import validate_sampling_method
+
+    if err is None:
+        validate_sampling_method(platform, method, **kwargs)
+    else:
+        with pytest.raises(ValueError) as e:
+            validate_sampling_method(platform, method, **kwargs)
+        assert err in str(e.value), (err, str(e.value))
