@@ -11,6 +11,7 @@ Sources:
 - /root/reference/telegramhelper/username_filter_test.go:5-68
 - /root/reference/telegramhelper/channelvalidator_test.go:15-85
 - /root/reference/telegramhelper/tdutils.go:1005-1031 (msg-id >> 20)
+- /root/reference/main_test.go:9-205 (sampling validation matrix)
 """
 import pytest
 
