@@ -48,20 +48,23 @@ class ChannelResult:
 
 def fetch_channel_messages(client, chat_id: int, cfg,
                            rng: Optional[random.Random] = None) -> list:
-    """FetchChannelMessagesWithSampling (telegramutils.go:25-157)."""
+    """FetchChannelMessagesWithSampling (telegramutils.go:25-157).
+
+    Filter precedence matches crawl/runner.go:909-912 exactly: when
+    BOTH date-between bounds are set the fetch uses that window and
+    min_post_date is IGNORED; otherwise min_post_date alone bounds the
+    walk. post_recency NEVER filters messages — the reference uses it
+    only for the channel-activity deadend check
+    (isChannelActiveWithinPeriod, runner.go:628)."""
     out = []
     from_id = 0
     max_posts = cfg.max_posts if cfg.max_posts and cfg.max_posts > 0 else None
-    min_ts = None
-    if cfg.min_post_date is not None:
-        min_ts = cfg.min_post_date.timestamp()
-    if cfg.post_recency is not None:
-        ts = cfg.post_recency.timestamp()
-        min_ts = max(min_ts, ts) if min_ts else ts
-    lo = hi = None
+    lo = hi = min_ts = None
     if cfg.date_between_min is not None and cfg.date_between_max is not None:
         lo = cfg.date_between_min.timestamp()
         hi = cfg.date_between_max.timestamp()
+    elif cfg.min_post_date is not None:
+        min_ts = cfg.min_post_date.timestamp()
     while True:
         page = client.get_chat_history(chat_id, from_message_id=from_id,
                                        limit=100)

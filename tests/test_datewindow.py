@@ -75,19 +75,34 @@ def test_min_post_date_stops_at_boundary():
     assert c.calls == 1  # stopped inside the first page, no more fetches
 
 
-def test_post_recency_tightens_min_date():
-    """min_ts = max(min_post_date, post_recency) — the tighter bound
-    wins (telegramutils.go:34-44)."""
+def test_post_recency_never_filters_messages():
+    """post_recency is the ACTIVITY check's bound only
+    (isChannelActiveWithinPeriod, runner.go:628); the message fetch
+    ignores it — only min_post_date bounds the walk
+    (runner.go:909-912)."""
     c = PagedClient(300)
     loose = BASE - 100 * DAY
     tight = BASE - 10 * DAY
     out = fetch_channel_messages(
         c, 1, cfg(min_post_date=ts2dt(loose), post_recency=ts2dt(tight)))
-    assert len(out) == 11
+    assert len(out) == 101  # min_post_date governs, recency ignored
     c2 = PagedClient(300)
     out2 = fetch_channel_messages(
         c2, 1, cfg(min_post_date=ts2dt(tight), post_recency=ts2dt(loose)))
-    assert len(out2) == 11  # symmetric: max() either way
+    assert len(out2) == 11
+
+
+def test_date_between_overrides_min_post_date():
+    """When BOTH date-between bounds are set, min_post_date is ignored
+    (runner.go:909-912 picks the sampling fetch exclusively)."""
+    c = PagedClient(300)
+    lo = BASE - 60 * DAY
+    hi = BASE - 30 * DAY
+    out = fetch_channel_messages(
+        c, 1, cfg(date_between_min=ts2dt(lo), date_between_max=ts2dt(hi),
+                  min_post_date=ts2dt(BASE - 40 * DAY)))
+    # min_post_date (40d) would cut the window short; it must NOT
+    assert len(out) == 31
 
 
 def test_date_between_window_clamps_both_sides():
