@@ -186,6 +186,21 @@ def validate_sampling_method(
         )
 
 
+def calculate_date_filters(cfg, now=None):
+    """Crawl date window with the reference's STRICT precedence
+    (dapr/standalone.go:1092-1117 CalculateDateFilters): date-between
+    (both bounds set) > post_recency > min_post_date; lower-priority
+    fields are IGNORED, not combined. Returns (from_time, to_time);
+    from_time may be None (the reference's zero time) and to_time
+    defaults to now for the non-date-between branches."""
+    now = now or _dt.datetime.now(_dt.timezone.utc)
+    if cfg.date_between_min is not None and cfg.date_between_max is not None:
+        return cfg.date_between_min, cfg.date_between_max
+    if cfg.post_recency is not None:
+        return cfg.post_recency, now
+    return cfg.min_post_date, now
+
+
 def config_from_env(cfg: CrawlerConfig, env=os.environ) -> CrawlerConfig:
     """Apply CRAWLER_-prefixed environment overrides (viper env parity:
     main.go:231-261 — dots/dashes map to underscores, prefix CRAWLER_)."""

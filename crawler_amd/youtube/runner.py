@@ -49,6 +49,17 @@ def run_youtube(cfg, urls: List[str], client=None, sm=None,
             client.get_channel_info(cid)
             videos.extend(client.get_channel_videos(cid, limit))
 
+    # date window (FetchYoutubeChannelInfoAndVideos passes the
+    # CalculateDateFilters range into the crawl job,
+    # dapr/standalone.go:1092-1147): strict precedence, not combined
+    from ..config import calculate_date_filters
+
+    from_t, to_t = calculate_date_filters(cfg)
+    if from_t is not None:
+        videos = [v for v in videos
+                  if v.published_at is not None
+                  and from_t <= v.published_at <= to_t]
+
     # 10-worker post conversion pool (youtube_crawler.go:354)
     def conv(v):
         ch = None

@@ -256,3 +256,37 @@ def test_sampling_validation_reference_vectors(platform, method, kwargs,
         with pytest.raises(ValueError) as e:
             validate_sampling_method(platform, method, **kwargs)
         assert err in str(e.value), (err, str(e.value))
+
+
+# --- dapr/calculate_date_filters_test.go:12-112 literal cases ---
+
+def test_calculate_date_filters_reference_vectors():
+    import datetime as dt
+
+    from crawler_amd.config import CrawlerConfig, calculate_date_filters
+
+    UTC = dt.timezone.utc
+    d = lambda *a: dt.datetime(*a, tzinfo=UTC)
+    now = d(2026, 6, 1)
+
+    # DateBetween takes precedence over everything
+    cfg = CrawlerConfig(crawl_id="c", date_between_min=d(2024, 1, 1),
+                        date_between_max=d(2024, 12, 31),
+                        post_recency=d(2023, 6, 1),
+                        min_post_date=d(2022, 1, 1))
+    assert calculate_date_filters(cfg, now) == (d(2024, 1, 1),
+                                                d(2024, 12, 31))
+    # only DateBetweenMin set -> branch NOT taken; PostRecency used
+    cfg = CrawlerConfig(crawl_id="c", date_between_min=d(2024, 1, 1),
+                        post_recency=d(2023, 6, 1))
+    assert calculate_date_filters(cfg, now) == (d(2023, 6, 1), now)
+    # PostRecency branch ignores MinPostDate; to == now
+    cfg = CrawlerConfig(crawl_id="c", post_recency=d(2024, 6, 1),
+                        min_post_date=d(2022, 1, 1))
+    assert calculate_date_filters(cfg, now) == (d(2024, 6, 1), now)
+    # MinPostDate fallback
+    cfg = CrawlerConfig(crawl_id="c", min_post_date=d(2022, 1, 1))
+    assert calculate_date_filters(cfg, now) == (d(2022, 1, 1), now)
+    # all zero -> (None, ~now)
+    cfg = CrawlerConfig(crawl_id="c")
+    assert calculate_date_filters(cfg, now) == (None, now)
