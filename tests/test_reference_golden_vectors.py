@@ -290,3 +290,59 @@ def test_calculate_date_filters_reference_vectors():
     # all zero -> (None, ~now)
     cfg = CrawlerConfig(crawl_id="c")
     assert calculate_date_filters(cfg, now) == (None, now)
+
+
+# --- common/utils_coverage_test.go:98-200 + utils_test.go:13-40 ---
+
+def test_read_urls_from_file_reference_vectors(tmp_path):
+    """ReadURLsFromFile: trims whitespace, drops empty lines, keeps
+    order (the CLI's --url-file path)."""
+    from crawler_amd.cli import resolve_urls
+
+    class A:
+        urls = ""
+        url_file = ""
+        url_file_url = ""
+
+    f = tmp_path / "urls.txt"
+    f.write_text(
+        "https://example.com/channel1\n\t\n"
+        "https://example.com/channel2\n"
+        "   https://example.com/channel3   \n\n"
+        "https://example.com/channel4\n")
+    a = A()
+    a.url_file = str(f)
+    assert resolve_urls(a) == [
+        "https://example.com/channel1",
+        "https://example.com/channel2",
+        "https://example.com/channel3",
+        "https://example.com/channel4",
+    ]
+    # empty file -> no urls
+    e = tmp_path / "empty.txt"
+    e.write_text("")
+    a.url_file = str(e)
+    assert resolve_urls(a) == []
+    # whitespace-only -> no urls
+    ws = tmp_path / "ws.txt"
+    ws.write_text("   \n\t\t\n   \t   \n")
+    a.url_file = str(ws)
+    assert resolve_urls(a) == []
+    # file:// form of --url-file-url resolves identically
+    a.url_file = ""
+    a.url_file_url = f"file://{f}"
+    assert len(resolve_urls(a)) == 4
+
+
+def test_generate_crawl_id_reference_format():
+    """utils_test.go:13-40: 14 digits, parseable as YYYYMMDDHHMMSS."""
+    import datetime as dt
+    import re
+
+    from crawler_amd.config import generate_crawl_id
+
+    cid = generate_crawl_id()
+    assert re.fullmatch(r"\d{14}", cid)
+    dt.datetime.strptime(cid, "%Y%m%d%H%M%S")  # parses back
+    fixed = generate_crawl_id(dt.datetime(2024, 3, 5, 6, 7, 8))
+    assert fixed == "20240305060708"
