@@ -346,3 +346,49 @@ def test_generate_crawl_id_reference_format():
     dt.datetime.strptime(cid, "%Y%m%d%H%M%S")  # parses back
     fixed = generate_crawl_id(dt.datetime(2024, 3, 5, 6, 7, 8))
     assert fixed == "20240305060708"
+
+
+# --- crawl/flood_wait_test.go:31-140 TestParseFloodWaitSecs matrix ---
+
+FLOOD_VECTORS = [
+    ("nil error", "", 0, False),
+    ("unrelated error", "connection refused", 0, False),
+    ("bare FLOOD_WAIT with seconds", "FLOOD_WAIT_72560", 72560, True),
+    ("prefixed as seen from TDLib", "[429] FLOOD_WAIT_300", 300, True),
+    ("short ban below retire threshold", "FLOOD_WAIT_30", 30, True),
+    ("exactly at retire threshold", "FLOOD_WAIT_300", 300, True),
+    ("FLOOD_WAIT_0", "FLOOD_WAIT_0", 0, True),
+    ("no trailing digits", "FLOOD_WAIT_", 0, True),
+    ("embedded in longer message",
+     "rpc error: code 429 FLOOD_WAIT_600 please wait", 600, True),
+    ("wrapped error containing FLOOD_WAIT",
+     "SearchPublicChat failed: FLOOD_WAIT_1800", 1800, True),
+    ("HTTP 429 retry-after format",
+     "429 Too Many Requests: retry after 72560", 72560, True),
+    ("retry after with no digits",
+     "429 Too Many Requests: retry after soon", 0, True),
+]
+
+
+@pytest.mark.parametrize("name,msg,want_secs,want_flood", FLOOD_VECTORS,
+                         ids=[v[0].replace(" ", "_")
+                              for v in FLOOD_VECTORS])
+def test_parse_flood_wait_reference_vectors(name, msg, want_secs,
+                                            want_flood):
+    from crawler_amd.engine.errors import parse_flood_wait_secs
+
+    secs, is_flood = parse_flood_wait_secs(msg)
+    assert is_flood == want_flood, (msg, is_flood)
+    assert secs == want_secs, (msg, secs)
+
+
+def test_flood_wait_retire_threshold_boundary():
+    """flood_wait_test.go:122-140 + crawl/runner.go:49: 300 s."""
+    from crawler_amd.engine.errors import (
+        FLOOD_WAIT_RETIRE_THRESHOLD_SECS, parse_flood_wait_secs)
+
+    assert FLOOD_WAIT_RETIRE_THRESHOLD_SECS == 300
+    below, _ = parse_flood_wait_secs("FLOOD_WAIT_299")
+    at, _ = parse_flood_wait_secs("FLOOD_WAIT_300")
+    above, _ = parse_flood_wait_secs("FLOOD_WAIT_301")
+    assert below < 300 <= at < above
