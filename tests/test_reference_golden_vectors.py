@@ -392,3 +392,28 @@ def test_flood_wait_retire_threshold_boundary():
     at, _ = parse_flood_wait_secs("FLOOD_WAIT_300")
     above, _ = parse_flood_wait_secs("FLOOD_WAIT_301")
     assert below < 300 <= at < above
+
+
+# --- crawl/runner_400_test.go:20-60, 841-900 isTDLib400 matrix ---
+
+TDLIB400_VECTORS = [
+    ("400 USERNAME_INVALID", True, "unbracketed prod-log format"),
+    ("400 USERNAME_NOT_OCCUPIED", True, "unbracketed prod-log format"),
+    ("[400] CHANNEL_INVALID", True, "bracketed TDLib format"),
+    ("no messages found in the chat", True, "empty/inaccessible channel"),
+    ("[404] USERNAME_NOT_OCCUPIED", False, "404 is not 400"),
+    ("FLOOD_WAIT_300", False, "flood wait is not a 400"),
+    ("network timeout", False, "unrelated error"),
+    ("", False, "empty string"),
+    ("getChannelInfo failed: 400 USERNAME_INVALID", True,
+     "wrapped error still detected"),
+]
+
+
+@pytest.mark.parametrize("msg,want,comment", TDLIB400_VECTORS,
+                         ids=[v[2].replace(" ", "_")
+                              for v in TDLIB400_VECTORS])
+def test_is_tdlib_400_reference_vectors(msg, want, comment):
+    from crawler_amd.engine.errors import is_tdlib_400
+
+    assert is_tdlib_400(msg) == want, (msg, comment)
