@@ -78,13 +78,19 @@ class WorkItem:
     config: Dict = dataclasses.field(default_factory=dict)
 
     def validate(self) -> None:
-        """messages.go validation: id, url and crawl_id are required."""
+        """messages.go validation: id, url, crawl_id and a SUPPORTED
+        platform are required (messages_test.go:103-170 matrix)."""
         if not self.id:
             raise ValueError("work item missing id")
         if not self.url:
             raise ValueError("work item missing url")
         if not self.crawl_id:
             raise ValueError("work item missing crawl_id")
+        if not self.platform:
+            raise ValueError("work item missing platform")
+        if self.platform not in ("telegram", "youtube"):
+            raise ValueError(
+                f"work item has unsupported platform: {self.platform}")
         if self.depth < 0:
             raise ValueError("work item depth must be >= 0")
 

@@ -417,3 +417,45 @@ def test_is_tdlib_400_reference_vectors(msg, want, comment):
     from crawler_amd.engine.errors import is_tdlib_400
 
     assert is_tdlib_400(msg) == want, (msg, comment)
+
+
+# --- distributed/messages_test.go:103-170, 365-399 ---
+
+def test_work_item_validation_reference_vectors():
+    from crawler_amd.parallel import messages as M
+
+    ok = M.WorkItem(id="test-id", url="https://t.me/testchannel",
+                    platform="telegram", crawl_id="test-crawl")
+    ok.validate()
+    cases = [
+        dict(url="https://t.me/testchannel", platform="telegram",
+             crawl_id="test-crawl"),                       # missing ID
+        dict(id="test-id", platform="telegram",
+             crawl_id="test-crawl"),                       # missing URL
+        dict(id="test-id", url="https://t.me/testchannel",
+             platform="", crawl_id="test-crawl"),          # missing platform
+        dict(id="test-id", url="https://example.com",
+             platform="unsupported", crawl_id="test-crawl"),
+    ]
+    for kw in cases:
+        with pytest.raises(ValueError):
+            M.WorkItem(**kw).validate()
+
+
+def test_trace_id_and_topic_constants_reference_shape():
+    from crawler_amd.parallel import messages as M
+
+    id1, id2 = M.new_trace_id(), M.new_trace_id()
+    assert id1 and id2 and id1 != id2
+    assert len(id1) > 16 and len(id2) > 16  # trace_YYYYMMDDHHMMSS_XXXX...
+    for const in (M.TOPIC_WORK_QUEUE, M.TOPIC_RESULTS,
+                  M.TOPIC_WORKER_STATUS, M.TOPIC_ORCHESTRATOR,
+                  M.MSG_WORKER_STARTED, M.MSG_WORKER_STOPPING,
+                  M.WORKER_IDLE, M.WORKER_BUSY, M.WORKER_ACTIVE,
+                  M.WORKER_OFFLINE):
+        assert const
+    # topic names are the reference's (messages.go:53-58)
+    assert M.TOPIC_WORK_QUEUE == "crawl-work-queue"
+    assert M.TOPIC_RESULTS == "crawl-results"
+    assert M.TOPIC_WORKER_STATUS == "worker-status"
+    assert M.TOPIC_ORCHESTRATOR == "orchestrator-commands"
