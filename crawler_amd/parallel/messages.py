@@ -46,8 +46,9 @@ TOPIC_ORCHESTRATOR = "orchestrator-commands"
 
 
 def new_trace_id() -> str:
-    """messages.go:239-241."""
-    return uuid.uuid4().hex[:16]
+    """trace_YYYYMMDDHHMMSS_XXXXXXXX (messages.go:239-241)."""
+    ts = _dt.datetime.now().strftime("%Y%m%d%H%M%S")
+    return f"trace_{ts}_{uuid.uuid4().hex[:8]}"
 
 
 def _iso(t: Optional[_dt.datetime]) -> Optional[str]:
