@@ -54,16 +54,17 @@ class Worker:
 
     @staticmethod
     def should_retry_error(err: Exception) -> bool:
-        """worker/worker.go:436 classification."""
+        """worker/worker.go:436-456 classification, exactly: permanent
+        for "not found"/"access denied"/"forbidden" (and our typed
+        TDLib 400), retry for connection/timeout/temporary, and RETRY
+        BY DEFAULT for unknown errors (the reference's fall-through)."""
         if isinstance(err, E.TDLib400):
             return False
-        if isinstance(err, (E.FloodWaitRetire, E.PoolExhausted)):
-            return True
-        msg = str(err)
-        _, is_flood = E.parse_flood_wait_secs(msg)
-        if is_flood:
-            return True
-        return "timeout" in msg.lower() or "connection" in msg.lower()
+        msg = str(err).lower()
+        if ("not found" in msg or "access denied" in msg
+                or "forbidden" in msg):
+            return False
+        return True
 
     def _merged_cfg(self, item: M.WorkItem):
         """WorkItemConfig overrides the worker's base config

@@ -459,3 +459,25 @@ def test_trace_id_and_topic_constants_reference_shape():
     assert M.TOPIC_RESULTS == "crawl-results"
     assert M.TOPIC_WORKER_STATUS == "worker-status"
     assert M.TOPIC_ORCHESTRATOR == "orchestrator-commands"
+
+
+# --- worker/worker_test.go TestShouldRetryError matrix ---
+
+RETRY_VECTORS = [
+    ("channel not found", False),
+    ("access denied", False),
+    ("forbidden", False),
+    ("connection failed", True),
+    ("timeout occurred", True),
+    ("temporary failure", True),
+    ("something went wrong", True),  # unknown errors default to retry
+]
+
+
+@pytest.mark.parametrize("msg,want", RETRY_VECTORS,
+                         ids=[v[0].replace(" ", "_")
+                              for v in RETRY_VECTORS])
+def test_should_retry_reference_vectors(msg, want):
+    from crawler_amd.parallel.worker import Worker
+
+    assert Worker.should_retry_error(Exception(msg)) == want
