@@ -481,3 +481,46 @@ def test_should_retry_reference_vectors(msg, want):
     from crawler_amd.parallel.worker import Worker
 
     assert Worker.should_retry_error(Exception(msg)) == want
+
+
+# --- BASELINE.md structural-constants table, pinned in one place ---
+
+def test_structural_constants_match_reference():
+    """Every hard-coded throttle/limit from BASELINE.md's table (the
+    reference's measurable performance envelope), pinned so drift is a
+    test failure, not a doc bug."""
+    from crawler_amd.config import CrawlerConfig, TelegramRateLimitConfig
+    from crawler_amd.engine import errors as E
+    from crawler_amd.engine import validator as V
+    from crawler_amd.engine.htmlvalidator import BODY_CAP
+    from crawler_amd.engine.media import MEDIA_SIZE_CAP_MB
+    from crawler_amd.engine.state import LocalStateManager, RandomWalkStore
+    from crawler_amd.parallel import orchestrator as O
+
+    rl = TelegramRateLimitConfig()
+    assert rl.get_chat_history_rate == 30.0          # utils.go:35
+    assert rl.search_public_chat_rate == 6.0         # utils.go:38
+    assert rl.get_supergroup_info_rate == 20.0       # utils.go:39
+    assert rl.get_message_server_hit_rate == 60.0    # utils.go:43
+    assert rl.get_chat_history_jitter_ms == 500
+    assert rl.search_public_chat_jitter_ms == 1500
+    assert rl.get_supergroup_info_jitter_ms == 800
+
+    cfg = CrawlerConfig(crawl_id="c")
+    assert cfg.concurrency == 1                       # main.go:758
+    assert cfg.max_pages == 108000                    # main.go:776
+    assert cfg.walkback_rate == 15                    # main.go:783
+    assert cfg.combine_trigger_size == 170            # main.go:800
+    assert cfg.combine_hard_cap == 200                # main.go:801
+
+    assert E.FLOOD_WAIT_RETIRE_THRESHOLD_SECS == 300  # runner.go:49
+    assert MEDIA_SIZE_CAP_MB == 150.0                 # tdutils.go:293
+    assert BODY_CAP == 64 * 1024                      # channelvalidator.go:103
+    assert V.BLOCKED_THRESHOLD == 5                   # validator.go:36
+    assert V.PROBE_INTERVAL_S == 300.0                # validator.go:37
+    assert RandomWalkStore.MAX_ATTEMPTS == 3          # daprstate.go poison
+    assert LocalStateManager.export_chunk_size_bytes == 100 * 1024 * 1024
+
+    assert O.DISTRIBUTE_INTERVAL_S == 5.0             # orchestrator.go:163
+    assert O.HEALTH_INTERVAL_S == 30.0                # orchestrator.go:475
+    assert O.DEFAULT_WORKER_TIMEOUT_S == 300.0        # orchestrator.go:498
