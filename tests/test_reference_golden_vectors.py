@@ -524,3 +524,4 @@ def test_structural_constants_match_reference():
     assert O.DISTRIBUTE_INTERVAL_S == 5.0             # orchestrator.go:163
     assert O.HEALTH_INTERVAL_S == 30.0                # orchestrator.go:475
     assert O.DEFAULT_WORKER_TIMEOUT_S == 300.0        # orchestrator.go:498
+    assert O.Orchestrator.MAX_RETRIES == 3            # orchestrator_test.go:308
