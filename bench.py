@@ -123,7 +123,7 @@ def main():
                 f"{time.time() - t_gen:.1f}s)")
         chunk_posts = chunk_ch * posts
     else:
-        from crawler_amd.youtube.batch import build_corpus
+        from crawler_amd.youtube.batch import build_corpus_fast
         from crawler_amd.youtube.synth import SyntheticYouTubeIndex
 
         idx = SyntheticYouTubeIndex(seed=99 + rank,
@@ -131,8 +131,8 @@ def main():
         n_chunks = max(1, args.videos // args.chunk_videos)
         for c in range(n_chunks):
             chunks.append(
-                build_corpus(idx, args.chunk_videos,
-                             crawl_label="yt-bench").to(device)
+                build_corpus_fast(idx, args.chunk_videos,
+                                  crawl_label="yt-bench").to(device)
             )
             log(f"yt chunk {c + 1}/{n_chunks} "
                 f"({args.chunk_videos} videos, "
