@@ -156,11 +156,11 @@ class GpuRandomWalk:
         self.timings["kernels"] += _time.perf_counter() - t0
         t0 = _time.perf_counter()
 
-        # Per-walker link dedup ON DEVICE: sort
-        # (walker, hash) pairs, keep the first of each run, gather ONE
-        # zero-padded name row per unique pair — the host never sees
-        # the ~1M raw link slots, only ~150k unique (walker, name)
-        # rows (5 MB instead of 65 MB D2H, no host np.unique).
+        # Per-walker link dedup ON DEVICE: an
+        # exact byte-lexicographic sort over (walker, name) pairs, keep
+        # the first of each run — the host never sees the ~1M raw link
+        # slots, only ~150k unique name rows (5 MB instead of 65 MB
+        # D2H, no host np.unique), already name-sorted per walker.
         lens_g = res.link_len
         names_g = res.link_name
         w = names_g.shape[2]
