@@ -12,7 +12,8 @@ import sys
 
 CSRC = os.path.dirname(os.path.abspath(__file__)) + "/csrc"
 SOURCES = ["parse_encode.hip", "dedup.hip", "feedgen.hip", "yt_encode.hip",
-           "aggregate.hip", "htmlclass.hip", "sample.hip"]
+           "yt_feedgen.hip", "aggregate.hip", "htmlclass.hip",
+           "sample.hip"]
 OUT = os.path.join(CSRC, "libcrawlhip.so")
 
 

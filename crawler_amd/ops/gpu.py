@@ -101,6 +101,18 @@ def _bind_v2(lib):
 
 
 def _bind_yt(lib):
+    for fn in ("crawl_yt_gen_meta", "crawl_yt_gen_fill",
+               "crawl_yt_gen_channels"):
+        getattr(lib, fn).restype = ctypes.c_int
+    lib.crawl_yt_gen_meta.argtypes = (
+        [ctypes.c_longlong] * 5 + [ctypes.c_void_p] * 11
+        + [ctypes.c_void_p])
+    lib.crawl_yt_gen_fill.argtypes = (
+        [ctypes.c_longlong] * 5 + [ctypes.c_void_p] * 6
+        + [ctypes.c_void_p])
+    lib.crawl_yt_gen_channels.argtypes = (
+        [ctypes.c_longlong] * 3 + [ctypes.c_void_p, ctypes.c_longlong]
+        + [ctypes.c_void_p] * 9 + [ctypes.c_void_p])
     lib.crawl_yt_ptr_count.restype = ctypes.c_int
     lib.crawl_yt_measure.restype = ctypes.c_int
     lib.crawl_yt_measure.argtypes = [

@@ -447,3 +447,139 @@ def build_corpus_fast(index: SyntheticYouTubeIndex, n_videos: int,
         ch_published=t(ch["published"]),
         crawl_label=crawl_label,
     )
+
+
+def build_corpus_device(index: SyntheticYouTubeIndex, n_videos: int,
+                        device, crawl_label: str = "",
+                        i0: int = 0) -> YouTubeBatch:
+    """GPU-resident corpus generation (csrc/yt_feedgen.hip) — the
+    YouTube half of the device synthetic-API engine. Byte-identical to
+    build_corpus_fast (GPU test pins it); generation never leaves HBM."""
+    import ctypes
+
+    from ..feed.synth import _splitmix64
+    from ..ops import gpu as _g
+
+    lib = _g.require_lib()
+    dev = torch.device(device)
+    n = n_videos
+    stream = ctypes.c_void_p(torch.cuda.current_stream().cuda_stream)
+    z64 = lambda m, dt_: torch.empty(m, dtype=dt_, device=dev)
+    vid_pool = z64(n * 11, torch.uint8)
+    published = z64(n, torch.int64)
+    views = z64(n, torch.int64)
+    likes = z64(n, torch.int32)
+    comments = z64(n, torch.int32)
+    duration_s = z64(n, torch.int32)
+    lang = z64(n, torch.int32)
+    n_chan = z64(n, torch.int64)
+    topic = z64(n, torch.int32)
+    tnum = z64(n, torch.int32)
+    desc_len = z64(n, torch.int32)
+    cv = ctypes.c_void_p
+    rc = lib.crawl_yt_gen_meta(
+        index.seed, index.universe, index.base_date, n, i0,
+        cv(vid_pool.data_ptr()), cv(published.data_ptr()),
+        cv(views.data_ptr()), cv(likes.data_ptr()),
+        cv(comments.data_ptr()), cv(duration_s.data_ptr()),
+        cv(lang.data_ptr()), cv(n_chan.data_ptr()),
+        cv(topic.data_ptr()), cv(tnum.data_ptr()),
+        cv(desc_len.data_ptr()), stream)
+    if rc != 0:
+        raise RuntimeError(f"crawl_yt_gen_meta failed: hip {rc}")
+
+    # layout: [vid n*11 | title n*27 | desc var | channel region]
+    vid_off = (torch.arange(n, dtype=torch.int64, device=dev) * 11)
+    title_off = n * 11 + torch.arange(n, dtype=torch.int64,
+                                      device=dev) * 27
+    desc_base = n * 11 + n * 27
+    dl64 = desc_len.to(torch.int64)
+    desc_off = torch.zeros(n, dtype=torch.int64, device=dev)
+    if n > 1:
+        torch.cumsum(dl64[:-1], 0, out=desc_off[1:])
+    desc_off += desc_base
+    desc_total = int(desc_base + dl64.sum().item())
+
+    # channel table in FIRST-APPEARANCE order
+    uniq, inverse = torch.unique(n_chan, sorted=True,
+                                 return_inverse=True)
+    K = uniq.numel()
+    first = torch.full((K,), n, dtype=torch.int64, device=dev)
+    first.scatter_reduce_(0, inverse,
+                          torch.arange(n, dtype=torch.int64, device=dev),
+                          reduce="amin")
+    order = torch.argsort(first)
+    rank = torch.empty_like(order)
+    rank[order] = torch.arange(K, dtype=torch.int64, device=dev)
+    channel_idx = rank[inverse].to(torch.int32)
+    chan_ns = uniq[order]
+
+    # channel string lengths (digits via comparisons; hc on host numpy
+    # for the country flag — torch lacks uint64 bit ops)
+    digs = torch.ones(K, dtype=torch.int64, device=dev)
+    p10 = 10
+    for _ in range(18):
+        digs += (chan_ns >= p10).to(torch.int64)
+        p10 *= 10
+    title_len_ch = 21 + digs
+    desc_len_ch = 8 + digs + 36
+    ns_host = chan_ns.cpu().numpy().astype(np.uint64)
+    a = np.uint64(index.seed ^ 0xC0FFEE)
+    for chf in "chanmeta":
+        a = _splitmix64(np.asarray(a ^ np.uint64(ord(chf))))
+    hc = _splitmix64(np.full(K, a, dtype=np.uint64) ^ ns_host)
+    country_len_np = np.where(hc % np.uint64(3) == 0, 2, 0)
+    country_len = torch.from_numpy(
+        country_len_np.astype(np.int64)).to(dev)
+    block = 24 + title_len_ch + desc_len_ch + country_len
+    ch_base = torch.zeros(K, dtype=torch.int64, device=dev)
+    if K > 1:
+        torch.cumsum(block[:-1], 0, out=ch_base[1:])
+    ch_base += desc_total
+    id_off = ch_base
+    t_off = id_off + 24
+    d_off = t_off + title_len_ch
+    c_off = d_off + desc_len_ch
+    total = int(desc_total + block.sum().item())
+
+    pool = torch.zeros(total, dtype=torch.uint8, device=dev)
+    rc = lib.crawl_yt_gen_fill(
+        index.seed, index.universe, index.base_date, n, i0,
+        cv(vid_pool.data_ptr()), cv(topic.data_ptr()),
+        cv(tnum.data_ptr()), cv(title_off.data_ptr()),
+        cv(desc_off.data_ptr()), cv(pool.data_ptr()), stream)
+    if rc != 0:
+        raise RuntimeError(f"crawl_yt_gen_fill failed: hip {rc}")
+    subs = z64(K, torch.int64)
+    videos_t = z64(K, torch.int32)
+    ch_views = z64(K, torch.int64)
+    ch_published = z64(K, torch.int64)
+    rc = lib.crawl_yt_gen_channels(
+        index.seed, index.universe, index.base_date,
+        cv(chan_ns.data_ptr()), K, cv(id_off.data_ptr()),
+        cv(t_off.data_ptr()), cv(d_off.data_ptr()),
+        cv(c_off.data_ptr()), cv(subs.data_ptr()),
+        cv(videos_t.data_ptr()), cv(ch_views.data_ptr()),
+        cv(ch_published.data_ptr()), cv(pool.data_ptr()), stream)
+    if rc != 0:
+        raise RuntimeError(f"crawl_yt_gen_channels failed: hip {rc}")
+
+    return YouTubeBatch(
+        n=n, vid_off=vid_off.to(torch.int32), channel_idx=channel_idx,
+        published=published, views=views, likes=likes,
+        comments=comments, duration_s=duration_s, lang=lang,
+        title_off=title_off, title_len=torch.full(
+            (n,), 27, dtype=torch.int32, device=dev),
+        desc_off=desc_off, desc_len=desc_len, pool=pool,
+        n_channels=int(K),
+        ch_id_off=id_off.to(torch.int32),
+        ch_title_off=t_off.to(torch.int32),
+        ch_title_len=title_len_ch.to(torch.int32),
+        ch_desc_off=d_off.to(torch.int32),
+        ch_desc_len=desc_len_ch.to(torch.int32),
+        ch_subs=subs, ch_videos=videos_t, ch_views=ch_views,
+        ch_country_off=c_off.to(torch.int32),
+        ch_country_len=country_len.to(torch.int32),
+        ch_published=ch_published,
+        crawl_label=crawl_label,
+    )
