@@ -123,16 +123,18 @@ def main():
                 f"{time.time() - t_gen:.1f}s)")
         chunk_posts = chunk_ch * posts
     else:
-        from crawler_amd.youtube.batch import build_corpus_fast
+        from crawler_amd.youtube.batch import build_corpus_device
         from crawler_amd.youtube.synth import SyntheticYouTubeIndex
 
         idx = SyntheticYouTubeIndex(seed=99 + rank,
                                     universe_channels=1_000_000)
         n_chunks = max(1, args.videos // args.chunk_videos)
         for c in range(n_chunks):
+            # device-side generation (yt_feedgen.hip) — the corpus
+            # never leaves HBM, like the telegram feedgen path
             chunks.append(
-                build_corpus_fast(idx, args.chunk_videos,
-                                  crawl_label="yt-bench").to(device)
+                build_corpus_device(idx, args.chunk_videos, device,
+                                    crawl_label="yt-bench")
             )
             log(f"yt chunk {c + 1}/{n_chunks} "
                 f"({args.chunk_videos} videos, "
