@@ -303,3 +303,18 @@ def test_plain_writer_forced_matches_staged(gpu_mod, feed,
     monkeypatch.delenv("CRAWL_NO_STAGED")
     assert bytes(res_staged.out.cpu().numpy()) == bytes(
         res_plain.out.cpu().numpy())
+
+
+def test_thousand_comment_posts_bytes_identical(gpu_mod):
+    """The full --max-comments 1000 shape (~57 KB lines): threads up to
+    1000 comments, byte-identical to the oracle (the comment tail reads
+    global memory — unaffected by the staged writer's LDS budget)."""
+    heavy = SyntheticFeed(FeedConfig(
+        seed=13, universe=2000, comment_rate=1.0,
+        max_comments_per_post=1000,
+    ))
+    batch = heavy.build_batch(np.arange(3), posts_per_channel=8)
+    max_c = int(batch.meta["com_cnt"].max())
+    assert max_c > 500, f"corpus too light: max {max_c} comments"
+    golden_lines, _, res = _roundtrip(gpu_mod, batch)
+    assert bytes(res.out.cpu().numpy()) == b"".join(golden_lines)
