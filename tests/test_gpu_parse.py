@@ -318,3 +318,14 @@ def test_thousand_comment_posts_bytes_identical(gpu_mod):
     assert max_c > 500, f"corpus too light: max {max_c} comments"
     golden_lines, _, res = _roundtrip(gpu_mod, batch)
     assert bytes(res.out.cpu().numpy()) == b"".join(golden_lines)
+
+
+def test_single_pass_variant_bytes_identical(gpu_mod, feed):
+    """The scratch+compact single-pass path (kept for experimentation)
+    stays byte-exact through emitter refactors."""
+    batch = feed.build_batch(np.arange(6), posts_per_channel=100)
+    golden_lines, _ = encode_batch(batch, now=NOW)
+    res = gpu_mod.parse_encode(batch.to("cuda:0"), now=NOW,
+                               single_pass=True)
+    torch.cuda.synchronize()
+    assert bytes(res.out.cpu().numpy()) == b"".join(golden_lines)
