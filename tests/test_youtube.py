@@ -227,3 +227,35 @@ def test_build_corpus_fast_matches_slow():
     assert encode_yt_batch(slow, now=now) == encode_yt_batch(fast,
                                                              now=now)
     assert fast.n_channels == slow.n_channels
+
+
+def test_youtube_runner_applies_date_window(tmp_path):
+    """run_youtube filters fetched videos through the
+    CalculateDateFilters window (dapr/standalone.go:1119-1147 passes
+    the range into the crawl job)."""
+    import datetime as dt
+
+    from crawler_amd.config import CrawlerConfig
+    from crawler_amd.engine import LocalStateManager
+    from crawler_amd.youtube.runner import run_youtube
+
+    UTC = dt.timezone.utc
+    base_cfg = dict(crawl_id="ytd", storage_root=str(tmp_path),
+                    platform="youtube", sampling_method="channel",
+                    min_users=1, max_posts=50)
+    cfg_all = CrawlerConfig(**base_cfg)
+    sm1 = LocalStateManager(cfg_all)
+    stats_all = run_youtube(cfg_all, ["3"], sm=sm1)
+
+    # a window far in the past excludes everything
+    cfg_none = CrawlerConfig(
+        **{**base_cfg, "crawl_id": "ytd2"},
+        date_between_min=dt.datetime(1990, 1, 1, tzinfo=UTC),
+        date_between_max=dt.datetime(1991, 1, 1, tzinfo=UTC))
+    sm2 = LocalStateManager(cfg_none)
+    stats_none = run_youtube(cfg_none, ["3"], sm=sm2)
+    assert stats_all["posts"] > 0
+    assert stats_none["posts"] == 0
+    # like the reference (the range rides in the crawl job), the video
+    # count reflects the filtered window
+    assert stats_none["videos"] == 0
