@@ -225,6 +225,11 @@ class GpuRandomWalk:
         h_g = res.link_hash[mi_g, si_g]
         hs_sel_g = h_g[sel]
         u_rows_g = all_rows[sel]
+        # validate the deduped rows ON DEVICE (oracle-equal torch port
+        # of FilterUsername + universe decode); the host miss path just
+        # indexes these instead of re-scanning bytes
+        ok_g, cid_ok_g, _ = vecvalidate.validate_names_torch(
+            u_rows_g, self.feed.cfg.universe)
 
         # D2H through a reusable pinned ring (2-deep: this hop's disk
         # writes overlap the next hop's kernels via the native sink)
@@ -245,6 +250,8 @@ class GpuRandomWalk:
         u_w = ws_g[keep].to(torch.int32).cpu().numpy()
         u_h = hs_sel_g.cpu().numpy()
         u_rows = u_rows_g.cpu().numpy()
+        ok_all = ok_g.cpu().numpy()
+        cid_ok_all = cid_ok_g.cpu().numpy()
         torch.cuda.synchronize()
         self.timings["d2h"] += _time.perf_counter() - t0
         t0 = _time.perf_counter()
@@ -294,8 +301,8 @@ class GpuRandomWalk:
         if miss.any():
             mh, m_first = np.unique(u_h[miss], return_index=True)
             m_rows = un[miss][m_first]
-            ok_v, cid_ok_v, _cids = vecvalidate.validate_names(
-                m_rows, self.feed.cfg.universe)
+            ok_v = ok_all[miss][m_first]
+            cid_ok_v = cid_ok_all[miss][m_first]
             adm_v = ok_v & cid_ok_v
             m_names = np.empty(len(mh), dtype=object)
             adm_i = np.nonzero(adm_v)[0]

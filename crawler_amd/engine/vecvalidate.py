@@ -68,6 +68,58 @@ def validate_names(unames: np.ndarray, universe: int):
     return ok, cid_ok, val
 
 
+def validate_names_torch(rows, universe: int):
+    """validate_names on a torch uint8 tensor [N, w] (any device).
+
+    Same decisions as validate_names (pinned by
+    tests/test_vecvalidate.py::test_torch_matches_numpy); used by the
+    GPU random-walk to validate the deduped name rows ON DEVICE, so
+    the host miss path only indexes precomputed results instead of
+    re-scanning ~150k rows of bytes per hop."""
+    import torch
+
+    n, w = rows.shape
+    dev = rows.device
+    if n == 0:
+        z = torch.zeros(0, dtype=torch.bool, device=dev)
+        return z, z, torch.zeros(0, dtype=torch.int64, device=dev)
+    U = rows.to(torch.int64)
+    L = (U != 0).sum(1)
+    first = U[:, 0]
+    is_letter = (((first >= 65) & (first <= 90))
+                 | ((first >= 97) & (first <= 122)))
+    valid_lut = torch.from_numpy(_VALID.copy()).to(dev)
+    colm = torch.arange(w, device=dev)[None, :] < L[:, None]
+    chars_ok = (valid_lut[U] | ~colm).all(1)
+    rows_i = torch.arange(n, device=dev)
+    lastpos = (L - 1).clamp(min=0)
+    no_trail_us = U[rows_i, lastpos] != 95
+
+    def low(c):
+        return torch.where((c >= 65) & (c <= 90), c + 32, c)
+
+    has3 = L >= 3
+    b1 = low(U[rows_i, (L - 3).clamp(min=0)])
+    b2 = low(U[rows_i, (L - 2).clamp(min=0)])
+    b3 = low(U[rows_i, lastpos])
+    bot = has3 & (b1 == 98) & (b2 == 111) & (b3 == 116)
+    ok = ((L >= 5) & (L <= 32) & is_letter & chars_ok
+          & no_trail_us & ~bot)
+
+    dig = (U[:, 1:] >= 48) & (U[:, 1:] <= 57)
+    digcols = torch.arange(1, w, device=dev)[None, :] < L[:, None]
+    all_dig = (dig | ~digcols).all(1)
+    nd = L - 1
+    cand = (first == 99) & all_dig & (nd >= 1) & (nd <= 12)
+    val = torch.zeros(n, dtype=torch.int64, device=dev)
+    for j in range(1, min(13, w)):
+        has = j < L
+        d = U[:, j] - 48
+        val = torch.where(has & cand, val * 10 + d, val)
+    cid_ok = cand & (val < universe)
+    return ok, cid_ok, val
+
+
 def fnv1a64_rows(rows: np.ndarray) -> np.ndarray:
     """Vectorized FNV-1a 64 over NUL-padded uint8[N, W] name rows —
     matches the device hash (csrc/common.h fnv1a64) and the scalar

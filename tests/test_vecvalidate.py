@@ -83,3 +83,29 @@ def test_fnv1a64_rows_matches_scalar_oracle():
         want = fnv1a64(nm.encode())
         # device/oracle hash is uint64; rows variant returns int64 view
         assert int(got.view(np.uint64)[i]) == want, nm
+
+
+def test_torch_matches_numpy():
+    """validate_names_torch (the device path the GPU random-walk uses)
+    decides identically to the numpy oracle, on the curated cases and
+    on random ASCII rows (CPU torch here; same code runs on cuda)."""
+    import torch
+
+    from crawler_amd.engine.vecvalidate import validate_names_torch
+
+    rng = np.random.default_rng(5)
+    rand = []
+    for _ in range(4000):
+        n = rng.integers(0, 33)
+        rand.append(bytes(rng.integers(32, 127, n).astype(np.uint8)))
+    for universe in (1, 1000, 10**6, 10**12):
+        for pool in (CASES, rand):
+            arr = np.array(pool, dtype="S40")
+            ok_n, cid_n, val_n = validate_names(arr, universe)
+            rows = torch.from_numpy(
+                np.ascontiguousarray(arr).view(np.uint8).reshape(
+                    len(arr), 40))
+            ok_t, cid_t, val_t = validate_names_torch(rows, universe)
+            assert (ok_t.numpy() == ok_n).all()
+            assert (cid_t.numpy() == cid_n).all()
+            assert (val_t.numpy()[cid_n] == val_n[cid_n]).all()
