@@ -169,6 +169,37 @@ def test_seed_channel_lifecycle():
     assert rw.load_seed_channels() == []  # fresh invalidation filtered
 
 
+def test_seed_channel_bulk_lazy_rows():
+    """Bulk-admitted names are stored as lazy sentinels; every accessor
+    sees full-row semantics (materialize-on-touch never escapes)."""
+    rw = RandomWalkStore()
+    rw.upsert_seed_channels_bulk(["a_chan", "b_chan", "c_chan"])
+    # reads on fresh rows behave like default rows
+    assert rw.get_channel_last_crawled("a_chan") is None
+    rows = rw.load_seed_channels()
+    assert {r["username"] for r in rows} == {"a_chan", "b_chan", "c_chan"}
+    assert all(r["chat_id"] == 0 and r["last_crawled_at"] is None
+               and r["invalidated_at"] is None for r in rows)
+    # mutations materialize
+    rw.mark_channel_crawled("a_chan", 42)
+    assert rw.get_channel_last_crawled("a_chan") is not None
+    rw.mark_seed_channel_invalid("b_chan")
+    left = {r["username"] for r in rw.load_seed_channels()}
+    assert left == {"a_chan", "c_chan"}  # fresh invalidation filtered
+    # invalidating an unknown name never creates a row
+    rw.mark_seed_channel_invalid("ghost_chan")
+    assert "ghost_chan" not in {r["username"]
+                                for r in rw.load_seed_channels()}
+    # re-upsert of an existing materialized row keeps its state
+    rw.upsert_seed_channels_bulk(["a_chan"])
+    assert rw.get_channel_last_crawled("a_chan") is not None
+    # chat_id upsert on a fresh sentinel materializes with the id
+    rw.upsert_seed_channel("c_chan", 7)
+    row = [r for r in rw.load_seed_channels()
+           if r["username"] == "c_chan"][0]
+    assert row["chat_id"] == 7
+
+
 def test_invalid_channel_ttl():
     rw = RandomWalkStore()
     rw.mark_invalid_channel("badchan")
