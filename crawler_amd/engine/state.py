@@ -1029,6 +1029,10 @@ class RandomWalkStore:
     def mark_invalid_channel(self, username: str) -> None:
         with self._lock:
             self.invalid_channels[username] = _now()
+            # version stamp for read-side caches (the GPU walk keeps a
+            # sorted snapshot keyed on this; direct dict edits in tests
+            # don't go through here and don't need the snapshot)
+            self.invalid_version = getattr(self, "invalid_version", 0) + 1
 
     def is_invalid_channel(self, username: str, ttl_days: int = 30) -> bool:
         with self._lock:

@@ -102,3 +102,36 @@ def test_vc_main_spill_preserves_entries():
     miss = g._vc_lookup(q, names, adm)
     assert not miss.any()
     assert all(names[i] == total[int(h)][0] for i, h in enumerate(q))
+
+
+def test_inv_snapshot_versioning():
+    """The invalid-name snapshot rebuilds only on store-version moves
+    (or TTL expiry) and answers membership exactly."""
+    import datetime as dt
+
+    from crawler_amd.engine.state import RandomWalkStore
+
+    g = GpuRandomWalk.__new__(GpuRandomWalk)
+    g.rw = RandomWalkStore()
+    now = dt.datetime.now(dt.timezone.utc)
+    a0 = g._inv_snapshot(now, 32)
+    assert a0.size == 0
+    g.rw.mark_invalid_channel("badchan")
+    g.rw.mark_invalid_channel("worse_chan")
+    a1 = g._inv_snapshot(now, 32)
+    assert sorted(a1.tolist()) == [b"badchan", b"worse_chan"]
+    # same version -> cached object reused
+    assert g._inv_snapshot(now, 32) is a1
+    # membership helper
+    q = np.array([b"badchan", b"goodchan", b"worse_chan"], dtype="S32")
+    assert GpuRandomWalk._in_sorted(a1, q).tolist() == [True, False, True]
+    # expired entries drop on rebuild after the TTL boundary
+    g.rw.invalid_channels["badchan"] = now - dt.timedelta(days=31)
+    g.rw.mark_invalid_channel("third")  # bump version
+    a2 = g._inv_snapshot(now, 32)
+    assert b"badchan" not in a2.tolist()
+    assert {b"worse_chan", b"third"} <= set(a2.tolist())
+    # version unchanged + TTL not crossed -> still cached even though
+    # the dict was edited behind the store's back (documented contract)
+    g.rw.invalid_channels["sneaky"] = now
+    assert g._inv_snapshot(now, 32) is a2
