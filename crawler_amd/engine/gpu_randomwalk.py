@@ -60,6 +60,7 @@ class GpuRandomWalk:
                      np.zeros(0, dtype=bool)] for _ in range(2)]
         import collections
         self.timings = collections.defaultdict(float)  # phase seconds
+        self.hop_log = []  # (pages, posts, seconds) per hop
 
     def seed(self, urls: List[str]):
         import uuid
@@ -393,7 +394,12 @@ class GpuRandomWalk:
             )
             if not pages:
                 break
-            self._hop(pages, now)
+            import time as _time
+            t0 = _time.perf_counter()
+            p0 = self.stats["posts"]
+            n = self._hop(pages, now)
+            self.hop_log.append((n, self.stats["posts"] - p0,
+                                 round(_time.perf_counter() - t0, 4)))
         for slot in (0, 1):
             self.sm.wait_post_write(self._spill_tickets[slot])
             self._spill_tickets[slot] = None

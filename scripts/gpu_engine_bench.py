@@ -116,6 +116,11 @@ def run_randomwalk(args):
         "edges": stats["edges"],
         "walkback_exhausted": stats["walkback_exhausted"],
         "phase_s": {k: round(v, 3) for k, v in eng.timings.items()},
+        "per_hop": [
+            {"pages": p, "posts": q,
+             "posts_per_s": round(q / t, 1) if t else None, "s": t}
+            for p, q, t in eng.hop_log
+        ],
     }))
 
 
