@@ -117,9 +117,10 @@ def run_randomwalk(args):
         "walkback_exhausted": stats["walkback_exhausted"],
         "phase_s": {k: round(v, 3) for k, v in eng.timings.items()},
         "per_hop": [
-            {"pages": p, "posts": q,
-             "posts_per_s": round(q / t, 1) if t else None, "s": t}
-            for p, q, t in eng.hop_log
+            {"pages": h[0], "posts": h[1],
+             "posts_per_s": round(h[1] / h[2], 1) if h[2] else None,
+             "s": h[2], "phases": (h[3] if len(h) > 3 else {})}
+            for h in eng.hop_log
         ],
     }))
 

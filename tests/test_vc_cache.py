@@ -1,18 +1,16 @@
 """Unit tests for the random-walk cross-hop validation cache
-(GpuRandomWalk._vc_*): pure-numpy two-level sorted cache with
-sorted-query probing and O(n) disjoint-key merges. CPU-only — the
-methods never touch the device, so the object is built via __new__."""
+(GpuRandomWalk._vc_*): a pure-numpy open-addressing hash table
+(vectorized linear probe, amortized growth). CPU-only — the methods
+never touch the device, so the object is built via __new__."""
 import numpy as np
 import pytest
 
 from crawler_amd.engine.gpu_randomwalk import GpuRandomWalk
 
 
-def make_cache():
+def make_cache(slots=1 << 10):
     g = GpuRandomWalk.__new__(GpuRandomWalk)
-    g._vc = [[np.zeros(0, dtype=np.int64),
-              np.zeros(0, dtype=object),
-              np.zeros(0, dtype=bool)] for _ in range(2)]
+    g._vc_init(slots)
     return g
 
 
@@ -50,40 +48,50 @@ def test_vc_cache_matches_dict_reference(seed):
             g._vc_insert(mh, mn, ma)
             for h, n, a in zip(mh, mn, ma):
                 ref[h] = (n, a)
-    # cache levels stay sorted and disjoint
-    for lvl in g._vc:
-        assert (np.diff(lvl[0]) > 0).all()
-    inter = np.intersect1d(g._vc[0][0], g._vc[1][0])
-    assert len(inter) == 0
+    # table bookkeeping: entry count matches the reference dict
+    assert g._vc_n == len(ref)
+    assert (g._vc_keys != g._VC_EMPTY).sum() == len(ref)
 
 
-def test_merge_sorted_interleave():
-    a = [np.array([1, 4, 9], dtype=np.int64),
-         np.array(["a", "d", "i"], dtype=object),
-         np.array([True, False, True])]
-    b = [np.array([2, 3, 10], dtype=np.int64),
-         np.array(["b", "c", "j"], dtype=object),
-         np.array([False, True, False])]
-    m = GpuRandomWalk._merge_sorted(a, b)
-    assert m[0].tolist() == [1, 2, 3, 4, 9, 10]
-    assert m[1].tolist() == ["a", "b", "c", "d", "i", "j"]
-    assert m[2].tolist() == [True, False, True, False, True, False]
+def test_vc_collision_chains():
+    """Keys engineered to land on the same initial slot still resolve
+    (linear probing), both within one insert batch and across
+    batches."""
+    g = make_cache(slots=1 << 4)
+    # craft keys with identical mixed slot: brute-force search
+    base_slot = None
+    ks = []
+    k = 0
+    while len(ks) < 6:
+        s = g._vc_slots_of(np.array([k], dtype=np.int64))[0]
+        if base_slot is None:
+            base_slot = s
+        if s == base_slot:
+            ks.append(k)
+        k += 1
+    ks = np.array(ks, dtype=np.int64)
+    names = np.array([f"n{v}" for v in ks], dtype=object)
+    adm = np.ones(len(ks), dtype=bool)
+    g._vc_insert(ks[:3], names[:3], adm[:3])   # same-slot batch
+    g._vc_insert(ks[3:], names[3:], adm[3:])   # cross-batch chain
+    out_n = np.empty(len(ks), dtype=object)
+    out_a = np.zeros(len(ks), dtype=bool)
+    miss = g._vc_lookup(ks, out_n, out_a)
+    assert not miss.any()
+    assert out_n.tolist() == names.tolist()
+    # absent key that hashes into the chain is still reported missing
+    probe = k
+    while g._vc_slots_of(np.array([probe], dtype=np.int64))[0] != base_slot:
+        probe += 1
+    miss2 = g._vc_lookup(np.array([probe], dtype=np.int64),
+                         np.empty(1, dtype=object),
+                         np.zeros(1, dtype=bool))
+    assert miss2.all()
 
 
-def test_merge_sorted_empty_sides():
-    e = [np.zeros(0, dtype=np.int64), np.zeros(0, dtype=object),
-         np.zeros(0, dtype=bool)]
-    b = [np.array([5], dtype=np.int64), np.array(["x"], dtype=object),
-         np.array([True])]
-    assert GpuRandomWalk._merge_sorted(e, b)[0].tolist() == [5]
-    assert GpuRandomWalk._merge_sorted(b, e)[0].tolist() == [5]
-    assert GpuRandomWalk._merge_sorted(e, e)[0].size == 0
-
-
-def test_vc_main_spill_preserves_entries():
-    g = make_cache()
-    # force several pending->main spills past the 1<<18 floor by
-    # shrinking the threshold: insert enough to trigger len>262144
+def test_vc_growth_preserves_entries():
+    g = make_cache(slots=1 << 10)
+    # force several table doublings
     total = {}
     rng = np.random.default_rng(7)
     for _ in range(3):
@@ -91,8 +99,8 @@ def test_vc_main_spill_preserves_entries():
         mn = np.array([str(h) for h in mh], dtype=object)
         ma = (mh % 2 == 0)
         # drop keys already cached (insert contract: misses only)
-        seen = np.concatenate([g._vc[0][0], g._vc[1][0]])
-        fresh = ~np.isin(mh, seen)
+        fresh = g._vc_lookup(mh, np.empty(len(mh), dtype=object),
+                             np.zeros(len(mh), dtype=bool))
         g._vc_insert(mh[fresh], mn[fresh], ma[fresh])
         for h, n, a in zip(mh[fresh], mn[fresh], ma[fresh]):
             total[h] = (n, a)
