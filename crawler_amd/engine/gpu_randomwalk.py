@@ -211,9 +211,19 @@ class GpuRandomWalk:
 
     def _hop(self, pages: List[Page],
              now: Optional[_dt.datetime] = None) -> int:
-        """One batched hop over up to W frontier pages."""
-        now = now or _dt.datetime.now(_dt.timezone.utc)
-        # 400s: invalid usernames get the replacement machinery
+        """One batched hop over up to W frontier pages (serial form:
+        claim -> GPU stage -> host tail)."""
+        live = self._claim_400(pages)
+        if not live:
+            return 0
+        slot = self._ring_slot()
+        payload = self._hop_gpu(live, slot, now)
+        self._hop_host(live, payload)
+        return len(live)
+
+    def _claim_400(self, pages: List[Page]) -> List[Page]:
+        """400s: invalid usernames get the replacement machinery
+        (host-side; consumes rng, so it stays on the main thread)."""
         live: List[Page] = []
         for p in pages:
             if self._cid_of(p.url) is None:
@@ -227,9 +237,26 @@ class GpuRandomWalk:
                 self.rw.delete_pages([p.id])
             else:
                 live.append(p)
-        if not live:
-            return 0
+        return live
 
+    def _ring_slot(self) -> int:
+        """Claim the next pinned-ring slot and retire its old spill
+        ticket. MAIN THREAD ONLY: the native sink's ticket wait also
+        evicts surplus fds and must not race a concurrent submit."""
+        slot = self._hop_idx % 2
+        self._hop_idx += 1
+        self.sm.wait_post_write(self._spill_tickets[slot])
+        self._spill_tickets[slot] = None
+        return slot
+
+    def _hop_gpu(self, live: List[Page], slot: int,
+                 now: Optional[_dt.datetime] = None) -> dict:
+        """Device stage: feedgen -> parse/encode -> seen claim ->
+        device dedup + validation -> host pulls + async JSONL D2H.
+        Returns the host-side payload for _hop_host. Safe to run on a
+        worker thread (GPU + numpy only; no rng, no store writes — the
+        pinned-ring slot is claimed by the caller)."""
+        now = now or _dt.datetime.now(_dt.timezone.utc)
         import time as _time
         t0 = _time.perf_counter()
         cids = np.array([self._cid_of(p.url) for p in live],
@@ -298,11 +325,20 @@ class GpuRandomWalk:
             u_rows_g, self.feed.cfg.universe)
 
         # D2H through a reusable pinned ring (2-deep: this hop's disk
-        # writes overlap the next hop's kernels via the native sink)
-        slot = self._hop_idx % 2
-        self._hop_idx += 1
-        self.sm.wait_post_write(self._spill_tickets[slot])
-        self._spill_tickets[slot] = None
+        # writes overlap the next hop's kernels via the native sink).
+        # Small control tensors pull FIRST (each .cpu() syncs the
+        # stream up to its producer); the big JSONL copy is issued
+        # LAST, non-blocking, behind a recorded event — the host tail
+        # waits on the EVENT, not a device-wide synchronize, so a
+        # pipelined run never serializes against the other pool's
+        # in-flight kernels.
+        line_off = res.line_off.cpu().numpy()
+        line_len = res.line_len.cpu().numpy()
+        u_w = ws_g[keep].to(torch.int32).cpu().numpy()
+        u_h = hs_sel_g.cpu().numpy()
+        u_rows = u_rows_g.cpu().numpy()
+        ok_all = ok_g.cpu().numpy()
+        cid_ok_all = cid_ok_g.cpu().numpy()
         need = int(res.out.numel())
         ring = self._pin_ring
         if ring[slot] is None or ring[slot].numel() < need:
@@ -311,14 +347,33 @@ class GpuRandomWalk:
                                      pin_memory=True)
         out_host = ring[slot][:need]
         out_host.copy_(res.out, non_blocking=True)
-        line_off = res.line_off.cpu().numpy()
-        line_len = res.line_len.cpu().numpy()
-        u_w = ws_g[keep].to(torch.int32).cpu().numpy()
-        u_h = hs_sel_g.cpu().numpy()
-        u_rows = u_rows_g.cpu().numpy()
-        ok_all = ok_g.cpu().numpy()
-        cid_ok_all = cid_ok_g.cpu().numpy()
-        torch.cuda.synchronize()
+        ev = torch.cuda.Event()
+        ev.record()
+        self.timings["d2h"] += _time.perf_counter() - t0
+        return {"slot": slot, "out_host": out_host, "ev": ev,
+                "res_out": res.out, "line_off": line_off,
+                "line_len": line_len, "u_w": u_w, "u_h": u_h,
+                "u_rows": u_rows, "ok_all": ok_all,
+                "cid_ok_all": cid_ok_all, "w": w, "now": now}
+
+    def _hop_host(self, live: List[Page], payload: dict) -> None:
+        """Host tail: spill, validation cache, admission, walk
+        decisions. MAIN THREAD ONLY (rng + store writes)."""
+        import time as _time
+        t0 = _time.perf_counter()
+        slot = payload["slot"]
+        out_host = payload["out_host"]
+        line_off = payload["line_off"]
+        line_len = payload["line_len"]
+        u_w = payload["u_w"]
+        u_h = payload["u_h"]
+        u_rows = payload["u_rows"]
+        ok_all = payload["ok_all"]
+        cid_ok_all = payload["cid_ok_all"]
+        w = payload["w"]
+        P = self.ppc
+        payload["ev"].synchronize()   # JSONL bytes landed in the ring
+        payload["res_out"] = None
         self.timings["d2h"] += _time.perf_counter() - t0
         t0 = _time.perf_counter()
 
@@ -422,9 +477,30 @@ class GpuRandomWalk:
             self.stats["pages"] += 1
         self.timings["walk-tail"] += _time.perf_counter() - t0
         self.stats["edges"] = self.rw.edge_count()
-        return len(live)
 
-    def run(self, max_pages: int, now: Optional[_dt.datetime] = None) -> dict:
+    def run(self, max_pages: int, now: Optional[_dt.datetime] = None,
+            pipelined: bool = False) -> dict:
+        """Walk until max_pages pages complete (or the frontier dries
+        up). pipelined=True runs two half-size walker pools in a
+        software pipeline: pool B's device stage (launched on a worker
+        thread — GPU + numpy only) overlaps pool A's host tail, hiding
+        the ~65 ms device phase under the ~135 ms host phase. The walk
+        REMAINS deterministic for a fixed seed, but consumes the rng in
+        a different (interleaved) order than the serial form, so it is
+        opt-in; both forms have identical per-page semantics."""
+        if pipelined:
+            self._run_pipelined(max_pages, now)
+        else:
+            self._run_serial(max_pages, now)
+        for slot in (0, 1):
+            self.sm.wait_post_write(self._spill_tickets[slot])
+            self._spill_tickets[slot] = None
+        self.sm.drain_post_writes()
+        self.sm.save_state()
+        self.sm.close()
+        return dict(self.stats)
+
+    def _run_serial(self, max_pages: int, now) -> None:
         while self.stats["pages"] < max_pages:
             pages = self.rw.get_pages(
                 min(self.walkers, max_pages - self.stats["pages"])
@@ -441,10 +517,41 @@ class GpuRandomWalk:
                                  {k: round(v - ph0.get(k, 0.0), 4)
                                   for k, v in self.timings.items()
                                   if v - ph0.get(k, 0.0) > 1e-4}))
-        for slot in (0, 1):
-            self.sm.wait_post_write(self._spill_tickets[slot])
-            self._spill_tickets[slot] = None
-        self.sm.drain_post_writes()
-        self.sm.save_state()
-        self.sm.close()
-        return dict(self.stats)
+
+    def _run_pipelined(self, max_pages: int, now) -> None:
+        import concurrent.futures as cf
+        import time as _time
+
+        half = max(1, self.walkers // 2)
+        pending = None  # (live_pages, future)
+        with cf.ThreadPoolExecutor(max_workers=1) as ex:
+            while True:
+                t0 = _time.perf_counter()
+                p0 = self.stats["posts"]
+                inflight = pending[0] if pending else []
+                budget = (max_pages - self.stats["pages"]
+                          - len(inflight))
+                pages: List[Page] = []
+                if budget > 0:
+                    want = min(half, budget)
+                    excl = {p.id for p in inflight}
+                    raw = self.rw.get_pages(want + len(excl))
+                    pages = [p for p in raw if p.id not in excl][:want]
+                live = self._claim_400(pages) if pages else []
+                fut = None
+                if live:
+                    slot = self._ring_slot()
+                    fut = ex.submit(self._hop_gpu, live, slot, now)
+                processed = pending is not None
+                if processed:
+                    plive, pfut = pending
+                    self._hop_host(plive, pfut.result())
+                    self.hop_log.append(
+                        (len(plive), self.stats["posts"] - p0,
+                         round(_time.perf_counter() - t0, 4), {}))
+                pending = (live, fut) if fut is not None else None
+                # stop only when nothing was claimed AND nothing was
+                # processed this iteration (a processed tail refills
+                # the frontier, so re-check before giving up)
+                if pending is None and not pages and not processed:
+                    break

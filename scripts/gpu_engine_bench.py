@@ -40,6 +40,8 @@ def main():
     ap.add_argument("--walkers", type=int, default=512,
                     help="randomwalk: concurrent walker chains per hop")
     ap.add_argument("--walkback-rate", type=int, default=15)
+    ap.add_argument("--pipelined", action="store_true",
+                    help="two half-pools: device stage overlaps host tail")
     args = ap.parse_args()
 
     if args.mode == "randomwalk":
@@ -104,7 +106,8 @@ def run_randomwalk(args):
                         walkers=args.walkers, rng=random.Random(11))
     eng.seed([feed.username_of(i) for i in range(args.walkers)])
     t0 = time.perf_counter()
-    stats = eng.run(max_pages=args.max_pages)
+    stats = eng.run(max_pages=args.max_pages,
+                    pipelined=bool(args.pipelined))
     elapsed = time.perf_counter() - t0
     print(json.dumps({
         "metric": "random-walk posts/sec (batched hops incl. disk)",

@@ -67,3 +67,26 @@ def test_gpu_walkback_rate_100(tmp_path):
     eng.run(max_pages=6, now=NOW)
     wb = [e for e in rw.edge_records if e.walkback]
     assert wb
+
+
+def test_gpu_walk_pipelined(tmp_path):
+    """Pipelined two-pool mode: same per-page semantics (pages, edges,
+    JSONL lines = posts stored), device stage overlapped on a worker
+    thread."""
+    cfg, sm, rw, eng = mk(tmp_path)
+    eng.seed(["c%010d" % i for i in range(1, 9)])
+    stats = eng.run(max_pages=24, now=NOW, pipelined=True)
+    assert stats["pages"] == 24
+    assert stats["posts"] == 24 * 64
+    followed = [e for e in rw.edge_records if not e.skipped]
+    assert followed
+    assert rw.buffer_size() >= 1
+    # every stored JSONL line is intact (count == posts, all parse)
+    import json as _json
+
+    n_lines = 0
+    for f in (tmp_path / "grw1").rglob("posts.jsonl"):
+        for line in f.read_bytes().splitlines():
+            _json.loads(line)
+            n_lines += 1
+    assert n_lines == stats["posts"]
