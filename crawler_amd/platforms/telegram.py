@@ -83,6 +83,26 @@ class TelegramCrawler(PlatformCrawler):
             print(f"gpu crawl complete: {stats}", file=sys.stderr)
             return stats
 
+        if cfg.sampling_method == "random-walk" and args.gpu:
+            # MI355X execution mode: batched walker hops through the
+            # HIP pipeline, pipelined two-pool form (the production
+            # default; engine/gpu_randomwalk.py)
+            from ..engine.gpu_randomwalk import GpuRandomWalk
+
+            rw = RandomWalkStore()
+            eng = GpuRandomWalk(
+                cfg, sm, rw, feed,
+                posts_per_hop=args.synthetic_posts,
+                walkers=max(cfg.concurrency, 16),
+            )
+            if cfg.seed_size and not urls:
+                urls = [feed.username_of(i)
+                        for i in range(cfg.seed_size)]
+            eng.seed(urls)
+            stats = eng.run(max_pages=cfg.max_pages, pipelined=True)
+            print(f"gpu random-walk complete: {stats}", file=sys.stderr)
+            return stats
+
         if cfg.sampling_method == "random-walk":
             rw = RandomWalkStore()
             runner = RandomWalkRunner(cfg, sm, rw, self.pool)

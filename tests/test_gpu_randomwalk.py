@@ -90,3 +90,21 @@ def test_gpu_walk_pipelined(tmp_path):
             _json.loads(line)
             n_lines += 1
     assert n_lines == stats["posts"]
+
+
+def test_cli_gpu_random_walk(tmp_path):
+    """--gpu --sampling random-walk through the CLI dispatch runs the
+    batched GPU engine (pipelined) end to end."""
+    from crawler_amd.cli import main
+
+    rc = main([
+        "--mode", "standalone", "--gpu", "--sampling", "random-walk",
+        "--seed-size", "8", "--max-pages", "24",
+        "--storage-root", str(tmp_path), "--crawl-id", "gcliwalk1",
+        "--synthetic-universe", "500", "--synthetic-posts", "32",
+        "--concurrency", "16", "--min-users", "1", "--skip-media",
+    ])
+    assert rc == 0
+    posts = list((tmp_path / "gcliwalk1").rglob("posts.jsonl"))
+    assert posts
+    assert sum(p.stat().st_size for p in posts) > 0
