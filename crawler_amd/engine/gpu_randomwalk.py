@@ -115,9 +115,12 @@ class GpuRandomWalk:
 
     # ---- validation cache: vectorized open-addressing hash table ----
     # Keys are the device fnv1a64 link hashes (int64); values index an
-    # append-only (names, adm) store. Inserts only ever add NEW keys
-    # (misses), so there are no updates or deletes; the sentinel key is
-    # int64.min, which fnv1a64 of a <=32-byte ASCII name never hits.
+    # append-only admission-bit store (the NAME need not be cached —
+    # every query carries its own bytes row, and validation is a pure
+    # function of the name, so a hit only answers "admitted?").
+    # Inserts only ever add NEW keys (misses), so there are no updates
+    # or deletes; the sentinel key is int64.min, which fnv1a64 of a
+    # <=32-byte ASCII name never hits.
 
     _VC_EMPTY = np.int64(-2 ** 63)
 
@@ -126,7 +129,6 @@ class GpuRandomWalk:
         self._vc_vals = np.zeros(slots, dtype=np.int64)
         self._vc_mask = slots - 1
         self._vc_n = 0
-        self._vc_names = np.zeros(0, dtype=object)
         self._vc_adm = np.zeros(0, dtype=bool)
 
     @staticmethod
@@ -143,9 +145,9 @@ class GpuRandomWalk:
         return (self._vc_mix(keys) & np.uint64(self._vc_mask)).astype(
             np.int64)
 
-    def _vc_lookup(self, hashes, out_names, out_adm):
-        """Fill out_names/out_adm for cached keys; returns the missing
-        mask. Vectorized linear probe: each round gathers one slot per
+    def _vc_lookup(self, hashes, out_adm):
+        """Fill out_adm for cached keys; returns the missing mask.
+        Vectorized linear probe: each round gathers one slot per
         still-active query."""
         n = len(hashes)
         miss = np.ones(n, dtype=bool)
@@ -160,21 +162,19 @@ class GpuRandomWalk:
             if hit.any():
                 ai = active[hit]
                 vi = self._vc_vals[slot[ai]]
-                out_names[ai] = self._vc_names[vi]
                 out_adm[ai] = self._vc_adm[vi]
                 miss[ai] = False
             active = active[~(hit | empty)]
             slot[active] = (slot[active] + 1) & self._vc_mask
         return miss
 
-    def _vc_insert(self, new_h, new_n, new_a):
+    def _vc_insert(self, new_h, new_a):
         """Insert NEW unique keys (the hop's cache misses). Grows the
         table at 50% load (rebuild is a bulk re-insert, amortized)."""
         k = len(new_h)
         if not k:
             return
-        base = len(self._vc_names)
-        self._vc_names = np.concatenate([self._vc_names, new_n])
+        base = len(self._vc_adm)
         self._vc_adm = np.concatenate([self._vc_adm, new_a])
         if (self._vc_n + k) * 2 > self._vc_mask + 1:
             slots = (self._vc_mask + 1) * 2
@@ -411,9 +411,8 @@ class GpuRandomWalk:
         hop_now = _dt.datetime.now(_dt.timezone.utc)
         U = len(u_h)
         un = np.ascontiguousarray(u_rows).view(f"S{w}").ravel()
-        res_names = np.empty(U, dtype=object)
         res_adm = np.zeros(U, dtype=bool)
-        miss = self._vc_lookup(u_h, res_names, res_adm)
+        miss = self._vc_lookup(u_h, res_adm)
         self.timings["t-cache"] += _time.perf_counter() - t0
         t0 = _time.perf_counter()
         inv_arr = self._inv_snapshot(hop_now, w)
@@ -423,18 +422,15 @@ class GpuRandomWalk:
             ok_v = ok_all[miss][m_first]
             cid_ok_v = cid_ok_all[miss][m_first]
             adm_v = ok_v & cid_ok_v
-            m_names = np.empty(len(mh), dtype=object)
             adm_i = np.nonzero(adm_v)[0]
             new_admitted = vecvalidate.decode_names(m_rows[adm_i])
-            m_names[adm_i] = new_admitted
             # filter-ok but not a live channel -> invalid-cache insert
             # (runner.go:1310-1383's 400 branch); marked once, cached
             for nm in vecvalidate.decode_names(m_rows[ok_v & ~cid_ok_v]):
                 self.rw.mark_invalid_channel(nm)
             mpos = np.searchsorted(mh, u_h[miss])
-            res_names[miss] = m_names[mpos]
             res_adm[miss] = adm_v[mpos]
-            self._vc_insert(mh, m_names, adm_v)
+            self._vc_insert(mh, adm_v)
             # discovered/seed admission only needs FIRST-SEEN names:
             # every admitted name from an earlier hop was admitted the
             # hop it was first validated (400-marked names are never
@@ -455,20 +451,16 @@ class GpuRandomWalk:
         owner_b = np.array([p.url for p in live], dtype=un.dtype)
         final = res_adm & not_inv & (un != owner_b[u_w])
         uw_s = u_w[final]
-        names_s = res_names[final]
+        names_s = un[final]   # names stay S<w> bytes end-to-end
         bounds = np.searchsorted(uw_s, np.arange(K + 1))
-        # per-walker lists arrive NAME-SORTED from the device sort
-        # (byte-lexicographic == python str sort for ASCII names), so
-        # no host re-sort (walk_tail_fast consumes sorted lists)
-        per_walker = [list(names_s[bounds[k]:bounds[k + 1]])
-                      for k in range(K)]
+        # per-walker slices arrive NAME-SORTED from the device sort
+        # (byte-lexicographic == python str sort for ASCII names)
         self.timings["t-group"] += _time.perf_counter() - t0
         t0 = _time.perf_counter()
         for k, p in enumerate(live):
             try:
-                randomwalk.walk_tail_fast(p, per_walker[k], self.sm,
-                                          self.rw, self.cfg, self.rng,
-                                          hop_now)
+                self._walk_tail_rows(
+                    p, names_s[bounds[k]:bounds[k + 1]], hop_now)
             except randomwalk.E.WalkbackExhausted:
                 self.stats["walkback_exhausted"] += 1
                 continue  # page left in buffer
@@ -477,6 +469,59 @@ class GpuRandomWalk:
             self.stats["pages"] += 1
         self.timings["walk-tail"] += _time.perf_counter() - t0
         self.stats["edges"] = self.rw.edge_count()
+
+    def _walk_tail_rows(self, owner: Page, rows, now) -> Page:
+        """randomwalk.walk_tail_fast with IDENTICAL decisions and rng
+        consumption, taking the walker's sorted S<w> byte rows instead
+        of decoded strings: only the followed name is decoded; skipped
+        names land in the O(1) edge block as the bytes slice and
+        decode lazily on edge_records materialization. Equivalence is
+        pinned by tests/test_gpu_randomwalk.py (CPU unit:
+        test_walk_tail_rows_matches_fast)."""
+        import uuid as _uuid
+
+        from .state import EdgeRecord
+
+        page = Page(
+            id=str(_uuid.uuid4()), parent_id=owner.id,
+            depth=owner.depth + 1, status="unfetched",
+        )
+        src = owner.url
+        seq = owner.sequence_id
+        walkback = len(rows) == 0
+        rnd = None
+        if not walkback:
+            rnd = self.rng.randint(1, 100)
+        if walkback or self.cfg.walkback_rate >= rnd:
+            # pick_walkback_channel (randomwalk.py:30-41) with a
+            # bytes-membership exclusion set (same rng consumption)
+            excl = set(rows.tolist())
+            url = None
+            for _ in range(randomwalk.MAX_WALKBACK_ATTEMPTS):
+                cand = self.sm.get_random_discovered_channel(self.rng)
+                if cand is None:
+                    raise randomwalk.E.WalkbackExhausted(src)
+                if cand == src or cand.encode() in excl:
+                    continue
+                url = cand
+                break
+            if url is None:
+                raise randomwalk.E.WalkbackExhausted(src)
+            page.url = url
+            page.sequence_id = str(_uuid.uuid4())  # fresh chain
+            skipped = rows
+            edge = EdgeRecord(url, now, src, True, False, seq, "")
+        else:
+            pick = self.rng.randrange(len(rows))
+            page.url = rows[pick].decode()
+            page.sequence_id = seq
+            skipped = (np.delete(rows, pick) if len(rows) > 1
+                       else rows[:0])
+            edge = EdgeRecord(page.url, now, src, False, False, seq, "")
+        self.rw.add_page(page)
+        self.rw.save_edge_records_fast([edge])
+        self.rw.save_skipped_edges_block(src, seq, skipped, now)
+        return page
 
     def run(self, max_pages: int, now: Optional[_dt.datetime] = None,
             pipelined: bool = False) -> dict:

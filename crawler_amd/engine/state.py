@@ -879,17 +879,22 @@ class RandomWalkStore:
         if not self._edge_blocks:
             return
         for (src, seq, t, names) in self._edge_blocks:
+            # names may be str lists (CPU path) or S<w> byte rows (the
+            # GPU walk stores the device slice verbatim; decode here,
+            # on first read — most runs only ever ask edge_count())
             self._edge_list.extend(
-                EdgeRecord(n, t, src, False, True, seq, "")
+                EdgeRecord(n.decode() if isinstance(n, bytes) else n,
+                           t, src, False, True, seq, "")
                 for n in names)
         self._edge_blocks = []
 
     def save_skipped_edges_block(self, source: str, sequence_id: str,
                                  names, now) -> None:
-        """O(1) append of one walker's skipped-edge set; expanded to
-        EdgeRecord rows on first edge_records read."""
+        """O(1) append of one walker's skipped-edge set (str list or
+        S<w> bytes array); expanded to EdgeRecord rows on first
+        edge_records read."""
         with self._lock:
-            if names:
+            if len(names):
                 self._edge_blocks.append((source, sequence_id, now,
                                           names))
 
