@@ -28,6 +28,8 @@ from __future__ import annotations
 import uuid
 from typing import Callable, List, Tuple
 
+import numpy as np
+
 from ..engine.state import LocalStateManager, Page
 from . import collectives as C
 from . import messages as M
@@ -94,7 +96,8 @@ class OrchestratedCrawl:
                 self.work_q.publish(item.to_json())
         dist.barrier()  # queue fully published before claims start
 
-        my_discovered: List[str] = []
+        my_discovered: List = []   # str names, or uint8[N,32] arrays
+        rows_mode = False
         my_dead: set = set()
         while True:
             raw = self.work_q.claim(timeout_s=0.0)
@@ -105,7 +108,15 @@ class OrchestratedCrawl:
             discovered, posts = self.process_fn(chans)
             if self.deadends_fn is not None:
                 my_dead |= set(self.deadends_fn())
-            my_discovered.extend(discovered)
+            if isinstance(discovered, np.ndarray):
+                # rows mode (engine as_arrays=True): discovery names
+                # stay zero-padded uint8[N, 32] rows end to end — the
+                # exchange, dedup and budget below never build ~1M
+                # python strings (only the admitted next layer decodes)
+                rows_mode = True
+                my_discovered.append(discovered)
+            else:
+                my_discovered.extend(discovered)
             self.stats["chunks"] += 1
             self.stats["pages"] += len(chans)
             self.stats["posts"] += posts
@@ -119,6 +130,16 @@ class OrchestratedCrawl:
         self.heartbeats.beat(f"rank{self.rank}", M.WORKER_IDLE)
         all_dead = (set(self._allgather_names(sorted(my_dead)))
                     if self.deadends_fn is not None else set())
+        if rows_mode:
+            import torch
+
+            rows_np = (np.concatenate(my_discovered)
+                       if my_discovered
+                       else np.zeros((0, 32), dtype=np.uint8))
+            out = C.allgather_rows(
+                torch.from_numpy(np.ascontiguousarray(rows_np)),
+                self.dist, self.world, device=self.device)
+            return out.cpu().numpy(), all_dead
         return self._allgather_names(my_discovered), all_dead
 
     # ---- the crawl ----
@@ -145,7 +166,7 @@ class OrchestratedCrawl:
             if (self.cfg.sampling_method == "snowball"
                     and (self.cfg.max_depth < 0
                          or depth < self.cfg.max_depth)
-                    and all_discovered):
+                    and len(all_discovered)):
                 # every rank derives the SAME next layer (sorted set ->
                 # deterministic admission order). Pre-apply add_layer's
                 # URL-dedup + MaxPages budget before building Page
@@ -160,14 +181,31 @@ class OrchestratedCrawl:
                     budget = max(0, max_pages - total) + deadends
                 existing = {p.url for p in sm.pages.values()}
                 url_dedup = getattr(sm, "url_dedup", {})
-                cand = []
-                for n in sorted(set(all_discovered)):
-                    if budget is not None and len(cand) >= budget:
-                        break
-                    if n in existing or n in url_dedup:
-                        continue
-                    existing.add(n)
-                    cand.append(n)
+                if isinstance(all_discovered, np.ndarray):
+                    # rows mode: sorted-unique on S32 views (ASCII byte
+                    # order == str sort order), vectorized membership,
+                    # decode ONLY the admitted budget-capped slice
+                    un = np.unique(np.ascontiguousarray(all_discovered)
+                                   .view("S32").ravel())
+                    known = {u.encode() for u in existing}
+                    known.update(u.encode() for u in url_dedup)
+                    if known:
+                        karr = np.array(sorted(known), dtype="S32")
+                        pos = np.searchsorted(karr, un)
+                        pos_c = np.clip(pos, 0, len(karr) - 1)
+                        un = un[karr[pos_c] != un]
+                    if budget is not None:
+                        un = un[:budget]
+                    cand = [b.decode() for b in un.tolist()]
+                else:
+                    cand = []
+                    for n in sorted(set(all_discovered)):
+                        if budget is not None and len(cand) >= budget:
+                            break
+                        if n in existing or n in url_dedup:
+                            continue
+                        existing.add(n)
+                        cand.append(n)
                 pages = [Page(url=n, depth=depth + 1, status="unfetched")
                          for n in cand]
                 sm.add_layer(pages)

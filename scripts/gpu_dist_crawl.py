@@ -80,7 +80,12 @@ def main():
         device = torch.device("cuda", local_rank)
     crawl = OrchestratedCrawl(
         cfg, sm, store, rank, world,
-        process_fn=lambda names: eng.process_channels(names),
+        process_fn=(
+            (lambda names: eng.process_channels(names))
+            if (args.fake_engine
+                or os.environ.get("CRAWL_DIST_STR_NAMES")) else
+            (lambda names: eng.process_channels(names, as_arrays=True))
+        ),
         chunk_channels=args.chunk_channels,
         dist=dist if world > 1 else _SoloDist(),
         device=device,
@@ -98,6 +103,8 @@ def main():
             "elapsed_s": round(elapsed, 2),
             "world": world,
             **stats,
+            "engine_phase_s": {k: round(v, 3) for k, v in
+                               getattr(eng, "timings", {}).items()},
         }))
     if world > 1:
         dist.destroy_process_group()

@@ -130,3 +130,43 @@ def test_orchestrated_marks_deadends(tmp_path):
     statuses = {p.url: p.status for p in sm.get_layer_by_depth(0)}
     assert statuses == {"chan001": "fetched", "deadbeef": "deadend",
                         "chan002": "fetched"}
+
+
+def test_single_rank_rows_mode(tmp_path):
+    """process_fn returning uint8[N,32] rows (the GPU engine's
+    as_arrays form): the exchange/dedup/budget path stays in rows and
+    admits the same next layer as the string path."""
+    import numpy as np
+
+    from crawler_amd.parallel import collectives as C
+
+    def rows_process(names):
+        out = []
+        for n in names:
+            for d in (n + "x1", n + "x2", "sharedchan"):
+                out.append(d)
+        rows = C.names_to_rows(out).numpy()
+        return rows, len(names) * 10
+
+    def str_process(names):
+        out = []
+        for n in names:
+            for d in (n + "x1", n + "x2", "sharedchan"):
+                out.append(d)
+        return out, len(names) * 10
+
+    res = {}
+    for key, fn in (("rows", rows_process), ("strs", str_process)):
+        cfg = CrawlerConfig(crawl_id=f"rm-{key}",
+                            storage_root=str(tmp_path / key),
+                            sampling_method="snowball", max_depth=1,
+                            max_pages=50, min_users=1)
+        sm = LocalStateManager(cfg)
+        store = InMemoryStore()
+        crawl = OrchestratedCrawl(cfg, sm, store, rank=0, world=1,
+                                  process_fn=fn, chunk_channels=2,
+                                  dist=FakeDist())
+        stats = crawl.run(["seedaa", "seedbb"])
+        res[key] = (stats["pages"], stats["posts"],
+                    sorted(p.url for p in sm.pages.values()))
+    assert res["rows"] == res["strs"]
