@@ -67,10 +67,15 @@ def _rank_main(rank, world, port, tmp_dir):
                         min_users=1)
     sm = LocalStateManager(cfg)
     processed = []
+    rows_mode = os.environ.get("ORCH_TEST_ROWS") == "1"
 
     def process(names):
         processed.extend(names)
         d, p = stub_process(names)
+        if rows_mode:
+            from crawler_amd.parallel import collectives as C
+
+            return C.names_to_rows(d).numpy(), p
         return d, p
 
     crawl = OrchestratedCrawl(cfg, sm, store, rank, world,
@@ -87,6 +92,28 @@ def test_two_rank_orchestrated_gloo(tmp_path):
 
     mp.spawn(_rank_main, args=(2, 29725, str(tmp_path)), nprocs=2,
              join=True)
+    _check_two_rank(tmp_path)
+
+
+def test_two_rank_orchestrated_gloo_rows(tmp_path):
+    """World-2 gloo with the GPU engine's rows-mode discovery payload:
+    uint8[N,32] rows travel through allgather_rows and the vectorized
+    admission — same split/coverage/totals as the string path."""
+    import torch.multiprocessing as mp
+
+    os.environ["ORCH_TEST_ROWS"] = "1"
+    try:
+        mp.spawn(_rank_main, args=(2, 29739, str(tmp_path)), nprocs=2,
+                 join=True)
+    finally:
+        os.environ.pop("ORCH_TEST_ROWS", None)
+    _check_two_rank(tmp_path)
+
+
+def _check_two_rank(tmp_path):
+    import pathlib
+
+    tmp_path = pathlib.Path(tmp_path)
     s0 = json.loads((tmp_path / "stats0.json").read_text())
     s1 = json.loads((tmp_path / "stats1.json").read_text())
     # work was split dynamically, no chunk processed twice, all covered
