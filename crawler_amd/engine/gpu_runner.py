@@ -70,7 +70,8 @@ class GpuCrawlEngine:
 
     def process_channels(self, usernames: List[str],
                          now: Optional[_dt.datetime] = None,
-                         as_arrays: bool = False):
+                         as_arrays: bool = False,
+                         drain: bool = True):
         """Process a list of channels; returns (newly discovered names,
         posts stored). as_arrays=True returns the names as ONE
         zero-padded uint8[M, 32] array instead of python strings — the
@@ -206,12 +207,24 @@ class GpuCrawlEngine:
                           else np.zeros((0, 32), dtype=np.uint8))
         self.stats["posts"] += posts_total
         self.stats["discovered"] += len(discovered)
-        # barrier: all spill writes down before the layer's save_state
+        if drain:
+            # barrier: all spill writes down before the layer's
+            # save_state. Callers that invoke process_channels once per
+            # work CHUNK (the orchestrated crawl) pass drain=False and
+            # call drain_spills() once per LAYER instead — otherwise
+            # every chunk waits out its own disk writes and the 2-deep
+            # spill pipeline never overlaps the next chunk's GPU work.
+            self.drain_spills()
+        return discovered, posts_total
+
+    def drain_spills(self) -> None:
+        """Wait out all in-flight ticketed sink writes (safe point for
+        save_state; truncate-then-write keeps re-processing
+        exactly-once if a crash lands before this)."""
         for slot in (0, 1):
             self.sm.wait_post_write(self._spill_tickets[slot])
             self._spill_tickets[slot] = None
         self.sm.drain_post_writes()
-        return discovered, posts_total
 
     # ---- BFS crawl (snowball / channel) ----
 

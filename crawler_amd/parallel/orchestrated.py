@@ -41,7 +41,8 @@ class OrchestratedCrawl:
                  world: int,
                  process_fn: Callable[[List[str]], Tuple[List[str], int]],
                  chunk_channels: int = 64, dist=None, device=None,
-                 deadends_fn: Callable[[], set] = None):
+                 deadends_fn: Callable[[], set] = None,
+                 flush_fn: Callable[[], None] = None):
         """process_fn(names) -> (discovered_names, posts_stored).
         `dist` is torch.distributed (injected so CPU tests can pass gloo
         and unit tests can fake it); `device` is the compute device used
@@ -49,7 +50,11 @@ class OrchestratedCrawl:
         `deadends_fn` (optional) reports the channels the LAST
         process_fn call classified as deadends, so page statuses and
         the MaxPages deadend-replacement budget (state/base.go:284)
-        stay correct — and identical across ranks via the exchange."""
+        stay correct — and identical across ranks via the exchange.
+        `flush_fn` (optional) is called once per layer after all
+        chunks complete, BEFORE statuses/checkpoint — engines that
+        pipeline disk writes across chunks (process_channels
+        drain=False) drain them here."""
         self.cfg = cfg
         self.sm = sm
         self.rank = rank
@@ -62,6 +67,7 @@ class OrchestratedCrawl:
         self.dist = dist
         self.device = device
         self.deadends_fn = deadends_fn
+        self.flush_fn = flush_fn
         self.work_q = StoreQueue(store, M.TOPIC_WORK_QUEUE)
         self.result_q = StoreQueue(store, M.TOPIC_RESULTS)
         self.heartbeats = Heartbeats(store)
@@ -128,6 +134,8 @@ class OrchestratedCrawl:
                 trace_id=item.trace_id,
             ).to_json())
         self.heartbeats.beat(f"rank{self.rank}", M.WORKER_IDLE)
+        if self.flush_fn is not None:
+            self.flush_fn()   # drain pipelined disk writes (layer safe point)
         all_dead = (set(self._allgather_names(sorted(my_dead)))
                     if self.deadends_fn is not None else set())
         if rows_mode:

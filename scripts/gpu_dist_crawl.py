@@ -84,8 +84,10 @@ def main():
             (lambda names: eng.process_channels(names))
             if (args.fake_engine
                 or os.environ.get("CRAWL_DIST_STR_NAMES")) else
-            (lambda names: eng.process_channels(names, as_arrays=True))
+            (lambda names: eng.process_channels(names, as_arrays=True,
+                                                drain=False))
         ),
+        flush_fn=(None if args.fake_engine else eng.drain_spills),
         chunk_channels=args.chunk_channels,
         dist=dist if world > 1 else _SoloDist(),
         device=device,
