@@ -142,16 +142,22 @@ DEV bool is_reserved_path(const unsigned char* name, int n) {
   return false;
 }
 
-// Extract username after "t.me/" at byte q (exclusive bound hi). Writes the
-// LOWERCASED name; returns length (5..32) or 0. Runs uniformly on all lanes.
-DEV int extract_name_at(const unsigned char* s, int hi, int q,
-                        unsigned char* name) {
+// Regex-group match after "t.me/" at byte q (exclusive bound hi).
+// Returns the MATCH length (5..32, lowercased into name) or 0 when the
+// regex does not match here; *reserved reports the reserved-path
+// filter separately, because the two failures behave differently:
+// no-regex-match means re.search keeps scanning for the NEXT t.me,
+// while a reserved MATCH ends a .search() (and, in the plaintext
+// FindAll walk, consumes its span). Runs uniformly on all lanes.
+DEV int match_name_at(const unsigned char* s, int hi, int q,
+                      unsigned char* name, bool* reserved) {
+  *reserved = false;
   if (q >= hi || !is_ascii_letter(s[q])) return 0;
   int len = 0;
   while (len < 32 && q + len < hi && is_word_char(s[q + len])) ++len;
   if (len < 5) return 0;
   for (int j = 0; j < len; ++j) name[j] = to_lower(s[q + j]);
-  if (is_reserved_path(name, len)) return 0;
+  *reserved = is_reserved_path(name, len);
   return len;
 }
 
@@ -232,13 +238,19 @@ DEV void extract_links(const BatchView& B, int i, LinkList& L, int lane) {
   for (int e = e0; e < e0 + ec; ++e) {
     const int* row = B.entities + (long)e * 5;
     int etype = row[0];
-    if (etype == 1) {  // text_url: scan the URL attribute
+    if (etype == 1) {  // text_url: .search() over the URL attribute
       const unsigned char* url = B.pool + row[3];
       int un = row[4];
-      int p = find_tme(url, 0, un, lane);
-      if (p >= 0) {
-        int len = extract_name_at(url, un, p + 5, name);
-        if (len) L.add(name, len, 1, lane);
+      for (int from = 0;;) {
+        int p = find_tme(url, from, un, lane);
+        if (p < 0) break;
+        bool rsv;
+        int len = match_name_at(url, un, p + 5, name, &rsv);
+        if (len) {  // first regex match decides the .search()
+          if (!rsv) L.add(name, len, 1, lane);
+          break;
+        }
+        from = p + 1;  // not a regex match here; search continues
       }
     } else {  // mention (0) / url (2): slice text at utf16 offsets
       int lo, hi;
@@ -248,16 +260,24 @@ DEV void extract_links(const BatchView& B, int i, LinkList& L, int lane) {
         int len = find_username(text, lo, hi, lane, name);
         if (len) L.add(name, len, 0, lane);
       } else {
-        int p = find_tme(text, lo, hi, lane);
-        if (p >= 0) {
-          int len = extract_name_at(text, hi, p + 5, name);
-          if (len) L.add(name, len, 2, lane);
+        for (int from = lo;;) {
+          int p = find_tme(text, from, hi, lane);
+          if (p < 0) break;
+          bool rsv;
+          int len = match_name_at(text, hi, p + 5, name, &rsv);
+          if (len) {
+            if (!rsv) L.add(name, len, 2, lane);
+            break;
+          }
+          from = p + 1;
         }
       }
     }
   }
 
-  // plaintext scan with non-overlap cursor (FindAllStringSubmatch semantics)
+  // plaintext scan with non-overlap cursor (FindAllStringSubmatch
+  // semantics: every regex match — reserved or not — consumes its
+  // span; only non-matches keep probing forward byte by byte)
   int cursor = 0;
   int from = 0;
   while (true) {
@@ -265,14 +285,12 @@ DEV void extract_links(const BatchView& B, int i, LinkList& L, int lane) {
     if (p < 0) break;
     from = p + 1;  // next candidate search position
     if (p < cursor) continue;
-    int len = extract_name_at(text, tn, p + 5, name);
+    bool rsv;
+    int len = match_name_at(text, tn, p + 5, name, &rsv);
     if (len) {
-      L.add(name, len, 3, lane);
+      if (!rsv) L.add(name, len, 3, lane);
       cursor = p + 5 + len;
       from = cursor;
-    } else {
-      // invalid name: Go regex has no match AT this t.me; try the next one
-      continue;
     }
   }
 }

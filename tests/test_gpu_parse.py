@@ -329,3 +329,57 @@ def test_single_pass_variant_bytes_identical(gpu_mod, feed):
                                single_pass=True)
     torch.cuda.synchronize()
     assert bytes(res.out.cpu().numpy()) == b"".join(golden_lines)
+
+
+def test_entity_search_retries_later_tme(gpu_mod):
+    """re.search semantics inside url/text_url entities (fuzz seed 5049
+    regression): a 't.me/' with no valid name after it does NOT end the
+    search — the next occurrence inside the same entity can still
+    match, truncated at the entity boundary ('t.me/joinc' out of
+    'joinchat' when the entity ends mid-word). Also pins: first regex
+    match DOES end the search even when reserved, and the plaintext
+    FindAll walk consumes reserved matches' spans."""
+    t1 = "plain e>f ñé test … t.me/    t.me/joinchat/xyz t.me/"
+    ents1 = [
+        G.Entity("mention", 14, 9),
+        G.Entity("url", 30, 13),
+        G.Entity("url", 20, 19),   # covers "t.me/    t.me/joinc"
+    ]
+    # url entity whose FIRST regex match is reserved: search stops
+    # there, so 'okname99' later in the entity is NOT found
+    t2 = "x t.me/share t.me/okname99 y"
+    ents2 = [G.Entity("url", 0, len(t2))]
+    # text_url attribute with a dead t.me/ before a live one
+    t3 = "click here"
+    ents3 = [G.Entity("text_url", 0, 5,
+                      url="https://t.me/ %% t.me/attrchan9 z")]
+    msgs = []
+    for k, (t, ents) in enumerate([(t1, ents1), (t2, ents2),
+                                   (t3, ents3)]):
+        msgs.append(G.SynthMessage(
+            chat_id=-100123, msg_id=(k + 1) << 20,
+            date=1_700_000_000 + k, content_type="messageText",
+            text=G.FormattedText(text=t, entities=ents),
+            views=k, forwards=k, poster_handle=f"user{k:04d}",
+        ))
+    ch = [B.ChannelRow(chat_id=-100123, username="entchan99",
+                       title="T", member_count=5, post_count=len(msgs),
+                       total_views=9)]
+    batch = B.pack(msgs, ch, [0] * len(msgs))
+    from crawler_amd.ops import gpu
+
+    golden_lines, golden_links = encode_batch(batch, now=NOW)
+    # oracle sanity: the exact expectations this test exists for
+    assert ("joinc", "url") in golden_links[0]
+    # the url-entity search stopped at the reserved first match, so
+    # okname99 is attributed to the PLAINTEXT scan, not the entity
+    assert ("okname99", "plaintext") in golden_links[1]
+    assert ("okname99", "url") not in golden_links[1]
+    assert ("attrchan9", "text_url") in golden_links[2]
+    res = gpu.parse_encode(batch.to("cuda:0"), now=NOW)
+    torch.cuda.synchronize()
+    out = bytes(res.out.cpu().numpy())
+    offs = res.line_off.cpu().numpy()
+    lens = res.line_len.cpu().numpy()
+    for i, gl in enumerate(golden_lines):
+        assert out[offs[i]: offs[i] + lens[i]] == gl, f"msg {i}"
