@@ -52,6 +52,10 @@ def rand_entities(rng, text):
             url = rng.choice([
                 "https://t.me/fuzz_target1", "https://example.com/x",
                 "t.me/abcd", "",
+                # dead t.me before a live one (search must retry)
+                "t.me/ %% t.me/deadlive1",
+                # reserved first match ends the search
+                "t.me/share t.me/after_rsv1",
             ])
         # occasionally overshoot the end (golden clamps)
         if rng.random() < 0.1:
