@@ -229,8 +229,17 @@ def handle_400_replacement(sm, rw: RandomWalkStore, p: Page, cfg,
 
     if edge is None:
         if sm.is_seed_channel(channel):
-            # seed replacement: walk back from the seed itself
-            walkback_from(channel)
+            # seed replacement (handle400SeedReplacement,
+            # runner.go:263-284): a random VALID seed channel becomes
+            # the new page — fresh sequence chain, NO edge record
+            seed_url = rw.get_random_seed_channel(rng)
+            if seed_url is None:
+                raise E.WalkbackExhausted(channel)
+            rw.add_page(Page(
+                id=str(uuid.uuid4()), parent_id=p.parent_id,
+                depth=p.depth, url=seed_url,
+                sequence_id=str(uuid.uuid4()), status="unfetched",
+            ))
             return
         walkback_from(channel)
         return

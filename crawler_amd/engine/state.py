@@ -1013,6 +1013,20 @@ class RandomWalkStore:
             if username in self.seed_channels:
                 self._seed_row(username)["invalidated_at"] = _now()
 
+    def get_random_seed_channel(self, rng=None,
+                                ttl_days: int = 30) -> Optional[str]:
+        """Random non-invalidated (or TTL-expired) seed username
+        (daprstate.go:4181-4196: invalidated_at IS NULL OR older than
+        30 days, ORDER BY RANDOM() LIMIT 1). Returns None when the
+        table has no eligible rows."""
+        r = rng or random
+        cutoff = _now() - _dt.timedelta(days=ttl_days)
+        with self._lock:
+            elig = [u for u, row in self.seed_channels.items()
+                    if row is None or row["invalidated_at"] is None
+                    or row["invalidated_at"] < cutoff]
+            return r.choice(elig) if elig else None
+
     def load_seed_channels(self, ttl_days: int = 30) -> List[dict]:
         """Rows whose invalidation is absent or older than the TTL
         (daprstate.go:3327-3429)."""

@@ -76,15 +76,39 @@ def test_400_no_edge_non_seed_walks_back_from_self(tmp_path):
 
 
 def test_400_seed_channel_replacement(tmp_path):
-    """Dead channel IS a seed: walk back from the seed (runner.go:197-216)."""
+    """Dead channel IS a seed: replaced by a random VALID seed from
+    seed_channels — fresh chain, NO edge record
+    (handle400SeedReplacement, runner.go:263-284; the seed pick
+    excludes invalidated rows per daprstate.go:4181-4196)."""
     cfg, sm, rw = mk(tmp_path)
     seed_discovered(sm, ["alpha", "bravo"])
     sm.upsert_seed_channel_chat_id("deadseed", 42)
+    for u in ("deadseed", "seedx", "seedy"):
+        rw.upsert_seed_channel(u)
+    n_edges = rw.edge_count()
     p = mk_page("deadseed", seq="seq-s")
     RW.handle_400_replacement(sm, rw, p, cfg, random.Random(3))
     pages = rw.get_pages(10)
-    assert len(pages) == 1 and pages[0].url in {"alpha", "bravo"}
+    # deadseed was invalidated by the handler itself, so only the
+    # other seed rows are eligible — never the discovered set
+    assert len(pages) == 1 and pages[0].url in {"seedx", "seedy"}
     assert pages[0].sequence_id != "seq-s"
+    assert rw.edge_count() == n_edges  # no edge record written
+
+
+def test_400_seed_replacement_empty_table_exhausts(tmp_path):
+    """No eligible seed rows -> the error path (reference
+    GetRandomSeedChannel errors on an empty table)."""
+    import pytest
+
+    from crawler_amd.engine import errors as E
+
+    cfg, sm, rw = mk(tmp_path)
+    seed_discovered(sm, ["alpha"])
+    sm.upsert_seed_channel_chat_id("deadseed", 42)
+    p = mk_page("deadseed", seq="seq-s")
+    with pytest.raises(E.WalkbackExhausted):
+        RW.handle_400_replacement(sm, rw, p, cfg, random.Random(3))
 
 
 def test_400_walkback_edge_rewalks_from_edge_source(tmp_path):

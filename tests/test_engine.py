@@ -183,6 +183,40 @@ def test_random_walk_400_replacement(tmp_path):
     assert "bad1" not in rw.page_buffer  # deleted after replacement
 
 
+def test_400_seed_replacement_picks_valid_seed(tmp_path):
+    """handle400SeedReplacement (runner.go:263-284): a 400'd SEED
+    channel (no incoming edge) is replaced by a random VALID seed —
+    fresh sequence chain, NO edge record, and never an invalidated
+    seed."""
+    import datetime as dt
+
+    from crawler_amd.engine import randomwalk as RW
+
+    cfg, feed, pool, sm, rw = rw_env(tmp_path)
+    for u in ("c0000000001", "c0000000002", "c0000000003"):
+        sm.add_discovered_channel(u)
+        rw.upsert_seed_channel(u)
+        sm._seed_channels[u] = True  # is_seed_channel
+    rw.mark_seed_channel_invalid("c0000000002")
+    bad = Page(id="seedbad", url="c0000000001", depth=0,
+               sequence_id="sq0", status="unfetched")
+    rw.add_page(bad)
+    n_edges = rw.edge_count()
+    RW.handle_400_replacement(sm, rw, bad, cfg, random.Random(3))
+    # replacement page exists, on a fresh chain, from the seed table
+    repl = [p for p in rw.page_buffer.values() if p.id != "seedbad"]
+    assert len(repl) == 1
+    # the failed seed was just invalidated, so only c...3 is eligible
+    assert repl[0].url == "c0000000003"
+    assert repl[0].sequence_id != "sq0"
+    # invalidated seeds are never picked
+    for _ in range(20):
+        pick = rw.get_random_seed_channel(random.Random(_))
+        assert pick != "c0000000002"
+    # NO edge record is written for seed replacement
+    assert rw.edge_count() == n_edges
+
+
 def test_tandem_mode_writes_pending_edges(tmp_path):
     cfg, feed, pool, sm, rw = rw_env(tmp_path, tandem_crawl=True)
     runner = RandomWalkRunner(cfg, sm, rw, pool, rng=random.Random(9))
