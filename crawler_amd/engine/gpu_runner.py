@@ -150,12 +150,10 @@ class GpuCrawlEngine:
                     # when a crash forces the layer to re-process
                     self.sm.truncate_posts(uname)
                     items.append((uname, lo, hi))
-                else:
-                    # valid channel but every post filtered out -> the
-                    # CPU pipeline's deadend status (runner.py status
-                    # machine; ref isChannelActiveWithinPeriod)
-                    self.last_deadends.add(uname)
-                    self.stats["deadends"] += 1
+                # an all-filtered window is NOT a deadend: the
+                # reference only deadends on inactivity/zero-message/
+                # min-users (runner.go:635) — synthetic channels always
+                # have messages, so only invalid usernames deadend here
             # one fan-out call for the whole chunk: the native sink
             # (crawler_amd/native) appends all channels in parallel with
             # the GIL released; the ticket is awaited when this ring slot

@@ -99,17 +99,24 @@ def _maybe_sample(msgs, cfg, rng):
     return msgs
 
 
-def is_channel_active(messages, member_count: int, cfg) -> Tuple[bool, str]:
-    """Activity filter (runner.go:628-643): stale channels and channels
-    below min-users become deadends."""
-    if member_count < cfg.min_users:
-        return False, f"member count {member_count} < min_users {cfg.min_users}"
-    if not messages:
-        return False, "no messages in window"
+def is_channel_active(latest_ts, member_count: int, cfg,
+                      message_count: int) -> Tuple[bool, str]:
+    """Deadend filter (runner.go:628-643 exactly):
+    `!active || messageCount == 0 || (sampling != "random-walk" &&
+    MinUsers > 0 && memberCount < MinUsers)` where active =
+    latestMessageTime.After(PostRecency) over the chat's actual
+    newest message (getLatestMessageTime — independent of the fetch
+    window, runner.go:724-732; a tie with the cutoff is INACTIVE).
+    min-users never applies in random-walk mode."""
     if cfg.post_recency is not None:
-        latest = max(m.date for m in messages)
-        if latest < cfg.post_recency.timestamp():
+        cutoff = cfg.post_recency.timestamp()
+        if latest_ts is None or latest_ts <= cutoff:
             return False, "latest message older than recency window"
+    if message_count == 0:
+        return False, "no messages in channel"
+    if (cfg.sampling_method != "random-walk" and cfg.min_users > 0
+            and member_count < cfg.min_users):
+        return False, f"member count {member_count} < min_users {cfg.min_users}"
     return True, ""
 
 
@@ -200,8 +207,20 @@ def run_for_channel(client, page: Page, sm, cfg, rw=None, seen=None,
         )
         cfg.null_validator.validate_channel_data(cd)
 
-    # -- activity filter -> deadend (runner.go:628-643) --
-    active, why = is_channel_active(messages, sg["member_count"], cfg)
+    # -- activity filter -> deadend (runner.go:628-643): the recency
+    # check uses the chat's NEWEST message (getLatestMessageTime, one
+    # GetChatHistory(limit=1) like the reference), not the filtered
+    # fetch window --
+    latest_ts = None
+    if messages:
+        latest_ts = max(m.date for m in messages)
+    else:
+        newest = client.get_chat_history(info.chat_id,
+                                         from_message_id=0, limit=1)
+        if newest:
+            latest_ts = newest[0].date
+    active, why = is_channel_active(latest_ts, sg["member_count"], cfg,
+                                    info.message_count)
     if not active:
         result.status = "deadend"
         result.error = why

@@ -152,18 +152,33 @@ def test_sampling_noop_when_fewer_than_sample_size():
 
 
 def test_is_channel_active_matrix():
+    """runner.go:635 exactly: !active || messageCount==0 ||
+    (sampling != random-walk && MinUsers>0 && members<MinUsers)."""
     conf = cfg(min_users=100,
                post_recency=ts2dt(BASE - 30 * DAY))
-    fresh = [Msg(1, BASE - 1 * DAY)]
-    stale = [Msg(1, BASE - 90 * DAY)]
-    ok, _ = is_channel_active(fresh, 500, conf)
+    fresh = BASE - 1 * DAY
+    stale = BASE - 90 * DAY
+    ok, _ = is_channel_active(fresh, 500, conf, 10)
     assert ok
-    ok, why = is_channel_active(fresh, 50, conf)
+    ok, why = is_channel_active(fresh, 50, conf, 10)
     assert not ok and "min_users" in why
-    ok, why = is_channel_active([], 500, conf)
-    assert not ok and "no messages" in why
-    ok, why = is_channel_active(stale, 500, conf)
+    # min-users NEVER applies in random-walk mode (runner.go:635)
+    ok, _ = is_channel_active(
+        fresh, 50, cfg(min_users=100, sampling_method="random-walk",
+                       post_recency=ts2dt(BASE - 30 * DAY)), 10)
+    assert ok
+    # min_users == 0 disables the member check entirely
+    ok, _ = is_channel_active(fresh, 0, cfg(
+        min_users=0, post_recency=ts2dt(BASE - 30 * DAY)), 10)
+    assert ok
+    ok, why = is_channel_active(stale, 500, conf, 10)
     assert not ok and "older than recency" in why
+    # exact tie with the cutoff is INACTIVE (time.After is strict)
+    ok, why = is_channel_active(BASE - 30 * DAY, 500, conf, 10)
+    assert not ok and "recency" in why
+    # channel with zero messages -> deadend even without recency
+    ok, why = is_channel_active(None, 500, cfg(min_users=100), 0)
+    assert not ok and "no messages" in why
     # no recency filter -> stale is fine
-    ok, _ = is_channel_active(stale, 500, cfg(min_users=100))
+    ok, _ = is_channel_active(stale, 500, cfg(min_users=100), 10)
     assert ok
