@@ -104,9 +104,23 @@ def main():
     chunks = []
     t_gen = time.time()
     if my_platform == "telegram":
-        n_chunks = max(1, args.channels // args.chunk_channels)
         chunk_ch = args.chunk_channels
         posts = args.posts
+        # keep each packed chunk's pools within int32 offsets: comment
+        # threads inflate pool bytes/post (~60B per comment incl.
+        # handles/reactions), so shrink the chunk until the estimate
+        # fits ~1.6 GB (the flagship shape is unaffected: 203 B/post)
+        est_post_bytes = 200
+        if args.comment_rate > 0:
+            est_post_bytes += int(args.comment_rate
+                                  * (args.max_comments_per_post / 2 + 1)
+                                  * 60)
+        max_chunk_posts = max(posts, int(1.6e9 / est_post_bytes))
+        if chunk_ch * posts > max_chunk_posts:
+            chunk_ch = max(1, max_chunk_posts // posts)
+            log(f"chunk_channels clamped to {chunk_ch} "
+                f"({est_post_bytes} est pool B/post keeps int32 offsets)")
+        n_chunks = max(1, args.channels // chunk_ch)
         feed = SyntheticFeed(FeedConfig(
             seed=1234 + rank, universe=1_000_000,
             comment_rate=args.comment_rate,
