@@ -102,6 +102,7 @@ def main():
         my_platform = "telegram" if rank % 2 == 0 else "youtube"
 
     chunks = []
+    eff_channels = args.channels
     t_gen = time.time()
     if my_platform == "telegram":
         chunk_ch = args.chunk_channels
@@ -120,7 +121,19 @@ def main():
             chunk_ch = max(1, max_chunk_posts // posts)
             log(f"chunk_channels clamped to {chunk_ch} "
                 f"({est_post_bytes} est pool B/post keeps int32 offsets)")
-        n_chunks = max(1, args.channels // chunk_ch)
+        channels = args.channels
+        # the whole corpus stays resident in HBM across steps; cap it
+        # so comment-heavy shapes fit the 288 GB card (weak scaling is
+        # per-step, so fewer resident channels only shortens the cycle)
+        max_total_posts = max(chunk_ch * posts,
+                              int(200e9 / est_post_bytes))
+        if channels * posts > max_total_posts:
+            channels = max(chunk_ch, max_total_posts // posts)
+            log(f"resident channels clamped to {channels} "
+                f"(~{est_post_bytes * channels * posts / 1e9:.0f} GB "
+                f"est pools fit HBM)")
+        n_chunks = max(1, channels // chunk_ch)
+        eff_channels = n_chunks * chunk_ch
         feed = SyntheticFeed(FeedConfig(
             seed=1234 + rank, universe=1_000_000,
             comment_rate=args.comment_rate,
@@ -410,7 +423,7 @@ def main():
                 "platform": args.platform,
                 "parallelism": f"dp{world} (channel-sharded, RCCL discovery "
                                "all-gather)" if world > 1 else "dp1",
-                "channels_per_gpu": args.channels,
+                "channels_per_gpu": eff_channels,
                 "posts_per_channel": args.posts,
                 "chunk_posts": chunk_posts,
                 "jsonl_bytes_per_step": total_out_bytes // max(
